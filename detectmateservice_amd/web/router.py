@@ -1,0 +1,53 @@
+"""Admin HTTP router.
+
+Reference parity (/root/reference/src/service/features/web/router.py:18-46):
+``POST /admin/start|stop|reconfigure|shutdown``, ``GET /admin/status``.
+The Service is injected via a FastAPI dependency override
+(web/server.py:31 in the reference).
+"""
+from __future__ import annotations
+
+from typing import Any, Dict, Optional
+
+from fastapi import APIRouter, Depends
+from pydantic import BaseModel
+
+router = APIRouter(prefix="/admin")
+
+
+def get_service():  # overridden by WebServer with the live Service
+    raise RuntimeError("service dependency not wired")
+
+
+class ReconfigPayload(BaseModel):
+    config: Dict[str, Any]
+    persist: bool = False
+
+
+@router.post("/start")
+def start(service=Depends(get_service)) -> Dict[str, Any]:
+    service.start()
+    return {"status": "started"}
+
+
+@router.post("/stop")
+def stop(service=Depends(get_service)) -> Dict[str, Any]:
+    service.stop()
+    return {"status": "stopped"}
+
+
+@router.get("/status")
+def status(service=Depends(get_service)) -> Dict[str, Any]:
+    return service.status()
+
+
+@router.post("/reconfigure")
+def reconfigure(payload: ReconfigPayload, service=Depends(get_service)) -> Dict[str, Any]:
+    configs = service.reconfigure(payload.config, payload.persist)
+    return {"status": "reconfigured", "configs": configs}
+
+
+@router.post("/shutdown")
+def shutdown(service=Depends(get_service)) -> Dict[str, Any]:
+    service.shutdown()
+    return {"status": "shutting down"}

@@ -1,0 +1,117 @@
+"""Detector tests: NewValue train/detect semantics, combo, dummy
+alternation, random determinism, checkpoint/resume."""
+import pytest
+
+from detectmateservice_amd.library.detectors import (
+    DummyDetector,
+    NewValueComboDetector,
+    NewValueDetector,
+    RandomDetector,
+)
+from detectmateservice_amd.schemas import DetectorSchema, ParserSchema
+
+
+def _parsed(url="/index.html", event_id=1, variables=None, log_id="l1"):
+    return ParserSchema(
+        EventID=event_id,
+        variables=variables or [],
+        logID=log_id,
+        logFormatVariables={"URL": url},
+    ).serialize()
+
+
+NV_CONFIG = {
+    "data_use_training": 2,
+    "global": {"global_instance": {"header_variables": [{"pos": "URL"}]}},
+}
+
+
+def test_new_value_detector_train_then_detect():
+    det = NewValueDetector(NV_CONFIG)
+    # first two frames are training (data_use_training: 2)
+    out = det.process_batch([_parsed("/a"), _parsed("/b")])
+    assert out == [None, None]
+    # known value: no alert
+    assert det.process(_parsed("/a")) is None
+    # unseen value: alert with the reference's description shape
+    alert_bytes = det.process(_parsed("/foobar", log_id="L9"))
+    assert alert_bytes is not None
+    alert = DetectorSchema.deserialize(alert_bytes)
+    assert alert.description == "Unknown value: '/foobar'"
+    assert "Global - URL" in alert.alertsObtain
+    assert alert.logIDs == ["L9"]
+    assert alert.score == 1.0
+
+
+def test_new_value_training_spans_batches():
+    det = NewValueDetector(NV_CONFIG)
+    assert det.process(_parsed("/a")) is None  # train 1
+    assert det.process(_parsed("/b")) is None  # train 2
+    assert det.process(_parsed("/c")) is not None  # detect: unseen
+
+
+def test_new_value_event_scoped_variables():
+    cfg = {
+        "data_use_training": 1,
+        "events": {
+            1: {"inst": {"variables": [{"pos": 0, "name": "var1"}]}}
+        },
+    }
+    det = NewValueDetector(cfg)
+    assert det.process(_parsed(variables=["ok"], event_id=1)) is None  # train
+    assert det.process(_parsed(variables=["ok"], event_id=1)) is None
+    assert det.process(_parsed(variables=["bad"], event_id=1)) is not None
+    # other events are not watched
+    assert det.process(_parsed(variables=["bad"], event_id=2)) is None
+
+
+def test_new_value_checkpoint_resume():
+    det = NewValueDetector(NV_CONFIG)
+    det.process_batch([_parsed("/a"), _parsed("/b")])
+    state = det.state_dict()
+
+    det2 = NewValueDetector(NV_CONFIG)
+    det2.load_state_dict(state)
+    assert det2.process(_parsed("/a")) is None
+    assert det2.process(_parsed("/zzz")) is not None
+
+
+def test_new_value_combo_detector():
+    cfg = {
+        "data_use_training": 1,
+        "global": {"g": {"header_variables": [{"pos": "URL"}],
+                          "variables": [{"pos": 0, "name": "v0"}]}},
+    }
+    det = NewValueComboDetector(cfg)
+    assert det.process(_parsed("/a", variables=["x"])) is None  # train
+    assert det.process(_parsed("/a", variables=["x"])) is None  # known combo
+    # known parts, new combination
+    out = det.process(_parsed("/a", variables=["y"]))
+    assert out is not None
+
+
+def test_dummy_detector_alternates():
+    det = DummyDetector()
+    results = [det.process(_parsed()) is not None for _ in range(4)]
+    assert results == [False, True, False, True]
+    alert = DetectorSchema.deserialize(det.process_batch([_parsed()] * 2)[1])
+    assert alert.description == "Dummy detection process"
+    assert alert.score == 1.0
+
+
+def test_random_detector_seeded():
+    cfg = {"params": {"seed": 42, "threshold": 0.5}}
+    a = RandomDetector(cfg)
+    b = RandomDetector(cfg)
+    frames = [_parsed(log_id=str(i)) for i in range(50)]
+    ra = [x is not None for x in a.process_batch(frames)]
+    rb = [x is not None for x in b.process_batch(frames)]
+    assert ra == rb
+    assert any(ra) and not all(ra)
+
+
+def test_random_detector_threshold_extremes():
+    never = RandomDetector({"params": {"seed": 1, "threshold": 1.1}})
+    assert all(x is None for x in never.process_batch([_parsed()] * 20))
+    always = RandomDetector({"params": {"seed": 1, "threshold": -0.1}})
+    assert all(x is not None for x in always.process_batch([_parsed()] * 20))

@@ -1,0 +1,122 @@
+"""MatcherParser tests: log_format extraction, template matching,
+normalization; synthetic audit-log end-to-end."""
+from detectmateservice_amd.library.parsers import DummyParser, MatcherParser
+from detectmateservice_amd.library.parsers.template_matcher import (
+    compile_log_format,
+    match_template,
+    split_template,
+)
+from detectmateservice_amd.schemas import LogSchema, ParserSchema
+from detectmateservice_amd.utils.synthetic import (
+    AUDIT_LOG_FORMAT,
+    AUDIT_TEMPLATES,
+    AuditLogGenerator,
+)
+
+
+def test_log_format_compile_and_match():
+    fmt = '<IP> - - [<Time>] "<Method> <URL> <Protocol>" <Status> <Bytes>'
+    rx = compile_log_format(fmt)
+    line = '1.2.3.4 - - [10/Oct/2000:13:55:36] "GET /index.html HTTP/1.0" 200 2326'
+    m = rx.match(line)
+    assert m
+    assert m.group("IP") == "1.2.3.4"
+    assert m.group("Method") == "GET"
+    assert m.group("URL") == "/index.html"
+    assert m.group("Status") == "200"
+    assert m.group("Bytes") == "2326"
+
+
+def test_template_match_basic():
+    segs = split_template("pid=<*> uid=<*> res=<*>")
+    caps = match_template("pid=123 uid=0 res=success", segs)
+    assert caps == ["123", "0", "success"]
+
+
+def test_template_match_anchoring():
+    segs = split_template("pid=<*> uid=<*>")
+    # must start with "pid=" and end after uid capture
+    assert match_template("xpid=1 uid=2", segs) is None
+    assert match_template("pid=1 uid=2", segs) == ["1", "2"]
+
+
+def test_template_no_match():
+    segs = split_template("completely=<*> different=<*>")
+    assert match_template("pid=1 uid=2", segs) is None
+
+
+def test_template_trailing_literal():
+    segs = split_template("a=<*> end")
+    assert match_template("a=1 end", segs) == ["1"]
+    assert match_template("a=1 endx", segs) is None
+
+
+def test_matcher_parser_on_synthetic_audit():
+    parser = MatcherParser(
+        {
+            "log_format": AUDIT_LOG_FORMAT,
+            "templates": AUDIT_TEMPLATES,
+        }
+    )
+    gen = AuditLogGenerator(seed=7)
+    n_match = 0
+    for _ in range(200):
+        line, _anom, tidx = gen.line()
+        frame = LogSchema(logID="x", log=line).serialize()
+        out = parser.process(frame)
+        parsed = ParserSchema.deserialize(out)
+        assert parsed.logFormatVariables.get("Type")
+        assert parsed.logFormatVariables.get("Content")
+        if parsed.EventID > 0:
+            n_match += 1
+            assert parsed.template == AUDIT_TEMPLATES[parsed.EventID - 1]
+            assert parsed.variables  # wildcards captured
+    assert n_match > 150  # most generated lines match their template
+
+
+def test_matcher_parser_batch_equals_single():
+    parser = MatcherParser({"templates": AUDIT_TEMPLATES})
+    gen = AuditLogGenerator(seed=11)
+    frames = [LogSchema(logID=str(i), log=gen.line()[0]).serialize() for i in range(20)]
+    singles = [parser.process(f) for f in frames]
+    batched = parser.process_batch(frames)
+    for s, b in zip(singles, batched):
+        ps, pb = ParserSchema.deserialize(s), ParserSchema.deserialize(b)
+        assert ps.EventID == pb.EventID
+        assert ps.variables == pb.variables
+
+
+def test_normalization_params():
+    parser = MatcherParser(
+        {
+            "templates": ["hello <*> world"],
+            "params": {"lowercase": True},
+        }
+    )
+    frame = LogSchema(log="HELLO Big WORLD").serialize()
+    parsed = ParserSchema.deserialize(parser.process(frame))
+    assert parsed.EventID == 1
+    assert parsed.variables == ["big"]
+
+
+def test_templates_from_file(tmp_path):
+    tf = tmp_path / "templates.txt"
+    tf.write_text("alpha=<*>\nbeta=<*> gamma=<*>\n")
+    parser = MatcherParser({"params": {"path_templates": str(tf)}})
+    parsed = ParserSchema.deserialize(
+        parser.process(LogSchema(log="beta=1 gamma=2").serialize())
+    )
+    assert parsed.EventID == 2
+    assert parsed.variables == ["1", "2"]
+
+
+def test_dummy_parser_contract():
+    p = DummyParser()
+    parsed = ParserSchema.deserialize(
+        p.process(LogSchema(logID="id9", log="whatever").serialize())
+    )
+    assert parsed.EventID == 2
+    assert parsed.template == "This is a dummy template"
+    assert parsed.variables == ["dummy_variable"]
+    assert parsed.log == "DummyParser"
+    assert parsed.logID == "id9"

@@ -1,0 +1,163 @@
+"""Transport tests: pair semantics, timeout, reconnect, late binding.
+
+Mirrors the reference's engine-level socket coverage
+(tests/test_engine_multi_output.py in /root/reference): real sockets over
+ipc://{tmp_path}, downstream-absent startup, late binding, 1 MiB frames.
+"""
+import threading
+import time
+
+import pytest
+
+from detectmateservice_amd.engine.sockets import (
+    InprocListener,
+    PairDialer,
+    PairListener,
+    PairSocketFactory,
+    RecvTimeout,
+)
+
+
+def test_ipc_roundtrip(ipc_addr):
+    listener = PairListener(ipc_addr)
+    dialer = PairDialer(ipc_addr)
+    try:
+        assert dialer.wait_connected(5.0)
+        assert dialer.send(b"hello")
+        assert listener.recv(timeout_ms=2000) == b"hello"
+        # reply path (request/reply mode)
+        assert listener.send(b"world")
+        assert dialer.recv(timeout_ms=2000) == b"world"
+    finally:
+        dialer.close()
+        listener.close()
+
+
+def test_recv_timeout(ipc_addr):
+    listener = PairListener(ipc_addr)
+    try:
+        with pytest.raises(RecvTimeout):
+            listener.recv(timeout_ms=50)
+    finally:
+        listener.close()
+
+
+def test_tcp_roundtrip():
+    listener = PairListener("tcp://127.0.0.1:0")
+    port = listener.bound_port
+    dialer = PairDialer(f"tcp://127.0.0.1:{port}")
+    try:
+        assert dialer.wait_connected(5.0)
+        dialer.send(b"x" * 100)
+        assert listener.recv(timeout_ms=2000) == b"x" * 100
+    finally:
+        dialer.close()
+        listener.close()
+
+
+def test_large_frame(ipc_addr):
+    """1 MiB messages (reference test_engine_multi_output.py:430)."""
+    listener = PairListener(ipc_addr)
+    dialer = PairDialer(ipc_addr)
+    big = b"A" * (1024 * 1024)
+    try:
+        assert dialer.wait_connected(5.0)
+        dialer.send(big)
+        assert listener.recv(timeout_ms=5000) == big
+    finally:
+        dialer.close()
+        listener.close()
+
+
+def test_late_binding(ipc_addr):
+    """Dialer starts before the listener exists; frames are buffered and
+    delivered once the peer appears (reference engine.py:173-179,
+    test_engine_multi_output.py:391-409)."""
+    dialer = PairDialer(ipc_addr, buffer_size=16)
+    try:
+        assert dialer.send(b"early", block=False)
+        time.sleep(0.3)  # dialer is retrying in the background
+        listener = PairListener(ipc_addr)
+        try:
+            assert listener.recv(timeout_ms=5000) == b"early"
+        finally:
+            listener.close()
+    finally:
+        dialer.close()
+
+
+def test_send_returns_false_when_buffer_full(ipc_addr):
+    """No peer + full buffer → non-blocking send fails (drop policy input)."""
+    dialer = PairDialer(ipc_addr, buffer_size=2)
+    try:
+        results = [dialer.send(b"f%d" % i, block=False) for i in range(10)]
+        assert not all(results)
+    finally:
+        dialer.close()
+
+
+def test_reconnect_after_listener_death(ipc_addr):
+    listener = PairListener(ipc_addr)
+    dialer = PairDialer(ipc_addr)
+    try:
+        assert dialer.wait_connected(5.0)
+        dialer.send(b"one")
+        assert listener.recv(timeout_ms=2000) == b"one"
+        listener.close()
+        time.sleep(0.3)
+        listener2 = PairListener(ipc_addr)
+        try:
+            deadline = time.monotonic() + 10
+            got = None
+            while time.monotonic() < deadline:
+                dialer.send(b"two", block=False)
+                try:
+                    got = listener2.recv(timeout_ms=300)
+                    break
+                except RecvTimeout:
+                    continue
+            assert got == b"two"
+        finally:
+            listener2.close()
+    finally:
+        dialer.close()
+
+
+def test_inproc_pair():
+    addr = "inproc://test-pair-1"
+    listener = InprocListener(addr)
+    dialer = PairDialer(addr)
+    try:
+        assert dialer.wait_connected(5.0)
+        dialer.send(b"ping")
+        assert listener.recv(timeout_ms=2000) == b"ping"
+        listener.send(b"pong")
+        assert dialer.recv(timeout_ms=2000) == b"pong"
+    finally:
+        dialer.close()
+        listener.close()
+
+
+def test_recv_many_batches(ipc_addr):
+    listener = PairListener(ipc_addr)
+    dialer = PairDialer(ipc_addr)
+    try:
+        assert dialer.wait_connected(5.0)
+        for i in range(20):
+            dialer.send(b"m%02d" % i)
+        got = []
+        deadline = time.monotonic() + 5
+        while len(got) < 20 and time.monotonic() < deadline:
+            got.extend(listener.recv_many(max_frames=64, timeout_ms=200, linger_ms=50))
+        assert got == [b"m%02d" % i for i in range(20)]
+    finally:
+        dialer.close()
+        listener.close()
+
+
+def test_factory_scheme_validation():
+    factory = PairSocketFactory()
+    with pytest.raises(ValueError):
+        factory.create("bogus://whatever")
+    with pytest.raises(ValueError):
+        factory.create("tcp://127.0.0.1")  # missing port
