@@ -1,0 +1,67 @@
+// Common helpers for detectmate-mi355x HIP kernels (gfx950 / CDNA4 only).
+//
+// Design references: /opt/skills/guides/cdna_hip_programming.md (wave64,
+// MFMA intrinsics, LDS banking, glds staging) — written for MI355X from
+// scratch; no CUDA compatibility paths.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <cstdint>
+
+#define DMX_WAVE 64
+
+#define DMX_HIP_CHECK(expr)                                                    \
+  do {                                                                         \
+    hipError_t _e = (expr);                                                    \
+    if (_e != hipSuccess) {                                                    \
+      printf("HIP error %s at %s:%d\n", hipGetErrorString(_e), __FILE__,       \
+             __LINE__);                                                        \
+      abort();                                                                 \
+    }                                                                          \
+  } while (0)
+
+// Vector types for wide loads (Guideline 13: always vectorize bf16 as
+// short4/short8 reinterpret — 8-16 B/lane).
+typedef __attribute__((ext_vector_type(4))) short short4v;
+typedef __attribute__((ext_vector_type(8))) short short8v;
+typedef __attribute__((ext_vector_type(2))) float float2v;
+typedef __attribute__((ext_vector_type(4))) float float4v;
+typedef __attribute__((ext_vector_type(4))) unsigned int uint4v;
+
+// MFMA fragment types for mfma_f32_16x16x32_bf16 (guide §3: 8 bf16 input
+// elements per lane = 4 VGPRs; 4 fp32 accumulator elements per lane).
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+static __device__ __forceinline__ float bf16_to_f32(short u) {
+  union { float f; unsigned int i; } v;
+  v.i = ((unsigned int)(unsigned short)u) << 16;
+  return v.f;
+}
+
+static __device__ __forceinline__ short f32_to_bf16(float f) {
+  union { float f; unsigned int i; } v;
+  v.f = f;
+  // round-to-nearest-even
+  unsigned int lsb = (v.i >> 16) & 1;
+  v.i += 0x7fff + lsb;
+  return (short)(v.i >> 16);
+}
+
+// GELU (erf form, matches torch.nn.functional.gelu default)
+static __device__ __forceinline__ float gelu_f32(float x) {
+  return 0.5f * x * (1.0f + erff(x * 0.70710678118654752440f));
+}
+
+static __device__ __forceinline__ float warp_reduce_sum_f32(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+  return v;
+}
+
+static __device__ __forceinline__ float warp_reduce_max_f32(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_down(v, off, 64));
+  return v;
+}

@@ -1,0 +1,146 @@
+"""GpuPipeline: fused reader→parser→detector on one device.
+
+The MI355X-native fast path: when pipeline stages are
+co-located, frames never round-trip through protobuf between stages —
+lines live in a [B, max_len] u8 SoA tensor on the GPU and flow
+parser-kernel → hash-probe / transformer-scoring as device tensors on one
+HIP stream. The socket Service path (core.py) remains for distributed /
+edge deployment; this class is what bench.py drives and what a
+TransformerDetector service uses internally per batch.
+
+Stage structure per batch (all on `device`):
+  1. template_match kernel: log_format header split + template match
+     (event_id + capture spans)
+  2. NewValue watch: hash watched spans (kernel) -> probe GPU hash sets
+     (training batches insert instead)
+  3. Transformer scorer (BERT-tiny bf16 MFMA) over content-span byte tokens
+Anomaly = NewValue unseen-value hit OR transformer score > threshold.
+"""
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Sequence
+
+import torch
+
+from . import ops
+from .models.bert_tiny import BertTinyConfig, BertTinyDetectorModel
+
+
+@dataclass
+class PipelineConfig:
+    templates: Sequence[str] = ()
+    log_format: Optional[str] = None
+    lowercase: bool = False
+    max_len: int = 512
+    #: watched fields: list of dicts {kind: "variable"|"header", pos: int,
+    #: event: int (-1 = any)} — mirrors NewValueDetector specs
+    watches: Sequence[dict] = ()
+    hashset_capacity: int = 1 << 16
+    #: transformer detector on/off + threshold
+    use_transformer: bool = True
+    score_threshold: float = 3.0
+    train_lines: int = 0  # NewValue training-phase length (data_use_training)
+    bert: BertTinyConfig = field(default_factory=BertTinyConfig)
+    seed: int = 1234
+
+
+class GpuPipeline:
+    def __init__(self, config: PipelineConfig, device: str | torch.device = "cpu") -> None:
+        self.config = config
+        self.device = torch.device(device)
+        self.matcher = ops.TemplateMatcher(
+            config.templates,
+            log_format=config.log_format,
+            lowercase=config.lowercase,
+            device=self.device,
+            max_len=config.max_len,
+        )
+        specs = []
+        for w in config.watches:
+            kind = 0 if w.get("kind", "variable") == "variable" else 1
+            specs.append([kind, int(w.get("event", -1)), int(w["pos"]), 0])
+        self.specs = (
+            torch.tensor(specs, dtype=torch.int32, device=self.device)
+            if specs
+            else torch.zeros((0, 4), dtype=torch.int32, device=self.device)
+        )
+        self.hashsets = (
+            ops.GpuHashSets(len(specs), config.hashset_capacity, device=self.device)
+            if specs
+            else None
+        )
+        self.model = (
+            BertTinyDetectorModel(config.bert, device=self.device, seed=config.seed)
+            if config.use_transformer
+            else None
+        )
+        self.seen_lines = 0
+
+    # ------------------------------------------------------------------
+    def process_packed(
+        self, lines: torch.Tensor, line_len: torch.Tensor
+    ) -> Dict[str, torch.Tensor]:
+        """lines [B, max_len] u8 + lengths on self.device → result tensors.
+
+        Returns event_id [B], anomaly [B] (bool), scores [B] (f32, 0 when
+        transformer off), nv_unseen [B, W]."""
+        B = lines.shape[0]
+        match = self.matcher.match_packed(lines, line_len)
+
+        n_train_left = max(0, self.config.train_lines - self.seen_lines)
+        train_upto = min(B, n_train_left)
+        self.seen_lines += B
+
+        nv_unseen = None
+        if self.hashsets is not None and self.specs.shape[0] > 0:
+            hashes = ops.watch_hashes(lines, match, self.specs, self.config.lowercase)
+            if train_upto > 0:
+                self.hashsets.insert(hashes[:train_upto])
+            if train_upto < B:
+                unseen_tail = self.hashsets.probe(hashes[train_upto:])
+                nv_unseen = torch.zeros(
+                    (B, self.specs.shape[0]), dtype=torch.int32, device=lines.device
+                )
+                nv_unseen[train_upto:] = unseen_tail
+            else:
+                nv_unseen = torch.zeros(
+                    (B, self.specs.shape[0]), dtype=torch.int32, device=lines.device
+                )
+
+        scores = torch.zeros(B, dtype=torch.float32, device=lines.device)
+        if self.model is not None:
+            # content span: last fmt capture when the header matched,
+            # else the whole line
+            nfc = match["n_fmt_caps"].long()
+            has_hdr = nfc > 0
+            last = (nfc - 1).clamp(min=0)
+            fc = match["fmt_caps"]
+            start = torch.where(
+                has_hdr,
+                fc.gather(1, last.view(-1, 1, 1).expand(-1, 1, 2))[:, 0, 0].long(),
+                torch.zeros_like(nfc),
+            )
+            end = torch.where(
+                has_hdr,
+                fc.gather(1, last.view(-1, 1, 1).expand(-1, 1, 2))[:, 0, 1].long(),
+                line_len.long(),
+            )
+            tokens = self.model.tokenize_spans(lines, start.int(), end.int())
+            scores = self.model(tokens)
+
+        anomaly = scores > self.config.score_threshold
+        if nv_unseen is not None:
+            anomaly = anomaly | (nv_unseen.sum(dim=1) > 0)
+        return {
+            "event_id": match["event_id"],
+            "anomaly": anomaly,
+            "scores": scores,
+            "nv_unseen": nv_unseen,
+            "match": match,
+        }
+
+    def process_lines(self, raw_lines: Sequence[bytes]) -> Dict[str, torch.Tensor]:
+        lines, lens = ops.pack_lines(raw_lines, self.config.max_len, device=self.device)
+        return self.process_packed(lines, lens)

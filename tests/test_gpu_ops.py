@@ -1,0 +1,217 @@
+"""GPU kernel numerics tests: every HIP kernel vs a plain PyTorch fp32
+reference of the same op (run on an MI355X with `-m gpu`)."""
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(
+    not torch.cuda.is_available(), reason="needs MI355X"
+)
+
+
+@pytest.fixture(scope="module", autouse=True)
+def _require_ext():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from detectmateservice_amd import ops
+
+    # On a GPU box the HIP extension MUST be present (no silent fallback).
+    assert ops.have_extension(), "HIP extension _dmx_C not built/loadable"
+    yield
+
+
+def test_mfma_layout_probe():
+    """Verify the assumed mfma_f32_16x16x32_bf16 fragment layout (guide §3:
+    layout tables unavailable offline — empirically pinned here).
+    Asymmetric operands so transposes cannot pass (guide §5.4 rule 16)."""
+    from detectmateservice_amd.ops import _dmx_C
+
+    torch.manual_seed(0)
+    A = (torch.randn(16, 32) * 0.5).bfloat16().cuda()
+    B = (torch.arange(32 * 16).reshape(32, 16).float() * 0.01 +
+         torch.randn(32, 16)).bfloat16().cuda()
+    ref = (A.float() @ B.float()).cpu()
+
+    results = {}
+    for a_l in (0, 1):
+        for b_l in (0, 1):
+            D = _dmx_C.probe_mfma(A, B, a_l, b_l).cpu()
+            results[(a_l, b_l)] = float((D - ref).abs().max())
+    best = min(results, key=results.get)
+    assert results[best] < 0.15, f"no layout matches: {results}"
+    assert best == (0, 0), (
+        f"kernel assumes contiguous-octet layout (0,0) but hardware wants "
+        f"{best}; errors: {results} — fix read_frag in gemm_bf16.hip"
+    )
+
+
+@pytest.mark.parametrize("M,N,K", [(128, 128, 128), (256, 384, 128),
+                                   (512, 128, 512), (200, 112, 128)])
+def test_fused_linear_vs_torch(M, N, K):
+    from detectmateservice_amd import ops
+
+    torch.manual_seed(1)
+    x = (torch.randn(M, K) * 0.5).bfloat16().cuda()
+    wt = (torch.randn(N, K) * 0.5).bfloat16().cuda()
+    bias = torch.randn(N).cuda()
+    y = ops.fused_linear(x, wt, bias, activation="none").float().cpu()
+    ref = torch.nn.functional.linear(x.float().cpu(), wt.float().cpu(), bias.cpu())
+    err = (y - ref).abs().max() / (ref.abs().max() + 1e-6)
+    assert err < 0.02, f"rel err {err}"
+
+
+def test_fused_linear_gelu():
+    from detectmateservice_amd import ops
+
+    torch.manual_seed(2)
+    x = (torch.randn(256, 128) * 0.5).bfloat16().cuda()
+    wt = (torch.randn(512, 128) * 0.5).bfloat16().cuda()
+    bias = torch.randn(512).cuda()
+    y = ops.fused_linear(x, wt, bias, activation="gelu").float().cpu()
+    ref = torch.nn.functional.gelu(
+        torch.nn.functional.linear(x.float().cpu(), wt.float().cpu(), bias.cpu())
+    )
+    err = (y - ref).abs().max() / (ref.abs().max() + 1e-6)
+    assert err < 0.02, f"rel err {err}"
+
+
+def test_layernorm_vs_torch():
+    from detectmateservice_amd import ops
+
+    torch.manual_seed(3)
+    for D in (128, 512):
+        x = torch.randn(333, D).bfloat16().cuda()
+        r = torch.randn(333, D).bfloat16().cuda()
+        g = torch.randn(D).bfloat16().cuda()
+        b = torch.randn(D).bfloat16().cuda()
+        y, xres = ops.layernorm(x, g, b, residual=r, return_xres=True)
+        ref = torch.nn.functional.layer_norm(
+            x.float().cpu() + r.float().cpu(), (D,), g.float().cpu(), b.float().cpu()
+        )
+        assert (y.float().cpu() - ref).abs().max() < 0.05
+        assert (xres.float().cpu() - (x.float().cpu() + r.float().cpu())).abs().max() < 0.02
+
+
+@pytest.mark.parametrize("S", [64, 48, 128])
+def test_attention_vs_torch(S):
+    from detectmateservice_amd import ops
+
+    torch.manual_seed(4)
+    BH, Dh = 12, 64
+    q = (torch.randn(BH, S, Dh) * 0.5).bfloat16().cuda()
+    k = (torch.randn(BH, S, Dh) * 0.5).bfloat16().cuda()
+    v = (torch.randn(BH, S, Dh) * 0.5).bfloat16().cuda()
+    o = ops.attention(q, k, v).float().cpu()
+    scale = 1.0 / Dh ** 0.5
+    s = torch.softmax(q.float().cpu() @ k.float().cpu().transpose(-1, -2) * scale, -1)
+    ref = s @ v.float().cpu()
+    assert (o - ref).abs().max() < 0.03
+
+
+def test_template_match_gpu_vs_cpu():
+    from detectmateservice_amd import ops
+    from detectmateservice_amd.utils.synthetic import (
+        AUDIT_LOG_FORMAT,
+        AUDIT_TEMPLATES,
+        AuditLogGenerator,
+    )
+
+    gen = AuditLogGenerator(seed=17, anomaly_rate=0.1)
+    raw = [gen.line()[0].encode() for _ in range(512)]
+    cpu_m = ops.TemplateMatcher(AUDIT_TEMPLATES, log_format=AUDIT_LOG_FORMAT, device="cpu")
+    gpu_m = ops.TemplateMatcher(AUDIT_TEMPLATES, log_format=AUDIT_LOG_FORMAT, device="cuda")
+    lines_c, lens_c = ops.pack_lines(raw, 512, device="cpu")
+    lines_g, lens_g = lines_c.cuda(), lens_c.cuda()
+    mc = cpu_m.match_packed(lines_c, lens_c)
+    mg = {k: v.cpu() for k, v in gpu_m.match_packed(lines_g, lens_g).items()}
+    assert torch.equal(mc["event_id"], mg["event_id"])
+    assert torch.equal(mc["n_caps"], mg["n_caps"])
+    assert torch.equal(mc["n_fmt_caps"], mg["n_fmt_caps"])
+    # compare capture spans for matched lines
+    for i in range(len(raw)):
+        nc = int(mc["n_caps"][i])
+        assert torch.equal(mc["caps"][i, :nc], mg["caps"][i, :nc]), raw[i]
+        nf = int(mc["n_fmt_caps"][i])
+        assert torch.equal(mc["fmt_caps"][i, :nf], mg["fmt_caps"][i, :nf]), raw[i]
+
+
+def test_watch_hashes_and_hashsets_gpu_vs_cpu():
+    from detectmateservice_amd import ops
+    from detectmateservice_amd.utils.synthetic import (
+        AUDIT_LOG_FORMAT,
+        AUDIT_TEMPLATES,
+        AuditLogGenerator,
+    )
+
+    gen = AuditLogGenerator(seed=23, anomaly_rate=0.2)
+    raw = [gen.line()[0].encode() for _ in range(256)]
+    lines_c, lens_c = ops.pack_lines(raw, 512)
+    cpu_m = ops.TemplateMatcher(AUDIT_TEMPLATES, log_format=AUDIT_LOG_FORMAT, device="cpu")
+    gpu_m = ops.TemplateMatcher(AUDIT_TEMPLATES, log_format=AUDIT_LOG_FORMAT, device="cuda")
+    mc = cpu_m.match_packed(lines_c, lens_c)
+    lines_g = lines_c.cuda()
+    mg = gpu_m.match_packed(lines_g, lens_c.cuda())
+
+    specs_c = torch.tensor([[0, 1, 5, 0], [0, -1, 0, 0]], dtype=torch.int32)
+    hc = ops.watch_hashes(lines_c, mc, specs_c)
+    hg = ops.watch_hashes(lines_g, mg, specs_c.cuda()).cpu()
+    assert torch.equal(hc, hg)
+
+    # insert first half, probe second half; CPU and GPU sets must agree
+    cpu_s = ops.GpuHashSets(2, 1 << 12, device="cpu")
+    gpu_s = ops.GpuHashSets(2, 1 << 12, device="cuda")
+    cpu_s.insert(hc[:128])
+    gpu_s.insert(hg[:128].cuda())
+    pc = cpu_s.probe(hc[128:])
+    pg = gpu_s.probe(hg[128:].cuda()).cpu()
+    assert torch.equal(pc, pg)
+
+
+def test_bert_tiny_gpu_vs_cpu():
+    """Full model forward: GPU kernels vs CPU fp32 reference path."""
+    from detectmateservice_amd.models.bert_tiny import (
+        BertTinyConfig,
+        BertTinyDetectorModel,
+    )
+
+    cfg = BertTinyConfig()
+    cpu_model = BertTinyDetectorModel(cfg, device="cpu", seed=7)
+    gpu_model = BertTinyDetectorModel(cfg, device="cuda", seed=7)
+    tokens = torch.randint(0, 259, (64, 64))
+    sc = cpu_model(tokens)
+    sg = gpu_model(tokens.cuda()).cpu()
+    assert (sc - sg).abs().max() < 0.05, (sc - sg).abs().max()
+
+
+def test_pipeline_gpu_end_to_end():
+    from detectmateservice_amd.pipeline import GpuPipeline, PipelineConfig
+    from detectmateservice_amd.utils.synthetic import (
+        AUDIT_LOG_FORMAT,
+        AUDIT_TEMPLATES,
+        AuditLogGenerator,
+    )
+
+    cfg = PipelineConfig(
+        templates=AUDIT_TEMPLATES,
+        log_format=AUDIT_LOG_FORMAT,
+        watches=[{"kind": "variable", "pos": 5, "event": 1}],
+        train_lines=256,
+        use_transformer=True,
+        score_threshold=1e9,
+    )
+    pipe = GpuPipeline(cfg, device="cuda")
+    gen = AuditLogGenerator(seed=5)
+    train = [gen.line()[0].encode() for _ in range(256)]
+    pipe.process_lines(train)
+    normal = [gen.line()[0].encode() for _ in range(64)]
+    out = pipe.process_lines(normal)
+    assert int(out["anomaly"].sum()) == 0
+    bad = (
+        "type=USER_ACCT msg=audit(1642723741.072:999): pid=1 uid=0 auid=1 ses=1 "
+        "msg='op=PAM:accounting acct=\"intruder\" exe=/usr/sbin/cron hostname=? "
+        "addr=? terminal=cron res=success'"
+    ).encode()
+    out = pipe.process_lines(normal[:3] + [bad])
+    assert out["anomaly"].cpu().tolist() == [False, False, False, True]
