@@ -95,17 +95,13 @@ class BertTinyDetectorModel:
         for layer in self.layers:
             x2 = x.view(M, h)
             qkv = ops.fused_linear(x2, layer["wqkv_t"], layer["bqkv"])  # [M, 3h]
-            qkv = qkv.view(B, S, 3, c.heads, c.head_dim)
-            q = qkv[:, :, 0].permute(0, 2, 1, 3).reshape(B * c.heads, S, c.head_dim).contiguous()
-            k = qkv[:, :, 1].permute(0, 2, 1, 3).reshape(B * c.heads, S, c.head_dim).contiguous()
-            v = qkv[:, :, 2].permute(0, 2, 1, 3).reshape(B * c.heads, S, c.head_dim).contiguous()
-            attn = ops.attention(q, k, v)  # [B*H, S, Dh]
-            attn = (
-                attn.view(B, c.heads, S, c.head_dim)
-                .permute(0, 2, 1, 3)
-                .reshape(M, h)
-                .contiguous()
-            )
+            # Fused-QKV attention: the GPU kernel reads [B, S, 3*H*Dh]
+            # directly (no permutes). NOTE the qkv linear emits q|k|v
+            # CONCATENATED over N: row = [q(h0..hH), k(...), v(...)] which
+            # is exactly the 3*H*Dh layout the kernel indexes.
+            attn = ops.attention_qkv(
+                qkv.view(B, S, 3 * h), S, c.heads, c.head_dim
+            ).view(M, h)
             proj = ops.fused_linear(attn, layer["wo_t"], layer["bo"])
             x1 = ops.layernorm(proj, layer["ln1_g"], layer["ln1_b"], residual=x2)
             ffn = ops.fused_linear(x1, layer["w1_t"], layer["b1"], activation="gelu")

@@ -100,6 +100,30 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, scale: Optional
     return (s @ v.float()).to(q.dtype)
 
 
+def attention_qkv(qkv: torch.Tensor, S: int, H: int, Dh: int, scale: Optional[float] = None) -> torch.Tensor:
+    """MHA over the fused QKV projection output.
+
+    qkv [B, S, 3*H*Dh] (q|k|v interleaved per row as the qkv linear emits
+    them) -> O [B, S, H*Dh]. GPU: single MFMA kernel with no permutes
+    (ops/csrc/attention_mfma.hip) when S%32==0, S<=128, Dh in {32,64};
+    otherwise splits + falls back to :func:`attention`.
+    """
+    if scale is None:
+        scale = 1.0 / math.sqrt(Dh)
+    B = qkv.shape[0]
+    if qkv.is_cuda and S % 32 == 0 and S <= 128 and Dh in (32, 64):
+        return _require_ext().attention_qkv_bf16(qkv, S, H, Dh, scale)
+    # reference path: split + permute
+    q, k, v = qkv.view(B, S, 3, H, Dh).unbind(dim=2)
+    q = q.permute(0, 2, 1, 3).reshape(B * H, S, Dh).contiguous()
+    k = k.permute(0, 2, 1, 3).reshape(B * H, S, Dh).contiguous()
+    v = v.permute(0, 2, 1, 3).reshape(B * H, S, Dh).contiguous()
+    o = attention(q, k, v, scale)
+    return (
+        o.view(B, H, S, Dh).permute(0, 2, 1, 3).reshape(B, S, H * Dh).contiguous()
+    )
+
+
 # ---------------------------------------------------------------------------
 # parser: batched template matching
 # ---------------------------------------------------------------------------

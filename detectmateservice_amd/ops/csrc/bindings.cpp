@@ -19,6 +19,8 @@ void dmx_launch_layernorm_bf16(const void*, const void*, const void*,
                                hipStream_t);
 void dmx_launch_attention_bf16(const void*, const void*, const void*, void*,
                                int, int, int, float, hipStream_t);
+void dmx_launch_attention_mfma_bf16(const void*, void*, int, int, int, int,
+                                    float, hipStream_t);
 void dmx_launch_template_match(const void*, const void*, int, int,
                                const void*, const void*, int, const void*,
                                int, const void*, const void*, int, int,
@@ -123,6 +125,22 @@ torch::Tensor attention_bf16(torch::Tensor q, torch::Tensor k,
   return o;
 }
 
+torch::Tensor attention_qkv_bf16(torch::Tensor qkv, int64_t S, int64_t H,
+                                 int64_t Dh, double scale) {
+  TORCH_CHECK(qkv.is_cuda() && qkv.dtype() == torch::kBFloat16 &&
+                  qkv.is_contiguous(),
+              "qkv must be contiguous bf16 on GPU");
+  TORCH_CHECK(qkv.dim() == 3 && qkv.size(2) == 3 * H * Dh,
+              "qkv must be [B, S, 3*H*Dh]");
+  TORCH_CHECK(S % 32 == 0 && S <= 128, "MFMA attention needs S%32==0, S<=128");
+  TORCH_CHECK(Dh == 32 || Dh == 64, "MFMA attention needs Dh in {32,64}");
+  const auto B = qkv.size(0);
+  auto O = torch::empty({B, S, H * Dh}, qkv.options());
+  dmx_launch_attention_mfma_bf16(qkv.data_ptr(), O.data_ptr(), (int)B, (int)S,
+                                 (int)H, (int)Dh, (float)scale, cur_stream());
+  return O;
+}
+
 std::vector<torch::Tensor> template_match(
     torch::Tensor lines, torch::Tensor line_len, torch::Tensor fmt_bytes,
     torch::Tensor fmt_seg_off, torch::Tensor seg_bytes, torch::Tensor seg_off,
@@ -197,6 +215,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("probe_mfma", &probe_mfma, "MFMA 16x16x32 bf16 layout probe");
   m.def("layernorm_bf16", &layernorm_bf16, "fused residual+LayerNorm bf16");
   m.def("attention_bf16", &attention_bf16, "fused short-seq MHA bf16");
+  m.def("attention_qkv_bf16", &attention_qkv_bf16,
+        "MFMA MHA reading fused QKV layout [B,S,3*H*Dh] -> [B,S,H*Dh]");
   m.def("template_match", &template_match, "batched wildcard template match");
   m.def("watch_hashes", &watch_hashes, "hash watched capture spans");
   m.def("hashset_insert", &hashset_insert, "insert hashes into GPU sets");

@@ -21,6 +21,7 @@ sources = [
     str(CSRC / "gemm_bf16.hip"),
     str(CSRC / "layernorm.hip"),
     str(CSRC / "attention.hip"),
+    str(CSRC / "attention_mfma.hip"),
     str(CSRC / "template_match.hip"),
     str(CSRC / "hashset.hip"),
 ]

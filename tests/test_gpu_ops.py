@@ -215,3 +215,36 @@ def test_pipeline_gpu_end_to_end():
     ).encode()
     out = pipe.process_lines(normal[:3] + [bad])
     assert out["anomaly"].cpu().tolist() == [False, False, False, True]
+
+
+def test_attention_qkv_mfma_vs_reference():
+    """MFMA fused-QKV attention vs the permute+torch reference path."""
+    from detectmateservice_amd import ops
+
+    torch.manual_seed(5)
+    B, S, H, Dh = 32, 64, 2, 64
+    qkv = (torch.randn(B, S, 3 * H * Dh) * 0.5).bfloat16().cuda()
+    o = ops.attention_qkv(qkv, S, H, Dh).float().cpu()
+    # reference on CPU fp32
+    q, k, v = qkv.float().cpu().view(B, S, 3, H, Dh).unbind(dim=2)
+    q = q.permute(0, 2, 1, 3)
+    k = k.permute(0, 2, 1, 3)
+    v = v.permute(0, 2, 1, 3)
+    s = torch.softmax(q @ k.transpose(-1, -2) / Dh ** 0.5, -1)
+    ref = (s @ v).permute(0, 2, 1, 3).reshape(B, S, H * Dh)
+    assert (o - ref).abs().max() < 0.03, (o - ref).abs().max()
+
+
+def test_attention_qkv_mfma_s32():
+    from detectmateservice_amd import ops
+
+    torch.manual_seed(6)
+    B, S, H, Dh = 16, 32, 4, 32
+    qkv = (torch.randn(B, S, 3 * H * Dh) * 0.5).bfloat16().cuda()
+    o = ops.attention_qkv(qkv, S, H, Dh).float().cpu()
+    q, k, v = qkv.float().cpu().view(B, S, 3, H, Dh).unbind(dim=2)
+    s = torch.softmax(
+        q.permute(0, 2, 1, 3) @ k.permute(0, 2, 1, 3).transpose(-1, -2) / Dh ** 0.5, -1
+    )
+    ref = (s @ v.permute(0, 2, 1, 3)).permute(0, 2, 1, 3).reshape(B, S, H * Dh)
+    assert (o - ref).abs().max() < 0.03
