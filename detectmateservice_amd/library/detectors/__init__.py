@@ -15,3 +15,6 @@ __all__ = [
     "DummyDetector",
     "DummyDetectorConfig",
 ]
+from .transformer import TransformerDetector, TransformerDetectorConfig
+
+__all__ += ["TransformerDetector", "TransformerDetectorConfig"]

@@ -169,9 +169,22 @@ class MatcherParser(CoreComponent):
         return header, -1, "", []
 
     def process(self, data: bytes) -> Optional[bytes]:
-        return self.process_batch([data])[0]
+        return self._process_python([data])[0]
 
     def process_batch(self, frames: List[bytes]) -> List[Optional[bytes]]:
+        """Batched path: C++ codec (decode N frames → SoA) + template-match
+        kernel (GPU when available, CPU span matcher otherwise) + C++ batch
+        encode. Falls back to the per-message Python path when the
+        extension is absent or normalization params the kernel does not
+        implement (remove_spaces / remove_punctuation) are set."""
+        if len(frames) >= 8 and self._remove_spaces is False and self._remove_punct is False:
+            try:
+                return self._process_batched(frames)
+            except RuntimeError:
+                pass  # extension missing: python fallback
+        return self._process_python(frames)
+
+    def _process_python(self, frames: List[bytes]) -> List[Optional[bytes]]:
         now = int(time.time())
         out: List[Optional[bytes]] = []
         for raw in frames:
@@ -191,3 +204,57 @@ class MatcherParser(CoreComponent):
             )
             out.append(parsed.serialize())
         return out
+
+    # -- batched SoA path ----------------------------------------------
+    _BATCH_MAX_LEN = 512
+
+    def _get_batch_matcher(self):
+        if getattr(self, "_batch_matcher", None) is None:
+            import torch
+
+            from ... import ops
+
+            if not ops.have_extension():
+                raise RuntimeError("no extension")
+            device = (self.config.params or {}).get("device")
+            if device is None:
+                device = "cuda" if torch.cuda.is_available() else "cpu"
+            self._batch_device = device
+            self._batch_matcher = ops.TemplateMatcher(
+                self.templates,
+                log_format=self.config.log_format,
+                lowercase=self._lowercase,
+                device=device,
+                max_len=self._BATCH_MAX_LEN,
+            )
+        return self._batch_matcher
+
+    def _process_batched(self, frames: List[bytes]) -> List[Optional[bytes]]:
+        from ... import ops
+        from ...ops import _dmx_C  # type: ignore[attr-defined]
+
+        matcher = self._get_batch_matcher()
+        lines, lens, log_ids, _src, _host = _dmx_C.decode_log_batch(
+            list(frames), self._BATCH_MAX_LEN
+        )
+        dev_lines = lines.to(self._batch_device)
+        dev_lens = lens.to(self._batch_device)
+        match = matcher.match_packed(dev_lines, dev_lens)
+        match_cpu = {k: v.cpu() for k, v in match.items()}
+        encoded = _dmx_C.encode_parser_batch(
+            lines,
+            lens,
+            match_cpu["event_id"],
+            match_cpu["caps"],
+            match_cpu["n_caps"],
+            match_cpu["fmt_caps"],
+            match_cpu["n_fmt_caps"],
+            list(matcher.fmt_field_names),
+            list(self.templates),
+            list(log_ids),
+            "matcher_parser",
+            self.parser_id,
+            int(time.time()),
+            "0.3",
+        )
+        return list(encoded)

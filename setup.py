@@ -18,6 +18,7 @@ CSRC = ROOT / "detectmateservice_amd" / "ops" / "csrc"
 
 sources = [
     str(CSRC / "bindings.cpp"),
+    str(CSRC / "codec.cpp"),
     str(CSRC / "gemm_bf16.hip"),
     str(CSRC / "layernorm.hip"),
     str(CSRC / "attention.hip"),

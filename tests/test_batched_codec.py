@@ -1,0 +1,109 @@
+"""Batched C++ codec + batched MatcherParser path vs the per-message
+Python path (both must produce semantically identical ParserSchema)."""
+import pytest
+import torch
+
+from detectmateservice_amd import ops
+from detectmateservice_amd.library.parsers import MatcherParser
+from detectmateservice_amd.schemas import LogSchema, ParserSchema
+from detectmateservice_amd.utils.synthetic import (
+    AUDIT_LOG_FORMAT,
+    AUDIT_TEMPLATES,
+    AuditLogGenerator,
+)
+
+needs_ext = pytest.mark.skipif(
+    not ops.have_extension(), reason="C++ extension not built"
+)
+
+
+@needs_ext
+def test_decode_log_batch():
+    from detectmateservice_amd.ops import _dmx_C
+
+    frames = [
+        LogSchema(logID=f"id{i}", log=f"line number {i}", logSource="s",
+                  hostname="h").serialize()
+        for i in range(5)
+    ]
+    lines, lens, ids, sources, hosts = _dmx_C.decode_log_batch(frames, 64)
+    assert lines.shape == (5, 64)
+    for i in range(5):
+        text = bytes(lines[i, : int(lens[i])].numpy().tobytes()).decode()
+        assert text == f"line number {i}"
+        assert bytes(ids[i]) == f"id{i}".encode()
+
+
+@needs_ext
+def test_batched_parser_equals_python_parser():
+    gen = AuditLogGenerator(seed=31)
+    frames = [
+        LogSchema(logID=f"L{i}", log=gen.line()[0]).serialize() for i in range(64)
+    ]
+    parser = MatcherParser(
+        {"log_format": AUDIT_LOG_FORMAT, "templates": list(AUDIT_TEMPLATES)}
+    )
+    batched = parser.process_batch(frames)      # C++ codec path (>=8 frames)
+    python = parser._process_python(frames)
+
+    for i, (b, p) in enumerate(zip(batched, python)):
+        pb = ParserSchema.deserialize(b)
+        pp = ParserSchema.deserialize(p)
+        assert pb.EventID == pp.EventID, i
+        assert pb.variables == pp.variables, i
+        assert pb.logID == pp.logID
+        assert pb.log == pp.log
+        assert pb.template == pp.template
+        assert pb.logFormatVariables == pp.logFormatVariables, i
+        assert pb.parserType == pp.parserType
+
+
+@needs_ext
+def test_batched_parser_negative_event_id():
+    parser = MatcherParser({"templates": ["nomatch=<*>"]})
+    frames = [LogSchema(logID=str(i), log="zzz unmatched").serialize() for i in range(10)]
+    out = parser.process_batch(frames)
+    for b in out:
+        pb = ParserSchema.deserialize(b)
+        assert pb.EventID == -1
+        assert pb.template == ""
+        assert pb.variables == []
+
+
+def test_transformer_detector_cpu():
+    from detectmateservice_amd.library.detectors import TransformerDetector
+
+    det = TransformerDetector(
+        {"data_use_training": 32, "z_threshold": 4.0, "device": "cpu"}
+    )
+    gen = AuditLogGenerator(seed=41)
+
+    def frame(line, lid):
+        return ParserSchema(logID=lid, log=line, EventID=1).serialize()
+
+    train = [frame(gen.line()[0], f"t{i}") for i in range(32)]
+    out = det.process_batch(train)
+    assert all(o is None for o in out)
+    assert det.threshold is not None
+
+    normal = [frame(gen.line()[0], f"n{i}") for i in range(16)]
+    out = det.process_batch(normal)
+    # calibrated at 4 sigma: normal traffic should mostly pass
+    assert sum(o is not None for o in out) <= 2
+
+    # wildly different line content should push the score distribution
+    state = det.state_dict()
+    det2 = TransformerDetector(
+        {"data_use_training": 0, "z_threshold": 4.0, "device": "cpu"}
+    )
+    det2.load_state_dict(state)
+    out2 = det2.process_batch(normal)
+    assert [o is None for o in out] == [o is None for o in out2]
+
+
+def test_transformer_detector_resolvable():
+    from detectmateservice_amd.components.resolver import ComponentResolver
+
+    path, cfg = ComponentResolver().resolve("TransformerDetector")
+    assert path.endswith("TransformerDetector")
+    assert cfg.endswith("TransformerDetectorConfig")
