@@ -38,6 +38,16 @@ class DetectMateClient:
     def shutdown(self) -> dict:
         return requests.post(f"{self.url}/admin/shutdown", timeout=TIMEOUT_S).json()
 
+    def checkpoint(self, path: str) -> dict:
+        return requests.post(
+            f"{self.url}/admin/checkpoint", json={"path": path}, timeout=TIMEOUT_S
+        ).json()
+
+    def restore(self, path: str) -> dict:
+        return requests.post(
+            f"{self.url}/admin/restore", json={"path": path}, timeout=TIMEOUT_S
+        ).json()
+
     def reconfigure(self, config_file: str, persist: bool = False) -> dict:
         with open(config_file, "r", encoding="utf-8") as fh:
             config = yaml.safe_load(fh) or {}
@@ -59,6 +69,9 @@ def main(argv=None) -> int:
     rec = sub.add_parser("reconfigure")
     rec.add_argument("config_file")
     rec.add_argument("--persist", action="store_true")
+    for cmd in ("checkpoint", "restore"):
+        cp = sub.add_parser(cmd)
+        cp.add_argument("path")
     args = parser.parse_args(argv)
 
     client = DetectMateClient(args.url)
@@ -67,6 +80,8 @@ def main(argv=None) -> int:
             print(client.metrics())
         elif args.command == "reconfigure":
             print(json.dumps(client.reconfigure(args.config_file, args.persist), indent=2))
+        elif args.command in ("checkpoint", "restore"):
+            print(json.dumps(getattr(client, args.command)(args.path), indent=2))
         else:
             print(json.dumps(getattr(client, args.command)(), indent=2))
     except requests.RequestException as exc:

@@ -21,6 +21,7 @@ from __future__ import annotations
 import logging
 import threading
 import time
+from pathlib import Path
 from typing import Any, Dict, List, Optional
 
 from .components.base import CoreComponent
@@ -167,6 +168,49 @@ class Service:
 
     def shutdown(self) -> None:
         self._service_exit_event.set()
+
+    # ------------------------------------------------------------------
+    # checkpoint / resume (the reference has none; this
+    # framework's detector state — GPU hash sets, transformer weights,
+    # calibration — must survive restarts)
+    # ------------------------------------------------------------------
+    def checkpoint(self, path: str | Path) -> Dict[str, Any]:
+        import torch
+
+        path = Path(path)
+        path.parent.mkdir(parents=True, exist_ok=True)
+        state = {
+            "version": __version__,
+            "component_type": self.settings.component_type,
+            "component_id": self.settings.component_id,
+            "component_state": (
+                self.library_component.state_dict()
+                if self.library_component is not None
+                else {}
+            ),
+            "config": self.config_manager.get(),
+            "timestamp": time.time(),
+        }
+        tmp = path.with_suffix(path.suffix + ".tmp")
+        torch.save(state, tmp)
+        tmp.rename(path)  # atomic publish
+        self.logger.info("checkpoint written to %s", path)
+        return {"path": str(path), "timestamp": state["timestamp"]}
+
+    def restore(self, path: str | Path) -> Dict[str, Any]:
+        import torch
+
+        path = Path(path)
+        state = torch.load(path, map_location="cpu", weights_only=False)
+        if state.get("component_type") != self.settings.component_type:
+            raise ValueError(
+                f"checkpoint is for component_type={state.get('component_type')!r}, "
+                f"this service is {self.settings.component_type!r}"
+            )
+        if self.library_component is not None and state.get("component_state"):
+            self.library_component.load_state_dict(state["component_state"])
+        self.logger.info("restored component state from %s", path)
+        return {"path": str(path), "timestamp": state.get("timestamp")}
 
     # ------------------------------------------------------------------
     def _create_status_report(self, running: bool) -> Dict[str, Any]:

@@ -51,3 +51,17 @@ def reconfigure(payload: ReconfigPayload, service=Depends(get_service)) -> Dict[
 def shutdown(service=Depends(get_service)) -> Dict[str, Any]:
     service.shutdown()
     return {"status": "shutting down"}
+
+
+class CheckpointPayload(BaseModel):
+    path: str
+
+
+@router.post("/checkpoint")
+def checkpoint(payload: CheckpointPayload, service=Depends(get_service)) -> Dict[str, Any]:
+    return service.checkpoint(payload.path)
+
+
+@router.post("/restore")
+def restore(payload: CheckpointPayload, service=Depends(get_service)) -> Dict[str, Any]:
+    return service.restore(payload.path)

@@ -1,0 +1,94 @@
+"""Checkpoint/resume tests: Service-level save/restore of detector state
+over the admin API and directly (SURVEY.md §5.4 — new capability)."""
+import threading
+
+import httpx
+import pytest
+import yaml
+
+from detectmateservice_amd import Service, ServiceSettings
+from detectmateservice_amd.schemas import ParserSchema
+
+
+def _detector_settings(tmp_path, ipc_addr, port=0, http=False):
+    cfg_file = tmp_path / "det.yaml"
+    cfg_file.write_text(yaml.safe_dump({
+        "detectors": {"NewValueDetector": {
+            "method_type": "new_value_detector",
+            "data_use_training": 2,
+            "global": {"g": {"header_variables": [{"pos": "URL"}]}},
+        }}
+    }))
+    return ServiceSettings(
+        component_type="NewValueDetector",
+        engine_addr=ipc_addr,
+        config_file=cfg_file,
+        http_enabled=http,
+        http_port=port,
+        log_dir=tmp_path / "logs",
+    )
+
+
+def _frame(url, lid="x"):
+    return ParserSchema(EventID=1, logID=lid, logFormatVariables={"URL": url}).serialize()
+
+
+def test_service_checkpoint_restore_roundtrip(tmp_path, ipc_addr):
+    svc = Service(_detector_settings(tmp_path, ipc_addr))
+    try:
+        # train through the service processor
+        svc.process_batch([_frame("/a"), _frame("/b")])
+        assert svc.process(_frame("/a")) is None
+        ck = tmp_path / "ckpt" / "det.pt"
+        info = svc.checkpoint(ck)
+        assert ck.exists()
+
+        # new service instance: restores learned state
+        svc2 = Service(_detector_settings(tmp_path, ipc_addr + "-2"))
+        try:
+            # untrained service would still be in training phase; restore
+            # brings seen_lines and the known set back
+            svc2.restore(ck)
+            assert svc2.process(_frame("/a")) is None       # known
+            assert svc2.process(_frame("/evil")) is not None  # alert
+        finally:
+            svc2.engine.close()
+    finally:
+        svc.engine.close()
+
+
+def test_restore_rejects_wrong_component_type(tmp_path, ipc_addr):
+    svc = Service(_detector_settings(tmp_path, ipc_addr))
+    try:
+        ck = tmp_path / "det.pt"
+        svc.checkpoint(ck)
+        other = Service(ServiceSettings(
+            component_type="core", engine_addr=ipc_addr + "-3",
+            http_enabled=False, log_dir=tmp_path / "logs",
+        ))
+        try:
+            with pytest.raises(ValueError):
+                other.restore(ck)
+        finally:
+            other.engine.close()
+    finally:
+        svc.engine.close()
+
+
+def test_checkpoint_over_http(tmp_path, ipc_addr, free_port):
+    settings = _detector_settings(tmp_path, ipc_addr, port=free_port, http=True)
+    svc = Service(settings)
+    t = threading.Thread(target=svc.run, daemon=True)
+    t.start()
+    try:
+        assert svc.web_server.wait_started(10.0)
+        svc.process_batch([_frame("/a"), _frame("/b")])
+        url = f"http://127.0.0.1:{free_port}"
+        ck = str(tmp_path / "http_ck.pt")
+        r = httpx.post(f"{url}/admin/checkpoint", json={"path": ck}, timeout=10.0)
+        assert r.status_code == 200 and r.json()["path"] == ck
+        r = httpx.post(f"{url}/admin/restore", json={"path": ck}, timeout=10.0)
+        assert r.status_code == 200
+    finally:
+        svc.shutdown()
+        t.join(timeout=5.0)
