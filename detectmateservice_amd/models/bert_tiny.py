@@ -134,6 +134,60 @@ class BertTinyDetectorModel:
         toks = torch.where(valid, toks, torch.zeros_like(toks))
         return toks
 
+    # ------------------------------------------------------------------
+    # fused whole-model kernel path (ops/csrc/bert_fused.hip)
+    # ------------------------------------------------------------------
+    def _fused_ok(self) -> bool:
+        c = self.config
+        return (
+            self.device.type == "cuda"
+            and ops.have_extension()
+            and c.hidden == 128
+            and c.heads == 2
+            and c.ffn == 512
+            and c.max_seq == 64
+            and c.vocab_size == 259
+        )
+
+    def _fused_blobs(self):
+        """Pack weights into the (bf16, f32) blobs the fused kernel indexes
+        (layout MUST match bert_fused.hip WB_*/FB_* offsets)."""
+        if getattr(self, "_wb", None) is None:
+            parts = [self.tok_emb, self.pos_emb]
+            for layer in self.layers:
+                parts += [
+                    layer["wqkv_t"], layer["wo_t"], layer["w1_t"], layer["w2_t"],
+                    layer["ln1_g"], layer["ln1_b"], layer["ln2_g"], layer["ln2_b"],
+                ]
+            parts.append(self.w_score.reshape(-1))
+            self._wb = torch.cat([p.reshape(-1) for p in parts]).contiguous()
+            fparts = []
+            for layer in self.layers:
+                fparts += [layer["bqkv"], layer["bo"], layer["b1"], layer["b2"]]
+            fparts.append(self.b_score)
+            self._fb = torch.cat([p.reshape(-1) for p in fparts]).contiguous()
+        return self._wb, self._fb
+
+    def score_spans(
+        self, lines: torch.Tensor, start: torch.Tensor, end: torch.Tensor
+    ) -> torch.Tensor:
+        """lines [B, max_len] u8 + content spans -> anomaly scores [B] f32.
+
+        GPU + flagship geometry: ONE fused kernel (embed → N transformer
+        layers → pool → score, all in LDS). Otherwise: tokenize + layered
+        forward (same numerics at bf16 tolerance — tests/test_gpu_ops.py).
+        """
+        if self._fused_ok():
+            from ..ops import _dmx_C  # type: ignore[attr-defined]
+
+            wb, fb = self._fused_blobs()
+            return _dmx_C.bert_fused_bf16(
+                lines, start.int().contiguous(), end.int().contiguous(),
+                wb, fb, self.config.layers, 1e-5,
+            )
+        tokens = self.tokenize_spans(lines, start.int(), end.int())
+        return self.forward(tokens)
+
     def state_dict(self) -> Dict[str, torch.Tensor]:
         out = {"tok_emb": self.tok_emb, "pos_emb": self.pos_emb,
                "w_score": self.w_score, "b_score": self.b_score}
@@ -143,6 +197,7 @@ class BertTinyDetectorModel:
         return {k: v.cpu() for k, v in out.items()}
 
     def load_state_dict(self, state: Dict[str, torch.Tensor]) -> None:
+        self._wb = self._fb = None  # invalidate fused blobs
         self.tok_emb = state["tok_emb"].to(self.device)
         self.pos_emb = state["pos_emb"].to(self.device)
         self.w_score = state["w_score"].to(self.device)

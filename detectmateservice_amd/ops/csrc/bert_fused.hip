@@ -1,0 +1,327 @@
+// Fused whole-model BERT-tiny forward: ONE workgroup per log line.
+//
+// The flagship detector config (BASELINE.json config 5) is tiny enough
+// that the idiomatic MI355X mapping is a per-line megakernel: all
+// activations for one line (S=64 tokens, hidden 128) live in LDS for the
+// whole forward pass; the model's weights (~600 KB total) are read
+// through the XCD L2s and shared by every workgroup; HBM traffic is just
+// line bytes in + one f32 score out. The layered path
+// (gemm_bf16/attention_mfma/layernorm kernels) measured 74% of step time
+// in GEMM HBM round-trips (profiles/r02_*); this kernel removes the
+// inter-op HBM traffic entirely.
+//
+// Fixed geometry (asserted host-side): S=64 tokens, hidden H=128, 2 heads
+// (Dh=64), FFN=512, arbitrary layer count. 512 threads = 8 waves; LDS
+// arena ~118 KiB -> 1 workgroup/CU; grid = batch of lines (>> 256 CUs).
+//
+// Phases (per layer, barriers between):
+//   embed  -> x[64][136]
+//   qkv    -> buf[64][392]; V written TRANSPOSED to vt[128][72]
+//   attn   -> per-wave (head, 16 q-rows): QK^T/softmax/PV exactly as
+//             attention_mfma.hip, operands from LDS; out -> buf[64][136]
+//   proj   -> x += proj(buf) (residual in-place), then LayerNorm(x)
+//   ffn1   -> buf[64][520] (GELU fused)
+//   ffn2   -> x += ffn2(buf), LayerNorm(x)
+// pool+score: mean over S, dot with w_score.
+#include "common.h"
+
+#define BF_WAVES 8
+#define BF_THREADS (BF_WAVES * DMX_WAVE)
+#define BF_S 64
+#define BF_H 128
+#define BF_HEADS 2
+#define BF_DH 64
+#define BF_FFN 512
+
+// LDS strides (elements), +8 pad keeps rows 16-B aligned and bank-spread
+#define XS (BF_H + 8)            // 136
+#define BUFS (BF_FFN + 8)        // 520
+#define VTS (BF_S + 8)           // 72
+#define PS (BF_S + 8)            // 72
+
+// bf16 weight-blob element offsets (host packs identically: bert_tiny.py)
+#define WB_TOK 0
+#define WB_POS (WB_TOK + 259 * BF_H)
+#define WB_LAYER0 (WB_POS + BF_S * BF_H)
+#define LW_QKV 0
+#define LW_WO (LW_QKV + 3 * BF_H * BF_H)
+#define LW_W1 (LW_WO + BF_H * BF_H)
+#define LW_W2 (LW_W1 + BF_FFN * BF_H)
+#define LW_LN1G (LW_W2 + BF_H * BF_FFN)
+#define LW_LN1B (LW_LN1G + BF_H)
+#define LW_LN2G (LW_LN1B + BF_H)
+#define LW_LN2B (LW_LN2G + BF_H)
+#define LW_SIZE (LW_LN2B + BF_H)
+// f32 blob: per layer [bqkv 384 | bo 128 | b1 512 | b2 128], then b_score
+#define FB_BQKV 0
+#define FB_BO (FB_BQKV + 3 * BF_H)
+#define FB_B1 (FB_BO + BF_H)
+#define FB_B2 (FB_B1 + BF_FFN)
+#define FB_SIZE (FB_B2 + BF_H)
+
+// ---- in-block GEMM: out = act(in_lds[64][K] @ Wt[N][K] + bias) ----------
+// mode 0: write buf[m][n]; mode 1 (qkv): n<2H -> buf, else vt transposed;
+// mode 2: x[m][n] += val (residual); act 1 = GELU on write (mode 0 only)
+template <int K, int N, int MODE, int ACT>
+static __device__ __forceinline__ void block_gemm(
+    const short* in_lds, int in_stride, const short* __restrict__ Wt,
+    const float* __restrict__ bias, short* out_lds, int out_stride,
+    short* x_lds, short* vt_lds, int wid, int lane) {
+  constexpr int N16 = N / 16;
+  constexpr int TOTAL = 4 * N16;  // (64/16) m-frags * n-frags
+  constexpr int KS = K / 32;
+  for (int ff = wid; ff < TOTAL; ff += BF_WAVES) {
+    const int fm = ff & 3;
+    const int fn = ff >> 2;
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int ks = 0; ks < KS; ++ks) {
+      bf16x8 a = *(const bf16x8*)(
+          in_lds + (fm * 16 + (lane & 15)) * in_stride + ks * 32 +
+          (lane >> 4) * 8);
+      bf16x8 b = *(const bf16x8*)(
+          Wt + (long)(fn * 16 + (lane & 15)) * K + ks * 32 + (lane >> 4) * 8);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+    }
+    const int n = fn * 16 + (lane & 15);
+    const float bval = bias ? bias[n] : 0.f;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int m = fm * 16 + (lane >> 4) * 4 + r;
+      float v = acc[r] + bval;
+      if (MODE == 0) {
+        if (ACT == 1) v = gelu_f32(v);
+        out_lds[m * out_stride + n] = f32_to_bf16(v);
+      } else if (MODE == 1) {
+        if (n < 2 * BF_H) {
+          out_lds[m * out_stride + n] = f32_to_bf16(v);
+        } else {
+          vt_lds[(n - 2 * BF_H) * VTS + m] = f32_to_bf16(v);  // V transposed
+        }
+      } else {  // MODE 2: residual add into x
+        const float xv = bf16_to_f32(x_lds[m * XS + n]);
+        x_lds[m * XS + n] = f32_to_bf16(v + xv);
+      }
+    }
+  }
+}
+
+// ---- in-block LayerNorm on x (post-LN), row per wave-pass ----------------
+static __device__ __forceinline__ void block_layernorm(
+    short* x_lds, const short* __restrict__ gamma,
+    const short* __restrict__ beta, int wid, int lane, float eps) {
+  for (int row = wid * 8; row < wid * 8 + 8; ++row) {
+    // 64 lanes x 2 elements cover the 128-wide row
+    const int c0 = lane * 2;
+    float v0 = bf16_to_f32(x_lds[row * XS + c0]);
+    float v1 = bf16_to_f32(x_lds[row * XS + c0 + 1]);
+    float sum = v0 + v1;
+    sum = warp_reduce_sum_f32(sum);
+    const float mean = __shfl(sum, 0, 64) / BF_H;
+    float var = (v0 - mean) * (v0 - mean) + (v1 - mean) * (v1 - mean);
+    var = warp_reduce_sum_f32(var);
+    const float rstd = rsqrtf(__shfl(var, 0, 64) / BF_H + eps);
+    x_lds[row * XS + c0] = f32_to_bf16(
+        (v0 - mean) * rstd * bf16_to_f32(gamma[c0]) + bf16_to_f32(beta[c0]));
+    x_lds[row * XS + c0 + 1] = f32_to_bf16(
+        (v1 - mean) * rstd * bf16_to_f32(gamma[c0 + 1]) +
+        bf16_to_f32(beta[c0 + 1]));
+  }
+}
+
+extern "C" __global__ __launch_bounds__(BF_THREADS)
+void dmx_bert_fused_bf16(
+    const unsigned char* __restrict__ lines,  // [B, max_len]
+    const int* __restrict__ start,            // [B] content span start
+    const int* __restrict__ end,              // [B] content span end
+    const short* __restrict__ wb,             // bf16 weight blob
+    const float* __restrict__ fb,             // f32 bias blob
+    float* __restrict__ scores,               // [B]
+    int B, int max_len, int n_layers, float eps) {
+  const int line = blockIdx.x;
+  if (line >= B) return;
+  const int tid = threadIdx.x;
+  const int wid = tid / DMX_WAVE;
+  const int lane = tid % DMX_WAVE;
+
+  extern __shared__ __attribute__((aligned(16))) short smem[];
+  short* x_lds = smem;                       // [64][XS]
+  short* buf = x_lds + BF_S * XS;            // [64][BUFS]
+  short* vt = buf + BF_S * BUFS;             // [128][VTS]
+  short* p_lds = vt + 2 * BF_DH * VTS;       // [8][16][PS]
+  float* red = (float*)(p_lds + BF_WAVES * 16 * PS);  // [128] pooling scratch
+
+  // ---- embed: x[s][c] = tok_emb[byte+3 or 0][c] + pos_emb[s][c] ----
+  {
+    const int s0 = start[line], e0 = end[line];
+    for (int i = tid * 8; i < BF_S * BF_H; i += BF_THREADS * 8) {
+      const int s = i / BF_H, c = i % BF_H;
+      int tok = 0;
+      const int idx = s0 + s;
+      if (idx < e0 && idx < max_len)
+        tok = (int)lines[(long)line * max_len + idx] + 3;
+      short8v te = *(const short8v*)(wb + WB_TOK + (long)tok * BF_H + c);
+      short8v pe = *(const short8v*)(wb + WB_POS + (long)s * BF_H + c);
+      short8v xv;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        xv[j] = f32_to_bf16(bf16_to_f32(te[j]) + bf16_to_f32(pe[j]));
+      *(short8v*)(x_lds + s * XS + c) = xv;
+    }
+  }
+  __syncthreads();
+
+  for (int layer = 0; layer < n_layers; ++layer) {
+    const short* lw = wb + WB_LAYER0 + (long)layer * LW_SIZE;
+    const float* lf = fb + (long)layer * FB_SIZE;
+
+    // ---- qkv: buf[64][384] (+V transposed into vt) ----
+    block_gemm<BF_H, 3 * BF_H, 1, 0>(x_lds, XS, lw + LW_QKV, lf + FB_BQKV,
+                                     buf, BUFS, x_lds, vt, wid, lane);
+    __syncthreads();
+
+    // ---- attention: wave = (head, 16 q-rows) ----
+    {
+      const int hh = wid >> 2;
+      const int q0 = (wid & 3) * 16;
+      short* my_p = p_lds + wid * 16 * PS;
+      const float scale = 0.125f;  // 1/sqrt(64)
+
+      f32x4 acc_p[4];
+#pragma unroll
+      for (int f = 0; f < 4; ++f) acc_p[f] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ks = 0; ks < BF_DH / 32; ++ks) {
+        bf16x8 a = *(const bf16x8*)(
+            buf + (q0 + (lane & 15)) * BUFS + hh * BF_DH + ks * 32 +
+            (lane >> 4) * 8);
+#pragma unroll
+        for (int f = 0; f < 4; ++f) {
+          bf16x8 b = *(const bf16x8*)(
+              buf + (f * 16 + (lane & 15)) * BUFS + BF_H + hh * BF_DH +
+              ks * 32 + (lane >> 4) * 8);
+          acc_p[f] =
+              __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc_p[f], 0, 0, 0);
+        }
+      }
+      float inv_sum[4];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float m = -1e30f;
+#pragma unroll
+        for (int f = 0; f < 4; ++f) m = fmaxf(m, acc_p[f][r] * scale);
+#pragma unroll
+        for (int mask = 1; mask < 16; mask <<= 1)
+          m = fmaxf(m, __shfl_xor(m, mask, 64));
+        float sum = 0.f;
+#pragma unroll
+        for (int f = 0; f < 4; ++f) {
+          const float e = __expf(acc_p[f][r] * scale - m);
+          acc_p[f][r] = e;
+          sum += e;
+        }
+#pragma unroll
+        for (int mask = 1; mask < 16; mask <<= 1)
+          sum += __shfl_xor(sum, mask, 64);
+        inv_sum[r] = 1.f / sum;
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = (lane >> 4) * 4 + r;
+#pragma unroll
+        for (int f = 0; f < 4; ++f)
+          my_p[row * PS + f * 16 + (lane & 15)] = f32_to_bf16(acc_p[f][r]);
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+      f32x4 acc_o[4];
+#pragma unroll
+      for (int f = 0; f < 4; ++f) acc_o[f] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ks = 0; ks < BF_S / 32; ++ks) {
+        bf16x8 a = *(const bf16x8*)(
+            my_p + (lane & 15) * PS + ks * 32 + (lane >> 4) * 8);
+#pragma unroll
+        for (int f = 0; f < 4; ++f) {
+          bf16x8 b = *(const bf16x8*)(
+              vt + (hh * BF_DH + f * 16 + (lane & 15)) * VTS + ks * 32 +
+              (lane >> 4) * 8);
+          acc_o[f] =
+              __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc_o[f], 0, 0, 0);
+        }
+      }
+      // attn out -> buf[m][hh*64+d] (qkv no longer needed); barrier first
+      // so no wave overwrites Q/K/V while another still reads them.
+      __syncthreads();
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int q = q0 + (lane >> 4) * 4 + r;
+        const float inv = __shfl(inv_sum[r], (lane >> 4) * 4 + r, 64);
+#pragma unroll
+        for (int f = 0; f < 4; ++f) {
+          const int d = f * 16 + (lane & 15);
+          buf[q * BUFS + hh * BF_DH + d] = f32_to_bf16(acc_o[f][r] * inv);
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- proj: x += Wo(attn) ; LN1 ----
+    block_gemm<BF_H, BF_H, 2, 0>(buf, BUFS, lw + LW_WO, lf + FB_BO, nullptr,
+                                 0, x_lds, nullptr, wid, lane);
+    __syncthreads();
+    block_layernorm(x_lds, lw + LW_LN1G, lw + LW_LN1B, wid, lane, eps);
+    __syncthreads();
+
+    // ---- ffn1 (GELU) -> buf ----
+    block_gemm<BF_H, BF_FFN, 0, 1>(x_lds, XS, lw + LW_W1, lf + FB_B1, buf,
+                                   BUFS, nullptr, nullptr, wid, lane);
+    __syncthreads();
+
+    // ---- ffn2: x += W2(buf) ; LN2 ----
+    block_gemm<BF_FFN, BF_H, 2, 0>(buf, BUFS, lw + LW_W2, lf + FB_B2, nullptr,
+                                   0, x_lds, nullptr, wid, lane);
+    __syncthreads();
+    block_layernorm(x_lds, lw + LW_LN2G, lw + LW_LN2B, wid, lane, eps);
+    __syncthreads();
+  }
+
+  // ---- pool (mean over S) + score head ----
+  {
+    if (tid < BF_H) {
+      float s = 0.f;
+      for (int row = 0; row < BF_S; ++row)
+        s += bf16_to_f32(x_lds[row * XS + tid]);
+      const float pooled = s / BF_S;
+      const float w = bf16_to_f32(wb[WB_LAYER0 + (long)n_layers * LW_SIZE + tid]);
+      red[tid] = pooled * w;
+    }
+    __syncthreads();
+    if (wid == 0) {
+      float v = red[lane] + red[lane + 64];
+      v = warp_reduce_sum_f32(v);
+      if (lane == 0)
+        scores[line] = v + fb[(long)n_layers * FB_SIZE];  // b_score
+    }
+  }
+}
+
+extern "C" void dmx_launch_bert_fused_bf16(
+    const void* lines, const void* start, const void* end, const void* wb,
+    const void* fb, void* scores, int B, int max_len, int n_layers, float eps,
+    hipStream_t stream) {
+  const size_t lds =
+      ((size_t)BF_S * XS + (size_t)BF_S * BUFS + (size_t)2 * BF_DH * VTS +
+       (size_t)BF_WAVES * 16 * PS) * sizeof(short) + 128 * sizeof(float);
+  static bool attr_set = false;
+  if (!attr_set) {
+    hipFuncSetAttribute((const void*)dmx_bert_fused_bf16,
+                        hipFuncAttributeMaxDynamicSharedMemorySize,
+                        (int)lds);
+    attr_set = true;
+  }
+  hipLaunchKernelGGL(dmx_bert_fused_bf16, dim3(B), dim3(BF_THREADS), lds,
+                     stream, (const unsigned char*)lines, (const int*)start,
+                     (const int*)end, (const short*)wb, (const float*)fb,
+                     (float*)scores, B, max_len, n_layers, eps);
+}

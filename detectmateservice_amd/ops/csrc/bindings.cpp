@@ -21,6 +21,9 @@ void dmx_launch_attention_bf16(const void*, const void*, const void*, void*,
                                int, int, int, float, hipStream_t);
 void dmx_launch_attention_mfma_bf16(const void*, void*, int, int, int, int,
                                     float, hipStream_t);
+void dmx_launch_bert_fused_bf16(const void*, const void*, const void*,
+                                const void*, const void*, void*, int, int,
+                                int, float, hipStream_t);
 void dmx_launch_template_match(const void*, const void*, int, int,
                                const void*, const void*, int, const void*,
                                int, const void*, const void*, int, int,
@@ -141,6 +144,26 @@ torch::Tensor attention_qkv_bf16(torch::Tensor qkv, int64_t S, int64_t H,
   return O;
 }
 
+torch::Tensor bert_fused_bf16(torch::Tensor lines, torch::Tensor start,
+                              torch::Tensor end, torch::Tensor wb,
+                              torch::Tensor fb, int64_t n_layers,
+                              double eps) {
+  TORCH_CHECK(lines.is_cuda() && lines.dtype() == torch::kUInt8 &&
+                  lines.is_contiguous(),
+              "lines must be contiguous u8 on GPU");
+  TORCH_CHECK(wb.dtype() == torch::kBFloat16 && wb.is_contiguous());
+  TORCH_CHECK(fb.dtype() == torch::kFloat32 && fb.is_contiguous());
+  TORCH_CHECK(start.dtype() == torch::kInt32 && end.dtype() == torch::kInt32);
+  const auto B = lines.size(0), max_len = lines.size(1);
+  auto scores = torch::empty(
+      {B}, torch::TensorOptions().dtype(torch::kFloat32).device(lines.device()));
+  dmx_launch_bert_fused_bf16(lines.data_ptr(), start.data_ptr(),
+                             end.data_ptr(), wb.data_ptr(), fb.data_ptr(),
+                             scores.data_ptr(), (int)B, (int)max_len,
+                             (int)n_layers, (float)eps, cur_stream());
+  return scores;
+}
+
 std::vector<torch::Tensor> template_match(
     torch::Tensor lines, torch::Tensor line_len, torch::Tensor fmt_bytes,
     torch::Tensor fmt_seg_off, torch::Tensor seg_bytes, torch::Tensor seg_off,
@@ -220,6 +243,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attention_bf16", &attention_bf16, "fused short-seq MHA bf16");
   m.def("attention_qkv_bf16", &attention_qkv_bf16,
         "MFMA MHA reading fused QKV layout [B,S,3*H*Dh] -> [B,S,H*Dh]");
+  m.def("bert_fused_bf16", &bert_fused_bf16,
+        "whole-model BERT-tiny forward, one workgroup per line");
   m.def("template_match", &template_match, "batched wildcard template match");
   m.def("watch_hashes", &watch_hashes, "hash watched capture spans");
   m.def("hashset_insert", &hashset_insert, "insert hashes into GPU sets");

@@ -127,8 +127,7 @@ class GpuPipeline:
                 fc.gather(1, last.view(-1, 1, 1).expand(-1, 1, 2))[:, 0, 1].long(),
                 line_len.long(),
             )
-            tokens = self.model.tokenize_spans(lines, start.int(), end.int())
-            scores = self.model(tokens)
+            scores = self.model.score_spans(lines, start.int(), end.int())
 
         anomaly = scores > self.config.score_threshold
         if nv_unseen is not None:

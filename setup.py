@@ -23,6 +23,7 @@ sources = [
     str(CSRC / "layernorm.hip"),
     str(CSRC / "attention.hip"),
     str(CSRC / "attention_mfma.hip"),
+    str(CSRC / "bert_fused.hip"),
     str(CSRC / "template_match.hip"),
     str(CSRC / "hashset.hip"),
 ]

@@ -248,3 +248,31 @@ def test_attention_qkv_mfma_s32():
     )
     ref = (s @ v.permute(0, 2, 1, 3)).permute(0, 2, 1, 3).reshape(B, S, H * Dh)
     assert (o - ref).abs().max() < 0.03
+
+
+def test_bert_fused_vs_layered():
+    """Fused whole-model kernel vs the layered GPU path (same weights)."""
+    from detectmateservice_amd import ops
+    from detectmateservice_amd.models.bert_tiny import (
+        BertTinyConfig,
+        BertTinyDetectorModel,
+    )
+    from detectmateservice_amd.utils.synthetic import AuditLogGenerator
+
+    model = BertTinyDetectorModel(BertTinyConfig(), device="cuda", seed=13)
+    assert model._fused_ok()
+    gen = AuditLogGenerator(seed=99)
+    raw = [gen.line()[0].encode() for _ in range(128)]
+    lines, lens = ops.pack_lines(raw, 256, device="cuda")
+    start = torch.zeros(128, dtype=torch.int32, device="cuda")
+
+    fused = model.score_spans(lines, start, lens.int())
+    # layered path with identical weights
+    tokens = model.tokenize_spans(lines, start, lens.int())
+    layered = model.forward(tokens)
+    diff = (fused - layered).abs().max().item()
+    assert diff < 0.05, f"fused vs layered diff {diff}"
+    # and against the CPU fp32 reference
+    cpu = BertTinyDetectorModel(BertTinyConfig(), device="cpu", seed=13)
+    ref = cpu.forward(tokens.cpu())
+    assert (fused.cpu() - ref).abs().max() < 0.08
