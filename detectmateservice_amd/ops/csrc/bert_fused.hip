@@ -25,7 +25,7 @@
 // pool+score: mean over S, dot with w_score.
 #include "common.h"
 
-#define BF_WAVES 8
+#define BF_WAVES 16
 #define BF_THREADS (BF_WAVES * DMX_WAVE)
 #define BF_S 64
 #define BF_H 128
@@ -37,7 +37,8 @@
 #define XS (BF_H + 8)            // 136
 #define BUFS (BF_FFN + 8)        // 520
 #define VTS (BF_S + 8)           // 72
-#define PS (BF_S + 8)            // 72
+#define PS (BF_S + 8)
+#define P_WAVES 8  // only attention's 8 active waves keep a P tile            // 72
 
 // bf16 weight-blob element offsets (host packs identically: bert_tiny.py)
 #define WB_TOK 0
@@ -63,44 +64,61 @@
 // mode 0: write buf[m][n]; mode 1 (qkv): n<2H -> buf, else vt transposed;
 // mode 2: x[m][n] += val (residual); act 1 = GELU on write (mode 0 only)
 template <int K, int N, int MODE, int ACT>
-static __device__ __forceinline__ void block_gemm(
+static __device__ __attribute__((noinline)) void block_gemm(
     const short* in_lds, int in_stride, const short* __restrict__ Wt,
     const float* __restrict__ bias, short* out_lds, int out_stride,
     short* x_lds, short* vt_lds, int wid, int lane) {
   constexpr int N16 = N / 16;
   constexpr int TOTAL = 4 * N16;  // (64/16) m-frags * n-frags
   constexpr int KS = K / 32;
-  for (int ff = wid; ff < TOTAL; ff += BF_WAVES) {
-    const int fm = ff & 3;
+  constexpr int FPW = TOTAL / BF_WAVES;  // frags per wave (always even here)
+  static_assert(FPW >= 2 && FPW % 2 == 0, "pair-chunked assignment");
+  // Pair-chunked frag assignment: consecutive even ff pairs share fn, so
+  // ONE global (L2) weight-fragment load feeds TWO independent MFMA
+  // accumulator chains — halves the L2 load count and doubles the
+  // latency-hiding ILP (the first fused-kernel profile was L2-latency
+  // bound in exactly this loop).
+  for (int ff = wid * FPW; ff < wid * FPW + FPW; ff += 2) {
     const int fn = ff >> 2;
-    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
+    const int fm0 = ff & 3;  // 0 or 2
+    f32x4 acc0 = {0.f, 0.f, 0.f, 0.f};
+    f32x4 acc1 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll 4
     for (int ks = 0; ks < KS; ++ks) {
-      bf16x8 a = *(const bf16x8*)(
-          in_lds + (fm * 16 + (lane & 15)) * in_stride + ks * 32 +
-          (lane >> 4) * 8);
       bf16x8 b = *(const bf16x8*)(
           Wt + (long)(fn * 16 + (lane & 15)) * K + ks * 32 + (lane >> 4) * 8);
-      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+      bf16x8 a0 = *(const bf16x8*)(
+          in_lds + (fm0 * 16 + (lane & 15)) * in_stride + ks * 32 +
+          (lane >> 4) * 8);
+      bf16x8 a1 = *(const bf16x8*)(
+          in_lds + ((fm0 + 1) * 16 + (lane & 15)) * in_stride + ks * 32 +
+          (lane >> 4) * 8);
+      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b, acc0, 0, 0, 0);
+      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b, acc1, 0, 0, 0);
     }
     const int n = fn * 16 + (lane & 15);
     const float bval = bias ? bias[n] : 0.f;
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int m = fm * 16 + (lane >> 4) * 4 + r;
-      float v = acc[r] + bval;
-      if (MODE == 0) {
-        if (ACT == 1) v = gelu_f32(v);
-        out_lds[m * out_stride + n] = f32_to_bf16(v);
-      } else if (MODE == 1) {
-        if (n < 2 * BF_H) {
+    for (int half = 0; half < 2; ++half) {
+      const f32x4& acc = half ? acc1 : acc0;
+      const int fm = fm0 + half;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = fm * 16 + (lane >> 4) * 4 + r;
+        float v = acc[r] + bval;
+        if (MODE == 0) {
+          if (ACT == 1) v = gelu_f32(v);
           out_lds[m * out_stride + n] = f32_to_bf16(v);
-        } else {
-          vt_lds[(n - 2 * BF_H) * VTS + m] = f32_to_bf16(v);  // V transposed
+        } else if (MODE == 1) {
+          if (n < 2 * BF_H) {
+            out_lds[m * out_stride + n] = f32_to_bf16(v);
+          } else {
+            vt_lds[(n - 2 * BF_H) * VTS + m] = f32_to_bf16(v);  // V transposed
+          }
+        } else {  // MODE 2: residual add into x
+          const float xv = bf16_to_f32(x_lds[m * XS + n]);
+          x_lds[m * XS + n] = f32_to_bf16(v + xv);
         }
-      } else {  // MODE 2: residual add into x
-        const float xv = bf16_to_f32(x_lds[m * XS + n]);
-        x_lds[m * XS + n] = f32_to_bf16(v + xv);
       }
     }
   }
@@ -110,7 +128,7 @@ static __device__ __forceinline__ void block_gemm(
 static __device__ __forceinline__ void block_layernorm(
     short* x_lds, const short* __restrict__ gamma,
     const short* __restrict__ beta, int wid, int lane, float eps) {
-  for (int row = wid * 8; row < wid * 8 + 8; ++row) {
+  for (int row = wid * (BF_S / BF_WAVES); row < (wid + 1) * (BF_S / BF_WAVES); ++row) {
     // 64 lanes x 2 elements cover the 128-wide row
     const int c0 = lane * 2;
     float v0 = bf16_to_f32(x_lds[row * XS + c0]);
@@ -148,8 +166,8 @@ void dmx_bert_fused_bf16(
   short* x_lds = smem;                       // [64][XS]
   short* buf = x_lds + BF_S * XS;            // [64][BUFS]
   short* vt = buf + BF_S * BUFS;             // [128][VTS]
-  short* p_lds = vt + 2 * BF_DH * VTS;       // [8][16][PS]
-  float* red = (float*)(p_lds + BF_WAVES * 16 * PS);  // [128] pooling scratch
+  short* p_lds = vt + 2 * BF_DH * VTS;       // [P_WAVES][16][PS]
+  float* red = (float*)(p_lds + P_WAVES * 16 * PS);  // [128] pooling scratch
 
   // ---- embed: x[s][c] = tok_emb[byte+3 or 0][c] + pos_emb[s][c] ----
   {
@@ -180,9 +198,10 @@ void dmx_bert_fused_bf16(
                                      buf, BUFS, x_lds, vt, wid, lane);
     __syncthreads();
 
-    // ---- attention: wave = (head, 16 q-rows) ----
+    // ---- attention: waves 0..7 = (head, 16 q-rows); rest idle ----
     {
-      const int hh = wid >> 2;
+      const bool active = wid < 8;
+      const int hh = (wid >> 2) & 1;
       const int q0 = (wid & 3) * 16;
       short* my_p = p_lds + wid * 16 * PS;
       const float scale = 0.125f;  // 1/sqrt(64)
@@ -190,6 +209,7 @@ void dmx_bert_fused_bf16(
       f32x4 acc_p[4];
 #pragma unroll
       for (int f = 0; f < 4; ++f) acc_p[f] = {0.f, 0.f, 0.f, 0.f};
+      if (active)
 #pragma unroll
       for (int ks = 0; ks < BF_DH / 32; ++ks) {
         bf16x8 a = *(const bf16x8*)(
@@ -204,7 +224,8 @@ void dmx_bert_fused_bf16(
               __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc_p[f], 0, 0, 0);
         }
       }
-      float inv_sum[4];
+      float inv_sum[4] = {1.f, 1.f, 1.f, 1.f};
+      if (active)
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         float m = -1e30f;
@@ -225,18 +246,21 @@ void dmx_bert_fused_bf16(
           sum += __shfl_xor(sum, mask, 64);
         inv_sum[r] = 1.f / sum;
       }
+      if (active) {
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int row = (lane >> 4) * 4 + r;
+        for (int r = 0; r < 4; ++r) {
+          const int row = (lane >> 4) * 4 + r;
 #pragma unroll
-        for (int f = 0; f < 4; ++f)
-          my_p[row * PS + f * 16 + (lane & 15)] = f32_to_bf16(acc_p[f][r]);
+          for (int f = 0; f < 4; ++f)
+            my_p[row * PS + f * 16 + (lane & 15)] = f32_to_bf16(acc_p[f][r]);
+        }
       }
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
       f32x4 acc_o[4];
 #pragma unroll
       for (int f = 0; f < 4; ++f) acc_o[f] = {0.f, 0.f, 0.f, 0.f};
+      if (active)
 #pragma unroll
       for (int ks = 0; ks < BF_S / 32; ++ks) {
         bf16x8 a = *(const bf16x8*)(
@@ -253,6 +277,7 @@ void dmx_bert_fused_bf16(
       // attn out -> buf[m][hh*64+d] (qkv no longer needed); barrier first
       // so no wave overwrites Q/K/V while another still reads them.
       __syncthreads();
+      if (active)
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int q = q0 + (lane >> 4) * 4 + r;
@@ -312,7 +337,7 @@ extern "C" void dmx_launch_bert_fused_bf16(
     hipStream_t stream) {
   const size_t lds =
       ((size_t)BF_S * XS + (size_t)BF_S * BUFS + (size_t)2 * BF_DH * VTS +
-       (size_t)BF_WAVES * 16 * PS) * sizeof(short) + 128 * sizeof(float);
+       (size_t)P_WAVES * 16 * PS) * sizeof(short) + 128 * sizeof(float);
   static bool attr_set = false;
   if (!attr_set) {
     hipFuncSetAttribute((const void*)dmx_bert_fused_bf16,
