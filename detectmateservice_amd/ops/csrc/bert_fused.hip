@@ -3,42 +3,41 @@
 // The flagship detector config (BASELINE.json config 5) is tiny enough
 // that the idiomatic MI355X mapping is a per-line megakernel: all
 // activations for one line (S=64 tokens, hidden 128) live in LDS for the
-// whole forward pass; the model's weights (~600 KB total) are read
-// through the XCD L2s and shared by every workgroup; HBM traffic is just
-// line bytes in + one f32 score out. The layered path
-// (gemm_bf16/attention_mfma/layernorm kernels) measured 74% of step time
-// in GEMM HBM round-trips (profiles/r02_*); this kernel removes the
-// inter-op HBM traffic entirely.
+// whole forward pass; the model's weights (~600 KB) are read through the
+// XCD L2s and shared by every workgroup; HBM traffic is line bytes in +
+// one f32 score out.
 //
-// Fixed geometry (asserted host-side): S=64 tokens, hidden H=128, 2 heads
-// (Dh=64), FFN=512, arbitrary layer count. 512 threads = 8 waves; LDS
-// arena ~118 KiB -> 1 workgroup/CU; grid = batch of lines (>> 256 CUs).
+// v3 design notes (PMC-driven, profiles/*):
+//  * v1 (layered kernels) was HBM-bound on inter-op traffic (74% GEMM).
+//  * v2 (fused, 118 KiB arena, 1 block/CU) showed MemUnitStalled~0 and
+//    VALUBusy 33%: bulk-synchronous phases at 1 block/CU leave the CU
+//    idle at every barrier. This version shrinks the arena to ~70 KiB so
+//    TWO blocks co-reside per CU - one block's barrier idle overlaps the
+//    other block's compute. 8 waves/block (512 thr), VGPR<=128 keeps both
+//    resident (4 waves/SIMD from two independent barrier domains).
+//  * arena cuts: qkv buffer holds Q|K only (V written transposed to vt
+//    during the qkv GEMM epilogue); FFN runs in two K=256 halves that
+//    accumulate into x; the attention P tile aliases the (dead) Q|K area.
+//  * pair-chunked weight fragments: one L2 B-fragment load feeds two
+//    independent MFMA chains.
 //
-// Phases (per layer, barriers between):
-//   embed  -> x[64][136]
-//   qkv    -> buf[64][392]; V written TRANSPOSED to vt[128][72]
-//   attn   -> per-wave (head, 16 q-rows): QK^T/softmax/PV exactly as
-//             attention_mfma.hip, operands from LDS; out -> buf[64][136]
-//   proj   -> x += proj(buf) (residual in-place), then LayerNorm(x)
-//   ffn1   -> buf[64][520] (GELU fused)
-//   ffn2   -> x += ffn2(buf), LayerNorm(x)
-// pool+score: mean over S, dot with w_score.
+// Fixed geometry (asserted host-side): S=64 tokens, H=128, 2 heads
+// (Dh=64), FFN=512, arbitrary layer count.
 #include "common.h"
 
-#define BF_WAVES 16
+#define BF_WAVES 8
 #define BF_THREADS (BF_WAVES * DMX_WAVE)
 #define BF_S 64
 #define BF_H 128
-#define BF_HEADS 2
 #define BF_DH 64
 #define BF_FFN 512
 
-// LDS strides (elements), +8 pad keeps rows 16-B aligned and bank-spread
-#define XS (BF_H + 8)            // 136
-#define BUFS (BF_FFN + 8)        // 520
-#define VTS (BF_S + 8)           // 72
-#define PS (BF_S + 8)
-#define P_WAVES 8  // only attention's 8 active waves keep a P tile            // 72
+// LDS strides (elements); +8 pads keep rows 16-B aligned and bank-spread
+#define XS (BF_H + 8)       // 136, x rows
+#define QKS (2 * BF_H + 8)  // 264, Q|K rows and FFN-half rows
+#define VTS (BF_S + 8)      // 72, transposed-V and P rows
+#define O_OFF (BF_WAVES * 16 * VTS)    // 9216: attn-out after the P tiles
+#define BUF_ELEMS (BF_S * QKS + 1024)  // 17920 = max(QK, P+O, FFN-half)
 
 // bf16 weight-blob element offsets (host packs identically: bert_tiny.py)
 #define WB_TOK 0
@@ -60,52 +59,46 @@
 #define FB_B2 (FB_B1 + BF_FFN)
 #define FB_SIZE (FB_B2 + BF_H)
 
-// ---- in-block GEMM: out = act(in_lds[64][K] @ Wt[N][K] + bias) ----------
-// mode 0: write buf[m][n]; mode 1 (qkv): n<2H -> buf, else vt transposed;
-// mode 2: x[m][n] += val (residual); act 1 = GELU on write (mode 0 only)
-template <int K, int N, int MODE, int ACT>
+// ---- in-block GEMM: out = act(in_lds[64][K] @ Wt[N][WTS] + bias) ---------
+// MODE 0: write out_lds[m][n]; MODE 1 (qkv): n<2H -> out (Q|K), else vt
+// transposed; MODE 2: x[m][n] += v (residual-accumulate). ACT 1 = GELU.
+template <int K, int N, int MODE, int ACT, int WTS>
 static __device__ __attribute__((noinline)) void block_gemm(
     const short* in_lds, int in_stride, const short* __restrict__ Wt,
     const float* __restrict__ bias, short* out_lds, int out_stride,
     short* x_lds, short* vt_lds, int wid, int lane) {
   constexpr int N16 = N / 16;
-  constexpr int TOTAL = 4 * N16;  // (64/16) m-frags * n-frags
+  constexpr int TOTAL = 4 * N16;
   constexpr int KS = K / 32;
-  constexpr int FPW = TOTAL / BF_WAVES;  // frags per wave (always even here)
-  static_assert(FPW >= 2 && FPW % 2 == 0, "pair-chunked assignment");
-  // Pair-chunked frag assignment: consecutive even ff pairs share fn, so
-  // ONE global (L2) weight-fragment load feeds TWO independent MFMA
-  // accumulator chains — halves the L2 load count and doubles the
-  // latency-hiding ILP (the first fused-kernel profile was L2-latency
-  // bound in exactly this loop).
-  for (int ff = wid * FPW; ff < wid * FPW + FPW; ff += 2) {
+  constexpr int FPW = TOTAL / BF_WAVES;
+  static_assert(FPW >= 4 && FPW % 4 == 0, "quad-chunked assignment");
+  // quad-chunk: each iteration owns a FULL fn column (all 4 m-fragments):
+  // ONE L2 weight-fragment load feeds FOUR independent MFMA chains.
+  for (int ff = wid * FPW; ff < wid * FPW + FPW; ff += 4) {
     const int fn = ff >> 2;
-    const int fm0 = ff & 3;  // 0 or 2
-    f32x4 acc0 = {0.f, 0.f, 0.f, 0.f};
-    f32x4 acc1 = {0.f, 0.f, 0.f, 0.f};
+    f32x4 acc[4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll 4
     for (int ks = 0; ks < KS; ++ks) {
       bf16x8 b = *(const bf16x8*)(
-          Wt + (long)(fn * 16 + (lane & 15)) * K + ks * 32 + (lane >> 4) * 8);
-      bf16x8 a0 = *(const bf16x8*)(
-          in_lds + (fm0 * 16 + (lane & 15)) * in_stride + ks * 32 +
-          (lane >> 4) * 8);
-      bf16x8 a1 = *(const bf16x8*)(
-          in_lds + ((fm0 + 1) * 16 + (lane & 15)) * in_stride + ks * 32 +
-          (lane >> 4) * 8);
-      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b, acc0, 0, 0, 0);
-      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b, acc1, 0, 0, 0);
+          Wt + (long)(fn * 16 + (lane & 15)) * WTS + ks * 32 + (lane >> 4) * 8);
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm) {
+        bf16x8 a = *(const bf16x8*)(
+            in_lds + (fm * 16 + (lane & 15)) * in_stride + ks * 32 +
+            (lane >> 4) * 8);
+        acc[fm] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[fm], 0, 0, 0);
+      }
     }
     const int n = fn * 16 + (lane & 15);
     const float bval = bias ? bias[n] : 0.f;
 #pragma unroll
-    for (int half = 0; half < 2; ++half) {
-      const f32x4& acc = half ? acc1 : acc0;
-      const int fm = fm0 + half;
+    for (int fm = 0; fm < 4; ++fm) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int m = fm * 16 + (lane >> 4) * 4 + r;
-        float v = acc[r] + bval;
+        float v = acc[fm][r] + bval;
         if (MODE == 0) {
           if (ACT == 1) v = gelu_f32(v);
           out_lds[m * out_stride + n] = f32_to_bf16(v);
@@ -115,7 +108,7 @@ static __device__ __attribute__((noinline)) void block_gemm(
           } else {
             vt_lds[(n - 2 * BF_H) * VTS + m] = f32_to_bf16(v);  // V transposed
           }
-        } else {  // MODE 2: residual add into x
+        } else {  // MODE 2
           const float xv = bf16_to_f32(x_lds[m * XS + n]);
           x_lds[m * XS + n] = f32_to_bf16(v + xv);
         }
@@ -124,17 +117,16 @@ static __device__ __attribute__((noinline)) void block_gemm(
   }
 }
 
-// ---- in-block LayerNorm on x (post-LN), row per wave-pass ----------------
+// ---- in-block LayerNorm on x (post-LN) -----------------------------------
 static __device__ __forceinline__ void block_layernorm(
     short* x_lds, const short* __restrict__ gamma,
     const short* __restrict__ beta, int wid, int lane, float eps) {
-  for (int row = wid * (BF_S / BF_WAVES); row < (wid + 1) * (BF_S / BF_WAVES); ++row) {
-    // 64 lanes x 2 elements cover the 128-wide row
-    const int c0 = lane * 2;
+  for (int row = wid * (BF_S / BF_WAVES); row < (wid + 1) * (BF_S / BF_WAVES);
+       ++row) {
+    const int c0 = lane * 2;  // 64 lanes x 2 elements = 128-wide row
     float v0 = bf16_to_f32(x_lds[row * XS + c0]);
     float v1 = bf16_to_f32(x_lds[row * XS + c0 + 1]);
-    float sum = v0 + v1;
-    sum = warp_reduce_sum_f32(sum);
+    float sum = warp_reduce_sum_f32(v0 + v1);
     const float mean = __shfl(sum, 0, 64) / BF_H;
     float var = (v0 - mean) * (v0 - mean) + (v1 - mean) * (v1 - mean);
     var = warp_reduce_sum_f32(var);
@@ -147,7 +139,7 @@ static __device__ __forceinline__ void block_layernorm(
   }
 }
 
-extern "C" __global__ __launch_bounds__(BF_THREADS)
+extern "C" __global__ __launch_bounds__(BF_THREADS, 4)
 void dmx_bert_fused_bf16(
     const unsigned char* __restrict__ lines,  // [B, max_len]
     const int* __restrict__ start,            // [B] content span start
@@ -163,11 +155,10 @@ void dmx_bert_fused_bf16(
   const int lane = tid % DMX_WAVE;
 
   extern __shared__ __attribute__((aligned(16))) short smem[];
-  short* x_lds = smem;                       // [64][XS]
-  short* buf = x_lds + BF_S * XS;            // [64][BUFS]
-  short* vt = buf + BF_S * BUFS;             // [128][VTS]
-  short* p_lds = vt + 2 * BF_DH * VTS;       // [P_WAVES][16][PS]
-  float* red = (float*)(p_lds + P_WAVES * 16 * PS);  // [128] pooling scratch
+  short* x_lds = smem;             // [64][XS]
+  short* buf = x_lds + BF_S * XS;  // BUF_ELEMS: QK | P+O | FFN-half
+  short* vt = buf + BUF_ELEMS;     // [128][VTS]
+  float* red = (float*)(vt + 2 * BF_DH * VTS);  // [128] pooling scratch
 
   // ---- embed: x[s][c] = tok_emb[byte+3 or 0][c] + pos_emb[s][c] ----
   {
@@ -193,39 +184,36 @@ void dmx_bert_fused_bf16(
     const short* lw = wb + WB_LAYER0 + (long)layer * LW_SIZE;
     const float* lf = fb + (long)layer * FB_SIZE;
 
-    // ---- qkv: buf[64][384] (+V transposed into vt) ----
-    block_gemm<BF_H, 3 * BF_H, 1, 0>(x_lds, XS, lw + LW_QKV, lf + FB_BQKV,
-                                     buf, BUFS, x_lds, vt, wid, lane);
+    // ---- qkv: Q|K -> buf[64][QKS], V -> vt transposed ----
+    block_gemm<BF_H, 3 * BF_H, 1, 0, BF_H>(x_lds, XS, lw + LW_QKV,
+                                           lf + FB_BQKV, buf, QKS, x_lds, vt,
+                                           wid, lane);
     __syncthreads();
 
-    // ---- attention: waves 0..7 = (head, 16 q-rows); rest idle ----
+    // ---- attention: wave = (head hh, 16 q-rows) ----
     {
-      const bool active = wid < 8;
-      const int hh = (wid >> 2) & 1;
+      const int hh = wid >> 2;
       const int q0 = (wid & 3) * 16;
-      short* my_p = p_lds + wid * 16 * PS;
       const float scale = 0.125f;  // 1/sqrt(64)
 
       f32x4 acc_p[4];
 #pragma unroll
       for (int f = 0; f < 4; ++f) acc_p[f] = {0.f, 0.f, 0.f, 0.f};
-      if (active)
 #pragma unroll
       for (int ks = 0; ks < BF_DH / 32; ++ks) {
         bf16x8 a = *(const bf16x8*)(
-            buf + (q0 + (lane & 15)) * BUFS + hh * BF_DH + ks * 32 +
+            buf + (q0 + (lane & 15)) * QKS + hh * BF_DH + ks * 32 +
             (lane >> 4) * 8);
 #pragma unroll
         for (int f = 0; f < 4; ++f) {
           bf16x8 b = *(const bf16x8*)(
-              buf + (f * 16 + (lane & 15)) * BUFS + BF_H + hh * BF_DH +
+              buf + (f * 16 + (lane & 15)) * QKS + BF_H + hh * BF_DH +
               ks * 32 + (lane >> 4) * 8);
           acc_p[f] =
               __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc_p[f], 0, 0, 0);
         }
       }
-      float inv_sum[4] = {1.f, 1.f, 1.f, 1.f};
-      if (active)
+      float inv_sum[4];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         float m = -1e30f;
@@ -246,25 +234,25 @@ void dmx_bert_fused_bf16(
           sum += __shfl_xor(sum, mask, 64);
         inv_sum[r] = 1.f / sum;
       }
-      if (active) {
+      // P tiles alias the Q|K area: every wave must be done reading Q/K
+      __syncthreads();
+      short* my_p = buf + wid * 16 * VTS;
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int row = (lane >> 4) * 4 + r;
+      for (int r = 0; r < 4; ++r) {
+        const int row = (lane >> 4) * 4 + r;
 #pragma unroll
-          for (int f = 0; f < 4; ++f)
-            my_p[row * PS + f * 16 + (lane & 15)] = f32_to_bf16(acc_p[f][r]);
-        }
+        for (int f = 0; f < 4; ++f)
+          my_p[row * VTS + f * 16 + (lane & 15)] = f32_to_bf16(acc_p[f][r]);
       }
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // wave-local P
 
       f32x4 acc_o[4];
 #pragma unroll
       for (int f = 0; f < 4; ++f) acc_o[f] = {0.f, 0.f, 0.f, 0.f};
-      if (active)
 #pragma unroll
       for (int ks = 0; ks < BF_S / 32; ++ks) {
         bf16x8 a = *(const bf16x8*)(
-            my_p + (lane & 15) * PS + ks * 32 + (lane >> 4) * 8);
+            my_p + (lane & 15) * VTS + ks * 32 + (lane >> 4) * 8);
 #pragma unroll
         for (int f = 0; f < 4; ++f) {
           bf16x8 b = *(const bf16x8*)(
@@ -274,10 +262,7 @@ void dmx_bert_fused_bf16(
               __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc_o[f], 0, 0, 0);
         }
       }
-      // attn out -> buf[m][hh*64+d] (qkv no longer needed); barrier first
-      // so no wave overwrites Q/K/V while another still reads them.
-      __syncthreads();
-      if (active)
+      // attn out -> buf[O_OFF + q*XS + hh*64 + d] (disjoint from P tiles)
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int q = q0 + (lane >> 4) * 4 + r;
@@ -285,28 +270,36 @@ void dmx_bert_fused_bf16(
 #pragma unroll
         for (int f = 0; f < 4; ++f) {
           const int d = f * 16 + (lane & 15);
-          buf[q * BUFS + hh * BF_DH + d] = f32_to_bf16(acc_o[f][r] * inv);
+          buf[O_OFF + q * XS + hh * BF_DH + d] =
+              f32_to_bf16(acc_o[f][r] * inv);
         }
       }
     }
     __syncthreads();
 
     // ---- proj: x += Wo(attn) ; LN1 ----
-    block_gemm<BF_H, BF_H, 2, 0>(buf, BUFS, lw + LW_WO, lf + FB_BO, nullptr,
-                                 0, x_lds, nullptr, wid, lane);
+    block_gemm<BF_H, BF_H, 2, 0, BF_H>(buf + O_OFF, XS, lw + LW_WO,
+                                       lf + FB_BO, nullptr, 0, x_lds, nullptr,
+                                       wid, lane);
     __syncthreads();
     block_layernorm(x_lds, lw + LW_LN1G, lw + LW_LN1B, wid, lane, eps);
     __syncthreads();
 
-    // ---- ffn1 (GELU) -> buf ----
-    block_gemm<BF_H, BF_FFN, 0, 1>(x_lds, XS, lw + LW_W1, lf + FB_B1, buf,
-                                   BUFS, nullptr, nullptr, wid, lane);
-    __syncthreads();
-
-    // ---- ffn2: x += W2(buf) ; LN2 ----
-    block_gemm<BF_FFN, BF_H, 2, 0>(buf, BUFS, lw + LW_W2, lf + FB_B2, nullptr,
-                                   0, x_lds, nullptr, wid, lane);
-    __syncthreads();
+    // ---- FFN in two K=256 halves: buf = gelu(x@W1_h); x += buf@W2_h ----
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      block_gemm<BF_H, BF_FFN / 2, 0, 1, BF_H>(
+          x_lds, XS, lw + LW_W1 + (long)h * (BF_FFN / 2) * BF_H,
+          lf + FB_B1 + h * (BF_FFN / 2), buf, QKS, nullptr, nullptr, wid,
+          lane);
+      __syncthreads();
+      // bias b2 added once (half 0); half 1 adds only the partial product
+      block_gemm<BF_FFN / 2, BF_H, 2, 0, BF_FFN>(
+          buf, QKS, lw + LW_W2 + (long)h * (BF_FFN / 2),
+          h == 0 ? lf + FB_B2 : nullptr, nullptr, 0, x_lds, nullptr, wid,
+          lane);
+      __syncthreads();
+    }
     block_layernorm(x_lds, lw + LW_LN2G, lw + LW_LN2B, wid, lane, eps);
     __syncthreads();
   }
@@ -317,9 +310,9 @@ void dmx_bert_fused_bf16(
       float s = 0.f;
       for (int row = 0; row < BF_S; ++row)
         s += bf16_to_f32(x_lds[row * XS + tid]);
-      const float pooled = s / BF_S;
-      const float w = bf16_to_f32(wb[WB_LAYER0 + (long)n_layers * LW_SIZE + tid]);
-      red[tid] = pooled * w;
+      const float w =
+          bf16_to_f32(wb[WB_LAYER0 + (long)n_layers * LW_SIZE + tid]);
+      red[tid] = (s / BF_S) * w;
     }
     __syncthreads();
     if (wid == 0) {
@@ -336,13 +329,13 @@ extern "C" void dmx_launch_bert_fused_bf16(
     const void* fb, void* scores, int B, int max_len, int n_layers, float eps,
     hipStream_t stream) {
   const size_t lds =
-      ((size_t)BF_S * XS + (size_t)BF_S * BUFS + (size_t)2 * BF_DH * VTS +
-       (size_t)P_WAVES * 16 * PS) * sizeof(short) + 128 * sizeof(float);
+      ((size_t)BF_S * XS + BUF_ELEMS + (size_t)2 * BF_DH * VTS) *
+          sizeof(short) +
+      128 * sizeof(float);
   static bool attr_set = false;
   if (!attr_set) {
     hipFuncSetAttribute((const void*)dmx_bert_fused_bf16,
-                        hipFuncAttributeMaxDynamicSharedMemorySize,
-                        (int)lds);
+                        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
     attr_set = true;
   }
   hipLaunchKernelGGL(dmx_bert_fused_bf16, dim3(B), dim3(BF_THREADS), lds,
