@@ -1,0 +1,100 @@
+"""Two concurrent services in one process + full From.log single-process
+pipeline (reference shapes: test_service_multi_output_integration.py:266-292
+and tests/library_integration/test_one_pipe_to_rule_them_all.py)."""
+import threading
+import time
+
+import pytest
+
+from detectmateservice_amd import Service, ServiceSettings
+from detectmateservice_amd.engine.sockets import PairDialer, PairListener
+from detectmateservice_amd.library.detectors import NewValueDetector
+from detectmateservice_amd.library.parsers import MatcherParser
+from detectmateservice_amd.library.readers import FileReader, From
+from detectmateservice_amd.schemas import DetectorSchema, LogSchema, ParserSchema
+from detectmateservice_amd.utils.synthetic import (
+    AUDIT_LOG_FORMAT,
+    AUDIT_TEMPLATES,
+    AuditLogGenerator,
+)
+
+
+def test_two_concurrent_services(tmp_path):
+    """Two independent detector services with distinct identities run in
+    one process; each processes its own traffic."""
+    services, threads = [], []
+    addrs = []
+    for i in range(2):
+        addr = f"ipc://{tmp_path}/svc{i}.ipc"
+        addrs.append(addr)
+        s = Service(ServiceSettings(
+            component_type="core",
+            component_name=f"multi-{i}",
+            engine_addr=addr,
+            http_enabled=False,
+            log_dir=tmp_path / "logs",
+        ))
+        services.append(s)
+        t = threading.Thread(target=s.run, daemon=True)
+        t.start()
+        threads.append(t)
+    try:
+        assert services[0].settings.component_id != services[1].settings.component_id
+        time.sleep(0.2)
+        clients = [PairDialer(a) for a in addrs]
+        try:
+            for i, c in enumerate(clients):
+                assert c.wait_connected(5.0)
+                c.send(b"svc%d-msg" % i)
+            for i, c in enumerate(clients):
+                assert c.recv(timeout_ms=3000) == b"svc%d-msg" % i
+        finally:
+            for c in clients:
+                c.close()
+    finally:
+        for s, t in zip(services, threads):
+            s.shutdown()
+            t.join(timeout=5.0)
+
+
+def test_one_pipe_single_process(tmp_path):
+    """From.log streams a file through parser → detector in one process:
+    train on clean traffic, detect an injected anomaly."""
+    gen = AuditLogGenerator(seed=123)
+    log_file = tmp_path / "audit.log"
+    lines = gen.lines(50)
+    bad = (
+        "type=USER_ACCT msg=audit(1642723741.072:999): pid=1 uid=0 auid=1 ses=1 "
+        "msg='op=PAM:accounting acct=\"intruder\" exe=/usr/sbin/cron hostname=? "
+        "addr=? terminal=cron res=success'"
+    )
+    log_file.write_text("\n".join(lines + [bad]) + "\n")
+
+    parser = MatcherParser({
+        "log_format": AUDIT_LOG_FORMAT,
+        "templates": list(AUDIT_TEMPLATES),
+    })
+    detector = NewValueDetector({
+        "data_use_training": 40,
+        "events": {1: {"inst": {"variables": [{"pos": 5, "name": "acct"}]}}},
+    })
+
+    parsed_frames = From.log(parser, log_file, do_process=True)
+    assert len(parsed_frames) == 51
+    alerts = [a for a in detector.process_batch(parsed_frames) if a is not None]
+    assert len(alerts) >= 1
+    final = DetectorSchema.deserialize(alerts[-1])
+    assert "intruder" in final.description
+
+
+def test_file_reader_streaming(tmp_path):
+    f = tmp_path / "x.log"
+    f.write_text("alpha\nbeta\n\ngamma\n")
+    reader = FileReader({"path": str(f)})
+    logs = list(reader.read())
+    assert [l.log for l in logs] == ["alpha", "beta", "gamma"]
+    assert len({l.logID for l in logs}) == 3
+    # process() wraps raw bytes
+    frame = reader.process(b"delta\n")
+    assert LogSchema.deserialize(frame).log == "delta"
+    assert reader.process(b"\n") is None
