@@ -53,6 +53,18 @@ class CoreComponent(ABC):
             self.config = self.CONFIG_CLASS.model_validate(config)
         else:
             raise TypeError(f"config must be dict/CoreConfig/None, got {type(config)}")
+        # method_type check (reference library config pipeline,
+        # docs/interfaces.md:74-82 "check_type"): a config naming a
+        # different method_type than the component's declared default is a
+        # wiring error, not a silent override.
+        declared = self.CONFIG_CLASS.model_fields.get("method_type")
+        if declared is not None and declared.default:
+            given = getattr(self.config, "method_type", None)
+            if given and given != declared.default:
+                raise ValueError(
+                    f"config method_type {given!r} does not match "
+                    f"{type(self).__name__}'s {declared.default!r}"
+                )
 
     @abstractmethod
     def process(self, data: bytes) -> Optional[bytes]:

@@ -77,8 +77,9 @@ class TransformerDetector(CoreDetector):
             raw_lines, self.config.batch_max_len, device=self.device
         )
         start = torch.zeros(len(raw_lines), dtype=torch.int32, device=self.device)
-        tokens = self.model.tokenize_spans(lines, start, lens.int())
-        return self.model(tokens)
+        # score_spans takes the fused whole-model kernel on GPU at the
+        # flagship geometry, layered kernels / CPU reference otherwise
+        return self.model.score_spans(lines, start, lens.int())
 
     def train(self, parsed_batch: List[ParserSchema]) -> None:
         raws = [(p.log or "").encode() for p in parsed_batch]

@@ -87,13 +87,22 @@ class GpuPipeline:
         Returns event_id [B], anomaly [B] (bool), scores [B] (f32, 0 when
         transformer off), nv_unseen [B, W]."""
         B = lines.shape[0]
+        # roctx-visible stage ranges (torch.cuda.nvtx maps to rocTracer
+        # markers on ROCm — SURVEY.md §5.1 tracing requirement)
+        _rng = torch.cuda.nvtx.range if lines.is_cuda else None
+        if _rng:
+            torch.cuda.nvtx.range_push("dmx::parse")
         match = self.matcher.match_packed(lines, line_len)
+        if _rng:
+            torch.cuda.nvtx.range_pop()
 
         n_train_left = max(0, self.config.train_lines - self.seen_lines)
         train_upto = min(B, n_train_left)
         self.seen_lines += B
 
         nv_unseen = None
+        if _rng:
+            torch.cuda.nvtx.range_push("dmx::new_value")
         if self.hashsets is not None and self.specs.shape[0] > 0:
             hashes = ops.watch_hashes(lines, match, self.specs, self.config.lowercase)
             if train_upto > 0:
@@ -109,7 +118,12 @@ class GpuPipeline:
                     (B, self.specs.shape[0]), dtype=torch.int32, device=lines.device
                 )
 
+        if _rng:
+            torch.cuda.nvtx.range_pop()
+
         scores = torch.zeros(B, dtype=torch.float32, device=lines.device)
+        if _rng:
+            torch.cuda.nvtx.range_push("dmx::transformer")
         if self.model is not None:
             # content span: last fmt capture when the header matched,
             # else the whole line
@@ -128,6 +142,8 @@ class GpuPipeline:
                 line_len.long(),
             )
             scores = self.model.score_spans(lines, start.int(), end.int())
+        if _rng:
+            torch.cuda.nvtx.range_pop()
 
         anomaly = scores > self.config.score_threshold
         if nv_unseen is not None:

@@ -115,3 +115,8 @@ def test_random_detector_threshold_extremes():
     assert all(x is None for x in never.process_batch([_parsed()] * 20))
     always = RandomDetector({"params": {"seed": 1, "threshold": -0.1}})
     assert all(x is not None for x in always.process_batch([_parsed()] * 20))
+
+
+def test_method_type_mismatch_raises():
+    with pytest.raises(ValueError):
+        NewValueDetector({"method_type": "wrong_type"})

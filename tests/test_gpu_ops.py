@@ -276,3 +276,30 @@ def test_bert_fused_vs_layered():
     cpu = BertTinyDetectorModel(BertTinyConfig(), device="cpu", seed=13)
     ref = cpu.forward(tokens.cpu())
     assert (fused.cpu() - ref).abs().max() < 0.08
+
+
+def test_multi_gpu_dp_pipeline():
+    """Single-node multi-GPU DP (SURVEY §4.2's missing 4th tier); skips on
+    1-GPU boxes — the 8-GPU scaling bench is driver-run."""
+    if torch.cuda.device_count() < 2:
+        pytest.skip("needs >=2 GPUs")
+    from detectmateservice_amd.pipeline import GpuPipeline, PipelineConfig
+    from detectmateservice_amd.utils.synthetic import (
+        AUDIT_LOG_FORMAT,
+        AUDIT_TEMPLATES,
+        AuditLogGenerator,
+    )
+    from detectmateservice_amd import ops
+
+    outs = []
+    for dev in ("cuda:0", "cuda:1"):
+        cfg = PipelineConfig(
+            templates=AUDIT_TEMPLATES, log_format=AUDIT_LOG_FORMAT,
+            use_transformer=True, train_lines=0, seed=1,
+        )
+        pipe = GpuPipeline(cfg, device=dev)
+        gen = AuditLogGenerator(seed=3)
+        raw = [gen.line()[0].encode() for _ in range(64)]
+        out = pipe.process_lines(raw)
+        outs.append(out["scores"].cpu())
+    assert torch.allclose(outs[0], outs[1], atol=1e-3)
