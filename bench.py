@@ -42,8 +42,8 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--batch", type=int, default=8192, help="lines per rank per step")
-    p.add_argument("--pool", type=int, default=8, help="distinct pre-generated batches")
+    p.add_argument("--batch", type=int, default=32768, help="lines per rank per step")
+    p.add_argument("--pool", type=int, default=4, help="distinct pre-generated batches")
     p.add_argument("--max-len", type=int, default=256)
     p.add_argument("--no-transformer", action="store_true")
     p.add_argument("--device", default=None)
