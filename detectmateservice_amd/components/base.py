@@ -107,11 +107,16 @@ class CoreDetectorConfig(CoreConfig):
     """Detector base config (reference docs/interfaces.md:139-167).
 
     ``data_use_training: N`` = train on the first N lines then switch to
-    detect (reference docs/getting_started.md:426-435).
+    detect (reference docs/getting_started.md:426-435). ``buffer_mode``
+    (interfaces.md:167): NO_BUF detects per line; FIXED collects
+    ``buffer_size`` lines and detects once per full window; SLIDING keeps
+    the last ``buffer_size`` lines as context and detects per line once
+    the window is primed.
     """
 
     data_use_training: int = 0
     buffer_mode: BufferMode = BufferMode.NO_BUF
+    buffer_size: int = 16
 
 
 class CoreDetector(CoreComponent):
@@ -128,6 +133,7 @@ class CoreDetector(CoreComponent):
     def __init__(self, config: Union[Dict[str, Any], CoreConfig, None] = None) -> None:
         super().__init__(config)
         self._seen_lines = 0
+        self._buffer: List[Any] = []  # FIXED/SLIDING window of ParserSchema
 
     # subclass API ------------------------------------------------------
     def train(self, parsed_batch: List[Any]) -> None:  # List[ParserSchema]
@@ -135,6 +141,11 @@ class CoreDetector(CoreComponent):
 
     def detect(self, parsed: Any, alert: Any) -> bool:  # (ParserSchema, DetectorSchema)
         raise NotImplementedError
+
+    def detect_window(self, window: List[Any], alert: Any) -> bool:
+        """Windowed detection hook (FIXED/SLIDING buffer modes). Default:
+        per-line detect on the newest window entry."""
+        return self.detect(window[-1], alert)
 
     # framework-provided plumbing --------------------------------------
     def process(self, data: bytes) -> Optional[bytes]:
@@ -154,8 +165,28 @@ class CoreDetector(CoreComponent):
             self.train(parsed[:train_upto])
         self._seen_lines += len(parsed)
 
+        mode = getattr(self.config, "buffer_mode", BufferMode.NO_BUF)
+        if isinstance(mode, str):
+            mode = BufferMode(mode)
+        size = max(1, int(getattr(self.config, "buffer_size", 16)))
+
         for i in range(train_upto, len(parsed)):
             alert = DetectorSchema()
-            if self.detect(parsed[i], alert):
-                results[i] = alert.serialize()
+            if mode == BufferMode.NO_BUF:
+                if self.detect(parsed[i], alert):
+                    results[i] = alert.serialize()
+            elif mode == BufferMode.FIXED:
+                self._buffer.append(parsed[i])
+                if len(self._buffer) >= size:
+                    window, self._buffer = self._buffer, []
+                    if self.detect_window(window, alert):
+                        results[i] = alert.serialize()
+            else:  # SLIDING
+                self._buffer.append(parsed[i])
+                if len(self._buffer) > size:
+                    self._buffer.pop(0)
+                if len(self._buffer) == size and self.detect_window(
+                    list(self._buffer), alert
+                ):
+                    results[i] = alert.serialize()
         return results

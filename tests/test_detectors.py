@@ -120,3 +120,46 @@ def test_random_detector_threshold_extremes():
 def test_method_type_mismatch_raises():
     with pytest.raises(ValueError):
         NewValueDetector({"method_type": "wrong_type"})
+
+
+def test_buffer_modes():
+    """FIXED emits once per full window; SLIDING detects per line once
+    primed (reference BufferMode, interfaces.md:167)."""
+    from detectmateservice_amd.components.base import CoreDetector, CoreDetectorConfig
+
+    class WindowCounter(CoreDetector):
+        CONFIG_CLASS = CoreDetectorConfig
+
+        def __init__(self, config=None):
+            super().__init__(config)
+            self.windows = []
+
+        def train(self, batch):
+            pass
+
+        def detect(self, parsed, alert):
+            alert.description = "line"
+            return True
+
+        def detect_window(self, window, alert):
+            self.windows.append([p.logID for p in window])
+            alert.description = f"window of {len(window)}"
+            return True
+
+    frames = [_parsed(log_id=f"l{i}") for i in range(8)]
+
+    fixed = WindowCounter({"buffer_mode": "fixed", "buffer_size": 4})
+    out = fixed.process_batch(frames)
+    assert [o is not None for o in out] == [False] * 3 + [True] + [False] * 3 + [True]
+    assert fixed.windows == [["l0", "l1", "l2", "l3"], ["l4", "l5", "l6", "l7"]]
+
+    sliding = WindowCounter({"buffer_mode": "sliding", "buffer_size": 4})
+    out = sliding.process_batch(frames)
+    assert [o is not None for o in out] == [False] * 3 + [True] * 5
+    assert sliding.windows[0] == ["l0", "l1", "l2", "l3"]
+    assert sliding.windows[-1] == ["l4", "l5", "l6", "l7"]
+
+    nobuf = WindowCounter({})
+    out = nobuf.process_batch(frames)
+    assert all(o is not None for o in out)
+    assert nobuf.windows == []
