@@ -36,7 +36,11 @@ from ...schemas import LogSchema, ParserSchema
 
 class MatcherParserConfig(CoreConfig):
     method_type: str = "matcher_parser"
+    #: mine templates from the first ``auto_config_lines`` lines instead of
+    #: requiring a template file (reference parser_config.yaml:4 flag; the
+    #: mining itself runs on the GPU edit-distance kernel — template_miner.py)
     auto_config: bool = False
+    auto_config_lines: int = 1000
     log_format: Optional[str] = None
     time_format: Optional[str] = None
     params: Dict = {}
@@ -131,6 +135,11 @@ class MatcherParser(CoreComponent):
         self.templates = templates
         self._segments = [split_template(t) for t in templates]
         self.parser_id = f"matcher_parser-{id(self):x}"
+        # auto_config: buffer content lines until the mining threshold,
+        # then derive templates (template_miner.py) and start matching.
+        self._auto_pending: Optional[List[str]] = (
+            [] if (cfg.auto_config and not templates) else None
+        )
         # Which extracted header field carries the free-text content that
         # template matching applies to. Defaults to the LAST token of the
         # log_format (e.g. ``<Content>`` in the audit format); when the
@@ -150,6 +159,38 @@ class MatcherParser(CoreComponent):
         if self._remove_spaces:
             content = content.replace(" ", "")
         return content
+
+    # -- auto_config mining --------------------------------------------
+    def _auto_observe(self, lines: List[str]) -> None:
+        """Collect content lines; once the threshold is reached, mine
+        templates and switch to matching mode."""
+        if self._auto_pending is None:
+            return
+        for line in lines:
+            if self._format_re is not None:
+                m = self._format_re.match(line)
+                if m and self._content_field and self._content_field in m.groupdict():
+                    line = m.group(self._content_field)
+            self._auto_pending.append(self._normalize(line))
+        if len(self._auto_pending) >= self.config.auto_config_lines:
+            self.mine_templates()
+
+    def mine_templates(self) -> List[str]:
+        """Run the miner on buffered lines and install the templates."""
+        from .template_miner import TemplateMiner
+        import torch
+
+        pending = self._auto_pending or []
+        self._auto_pending = None
+        device = (self.config.params or {}).get("device")
+        if device is None:
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+        miner = TemplateMiner(device=device)
+        mined = miner.fit(pending)
+        self.templates = mined
+        self._segments = [split_template(t) for t in mined]
+        self._batch_matcher = None  # rebuild packed tables on next batch
+        return mined
 
     # -- per-line parse (semantic reference for the GPU kernel) ---------
     def parse_line(self, line: str) -> Tuple[Dict[str, str], int, str, List[str]]:
@@ -177,6 +218,10 @@ class MatcherParser(CoreComponent):
         encode. Falls back to the per-message Python path when the
         extension is absent or normalization params the kernel does not
         implement (remove_spaces / remove_punctuation) are set."""
+        if self._auto_pending is not None:
+            self._auto_observe(
+                [LogSchema.deserialize(f).log for f in frames]
+            )
         if len(frames) >= 8 and self._remove_spaces is False and self._remove_punct is False:
             try:
                 return self._process_batched(frames)

@@ -37,6 +37,8 @@ void dmx_launch_hashset_insert(const void*, void*, int, int, int,
                                hipStream_t);
 void dmx_launch_hashset_probe(const void*, const void*, int, int, int,
                               void*, hipStream_t);
+void dmx_launch_edit_distance(const void*, const void*, int, const void*,
+                              const void*, int, int, void*, hipStream_t);
 }
 
 namespace {
@@ -193,6 +195,21 @@ std::vector<torch::Tensor> template_match(
   return {event_id, fmt_caps, n_fmt_caps, caps, n_caps};
 }
 
+torch::Tensor edit_distance(torch::Tensor A, torch::Tensor a_len,
+                            torch::Tensor B, torch::Tensor b_len) {
+  TORCH_CHECK(A.is_cuda() && A.dtype() == torch::kUInt8 && A.is_contiguous());
+  TORCH_CHECK(B.is_cuda() && B.dtype() == torch::kUInt8 && B.is_contiguous());
+  TORCH_CHECK(A.size(1) == B.size(1), "A and B must share max_len");
+  TORCH_CHECK(A.size(1) <= 256, "edit_distance supports max_len<=256");
+  const auto Na = A.size(0), Nb = B.size(0), max_len = A.size(1);
+  auto dist = torch::empty(
+      {Na, Nb}, torch::TensorOptions().dtype(torch::kInt32).device(A.device()));
+  dmx_launch_edit_distance(A.data_ptr(), a_len.data_ptr(), (int)Na,
+                           B.data_ptr(), b_len.data_ptr(), (int)Nb,
+                           (int)max_len, dist.data_ptr(), cur_stream());
+  return dist;
+}
+
 torch::Tensor watch_hashes(torch::Tensor lines, torch::Tensor event_id,
                            torch::Tensor caps, torch::Tensor n_caps,
                            torch::Tensor fmt_caps, torch::Tensor n_fmt_caps,
@@ -249,4 +266,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("watch_hashes", &watch_hashes, "hash watched capture spans");
   m.def("hashset_insert", &hashset_insert, "insert hashes into GPU sets");
   m.def("hashset_probe", &hashset_probe, "probe hashes against GPU sets");
+  m.def("edit_distance", &edit_distance,
+        "batched Levenshtein distances (wavefront DP in LDS)");
 }

@@ -26,6 +26,7 @@ sources = [
     str(CSRC / "bert_fused.hip"),
     str(CSRC / "template_match.hip"),
     str(CSRC / "hashset.hip"),
+    str(CSRC / "edit_distance.hip"),
 ]
 
 setup(

@@ -303,3 +303,43 @@ def test_multi_gpu_dp_pipeline():
         out = pipe.process_lines(raw)
         outs.append(out["scores"].cpu())
     assert torch.allclose(outs[0], outs[1], atol=1e-3)
+
+
+def test_edit_distance_kernel_vs_python():
+    """Wavefront-DP Levenshtein kernel vs the CPU reference on random and
+    structured strings."""
+    import random
+
+    from detectmateservice_amd import ops
+    from detectmateservice_amd.library.parsers.template_miner import (
+        levenshtein_py,
+    )
+    from detectmateservice_amd.ops import _dmx_C
+
+    rng = random.Random(7)
+    alpha = b"abcdefgh "
+    qs, rs = [], []
+    for _ in range(12):
+        n = rng.randrange(0, 200)
+        qs.append(bytes(rng.choice(alpha) for _ in range(n)))
+    for _ in range(9):
+        n = rng.randrange(0, 250)
+        rs.append(bytes(rng.choice(alpha) for _ in range(n)))
+    qs.append(b"kitten"); rs.append(b"sitting")
+    qa, ql = ops.pack_lines(qs, 256, device="cuda")
+    ra, rl = ops.pack_lines(rs, 256, device="cuda")
+    d = _dmx_C.edit_distance(qa, ql, ra, rl).cpu()
+    for i, q in enumerate(qs):
+        for j, r in enumerate(rs):
+            assert int(d[i, j]) == levenshtein_py(q, r), (i, j, len(q), len(r))
+
+
+def test_template_miner_gpu_matches_cpu():
+    from detectmateservice_amd.library.parsers.template_miner import TemplateMiner
+    from detectmateservice_amd.utils.synthetic import AuditLogGenerator
+
+    gen = AuditLogGenerator(seed=5)
+    contents = [gen.line()[0].split("): ", 1)[1] for _ in range(200)]
+    cpu_t = TemplateMiner(device="cpu").fit(contents)
+    gpu_t = TemplateMiner(device="cuda").fit(contents)
+    assert cpu_t == gpu_t
