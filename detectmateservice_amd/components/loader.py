@@ -23,6 +23,24 @@ DEFAULT_ROOT = "detectmateservice_amd.library"
 BASE_PACKAGE = "detectmateservice_amd.library"
 
 
+def normalize_component_config(config):
+    """Library-side config pipeline (reference docs/interfaces.md:74-82):
+    flatten ``params`` into the top level (without removing the original
+    block) and strip the ``all_`` prefix from keys (an ``all_x`` key
+    applies ``x`` to every instance)."""
+    if not isinstance(config, dict):
+        return config
+    out = dict(config)
+    params = out.get("params")
+    if isinstance(params, dict):
+        for k, v in params.items():
+            out.setdefault(k, v)
+    for k in list(out.keys()):
+        if isinstance(k, str) and k.startswith("all_"):
+            out.setdefault(k[4:], out[k])
+    return out
+
+
 class ComponentLoadError(Exception):
     pass
 
@@ -56,6 +74,7 @@ class ComponentLoader:
             raise ComponentLoadError(
                 f"module {module.__name__!r} has no class {cls_name!r}"
             )
+        config = normalize_component_config(config)
         try:
             instance = cls(config=config) if config else cls()
         except TypeError as exc:

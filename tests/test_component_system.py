@@ -129,3 +129,20 @@ def test_config_manager_rejects_non_mapping(tmp_path):
     cfg_file.write_text("- just\n- a list\n")
     with pytest.raises(Exception):
         ConfigManager(cfg_file)
+
+
+def test_normalize_component_config():
+    """params flattening + all_ prefix stripping (reference
+    interfaces.md:74-82 config pipeline)."""
+    from detectmateservice_amd.components.loader import normalize_component_config
+
+    cfg = normalize_component_config({
+        "params": {"threshold": 0.5, "seed": 1},
+        "all_window": 8,
+        "threshold": 0.9,  # explicit top-level wins over params
+    })
+    assert cfg["threshold"] == 0.9
+    assert cfg["seed"] == 1
+    assert cfg["window"] == 8
+    assert cfg["all_window"] == 8  # original key kept
+    assert normalize_component_config(None) is None
