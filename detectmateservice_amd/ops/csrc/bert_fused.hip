@@ -25,6 +25,16 @@
 // (Dh=64), FFN=512, arbitrary layer count.
 #include "common.h"
 
+// Explicit LDS address-space typing: the noinline block_gemm receives the
+// arena pointers as arguments, and a generic (flat) short* there makes
+// hipcc emit flat_load for what are really LDS reads (35 flat_load in the
+// .s before this change). AS(3) pointers guarantee ds_* lowering.
+typedef __attribute__((address_space(3))) short lds_short;
+typedef __attribute__((address_space(1))) const short glob_cshort;
+typedef __attribute__((address_space(1))) const float glob_cfloat;
+typedef __attribute__((address_space(3))) const short lds_cshort;
+typedef __attribute__((address_space(3))) float lds_float;
+
 #define BF_WAVES 8
 #define BF_THREADS (BF_WAVES * DMX_WAVE)
 #define BF_S 64
@@ -64,9 +74,9 @@
 // transposed; MODE 2: x[m][n] += v (residual-accumulate). ACT 1 = GELU.
 template <int K, int N, int MODE, int ACT, int WTS>
 static __device__ __attribute__((noinline)) void block_gemm(
-    const short* in_lds, int in_stride, const short* __restrict__ Wt,
-    const float* __restrict__ bias, short* out_lds, int out_stride,
-    short* x_lds, short* vt_lds, int wid, int lane) {
+    const lds_short* in_lds, int in_stride, const glob_cshort* __restrict__ Wt,
+    const glob_cfloat* __restrict__ bias, lds_short* out_lds, int out_stride,
+    lds_short* x_lds, lds_short* vt_lds, int wid, int lane) {
   constexpr int N16 = N / 16;
   constexpr int TOTAL = 4 * N16;
   constexpr int KS = K / 32;
@@ -74,6 +84,11 @@ static __device__ __attribute__((noinline)) void block_gemm(
   static_assert(FPW >= 4 && FPW % 4 == 0, "quad-chunked assignment");
   // quad-chunk: each iteration owns a FULL fn column (all 4 m-fragments):
   // ONE L2 weight-fragment load feeds FOUR independent MFMA chains.
+  // unroll 2 quads: the second quad's (independent) weight loads issue
+  // under the first quad's MFMA chains, hiding the L2 latency that
+  // dominated the per-quad cost (phase probe: GEMMs ~3.9 of 4.5 ms while
+  // pure MFMA issue accounts for <10% of that)
+#pragma unroll 2
   for (int ff = wid * FPW; ff < wid * FPW + FPW; ff += 4) {
     const int fn = ff >> 2;
     f32x4 acc[4];
@@ -81,11 +96,11 @@ static __device__ __attribute__((noinline)) void block_gemm(
     for (int i = 0; i < 4; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll 4
     for (int ks = 0; ks < KS; ++ks) {
-      bf16x8 b = *(const bf16x8*)(
+      bf16x8 b = *(const __attribute__((address_space(1))) bf16x8*)(
           Wt + (long)(fn * 16 + (lane & 15)) * WTS + ks * 32 + (lane >> 4) * 8);
 #pragma unroll
       for (int fm = 0; fm < 4; ++fm) {
-        bf16x8 a = *(const bf16x8*)(
+        bf16x8 a = *(const __attribute__((address_space(3))) bf16x8*)(
             in_lds + (fm * 16 + (lane & 15)) * in_stride + ks * 32 +
             (lane >> 4) * 8);
         acc[fm] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[fm], 0, 0, 0);
@@ -118,29 +133,65 @@ static __device__ __attribute__((noinline)) void block_gemm(
 }
 
 // ---- in-block LayerNorm on x (post-LN) -----------------------------------
+// All 8 of the wave's rows in ONE parallel pass: 8 lanes per row, 16
+// elements per lane, 3-step shfl_xor reduction within each 8-lane row
+// group. (The serial per-row loop with full-wave reductions measured
+// +1.12 ms of the 5.2 ms step — as costly as the whole QKV GEMM.)
 static __device__ __forceinline__ void block_layernorm(
-    short* x_lds, const short* __restrict__ gamma,
+    lds_short* x_lds, const short* __restrict__ gamma,
     const short* __restrict__ beta, int wid, int lane, float eps) {
-  for (int row = wid * (BF_S / BF_WAVES); row < (wid + 1) * (BF_S / BF_WAVES);
-       ++row) {
-    const int c0 = lane * 2;  // 64 lanes x 2 elements = 128-wide row
-    float v0 = bf16_to_f32(x_lds[row * XS + c0]);
-    float v1 = bf16_to_f32(x_lds[row * XS + c0 + 1]);
-    float sum = warp_reduce_sum_f32(v0 + v1);
-    const float mean = __shfl(sum, 0, 64) / BF_H;
-    float var = (v0 - mean) * (v0 - mean) + (v1 - mean) * (v1 - mean);
-    var = warp_reduce_sum_f32(var);
-    const float rstd = rsqrtf(__shfl(var, 0, 64) / BF_H + eps);
-    x_lds[row * XS + c0] = f32_to_bf16(
-        (v0 - mean) * rstd * bf16_to_f32(gamma[c0]) + bf16_to_f32(beta[c0]));
-    x_lds[row * XS + c0 + 1] = f32_to_bf16(
-        (v1 - mean) * rstd * bf16_to_f32(gamma[c0 + 1]) +
-        bf16_to_f32(beta[c0 + 1]));
+  const int row = wid * 8 + (lane >> 3);
+  const int c0 = (lane & 7) * 16;
+  short8v va = *(const __attribute__((address_space(3))) short8v*)(x_lds + row * XS + c0);
+  short8v vb = *(const __attribute__((address_space(3))) short8v*)(x_lds + row * XS + c0 + 8);
+  float v[16];
+  float sum = 0.f;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    v[j] = bf16_to_f32(va[j]);
+    v[8 + j] = bf16_to_f32(vb[j]);
+    sum += v[j] + v[8 + j];
   }
+#pragma unroll
+  for (int mask = 1; mask < 8; mask <<= 1) sum += __shfl_xor(sum, mask, 64);
+  const float mean = sum / BF_H;
+  float var = 0.f;
+#pragma unroll
+  for (int j = 0; j < 16; ++j) {
+    const float d = v[j] - mean;
+    var += d * d;
+  }
+#pragma unroll
+  for (int mask = 1; mask < 8; mask <<= 1) var += __shfl_xor(var, mask, 64);
+  const float rstd = rsqrtf(var / BF_H + eps);
+  short8v ga = *(const short8v*)(gamma + c0);
+  short8v gb = *(const short8v*)(gamma + c0 + 8);
+  short8v ba = *(const short8v*)(beta + c0);
+  short8v bb = *(const short8v*)(beta + c0 + 8);
+  short8v oa, ob;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    oa[j] = f32_to_bf16((v[j] - mean) * rstd * bf16_to_f32(ga[j]) +
+                        bf16_to_f32(ba[j]));
+    ob[j] = f32_to_bf16((v[8 + j] - mean) * rstd * bf16_to_f32(gb[j]) +
+                        bf16_to_f32(bb[j]));
+  }
+  *(__attribute__((address_space(3))) short8v*)(x_lds + row * XS + c0) = oa;
+  *(__attribute__((address_space(3))) short8v*)(x_lds + row * XS + c0 + 8) = ob;
 }
 
-extern "C" __global__ __launch_bounds__(BF_THREADS, 4)
-void dmx_bert_fused_bf16(
+// Phase bits for the profiling probe (production uses PH_ALL; skipped
+// phases cannot be dead-code-eliminated backwards because every phase
+// ends in LDS stores).
+#define PH_QKV 1
+#define PH_ATTN 2
+#define PH_PROJ 4
+#define PH_LN 8
+#define PH_FFN 16
+#define PH_ALL 31
+
+template <int PHASES>
+static __device__ __forceinline__ void bert_fused_body(
     const unsigned char* __restrict__ lines,  // [B, max_len]
     const int* __restrict__ start,            // [B] content span start
     const int* __restrict__ end,              // [B] content span end
@@ -154,11 +205,12 @@ void dmx_bert_fused_bf16(
   const int wid = tid / DMX_WAVE;
   const int lane = tid % DMX_WAVE;
 
-  extern __shared__ __attribute__((aligned(16))) short smem[];
-  short* x_lds = smem;             // [64][XS]
-  short* buf = x_lds + BF_S * XS;  // BUF_ELEMS: QK | P+O | FFN-half
-  short* vt = buf + BUF_ELEMS;     // [128][VTS]
-  float* red = (float*)(vt + 2 * BF_DH * VTS);  // [128] pooling scratch
+  extern __shared__ __attribute__((aligned(16))) short smem_raw[];
+  lds_short* smem = (lds_short*)smem_raw;
+  lds_short* x_lds = smem;             // [64][XS]
+  lds_short* buf = x_lds + BF_S * XS;  // BUF_ELEMS: QK | P+O | FFN-half
+  lds_short* vt = buf + BUF_ELEMS;     // [128][VTS]
+  lds_float* red = (lds_float*)(vt + 2 * BF_DH * VTS);  // [128] pooling
 
   // ---- embed: x[s][c] = tok_emb[byte+3 or 0][c] + pos_emb[s][c] ----
   {
@@ -175,7 +227,7 @@ void dmx_bert_fused_bf16(
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         xv[j] = f32_to_bf16(bf16_to_f32(te[j]) + bf16_to_f32(pe[j]));
-      *(short8v*)(x_lds + s * XS + c) = xv;
+      *(__attribute__((address_space(3))) short8v*)(x_lds + s * XS + c) = xv;
     }
   }
   __syncthreads();
@@ -185,13 +237,14 @@ void dmx_bert_fused_bf16(
     const float* lf = fb + (long)layer * FB_SIZE;
 
     // ---- qkv: Q|K -> buf[64][QKS], V -> vt transposed ----
-    block_gemm<BF_H, 3 * BF_H, 1, 0, BF_H>(x_lds, XS, lw + LW_QKV,
-                                           lf + FB_BQKV, buf, QKS, x_lds, vt,
-                                           wid, lane);
+    if (PHASES & PH_QKV)
+      block_gemm<BF_H, 3 * BF_H, 1, 0, BF_H>(x_lds, XS, (glob_cshort*)(lw + LW_QKV),
+                                             (glob_cfloat*)(lf + FB_BQKV), buf, QKS, x_lds, vt,
+                                             wid, lane);
     __syncthreads();
 
     // ---- attention: wave = (head hh, 16 q-rows) ----
-    {
+    if (PHASES & PH_ATTN) {
       const int hh = wid >> 2;
       const int q0 = (wid & 3) * 16;
       const float scale = 0.125f;  // 1/sqrt(64)
@@ -201,12 +254,12 @@ void dmx_bert_fused_bf16(
       for (int f = 0; f < 4; ++f) acc_p[f] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int ks = 0; ks < BF_DH / 32; ++ks) {
-        bf16x8 a = *(const bf16x8*)(
+        bf16x8 a = *(const __attribute__((address_space(3))) bf16x8*)(
             buf + (q0 + (lane & 15)) * QKS + hh * BF_DH + ks * 32 +
             (lane >> 4) * 8);
 #pragma unroll
         for (int f = 0; f < 4; ++f) {
-          bf16x8 b = *(const bf16x8*)(
+          bf16x8 b = *(const __attribute__((address_space(3))) bf16x8*)(
               buf + (f * 16 + (lane & 15)) * QKS + BF_H + hh * BF_DH +
               ks * 32 + (lane >> 4) * 8);
           acc_p[f] =
@@ -236,7 +289,7 @@ void dmx_bert_fused_bf16(
       }
       // P tiles alias the Q|K area: every wave must be done reading Q/K
       __syncthreads();
-      short* my_p = buf + wid * 16 * VTS;
+      lds_short* my_p = buf + wid * 16 * VTS;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int row = (lane >> 4) * 4 + r;
@@ -251,11 +304,11 @@ void dmx_bert_fused_bf16(
       for (int f = 0; f < 4; ++f) acc_o[f] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int ks = 0; ks < BF_S / 32; ++ks) {
-        bf16x8 a = *(const bf16x8*)(
+        bf16x8 a = *(const __attribute__((address_space(3))) bf16x8*)(
             my_p + (lane & 15) * VTS + ks * 32 + (lane >> 4) * 8);
 #pragma unroll
         for (int f = 0; f < 4; ++f) {
-          bf16x8 b = *(const bf16x8*)(
+          bf16x8 b = *(const __attribute__((address_space(3))) bf16x8*)(
               vt + (hh * BF_DH + f * 16 + (lane & 15)) * VTS + ks * 32 +
               (lane >> 4) * 8);
           acc_o[f] =
@@ -278,29 +331,33 @@ void dmx_bert_fused_bf16(
     __syncthreads();
 
     // ---- proj: x += Wo(attn) ; LN1 ----
-    block_gemm<BF_H, BF_H, 2, 0, BF_H>(buf + O_OFF, XS, lw + LW_WO,
-                                       lf + FB_BO, nullptr, 0, x_lds, nullptr,
-                                       wid, lane);
+    if (PHASES & PH_PROJ)
+      block_gemm<BF_H, BF_H, 2, 0, BF_H>(buf + O_OFF, XS, (glob_cshort*)(lw + LW_WO),
+                                         (glob_cfloat*)(lf + FB_BO), nullptr, 0, x_lds,
+                                         nullptr, wid, lane);
     __syncthreads();
-    block_layernorm(x_lds, lw + LW_LN1G, lw + LW_LN1B, wid, lane, eps);
+    if (PHASES & PH_LN)
+      block_layernorm(x_lds, lw + LW_LN1G, lw + LW_LN1B, wid, lane, eps);
     __syncthreads();
 
     // ---- FFN in two K=256 halves: buf = gelu(x@W1_h); x += buf@W2_h ----
+    if (PHASES & PH_FFN)
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
       block_gemm<BF_H, BF_FFN / 2, 0, 1, BF_H>(
-          x_lds, XS, lw + LW_W1 + (long)h * (BF_FFN / 2) * BF_H,
-          lf + FB_B1 + h * (BF_FFN / 2), buf, QKS, nullptr, nullptr, wid,
-          lane);
+          x_lds, XS, (glob_cshort*)(lw + LW_W1 + (long)h * (BF_FFN / 2) * BF_H),
+          (glob_cfloat*)(lf + FB_B1 + h * (BF_FFN / 2)), buf, QKS, nullptr,
+          nullptr, wid, lane);
       __syncthreads();
       // bias b2 added once (half 0); half 1 adds only the partial product
       block_gemm<BF_FFN / 2, BF_H, 2, 0, BF_FFN>(
-          buf, QKS, lw + LW_W2 + (long)h * (BF_FFN / 2),
-          h == 0 ? lf + FB_B2 : nullptr, nullptr, 0, x_lds, nullptr, wid,
-          lane);
+          buf, QKS, (glob_cshort*)(lw + LW_W2 + (long)h * (BF_FFN / 2)),
+          h == 0 ? (glob_cfloat*)(lf + FB_B2) : nullptr, nullptr, 0, x_lds,
+          nullptr, wid, lane);
       __syncthreads();
     }
-    block_layernorm(x_lds, lw + LW_LN2G, lw + LW_LN2B, wid, lane, eps);
+    if (PHASES & PH_LN)
+      block_layernorm(x_lds, lw + LW_LN2G, lw + LW_LN2B, wid, lane, eps);
     __syncthreads();
   }
 
@@ -322,6 +379,63 @@ void dmx_bert_fused_bf16(
         scores[line] = v + fb[(long)n_layers * FB_SIZE];  // b_score
     }
   }
+}
+
+extern "C" __global__ __launch_bounds__(BF_THREADS, 4)
+void dmx_bert_fused_bf16(const unsigned char* __restrict__ lines,
+                         const int* __restrict__ start,
+                         const int* __restrict__ end,
+                         const short* __restrict__ wb,
+                         const float* __restrict__ fb,
+                         float* __restrict__ scores, int B, int max_len,
+                         int n_layers, float eps) {
+  bert_fused_body<PH_ALL>(lines, start, end, wb, fb, scores, B, max_len,
+                          n_layers, eps);
+}
+
+// probe variants (in-kernel phase ablation; guide §5.4 rule 19: co-compiled
+// variants can perturb codegen by a few % — read the deltas, not absolutes)
+template <int PHASES>
+__global__ __launch_bounds__(BF_THREADS, 4) void dmx_bert_fused_probe(
+    const unsigned char* lines, const int* start, const int* end,
+    const short* wb, const float* fb, float* scores, int B, int max_len,
+    int n_layers, float eps) {
+  bert_fused_body<PHASES>(lines, start, end, wb, fb, scores, B, max_len,
+                          n_layers, eps);
+}
+
+extern "C" void dmx_launch_bert_fused_probe(
+    const void* lines, const void* start, const void* end, const void* wb,
+    const void* fb, void* scores, int B, int max_len, int n_layers,
+    float eps, int phase_mask, hipStream_t stream) {
+  const size_t lds =
+      ((size_t)BF_S * XS + BUF_ELEMS + (size_t)2 * BF_DH * VTS) *
+          sizeof(short) +
+      128 * sizeof(float);
+#define LAUNCH_PROBE(MASK)                                                   \
+  case MASK:                                                                 \
+    hipFuncSetAttribute((const void*)dmx_bert_fused_probe<MASK>,             \
+                        hipFuncAttributeMaxDynamicSharedMemorySize,          \
+                        (int)lds);                                           \
+    hipLaunchKernelGGL(dmx_bert_fused_probe<MASK>, dim3(B),                  \
+                       dim3(BF_THREADS), lds, stream,                        \
+                       (const unsigned char*)lines, (const int*)start,       \
+                       (const int*)end, (const short*)wb, (const float*)fb,  \
+                       (float*)scores, B, max_len, n_layers, eps);           \
+    break;
+  switch (phase_mask) {
+    LAUNCH_PROBE(0)
+    LAUNCH_PROBE(PH_QKV)
+    LAUNCH_PROBE(PH_QKV | PH_ATTN)
+    LAUNCH_PROBE(PH_QKV | PH_ATTN | PH_PROJ)
+    LAUNCH_PROBE(PH_QKV | PH_ATTN | PH_PROJ | PH_LN)
+    LAUNCH_PROBE(PH_ALL)
+    LAUNCH_PROBE(PH_FFN)
+    LAUNCH_PROBE(PH_LN)
+    default:
+      break;
+  }
+#undef LAUNCH_PROBE
 }
 
 extern "C" void dmx_launch_bert_fused_bf16(

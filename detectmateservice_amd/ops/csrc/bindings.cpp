@@ -24,6 +24,9 @@ void dmx_launch_attention_mfma_bf16(const void*, void*, int, int, int, int,
 void dmx_launch_bert_fused_bf16(const void*, const void*, const void*,
                                 const void*, const void*, void*, int, int,
                                 int, float, hipStream_t);
+void dmx_launch_bert_fused_probe(const void*, const void*, const void*,
+                                 const void*, const void*, void*, int, int,
+                                 int, float, int, hipStream_t);
 void dmx_launch_template_match(const void*, const void*, int, int,
                                const void*, const void*, int, const void*,
                                int, const void*, const void*, int, int,
@@ -195,6 +198,21 @@ std::vector<torch::Tensor> template_match(
   return {event_id, fmt_caps, n_fmt_caps, caps, n_caps};
 }
 
+torch::Tensor bert_fused_probe(torch::Tensor lines, torch::Tensor start,
+                               torch::Tensor end, torch::Tensor wb,
+                               torch::Tensor fb, int64_t n_layers, double eps,
+                               int64_t phase_mask) {
+  const auto B = lines.size(0), max_len = lines.size(1);
+  auto scores = torch::empty(
+      {B}, torch::TensorOptions().dtype(torch::kFloat32).device(lines.device()));
+  dmx_launch_bert_fused_probe(lines.data_ptr(), start.data_ptr(),
+                              end.data_ptr(), wb.data_ptr(), fb.data_ptr(),
+                              scores.data_ptr(), (int)B, (int)max_len,
+                              (int)n_layers, (float)eps, (int)phase_mask,
+                              cur_stream());
+  return scores;
+}
+
 torch::Tensor edit_distance(torch::Tensor A, torch::Tensor a_len,
                             torch::Tensor B, torch::Tensor b_len) {
   TORCH_CHECK(A.is_cuda() && A.dtype() == torch::kUInt8 && A.is_contiguous());
@@ -262,6 +280,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "MFMA MHA reading fused QKV layout [B,S,3*H*Dh] -> [B,S,H*Dh]");
   m.def("bert_fused_bf16", &bert_fused_bf16,
         "whole-model BERT-tiny forward, one workgroup per line");
+  m.def("bert_fused_probe", &bert_fused_probe,
+        "phase-masked probe variant of the fused BERT kernel");
   m.def("template_match", &template_match, "batched wildcard template match");
   m.def("watch_hashes", &watch_hashes, "hash watched capture spans");
   m.def("hashset_insert", &hashset_insert, "insert hashes into GPU sets");
