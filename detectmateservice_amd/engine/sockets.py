@@ -20,8 +20,12 @@ same capability surface as the reference's socket layer without NNG:
   (``engine_buffer_size``, engine.py:153-154) and report send failure so the
   engine can retry-then-drop (engine.py:281-301).
 
-Wire framing: 4-byte big-endian length prefix + payload. This framing plus
-the proto3 schemas in ``detectmateservice_amd.schemas`` is the complete wire
+Wire framing: ``tcp://`` and ``tls+tcp://`` speak the NNG SP mapping
+(8-octet ``\x00SP\x00`` + pair0 protocol-id connection header, then
+64-bit BE length-prefixed messages) so NNG/pynng/fluentd-nng peers
+interoperate at the pipeline edges; ``ipc://`` and ``inproc://`` are
+intra-node and use a compact 4-byte length prefix. These framings plus the
+proto3 schemas in ``detectmateservice_amd.schemas`` are the complete wire
 contract between stages.
 """
 from __future__ import annotations
@@ -41,6 +45,20 @@ from ..settings import EngineAddr, TlsInputConfig, TlsOutputConfig
 
 MAX_FRAME_BYTES = 64 * 1024 * 1024
 _LEN = struct.Struct(">I")
+
+#: NNG/nanomsg SP wire compatibility for tcp:// and tls+tcp:// (the
+#: reference's fluentd plugins dial NNG pair0 over TCP — SURVEY.md §2.4):
+#: connections open with the 8-octet SP header ``\x00SP\x00`` + 16-bit BE
+#: protocol id + 2 reserved octets, and every message is prefixed with a
+#: 64-bit BE length (SP TCP mapping). ipc:// and inproc:// are intra-node
+#: and use this framework's compact 4-byte framing.
+SP_PAIR0_PROTO = 0x10
+_SP_HEADER = struct.Struct(">ccBBHH")  # \x00, S, P, ver, proto, reserved
+_LEN64 = struct.Struct(">Q")
+
+
+def _sp_header_bytes(proto: int = SP_PAIR0_PROTO) -> bytes:
+    return b"\x00SP\x00" + struct.pack(">H", proto) + b"\x00\x00"
 
 
 class RecvTimeout(Exception):
@@ -65,10 +83,13 @@ class EngineSocket(Protocol):
 # ---------------------------------------------------------------------------
 
 
-def _send_frame(sock: socket.socket, data: bytes) -> None:
+def _send_frame(sock: socket.socket, data: bytes, sp: bool = False) -> None:
     if len(data) > MAX_FRAME_BYTES:
         raise ValueError(f"frame of {len(data)} bytes exceeds MAX_FRAME_BYTES")
-    sock.sendall(_LEN.pack(len(data)) + data)
+    if sp:
+        sock.sendall(_LEN64.pack(len(data)) + data)
+    else:
+        sock.sendall(_LEN.pack(len(data)) + data)
 
 
 def _recv_exact(sock: socket.socket, n: int) -> Optional[bytes]:
@@ -84,16 +105,42 @@ def _recv_exact(sock: socket.socket, n: int) -> Optional[bytes]:
     return bytes(buf)
 
 
-def _recv_frame(sock: socket.socket) -> Optional[bytes]:
-    header = _recv_exact(sock, 4)
-    if header is None:
-        return None
-    (length,) = _LEN.unpack(header)
+def _recv_frame(sock: socket.socket, sp: bool = False) -> Optional[bytes]:
+    if sp:
+        header = _recv_exact(sock, 8)
+        if header is None:
+            return None
+        (length,) = _LEN64.unpack(header)
+    else:
+        header = _recv_exact(sock, 4)
+        if header is None:
+            return None
+        (length,) = _LEN.unpack(header)
     if length > MAX_FRAME_BYTES:
         raise ValueError(f"peer announced oversize frame ({length} bytes)")
     if length == 0:
         return b""
     return _recv_exact(sock, length)
+
+
+def _sp_handshake(sock: socket.socket, logger: logging.Logger) -> bool:
+    """Exchange the 8-octet SP connection header (both sides send theirs;
+    any peer protocol id is accepted — pair0 expected)."""
+    try:
+        sock.sendall(_sp_header_bytes())
+        peer = _recv_exact(sock, 8)
+        if peer is None:
+            return False
+        if peer[:4] != b"\x00SP\x00":
+            logger.warning("peer sent a non-SP header %r; closing", peer)
+            return False
+        proto = int.from_bytes(peer[4:6], "big")
+        if proto != SP_PAIR0_PROTO:
+            logger.debug("peer SP protocol id %d (expected pair0=%d)",
+                         proto, SP_PAIR0_PROTO)
+        return True
+    except (OSError, ssl.SSLError):
+        return False
 
 
 # ---------------------------------------------------------------------------
@@ -136,6 +183,8 @@ class PairListener:
         self._ssl_ctx: Optional[ssl.SSLContext] = None
 
         scheme, rest = self.addr.scheme, self.addr.rest
+        #: SP (NNG pair0) wire mapping on tcp/tls+tcp edges
+        self._sp = scheme in ("tcp", "tls+tcp")
         if scheme == "inproc":
             raise ValueError("use InprocListener for inproc:// addresses")
         if scheme == "ipc":
@@ -193,6 +242,9 @@ class PairListener:
                     self._log.warning("TLS handshake failed: %s", exc)
                     conn.close()
                     continue
+            if self._sp and not _sp_handshake(conn, self._log):
+                conn.close()
+                continue
             with self._peers_lock:
                 self._peers.append(conn)
             threading.Thread(
@@ -203,7 +255,7 @@ class PairListener:
     def _reader_loop(self, conn: socket.socket) -> None:
         try:
             while not self._closed.is_set():
-                frame = _recv_frame(conn)
+                frame = _recv_frame(conn, self._sp)
                 if frame is None:
                     break
                 self._recv_q.put((conn, frame))
@@ -268,7 +320,7 @@ class PairListener:
         if conn is None:
             return False
         try:
-            _send_frame(conn, data)
+            _send_frame(conn, data, self._sp)
             return True
         except OSError:
             return False
@@ -373,6 +425,7 @@ class PairDialer:
         self._tls_config = tls_config
         self._conn: Optional[socket.socket] = None
         self._inproc: Optional[InprocListener] = None
+        self._sp = self.addr.scheme in ("tcp", "tls+tcp")
         if self.addr.scheme == "inproc":
             # resolved lazily in the worker loop so late binding works
             pass
@@ -405,6 +458,9 @@ class PairDialer:
                     s = ctx.wrap_socket(
                         s, server_hostname=self._tls_config.server_name
                     )
+                if self._sp and not _sp_handshake(s, self._log):
+                    s.close()
+                    return False
             s.settimeout(None)
             self._conn = s
             threading.Thread(
@@ -418,7 +474,7 @@ class PairDialer:
     def _reader_loop(self, conn: socket.socket) -> None:
         try:
             while not self._closed.is_set():
-                frame = _recv_frame(conn)
+                frame = _recv_frame(conn, self._sp)
                 if frame is None:
                     break
                 try:
@@ -473,7 +529,7 @@ class PairDialer:
                 self._requeue_front(data)
                 continue
             try:
-                _send_frame(conn, data)
+                _send_frame(conn, data, self._sp)
             except OSError:
                 self._log.debug("send to %s failed; reconnecting", self.addr)
                 try:

@@ -161,3 +161,53 @@ def test_factory_scheme_validation():
         factory.create("bogus://whatever")
     with pytest.raises(ValueError):
         factory.create("tcp://127.0.0.1")  # missing port
+
+
+def test_sp_wire_format_on_tcp():
+    """tcp:// speaks the NNG SP mapping: 8-octet \\x00SP\\x00 header with
+    pair0 proto id, then 64-bit BE length-prefixed messages (fluentd
+    interop edge)."""
+    import socket
+    import struct
+
+    listener = PairListener("tcp://127.0.0.1:0")
+    port = listener.bound_port
+    raw = socket.create_connection(("127.0.0.1", port), timeout=5.0)
+    try:
+        # we act as a raw NNG-style peer
+        raw.sendall(b"\x00SP\x00" + struct.pack(">H", 0x10) + b"\x00\x00")
+        hdr = b""
+        while len(hdr) < 8:
+            hdr += raw.recv(8 - len(hdr))
+        assert hdr[:4] == b"\x00SP\x00"
+        assert struct.unpack(">H", hdr[4:6])[0] == 0x10
+        payload = b"from-nng-peer"
+        raw.sendall(struct.pack(">Q", len(payload)) + payload)
+        assert listener.recv(timeout_ms=5000) == payload
+        listener.send(b"reply")
+        resp = b""
+        while len(resp) < 8 + 5:
+            chunk = raw.recv(64)
+            assert chunk
+            resp += chunk
+        assert struct.unpack(">Q", resp[:8])[0] == 5
+        assert resp[8:] == b"reply"
+    finally:
+        raw.close()
+        listener.close()
+
+
+def test_non_sp_peer_rejected_on_tcp():
+    listener = PairListener("tcp://127.0.0.1:0")
+    port = listener.bound_port
+    import socket
+
+    raw = socket.create_connection(("127.0.0.1", port), timeout=5.0)
+    try:
+        raw.sendall(b"GET / HTTP/1.1\r\n\r\n")  # not an SP header
+        time.sleep(0.3)
+        with pytest.raises(RecvTimeout):
+            listener.recv(timeout_ms=200)
+    finally:
+        raw.close()
+        listener.close()
