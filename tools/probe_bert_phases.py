@@ -1,3 +1,5 @@
+import sys, os
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch, time
 from detectmateservice_amd import ops
 from detectmateservice_amd.models.bert_tiny import BertTinyDetectorModel, BertTinyConfig
