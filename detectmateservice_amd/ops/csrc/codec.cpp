@@ -198,6 +198,231 @@ py::list encode_parser_batch(
   return out;
 }
 
+// ---- CPU template matcher (same algorithm as template_match.hip) ----------
+
+// Greedy-anchored wildcard match over [start, end) of `line`; returns
+// capture count or -1. Mirrors match_segments in template_match.hip and
+// _span_match in ops/__init__.py exactly.
+int match_segments_cpu(const unsigned char* line, int start, int end,
+                       const unsigned char* seg_bytes, const int32_t* seg_off,
+                       int s_begin, int s_end, int32_t* caps, int max_caps,
+                       bool lower) {
+  int pos = start;
+  int ncap = 0;
+  const int nseg = s_end - s_begin;
+  for (int i = 0; i < nseg; ++i) {
+    const int so = seg_off[s_begin + i];
+    const int sl = seg_off[s_begin + i + 1] - so;
+    if (sl == 0) {
+      if (i == nseg - 1) {
+        if (ncap < max_caps) {
+          caps[ncap * 2] = pos;
+          caps[ncap * 2 + 1] = end;
+        }
+        return ncap + 1;
+      }
+      continue;
+    }
+    // find seg in line[pos, end)
+    int idx = -1;
+    for (int cand = pos; cand + sl <= end; ++cand) {
+      bool ok = true;
+      for (int b = 0; b < sl; ++b) {
+        unsigned char c = line[cand + b];
+        if (lower && c >= 'A' && c <= 'Z') c += 32;
+        if (c != seg_bytes[so + b]) { ok = false; break; }
+      }
+      if (ok) { idx = cand; break; }
+    }
+    if (idx < 0) return -1;
+    if (i == 0 && idx != start) return -1;
+    if (i > 0) {
+      if (ncap < max_caps) {
+        caps[ncap * 2] = pos;
+        caps[ncap * 2 + 1] = idx;
+      }
+      ++ncap;
+    }
+    pos = idx + sl;
+  }
+  if (pos != end) return -1;
+  return ncap;
+}
+
+// CPU twin of the dmx_template_match kernel, identical outputs (the CPU
+// service path was Python-matcher-bound at ~9.5k lines/s).
+std::vector<torch::Tensor> template_match_cpu(
+    torch::Tensor lines, torch::Tensor line_len, torch::Tensor fmt_bytes,
+    torch::Tensor fmt_seg_off, torch::Tensor seg_bytes, torch::Tensor seg_off,
+    torch::Tensor tpl_seg_start, bool lower, int64_t max_fmt_caps,
+    int64_t max_caps) {
+  TORCH_CHECK(!lines.is_cuda() && lines.dtype() == torch::kUInt8 &&
+              lines.is_contiguous());
+  const int64_t B = lines.size(0), max_len = lines.size(1);
+  const int nf_seg =
+      fmt_seg_off.numel() > 0 ? (int)fmt_seg_off.numel() - 1 : 0;
+  const int n_tpl = (int)tpl_seg_start.numel() - 1;
+  auto opts = torch::TensorOptions().dtype(torch::kInt32);
+  auto event_id = torch::empty({B}, opts);
+  auto fmt_caps = torch::zeros({B, max_fmt_caps, 2}, opts);
+  auto n_fmt_caps = torch::zeros({B}, opts);
+  auto caps = torch::zeros({B, max_caps, 2}, opts);
+  auto n_caps = torch::zeros({B}, opts);
+
+  const unsigned char* lbuf = lines.data_ptr<uint8_t>();
+  const int32_t* lenp = line_len.data_ptr<int32_t>();
+  const unsigned char* fb =
+      nf_seg ? fmt_bytes.data_ptr<uint8_t>() : nullptr;
+  const int32_t* fo = nf_seg ? fmt_seg_off.data_ptr<int32_t>() : nullptr;
+  const unsigned char* sb = seg_bytes.data_ptr<uint8_t>();
+  const int32_t* so = seg_off.data_ptr<int32_t>();
+  const int32_t* ts = tpl_seg_start.data_ptr<int32_t>();
+  int32_t* evp = event_id.data_ptr<int32_t>();
+  int32_t* fcp = fmt_caps.data_ptr<int32_t>();
+  int32_t* nfp = n_fmt_caps.data_ptr<int32_t>();
+  int32_t* cp = caps.data_ptr<int32_t>();
+  int32_t* ncp = n_caps.data_ptr<int32_t>();
+
+  for (int64_t i = 0; i < B; ++i) {
+    const unsigned char* line = lbuf + i * max_len;
+    const int len = std::min((int)lenp[i], (int)max_len);
+    int content_start = 0, content_end = len;
+    int nfc = 0;
+    if (nf_seg > 0) {
+      nfc = match_segments_cpu(line, 0, len, fb, fo, 0, nf_seg,
+                               fcp + i * max_fmt_caps * 2, (int)max_fmt_caps,
+                               false);
+      if (nfc > 0) {
+        const int last = std::min<int>(nfc, (int)max_fmt_caps) - 1;
+        content_start = fcp[i * max_fmt_caps * 2 + last * 2];
+        content_end = fcp[i * max_fmt_caps * 2 + last * 2 + 1];
+      } else {
+        nfc = 0;
+      }
+    }
+    nfp[i] = nfc;
+    int eid = -1, nc = 0;
+    for (int t = 0; t < n_tpl; ++t) {
+      const int r = match_segments_cpu(line, content_start, content_end, sb,
+                                       so, ts[t], ts[t + 1], cp + i * max_caps * 2,
+                                       (int)max_caps, lower);
+      if (r >= 0) {
+        eid = t + 1;
+        nc = r;
+        break;
+      }
+    }
+    evp[i] = eid;
+    ncp[i] = nc;
+  }
+  return {event_id, fmt_caps, n_fmt_caps, caps, n_caps};
+}
+
+// ---- batched ParserSchema watch-hash extraction ---------------------------
+
+inline uint64_t fnv1a64_host(const char* p, size_t n, bool lower) {
+  uint64_t h = 1469598103934665603ull;
+  for (size_t i = 0; i < n; ++i) {
+    unsigned char c = (unsigned char)p[i];
+    if (lower && c >= 'A' && c <= 'Z') c += 32;
+    h ^= (uint64_t)c;
+    h *= 1099511628211ull;
+  }
+  return h | 1ull;
+}
+
+// For each frame: EventID + FNV-1a hashes of the watched fields, without
+// building any per-message Python object (the NewValueDetector service
+// path's hot loop). var_specs = [(event, pos)] index `variables` (field 6)
+// by position; hdr_names name keys of `logFormatVariables` (field 10).
+// Returns (hashes i64 [B, Wv+Wh], event_ids i32 [B], logIDs list).
+py::tuple parser_watch_hashes(
+    const std::vector<py::bytes>& frames,
+    const std::vector<std::pair<int64_t, int64_t>>& var_specs,
+    const std::vector<std::string>& hdr_names, bool lower) {
+  const int64_t B = (int64_t)frames.size();
+  const int64_t Wv = (int64_t)var_specs.size();
+  const int64_t Wh = (int64_t)hdr_names.size();
+  auto hashes = torch::zeros({B, Wv + Wh}, torch::kInt64);
+  auto event_ids = torch::zeros({B}, torch::kInt32);
+  int64_t* hp = hashes.data_ptr<int64_t>();
+  int32_t* ep = event_ids.data_ptr<int32_t>();
+  py::list log_ids;
+
+  std::vector<std::pair<const char*, size_t>> variables;
+  for (int64_t i = 0; i < B; ++i) {
+    char* fptr;
+    Py_ssize_t flen;
+    PyBytes_AsStringAndSize(frames[i].ptr(), &fptr, &flen);
+    const uint8_t* p = (const uint8_t*)fptr;
+    const size_t n = (size_t)flen;
+    size_t pos = 0;
+    int32_t event_id = 0;
+    std::string log_id;
+    variables.clear();
+    while (pos < n) {
+      uint64_t key;
+      if (!get_varint(p, n, pos, key)) break;
+      const int field = (int)(key >> 3), wt = (int)(key & 7);
+      if (field == 4 && wt == 0) {
+        uint64_t v;
+        if (!get_varint(p, n, pos, v)) break;
+        event_id = (int32_t)(uint32_t)v;
+      } else if (wt == 2) {
+        uint64_t sl;
+        if (!get_varint(p, n, pos, sl) || pos + sl > n) break;
+        const char* s = (const char*)(p + pos);
+        if (field == 6) {
+          variables.emplace_back(s, (size_t)sl);
+        } else if (field == 8) {
+          log_id.assign(s, sl);
+        } else if (field == 10 && Wh > 0) {
+          // map entry {1: key, 2: value}
+          size_t epos = 0;
+          const uint8_t* e = (const uint8_t*)s;
+          std::pair<const char*, size_t> k{nullptr, 0}, v{nullptr, 0};
+          while (epos < sl) {
+            uint64_t ekey;
+            if (!get_varint(e, sl, epos, ekey)) break;
+            if ((ekey & 7) == 2) {
+              uint64_t el;
+              if (!get_varint(e, sl, epos, el) || epos + el > sl) break;
+              if ((ekey >> 3) == 1) k = {(const char*)e + epos, (size_t)el};
+              else if ((ekey >> 3) == 2) v = {(const char*)e + epos, (size_t)el};
+              epos += el;
+            } else if (!skip_field(e, sl, epos, (int)(ekey & 7))) {
+              break;
+            }
+          }
+          if (k.first) {
+            for (int64_t w = 0; w < Wh; ++w) {
+              const std::string& name = hdr_names[w];
+              if (k.second == name.size() &&
+                  std::memcmp(k.first, name.data(), k.second) == 0 && v.first)
+                hp[i * (Wv + Wh) + Wv + w] =
+                    (int64_t)fnv1a64_host(v.first, v.second, lower);
+            }
+          }
+        }
+        pos += sl;
+      } else {
+        if (!skip_field(p, n, pos, wt)) break;
+      }
+    }
+    ep[i] = event_id;
+    for (int64_t w = 0; w < Wv; ++w) {
+      const auto& sp = var_specs[w];
+      if (sp.first >= 0 && (int32_t)sp.first != event_id) continue;
+      const int64_t vi = sp.second;
+      if (vi >= 0 && vi < (int64_t)variables.size())
+        hp[i * (Wv + Wh) + w] = (int64_t)fnv1a64_host(
+            variables[vi].first, variables[vi].second, lower);
+    }
+    log_ids.append(py::bytes(log_id));
+  }
+  return py::make_tuple(hashes, event_ids, log_ids);
+}
+
 }  // namespace
 
 void register_codec(py::module_& m) {
@@ -206,4 +431,8 @@ void register_codec(py::module_& m) {
         "sources, hostnames)");
   m.def("encode_parser_batch", &encode_parser_batch,
         "encode N ParserSchema frames from parser-kernel span outputs");
+  m.def("parser_watch_hashes", &parser_watch_hashes,
+        "batched ParserSchema -> watched-field FNV hashes + event ids");
+  m.def("template_match_cpu", &template_match_cpu,
+        "CPU twin of the dmx_template_match kernel (identical outputs)");
 }

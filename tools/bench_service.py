@@ -1,0 +1,175 @@
+#!/usr/bin/env python3
+"""Service-mode benchmark: lines/s through REAL service processes.
+
+Unlike bench.py (the fused in-process GPU pipeline), this drives the full
+deployment path: LogSchema frames over the framed ipc socket → parser
+Service (batched engine + C++ codec + template-match kernel) → detector
+Service (NewValueDetector) → sink. Measures end-to-end service throughput
+including protobuf, sockets and the admin plane.
+
+Usage: python tools/bench_service.py [--lines 200000] [--batch 4096]
+Prints one JSON line (same shape as bench.py, metric service_lines_per_sec).
+"""
+import argparse
+import json
+import os
+import subprocess
+import sys
+import tempfile
+import time
+import uuid
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import yaml  # noqa: E402
+
+from detectmateservice_amd.engine.sockets import PairDialer, PairListener, RecvTimeout  # noqa: E402
+from detectmateservice_amd.schemas import LogSchema  # noqa: E402
+from detectmateservice_amd.utils.synthetic import (  # noqa: E402
+    AUDIT_LOG_FORMAT,
+    AUDIT_TEMPLATES,
+    AuditLogGenerator,
+)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--lines", type=int, default=200_000)
+    ap.add_argument("--batch", type=int, default=4096, help="engine batch size")
+    ap.add_argument("--warmup-lines", type=int, default=20_000)
+    args = ap.parse_args()
+
+    tmp = tempfile.mkdtemp(prefix="dmx-bench-")
+    uid = uuid.uuid4().hex[:6]
+    parser_in = f"ipc://{tmp}/parser-{uid}.ipc"
+    detector_in = f"ipc://{tmp}/det-{uid}.ipc"
+    sink_addr = f"ipc://{tmp}/sink-{uid}.ipc"
+
+    tpl = os.path.join(tmp, "templates.txt")
+    with open(tpl, "w") as fh:
+        fh.write("\n".join(AUDIT_TEMPLATES) + "\n")
+
+    def write_yaml(name, data):
+        p = os.path.join(tmp, name)
+        with open(p, "w") as fh:
+            yaml.safe_dump(data, fh)
+        return p
+
+    parser_settings = write_yaml("ps.yaml", {
+        "component_type": "MatcherParser",
+        "engine_addr": parser_in,
+        "out_addr": [detector_in],
+        "http_enabled": False,
+        "engine_batch_size": args.batch,
+        "engine_batch_linger_ms": 3.0,
+        "engine_buffer_size": 8192,
+        "config_file": write_yaml("pc.yaml", {"parsers": {"MatcherParser": {
+            "log_format": AUDIT_LOG_FORMAT,
+            "params": {"path_templates": tpl},
+        }}}),
+        "log_dir": os.path.join(tmp, "logs"),
+    })
+    detector_settings = write_yaml("ds.yaml", {
+        "component_type": "NewValueDetector",
+        "engine_addr": detector_in,
+        "out_addr": [sink_addr],
+        "http_enabled": False,
+        "engine_batch_size": args.batch,
+        "engine_batch_linger_ms": 3.0,
+        "engine_buffer_size": 8192,
+        "config_file": write_yaml("dc.yaml", {"detectors": {"NewValueDetector": {
+            "data_use_training": args.warmup_lines,
+            "global": {"g": {"header_variables": [{"pos": "Type"}]}},
+        }}}),
+        "log_dir": os.path.join(tmp, "logs"),
+    })
+
+    procs = [
+        subprocess.Popen([sys.executable, "-m", "detectmateservice_amd.cli",
+                          "--settings", s],
+                         stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+        for s in (parser_settings, detector_settings)
+    ]
+    sink = PairListener(sink_addr)
+    feeder = PairDialer(parser_in, buffer_size=8192)
+    try:
+        assert feeder.wait_connected(20.0), "parser service did not come up"
+        gen = AuditLogGenerator(seed=7, anomaly_rate=0.0)
+        # pre-serialize frames so the feeder isn't the bottleneck
+        def make_frames(n, tag):
+            return [
+                LogSchema(logID=f"{tag}{i}", log=gen.line()[0]).serialize()
+                for i in range(n)
+            ]
+
+        warmup = make_frames(args.warmup_lines, "w")
+        frames = make_frames(args.lines, "m")
+
+        def pump(batch):
+            sent = 0
+            for f in batch:
+                while not feeder.send(f, block=False):
+                    time.sleep(0.0005)
+                sent += 1
+            return sent
+
+        pump(warmup)
+        time.sleep(2.0)  # drain training frames through both stages
+
+        # measure: parser's processed-lines counter via the detector's
+        # side effect is invisible (no alerts on clean traffic), so track
+        # the PARSER stage drain by timing the feed + drain of the
+        # detector input: we time until the parser has accepted all
+        # frames AND the pipeline is idle (sink quiet + sockets drained).
+        t0 = time.perf_counter()
+        pump(frames)
+        # wait until pipeline is idle: detector emits nothing for clean
+        # traffic; send one marked anomalous line and wait for its alert
+        # (it can only arrive after everything queued before it).
+        bad = (
+            "type=ZZZ_NEVER_SEEN msg=audit(1.0:1): pid=1 uid=0 auid=1 ses=1 "
+            "msg='op=PAM:x acct=\"x\" exe=/bin/x hostname=? addr=? "
+            "terminal=x res=success'"
+        )
+        while not feeder.send(LogSchema(logID="sentinel", log=bad).serialize(),
+                              block=False):
+            time.sleep(0.0005)
+        while True:
+            try:
+                sink.recv(timeout_ms=30000)
+                break
+            except RecvTimeout:
+                raise SystemExit("sentinel alert never arrived")
+        elapsed = time.perf_counter() - t0
+
+        total = args.lines + 1
+        print(json.dumps({
+            "metric": "service_lines_per_sec",
+            "value": round(total / elapsed, 1),
+            "unit": "lines/s",
+            "n_gpus": 1,
+            "higher_is_better": True,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "mode": "3-stage service processes over ipc (SP-framed engine sockets)",
+                "pipeline": "feeder->MatcherParser svc->NewValueDetector svc->sink",
+                "engine_batch_size": args.batch,
+                "lines": total,
+                "elapsed_s": round(elapsed, 3),
+            },
+        }))
+    finally:
+        feeder.close()
+        sink.close()
+        for p in procs:
+            p.terminate()
+        for p in procs:
+            try:
+                p.wait(timeout=5)
+            except subprocess.TimeoutExpired:
+                p.kill()
+
+
+if __name__ == "__main__":
+    main()
