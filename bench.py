@@ -46,6 +46,7 @@ def parse_args():
     p.add_argument("--pool", type=int, default=4, help="distinct pre-generated batches")
     p.add_argument("--max-len", type=int, default=256)
     p.add_argument("--no-transformer", action="store_true")
+    p.add_argument("--no-graph", action="store_true", help="disable hipGraph capture")
     p.add_argument("--device", default=None)
     return p.parse_args()
 
@@ -113,9 +114,14 @@ def main() -> None:
         if dist is not None:
             dist.barrier()
 
+    use_graph = [False]
+
     def step(i: int) -> None:
         lines, lens = pool[i % len(pool)]
-        out = pipe.process_packed(lines, lens)
+        if use_graph[0]:
+            out = pipe.process_packed_graph(lines, lens)
+        else:
+            out = pipe.process_packed(lines, lens)
         if dist is not None:
             # DP aggregation over RCCL/xGMI: per-rank anomaly summary
             summary = torch.stack(
@@ -127,6 +133,16 @@ def main() -> None:
     # ---- warmup (also trains the NewValue hash sets on the first batch) ----
     for i in range(args.warmup):
         step(i)
+    sync()
+    # capture the steady-state detect path as ONE hipGraph (HIP graphs for
+    # the launch-bound loop); fall back silently when capture is unsupported
+    if use_gpu and not args.no_graph:
+        try:
+            if pipe.enable_graph(args.batch):
+                use_graph[0] = True
+        except Exception as exc:  # noqa: BLE001
+            if rank == 0:
+                print(f"# graph capture disabled: {exc}", flush=True)
     sync()
     barrier()
     sync()
@@ -178,6 +194,7 @@ def main() -> None:
                 "p50_detect_latency_us_per_line": round(p50_line_us, 3),
                 "p50_batch_ms": round(p50_step * 1000.0, 3),
                 "transformer": not args.no_transformer,
+                "hip_graph": use_graph[0],
                 "device": str(device),
             },
         }

@@ -77,6 +77,10 @@ class GpuPipeline:
             else None
         )
         self.seen_lines = 0
+        # hipGraph capture state (enable_graph(); steady-state detect only)
+        self._graph = None
+        self._graph_in: Optional[tuple] = None
+        self._graph_out: Optional[Dict[str, torch.Tensor]] = None
 
     # ------------------------------------------------------------------
     def process_packed(
@@ -159,3 +163,44 @@ class GpuPipeline:
     def process_lines(self, raw_lines: Sequence[bytes]) -> Dict[str, torch.Tensor]:
         lines, lens = ops.pack_lines(raw_lines, self.config.max_len, device=self.device)
         return self.process_packed(lines, lens)
+
+    # ------------------------------------------------------------------
+    # hipGraph capture (HIP graphs for the launch-bound steady state —
+    # SURVEY.md build mandate). Valid only AFTER the training phase: the
+    # captured graph replays the detect-only path at fixed batch size.
+    # ------------------------------------------------------------------
+    def enable_graph(self, batch_size: int) -> bool:
+        if self.device.type != "cuda":
+            return False
+        if self.seen_lines < self.config.train_lines:
+            raise RuntimeError("enable_graph() requires the training phase done")
+        B = batch_size
+        static_lines = torch.zeros(
+            (B, self.config.max_len), dtype=torch.uint8, device=self.device
+        )
+        static_lens = torch.zeros((B,), dtype=torch.int32, device=self.device)
+        # warm up allocator/kernels on the same shapes, then capture
+        for _ in range(2):
+            self.process_packed(static_lines, static_lens)
+            self.seen_lines -= B  # warmups must not advance stream state
+        torch.cuda.synchronize()
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            out = self.process_packed(static_lines, static_lens)
+        self.seen_lines -= B
+        self._graph = graph
+        self._graph_in = (static_lines, static_lens)
+        self._graph_out = out
+        return True
+
+    def process_packed_graph(
+        self, lines: torch.Tensor, lens: torch.Tensor
+    ) -> Dict[str, torch.Tensor]:
+        """Replay the captured graph on a new batch (same B, max_len)."""
+        assert self._graph is not None, "call enable_graph() first"
+        sl, sn = self._graph_in
+        sl.copy_(lines, non_blocking=True)
+        sn.copy_(lens, non_blocking=True)
+        self._graph.replay()
+        self.seen_lines += lines.shape[0]
+        return self._graph_out

@@ -343,3 +343,46 @@ def test_template_miner_gpu_matches_cpu():
     cpu_t = TemplateMiner(device="cpu").fit(contents)
     gpu_t = TemplateMiner(device="cuda").fit(contents)
     assert cpu_t == gpu_t
+
+
+def test_pipeline_graph_replay_matches_eager():
+    """hipGraph-captured steady-state pipeline must produce identical
+    results to the eager path on fresh data."""
+    from detectmateservice_amd import ops
+    from detectmateservice_amd.pipeline import GpuPipeline, PipelineConfig
+    from detectmateservice_amd.utils.synthetic import (
+        AUDIT_LOG_FORMAT,
+        AUDIT_TEMPLATES,
+        AuditLogGenerator,
+    )
+
+    B = 512
+    cfg = PipelineConfig(
+        templates=AUDIT_TEMPLATES, log_format=AUDIT_LOG_FORMAT,
+        watches=[{"kind": "variable", "pos": 5, "event": 1}],
+        train_lines=B, use_transformer=True, score_threshold=3.0,
+    )
+    pipe = GpuPipeline(cfg, device="cuda")
+    gen = AuditLogGenerator(seed=3, anomaly_rate=0.05)
+    train = [gen.line()[0].encode() for _ in range(B)]
+    pipe.process_lines(train)  # training batch
+
+    batch = [gen.line()[0].encode() for _ in range(B)]
+    lines, lens = ops.pack_lines(batch, cfg.max_len, device="cuda")
+    eager = pipe.process_packed(lines.clone(), lens.clone())
+    eager_scores = eager["scores"].clone().cpu()
+    eager_anom = eager["anomaly"].clone().cpu()
+    seen_before = pipe.seen_lines
+
+    assert pipe.enable_graph(B)
+    out = pipe.process_packed_graph(lines, lens)
+    torch.cuda.synchronize()
+    assert torch.allclose(out["scores"].cpu(), eager_scores, atol=1e-4)
+    assert torch.equal(out["anomaly"].cpu(), eager_anom)
+    # replay again on different data
+    batch2 = [gen.line()[0].encode() for _ in range(B)]
+    l2, n2 = ops.pack_lines(batch2, cfg.max_len, device="cuda")
+    out2 = pipe.process_packed_graph(l2, n2)
+    torch.cuda.synchronize()
+    ref2 = pipe.process_packed(l2, n2)
+    assert torch.allclose(out2["scores"].cpu(), ref2["scores"].cpu(), atol=1e-4)
