@@ -54,7 +54,12 @@ class WebServer(threading.Thread):
 
     def stop(self) -> None:
         self._server.should_exit = True
-        self.join(timeout=5.0)
+        self.join(timeout=3.0)
+        if self.is_alive():
+            # keep-alive connections can hold graceful shutdown open;
+            # force-close them (uvicorn's force_exit path)
+            self._server.force_exit = True
+            self.join(timeout=3.0)
 
     def wait_started(self, timeout_s: float = 10.0) -> bool:
         deadline = time.monotonic() + timeout_s

@@ -37,3 +37,40 @@ def test_client_parser():
 
     with pytest.raises(SystemExit):
         client_main([])  # subcommand required
+
+
+def test_client_cli_against_live_service(free_port, tmp_path, capsys):
+    """detectmate-client subcommands against a running service."""
+    import threading
+
+    from detectmateservice_amd import Service, ServiceSettings
+    from detectmateservice_amd.client import main as client_main
+
+    settings = ServiceSettings(
+        component_type="core", engine_addr=f"ipc://{tmp_path}/cli.ipc",
+        http_port=free_port, log_dir=tmp_path / "logs",
+    )
+    svc = Service(settings)
+    t = threading.Thread(target=svc.run, daemon=True)
+    t.start()
+    try:
+        assert svc.web_server.wait_started(10.0)
+        url = f"http://127.0.0.1:{free_port}"
+        assert client_main(["--url", url, "status"]) == 0
+        out = capsys.readouterr().out
+        assert '"running": true' in out
+        assert client_main(["--url", url, "stop"]) == 0
+        assert client_main(["--url", url, "start"]) == 0
+        assert client_main(["--url", url, "metrics"]) == 0
+        assert "engine_running" in capsys.readouterr().out
+        cfg = tmp_path / "re.yaml"
+        cfg.write_text("detectors: {}\n")
+        assert client_main(["--url", url, "reconfigure", str(cfg)]) == 0
+        ck = tmp_path / "c.pt"
+        assert client_main(["--url", url, "checkpoint", str(ck)]) == 0
+        assert ck.exists()
+        assert client_main(["--url", url, "restore", str(ck)]) == 0
+        assert client_main(["--url", url, "shutdown"]) == 0
+    finally:
+        svc.shutdown()
+        t.join(timeout=10.0)
