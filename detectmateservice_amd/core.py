@@ -214,12 +214,27 @@ class Service:
 
     # ------------------------------------------------------------------
     def _create_status_report(self, running: bool) -> Dict[str, Any]:
+        m = self.metrics
+
+        def _val(c):
+            try:
+                return c._value.get()
+            except AttributeError:
+                return None
+
         return {
             "status": {
                 "running": running,
                 "engine_running": self.engine.running,
                 "version": __version__,
                 "timestamp": int(time.time()),
+            },
+            "metrics": {
+                "read_lines": _val(m.data_read_lines_total),
+                "processed_lines": _val(m.data_processed_lines_total),
+                "written_lines": _val(m.data_written_lines_total),
+                "dropped_lines": _val(m.data_dropped_lines_total),
+                "processing_errors": _val(m.processing_errors_total),
             },
             "settings": self.settings.model_dump(mode="json"),
             "configs": self.config_manager.get(),

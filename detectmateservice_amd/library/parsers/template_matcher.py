@@ -222,12 +222,34 @@ class MatcherParser(CoreComponent):
             self._auto_observe(
                 [LogSchema.deserialize(f).log for f in frames]
             )
-        if len(frames) >= 8 and self._remove_spaces is False and self._remove_punct is False:
+        if (len(frames) >= 8 and self._remove_spaces is False
+                and self._remove_punct is False and not self.config.time_format):
             try:
                 return self._process_batched(frames)
             except RuntimeError:
                 pass  # extension missing: python fallback
         return self._process_python(frames)
+
+    def _extracted_timestamp(self, header: Dict[str, str]) -> Optional[int]:
+        """Parse the header's time field with ``time_format`` (reference
+        config key, container/config/parser_config.yaml:5). Field name
+        defaults to "Time" (``params.time_field`` overrides)."""
+        fmt = self.config.time_format
+        if not fmt:
+            return None
+        field = (self.config.params or {}).get("time_field", "Time")
+        value = header.get(field)
+        if not value:
+            return None
+        import datetime
+
+        try:
+            dt = datetime.datetime.strptime(value, fmt)
+            if dt.tzinfo is None:
+                dt = dt.replace(tzinfo=datetime.timezone.utc)
+            return int(dt.timestamp())
+        except ValueError:
+            return None
 
     def _process_python(self, frames: List[bytes]) -> List[Optional[bytes]]:
         now = int(time.time())
@@ -235,6 +257,7 @@ class MatcherParser(CoreComponent):
         for raw in frames:
             log = LogSchema.deserialize(raw)
             header, event_id, template, variables = self.parse_line(log.log)
+            ts = self._extracted_timestamp(header)
             parsed = ParserSchema(
                 parserType="matcher_parser",
                 parserID=self.parser_id,
@@ -244,7 +267,7 @@ class MatcherParser(CoreComponent):
                 logID=log.logID,
                 log=log.log,
                 logFormatVariables=header,
-                receivedTimestamp=now,
+                receivedTimestamp=ts if ts is not None else now,
                 parsedTimestamp=now,
             )
             out.append(parsed.serialize())

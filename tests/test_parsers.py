@@ -120,3 +120,22 @@ def test_dummy_parser_contract():
     assert parsed.variables == ["dummy_variable"]
     assert parsed.log == "DummyParser"
     assert parsed.logID == "id9"
+
+
+def test_time_format_extraction():
+    """time_format parses the header time into receivedTimestamp
+    (reference config key, parser_config.yaml:5)."""
+    parser = MatcherParser({
+        "log_format": '<IP> - - [<Time>] "<Req>"',
+        "time_format": "%d/%b/%Y:%H:%M:%S",
+        "templates": [],
+    })
+    frame = LogSchema(
+        log='1.2.3.4 - - [10/Oct/2000:13:55:36] "GET /x"'
+    ).serialize()
+    parsed = ParserSchema.deserialize(parser.process(frame))
+    assert parsed.logFormatVariables["Time"] == "10/Oct/2000:13:55:36"
+    import datetime
+    expect = int(datetime.datetime(2000, 10, 10, 13, 55, 36,
+                                   tzinfo=datetime.timezone.utc).timestamp())
+    assert parsed.receivedTimestamp == expect
