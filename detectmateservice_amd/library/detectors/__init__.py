@@ -18,3 +18,6 @@ __all__ = [
 from .transformer import TransformerDetector, TransformerDetectorConfig
 
 __all__ += ["TransformerDetector", "TransformerDetectorConfig"]
+from .embedding import EmbeddingDetector, EmbeddingDetectorConfig
+
+__all__ += ["EmbeddingDetector", "EmbeddingDetectorConfig"]

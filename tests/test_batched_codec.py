@@ -107,3 +107,45 @@ def test_transformer_detector_resolvable():
     path, cfg = ComponentResolver().resolve("TransformerDetector")
     assert path.endswith("TransformerDetector")
     assert cfg.endswith("TransformerDetectorConfig")
+
+
+def test_embedding_detector_cpu():
+    from detectmateservice_amd.library.detectors import EmbeddingDetector
+
+    det = EmbeddingDetector({"data_use_training": 48, "z_threshold": 4.0,
+                             "device": "cpu"})
+    gen = AuditLogGenerator(seed=61)
+
+    def frame(line, lid):
+        return ParserSchema(logID=lid, log=line, EventID=1).serialize()
+
+    train = [frame(gen.line()[0], f"t{i}") for i in range(48)]
+    assert all(o is None for o in det.process_batch(train))
+    assert det._n == 48
+
+    normal = [frame(gen.line()[0], f"n{i}") for i in range(16)]
+    out = det.process_batch(normal)
+    assert sum(o is not None for o in out) <= 2  # 4-sigma on in-dist data
+
+    # a wildly out-of-distribution line
+    weird = frame("\x01\x02!!!" + "Z" * 180, "weird")
+    out = det.process_batch(normal[:3] + [weird])
+    # distance ordering: the weird line must have the max distance
+    emb = det.embed([b"normal audit line", b"\x01\x02!!!" + b"Z" * 180])
+    d = det._distance(emb)
+    assert d[1] > d[0]
+
+    # checkpoint roundtrip: restored detector scores `normal` identically
+    ref = det.process_batch(normal)
+    state = det.state_dict()
+    det2 = EmbeddingDetector({"data_use_training": 0, "device": "cpu"})
+    det2.load_state_dict(state)
+    out2 = det2.process_batch(normal)
+    assert [o is None for o in ref] == [o is None for o in out2]
+
+
+def test_embedding_detector_resolvable():
+    from detectmateservice_amd.components.resolver import ComponentResolver
+
+    path, cfg = ComponentResolver().resolve("EmbeddingDetector")
+    assert path.endswith("EmbeddingDetector")
