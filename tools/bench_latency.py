@@ -28,7 +28,7 @@ def main():
         cfg = PipelineConfig(
             templates=AUDIT_TEMPLATES, log_format=AUDIT_LOG_FORMAT,
             watches=[{"kind": "variable", "pos": 5, "event": 1}],
-            train_lines=B, use_transformer=True,
+            train_lines=B, use_transformer=True, max_len=256,
         )
         pipe = GpuPipeline(cfg, device=device)
         pool = []
