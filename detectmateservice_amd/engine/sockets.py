@@ -23,10 +23,12 @@ same capability surface as the reference's socket layer without NNG:
 Wire framing: ``tcp://`` and ``tls+tcp://`` speak the NNG SP mapping
 (8-octet ``\x00SP\x00`` + pair0 protocol-id connection header, then
 64-bit BE length-prefixed messages) so NNG/pynng/fluentd-nng peers
-interoperate at the pipeline edges; ``ipc://`` and ``inproc://`` are
-intra-node and use a compact 4-byte length prefix. These framings plus the
-proto3 schemas in ``detectmateservice_amd.schemas`` are the complete wire
-contract between stages.
+interoperate at the pipeline edges; ``ws://`` carries real RFC 6455
+binary frames (upgrade handshake, masked client frames, ping/pong);
+``ipc://`` and ``inproc://`` are intra-node and use a compact 4-byte
+length prefix. These framings plus the proto3 schemas in
+``detectmateservice_amd.schemas`` are the complete wire contract between
+stages.
 """
 from __future__ import annotations
 
@@ -178,6 +180,187 @@ class _FrameReader:
         return frames[0]
 
 
+# ---------------------------------------------------------------------------
+# ws:// — minimal RFC 6455 framing (binary messages, ping/pong, close).
+# The reference supports ws as an NNG transport scheme (settings.py:30-37
+# there); here ws carries the same pair semantics over real WebSocket
+# frames so browser/proxy-facing edges interoperate.
+# ---------------------------------------------------------------------------
+
+_WS_GUID = b"258EAFA5-E914-47DA-95CA-C5AB0DC85B11"
+
+
+def _ws_accept_key(key: bytes) -> bytes:
+    import base64
+    import hashlib
+
+    return base64.b64encode(hashlib.sha1(key + _WS_GUID).digest())
+
+
+def _ws_server_handshake(sock: socket.socket) -> bool:
+    """Read the HTTP upgrade request, reply 101. Blocking, pre-reader."""
+    try:
+        buf = b""
+        while b"\r\n\r\n" not in buf:
+            chunk = sock.recv(4096)
+            if not chunk:
+                return False
+            buf += chunk
+            if len(buf) > 16384:
+                return False
+        headers = {}
+        for line in buf.split(b"\r\n")[1:]:
+            if b":" in line:
+                k, v = line.split(b":", 1)
+                headers[k.strip().lower()] = v.strip()
+        key = headers.get(b"sec-websocket-key")
+        if key is None:
+            return False
+        resp = (
+            b"HTTP/1.1 101 Switching Protocols\r\n"
+            b"Upgrade: websocket\r\nConnection: Upgrade\r\n"
+            b"Sec-WebSocket-Accept: " + _ws_accept_key(key) + b"\r\n\r\n"
+        )
+        sock.sendall(resp)
+        return True
+    except OSError:
+        return False
+
+
+def _ws_client_handshake(sock: socket.socket, host: str) -> bool:
+    import base64
+    import os as _os
+
+    try:
+        key = base64.b64encode(_os.urandom(16))
+        req = (
+            b"GET / HTTP/1.1\r\nHost: " + host.encode() + b"\r\n"
+            b"Upgrade: websocket\r\nConnection: Upgrade\r\n"
+            b"Sec-WebSocket-Key: " + key + b"\r\nSec-WebSocket-Version: 13\r\n\r\n"
+        )
+        sock.sendall(req)
+        buf = b""
+        while b"\r\n\r\n" not in buf:
+            chunk = sock.recv(4096)
+            if not chunk:
+                return False
+            buf += chunk
+            if len(buf) > 16384:
+                return False
+        if b" 101 " not in buf.split(b"\r\n", 1)[0]:
+            return False
+        expect = _ws_accept_key(key)
+        return expect in buf
+    except OSError:
+        return False
+
+
+def _ws_control(opcode: int, payload: bytes, mask: bool) -> bytes:
+    import os as _os
+
+    header = bytearray([0x80 | opcode])
+    mask_bit = 0x80 if mask else 0
+    header.append(mask_bit | len(payload))
+    if mask:
+        mkey = _os.urandom(4)
+        header += mkey
+        payload = bytes(b ^ mkey[i % 4] for i, b in enumerate(payload))
+    return bytes(header) + payload
+
+
+def _ws_encode(data: bytes, mask: bool) -> bytes:
+    """One binary message frame (FIN=1, opcode 2); clients mask."""
+    import os as _os
+
+    header = bytearray([0x82])  # FIN + binary
+    n = len(data)
+    mask_bit = 0x80 if mask else 0
+    if n < 126:
+        header.append(mask_bit | n)
+    elif n < (1 << 16):
+        header.append(mask_bit | 126)
+        header += n.to_bytes(2, "big")
+    else:
+        header.append(mask_bit | 127)
+        header += n.to_bytes(8, "big")
+    if mask:
+        mkey = _os.urandom(4)
+        header += mkey
+        masked = bytes(b ^ mkey[i % 4] for i, b in enumerate(data))
+        return bytes(header) + masked
+    return bytes(header) + data
+
+
+class _WsFrameReader:
+    """Parse WebSocket frames into messages; answers pings, honors close."""
+
+    def __init__(self, sock: socket.socket, server_side: bool) -> None:
+        self.sock = sock
+        self.server_side = server_side
+        self.buf = bytearray()
+
+    def _fill(self, need: int) -> bool:
+        while len(self.buf) < need:
+            try:
+                chunk = self.sock.recv(262144)
+            except (ssl.SSLWantReadError, BlockingIOError):
+                continue
+            if not chunk:
+                return False
+            self.buf += chunk
+        return True
+
+    def next_frames(self) -> Optional[List[bytes]]:
+        out: List[bytes] = []
+        while True:
+            if not self._fill(2):
+                return out or None
+            b0, b1 = self.buf[0], self.buf[1]
+            opcode = b0 & 0x0F
+            masked = bool(b1 & 0x80)
+            n = b1 & 0x7F
+            pos = 2
+            if n == 126:
+                if not self._fill(4):
+                    return out or None
+                n = int.from_bytes(self.buf[2:4], "big")
+                pos = 4
+            elif n == 127:
+                if not self._fill(10):
+                    return out or None
+                n = int.from_bytes(self.buf[2:10], "big")
+                pos = 10
+            if n > MAX_FRAME_BYTES:
+                raise ValueError("oversize ws frame")
+            need = pos + (4 if masked else 0) + n
+            if not self._fill(need):
+                return out or None
+            if masked:
+                mkey = bytes(self.buf[pos:pos + 4])
+                payload = bytes(
+                    b ^ mkey[i % 4]
+                    for i, b in enumerate(self.buf[pos + 4:need])
+                )
+            else:
+                payload = bytes(self.buf[pos:need])
+            del self.buf[:need]
+            if opcode == 0x8:  # close
+                return out or None
+            if opcode == 0x9:  # ping -> pong (server unmasked, client masked)
+                try:
+                    self.sock.sendall(_ws_control(0xA, payload, not self.server_side))
+                except OSError:
+                    return out or None
+                continue
+            if opcode in (0x2, 0x1, 0x0):  # binary/text (no fragmentation)
+                out.append(payload)
+                if not self.buf:
+                    return out
+                continue
+            # pong / unknown control: ignore and keep parsing
+            continue
+
+
 def _sp_handshake(sock: socket.socket, logger: logging.Logger) -> bool:
     """Exchange the 8-octet SP connection header (both sides send theirs;
     any peer protocol id is accepted — pair0 expected)."""
@@ -240,6 +423,8 @@ class PairListener:
         scheme, rest = self.addr.scheme, self.addr.rest
         #: SP (NNG pair0) wire mapping on tcp/tls+tcp edges
         self._sp = scheme in ("tcp", "tls+tcp")
+        #: real RFC 6455 framing on ws:// edges
+        self._ws = scheme == "ws"
         if scheme == "inproc":
             raise ValueError("use InprocListener for inproc:// addresses")
         if scheme == "ipc":
@@ -300,6 +485,9 @@ class PairListener:
             if self._sp and not _sp_handshake(conn, self._log):
                 conn.close()
                 continue
+            if self._ws and not _ws_server_handshake(conn):
+                conn.close()
+                continue
             with self._peers_lock:
                 self._peers.append(conn)
             threading.Thread(
@@ -308,7 +496,8 @@ class PairListener:
             ).start()
 
     def _reader_loop(self, conn: socket.socket) -> None:
-        reader = _FrameReader(conn, self._sp)
+        reader = (_WsFrameReader(conn, server_side=True) if self._ws
+                  else _FrameReader(conn, self._sp))
         try:
             while not self._closed.is_set():
                 frames = reader.next_frames()
@@ -395,7 +584,10 @@ class PairListener:
         if conn is None:
             return False
         try:
-            _send_frame(conn, data, self._sp)
+            if self._ws:
+                conn.sendall(_ws_encode(data, mask=False))
+            else:
+                _send_frame(conn, data, self._sp)
             return True
         except OSError:
             return False
@@ -501,6 +693,7 @@ class PairDialer:
         self._conn: Optional[socket.socket] = None
         self._inproc: Optional[InprocListener] = None
         self._sp = self.addr.scheme in ("tcp", "tls+tcp")
+        self._ws = self.addr.scheme == "ws"
         if self.addr.scheme == "inproc":
             # resolved lazily in the worker loop so late binding works
             pass
@@ -536,6 +729,9 @@ class PairDialer:
                 if self._sp and not _sp_handshake(s, self._log):
                     s.close()
                     return False
+                if self._ws and not _ws_client_handshake(s, host):
+                    s.close()
+                    return False
             s.settimeout(None)
             self._conn = s
             threading.Thread(
@@ -547,16 +743,18 @@ class PairDialer:
             return False
 
     def _reader_loop(self, conn: socket.socket) -> None:
-        reader = _FrameReader(conn, self._sp)
+        reader = (_WsFrameReader(conn, server_side=False) if self._ws
+                  else _FrameReader(conn, self._sp))
         try:
             while not self._closed.is_set():
-                frame = reader.next_frame()
-                if frame is None:
+                frames = reader.next_frames()
+                if frames is None:
                     break
-                try:
-                    self._recv_q.put(frame, timeout=1.0)
-                except queue.Full:
-                    pass  # drop inbound overflow on the reply channel
+                for frame in frames:
+                    try:
+                        self._recv_q.put(frame, timeout=1.0)
+                    except queue.Full:
+                        pass  # drop inbound overflow on the reply channel
         except (OSError, ValueError):
             pass
         finally:
@@ -607,17 +805,28 @@ class PairDialer:
             # coalesce everything queued into ONE sendall (per-frame
             # sendall measured ~13k frames/s; batching removes the
             # syscall + GIL ping-pong per frame)
-            hdr = _LEN64 if self._sp else _LEN
-            chunks = [hdr.pack(len(data)), data]
-            n_coalesced = 1
-            while n_coalesced < 512:
-                try:
-                    more = self._send_q.get_nowait()
-                except queue.Empty:
-                    break
-                chunks.append(hdr.pack(len(more)))
-                chunks.append(more)
-                n_coalesced += 1
+            if self._ws:
+                chunks = [_ws_encode(data, mask=True)]
+                n_coalesced = 1
+                while n_coalesced < 512:
+                    try:
+                        more = self._send_q.get_nowait()
+                    except queue.Empty:
+                        break
+                    chunks.append(_ws_encode(more, mask=True))
+                    n_coalesced += 1
+            else:
+                hdr = _LEN64 if self._sp else _LEN
+                chunks = [hdr.pack(len(data)), data]
+                n_coalesced = 1
+                while n_coalesced < 512:
+                    try:
+                        more = self._send_q.get_nowait()
+                    except queue.Empty:
+                        break
+                    chunks.append(hdr.pack(len(more)))
+                    chunks.append(more)
+                    n_coalesced += 1
             payload = b"".join(chunks)
             try:
                 conn.sendall(payload)

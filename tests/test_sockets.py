@@ -211,3 +211,62 @@ def test_non_sp_peer_rejected_on_tcp():
     finally:
         raw.close()
         listener.close()
+
+
+def test_ws_roundtrip():
+    """ws:// carries real RFC6455 frames (binary, masked client→server)."""
+    listener = PairListener("ws://127.0.0.1:0")
+    port = listener.bound_port
+    dialer = PairDialer(f"ws://127.0.0.1:{port}")
+    try:
+        assert dialer.wait_connected(5.0)
+        dialer.send(b"over websocket")
+        assert listener.recv(timeout_ms=3000) == b"over websocket"
+        listener.send(b"reply-ws")
+        assert dialer.recv(timeout_ms=3000) == b"reply-ws"
+        # larger-than-125 payload exercises the 126 length form
+        big = b"B" * 70000
+        dialer.send(big)
+        assert listener.recv(timeout_ms=5000) == big
+    finally:
+        dialer.close()
+        listener.close()
+
+
+def test_ws_handshake_is_rfc6455():
+    """Raw client: upgrade request gets the correct Sec-WebSocket-Accept."""
+    import base64
+    import hashlib
+    import socket
+
+    listener = PairListener("ws://127.0.0.1:0")
+    port = listener.bound_port
+    raw = socket.create_connection(("127.0.0.1", port), timeout=5.0)
+    try:
+        key = base64.b64encode(b"0123456789abcdef")
+        raw.sendall(
+            b"GET / HTTP/1.1\r\nHost: x\r\nUpgrade: websocket\r\n"
+            b"Connection: Upgrade\r\nSec-WebSocket-Key: " + key +
+            b"\r\nSec-WebSocket-Version: 13\r\n\r\n"
+        )
+        resp = b""
+        while b"\r\n\r\n" not in resp:
+            chunk = raw.recv(4096)
+            assert chunk
+            resp += chunk
+        assert b" 101 " in resp.split(b"\r\n", 1)[0]
+        expect = base64.b64encode(
+            hashlib.sha1(key + b"258EAFA5-E914-47DA-95CA-C5AB0DC85B11").digest()
+        )
+        assert expect in resp
+        # masked binary frame from the raw client
+        payload = b"raw-ws-frame"
+        mask = b"\x01\x02\x03\x04"
+        frame = bytes([0x82, 0x80 | len(payload)]) + mask + bytes(
+            b ^ mask[i % 4] for i, b in enumerate(payload)
+        )
+        raw.sendall(frame)
+        assert listener.recv(timeout_ms=3000) == payload
+    finally:
+        raw.close()
+        listener.close()
