@@ -55,7 +55,6 @@ _LEN = struct.Struct(">I")
 #: 64-bit BE length (SP TCP mapping). ipc:// and inproc:// are intra-node
 #: and use this framework's compact 4-byte framing.
 SP_PAIR0_PROTO = 0x10
-_SP_HEADER = struct.Struct(">ccBBHH")  # \x00, S, P, ver, proto, reserved
 _LEN64 = struct.Struct(">Q")
 
 
