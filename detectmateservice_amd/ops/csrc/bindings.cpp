@@ -14,6 +14,7 @@ void dmx_launch_fused_linear_bf16(const void*, const void*, const void*,
                                   void*, int, int, int, int, hipStream_t);
 void dmx_launch_probe_mfma(const void*, const void*, void*, int, int,
                            hipStream_t);
+void dmx_launch_probe_mfma32(const void*, const void*, void*, hipStream_t);
 void dmx_launch_layernorm_bf16(const void*, const void*, const void*,
                                const void*, void*, void*, int, int, float,
                                hipStream_t);
@@ -76,6 +77,15 @@ torch::Tensor fused_linear_bf16(torch::Tensor x, torch::Tensor wt,
                                C.data_ptr(), (int)M, (int)N, (int)K,
                                (int)epilogue, cur_stream());
   return C;
+}
+
+torch::Tensor probe_mfma32(torch::Tensor A, torch::Tensor B) {
+  TORCH_CHECK(A.size(0) == 32 && A.size(1) == 16, "A must be 32x16");
+  TORCH_CHECK(B.size(0) == 16 && B.size(1) == 32, "B must be 16x32");
+  auto D = torch::zeros({32, 32}, A.options().dtype(torch::kFloat32));
+  dmx_launch_probe_mfma32(A.data_ptr(), B.data_ptr(), D.data_ptr(),
+                          cur_stream());
+  return D;
 }
 
 torch::Tensor probe_mfma(torch::Tensor A, torch::Tensor B, int64_t a_layout,
@@ -274,6 +284,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_linear_bf16", &fused_linear_bf16,
         "C = act(x @ wt^T + bias), bf16 MFMA (epilogue: 0 none, 1 gelu, 2 relu)");
   m.def("probe_mfma", &probe_mfma, "MFMA 16x16x32 bf16 layout probe");
+  m.def("probe_mfma32", &probe_mfma32, "MFMA 32x32x16 bf16 layout probe");
   m.def("layernorm_bf16", &layernorm_bf16, "fused residual+LayerNorm bf16");
   m.def("attention_bf16", &attention_bf16, "fused short-seq MHA bf16");
   m.def("attention_qkv_bf16", &attention_qkv_bf16,

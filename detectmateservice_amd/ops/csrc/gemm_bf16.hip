@@ -218,3 +218,41 @@ extern "C" void dmx_launch_probe_mfma(
                      (const short*)A, (const short*)B, (float*)D, a_layout,
                      b_layout);
 }
+
+// ---------------------------------------------------------------------------
+// Layout probe for mfma_f32_32x32x16_bf16 (A 32x16, B 16x32, D 32x32).
+// Assumed (verified empirically on gfx950 by tests/test_gpu_ops.py):
+//   A: lane l holds A[l&31][(l>>5)*8 + j], j = 0..7 (contiguous k-octet)
+//   B: lane l holds B[(l>>5)*8 + j][l&31]
+//   C/D (guide §3): col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+extern "C" __global__ void dmx_probe_mfma_32x32x16(
+    const short* __restrict__ A,  // [32][16] bf16
+    const short* __restrict__ B,  // [16][32] bf16
+    float* __restrict__ D) {      // [32][32] f32
+  const int lane = threadIdx.x & 63;
+  bf16x8 a, b;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    a[j] = A[(lane & 31) * 16 + (lane >> 5) * 8 + j];
+    b[j] = B[((lane >> 5) * 8 + j) * 32 + (lane & 31)];
+  }
+  f32x16 acc;
+#pragma unroll
+  for (int i = 0; i < 16; ++i) acc[i] = 0.f;
+  acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+    const int col = lane & 31;
+    D[row * 32 + col] = acc[r];
+  }
+}
+
+extern "C" void dmx_launch_probe_mfma32(const void* A, const void* B, void* D,
+                                        hipStream_t stream) {
+  hipLaunchKernelGGL(dmx_probe_mfma_32x32x16, dim3(1), dim3(64), 0, stream,
+                     (const short*)A, (const short*)B, (float*)D);
+}

@@ -386,3 +386,17 @@ def test_pipeline_graph_replay_matches_eager():
     torch.cuda.synchronize()
     ref2 = pipe.process_packed(l2, n2)
     assert torch.allclose(out2["scores"].cpu(), ref2["scores"].cpu(), atol=1e-4)
+
+
+def test_mfma_32x32x16_layout_probe():
+    """Pin the 32x32x16 bf16 fragment layout (used by block_gemm's wide
+    fragments). Asymmetric operands (rule 16)."""
+    from detectmateservice_amd.ops import _dmx_C
+
+    torch.manual_seed(9)
+    A = (torch.randn(32, 16) * 0.5 + torch.arange(32).view(32, 1) * 0.01).bfloat16().cuda()
+    B = (torch.randn(16, 32) * 0.5 + torch.arange(32).view(1, 32) * 0.02).bfloat16().cuda()
+    D = _dmx_C.probe_mfma32(A, B).cpu()
+    ref = A.float().cpu() @ B.float().cpu()
+    err = (D - ref).abs().max()
+    assert err < 0.1, f"32x32x16 layout mismatch: max err {err}"
