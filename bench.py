@@ -75,10 +75,13 @@ def main() -> None:
         dist = dist_mod
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29517")
+        import datetime
+
         dist.init_process_group(
             backend="nccl" if use_gpu else "gloo",
             rank=rank,
             world_size=world_size,
+            timeout=datetime.timedelta(seconds=300),
         )
 
     # ---- build the pipeline (random-init weights, synthetic shapes) ----

@@ -175,3 +175,59 @@ def test_stage_pipeline_p2p(free_port):
 def test_dp_pipeline_allgather(free_port):
     results = _launch("_body_dp_pipeline", 2, free_port)
     assert results[0] == results[1]
+
+
+def _body_elastic_peer_death(rank, world):
+    """Rank 1 exits mid-run; rank 0's guarded collective degrades to a
+    local-only result instead of hanging (drop-don't-block on the
+    collective path, SURVEY §7 hard part 1)."""
+    import torch.distributed as tdist
+
+    from detectmateservice_amd.parallel.elastic import ElasticComm
+
+    comm = ElasticComm(timeout_s=5.0)
+    s = torch.tensor([float(rank), 1.0])
+    # healthy round
+    out = comm.all_gather_summaries(s)
+    assert out.shape == (world, 2)
+    if rank == 1:
+        return "rank1-exits"  # dies without participating again
+    # rank 0: the next collective must fail fast (gloo timeout) and fall
+    # back to the local summary
+    out2 = comm.all_gather_summaries(s)
+    assert comm.degraded
+    assert comm.drops == 1
+    assert out2.shape == (1, 2)
+    assert out2[0, 0] == 0.0
+    # further collectives short-circuit locally (no more timeouts)
+    out3 = comm.all_gather_summaries(s)
+    assert out3.shape == (1, 2) and comm.drops == 1
+    return "rank0-degraded"
+
+
+def test_elastic_peer_death(free_port):
+    results = _launch("_body_elastic_peer_death", 2, free_port)
+    assert results[0] == "rank0-degraded"
+    assert results[1] == "rank1-exits"
+
+
+def _body_elastic_reform(rank, world):
+    """Communicator rebuild: both ranks reform the group on a new port and
+    collectives work again."""
+    from detectmateservice_amd.parallel.elastic import ElasticComm
+
+    comm = ElasticComm(timeout_s=10.0)
+    s = torch.tensor([float(rank + 1)])
+    out = comm.all_gather_summaries(s)
+    assert out.shape == (world, 1)
+    new_port = int(os.environ["MASTER_PORT"]) + 1
+    comm.reform_group(rank, world, master_port=new_port)
+    out2 = comm.all_gather_summaries(s)
+    assert out2.shape == (world, 1)
+    assert out2[:, 0].tolist() == [1.0, 2.0]
+    return "reformed"
+
+
+def test_elastic_reform_group(free_port):
+    results = _launch("_body_elastic_reform", 2, free_port)
+    assert results[0] == results[1] == "reformed"
