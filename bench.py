@@ -42,7 +42,7 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--batch", type=int, default=32768, help="lines per rank per step")
+    p.add_argument("--batch", type=int, default=65536, help="lines per rank per step")
     p.add_argument("--pool", type=int, default=4, help="distinct pre-generated batches")
     p.add_argument("--max-len", type=int, default=256)
     p.add_argument("--no-transformer", action="store_true")
