@@ -693,6 +693,7 @@ class PairDialer:
         self._inproc: Optional[InprocListener] = None
         self._sp = self.addr.scheme in ("tcp", "tls+tcp")
         self._ws = self.addr.scheme == "ws"
+        self._inflight: Optional[bytes] = None
         if self.addr.scheme == "inproc":
             # resolved lazily in the worker loop so late binding works
             pass
@@ -770,10 +771,13 @@ class PairDialer:
                         time.sleep(self.REDIAL_INTERVAL_S)
                         continue
                 self._connected.set()
-                try:
-                    data = self._send_q.get(timeout=0.2)
-                except queue.Empty:
-                    continue
+                if self._inflight is not None:
+                    data, self._inflight = self._inflight, None
+                else:
+                    try:
+                        data = self._send_q.get(timeout=0.2)
+                    except queue.Empty:
+                        continue
                 target = self._inproc
                 delivered = False
                 if target is not None and not target._closed.is_set():
@@ -783,7 +787,7 @@ class PairDialer:
                     except queue.Full:
                         pass
                 if not delivered:
-                    self._requeue_front(data)
+                    self._inflight = data
                 continue
 
             if self._conn is None:
@@ -793,13 +797,18 @@ class PairDialer:
                     continue
                 self._connected.set()
                 self._log.debug("dialer connected to %s", self.addr)
-            try:
-                data = self._send_q.get(timeout=0.2)
-            except queue.Empty:
-                continue
+            # the in-flight slot preserves ordering across reconnects (a
+            # plain re-put would append BEHIND frames queued meanwhile)
+            if self._inflight is not None:
+                data, self._inflight = self._inflight, None
+            else:
+                try:
+                    data = self._send_q.get(timeout=0.2)
+                except queue.Empty:
+                    continue
             conn = self._conn
             if conn is None:
-                self._requeue_front(data)
+                self._inflight = data
                 continue
             # coalesce everything queued into ONE sendall (per-frame
             # sendall measured ~13k frames/s; batching removes the
@@ -838,7 +847,7 @@ class PairDialer:
                 if self._conn is conn:
                     self._conn = None
                 self._connected.clear()
-                self._requeue_front(data)
+                self._inflight = data  # retried first after reconnect
         # drain on close
         conn = self._conn
         if conn is not None:
@@ -846,13 +855,6 @@ class PairDialer:
                 conn.close()
             except OSError:
                 pass
-
-    def _requeue_front(self, data: bytes) -> None:
-        """Best-effort put-back so a reconnect retries the in-flight frame."""
-        try:
-            self._send_q.put(data, block=False)
-        except queue.Full:
-            pass  # buffer full while down: frame is dropped (engine counts it)
 
     # -- EngineSocket API ----------------------------------------------
     def send(self, data: bytes, block: bool = True) -> bool:

@@ -270,3 +270,29 @@ def test_ws_handshake_is_rfc6455():
     finally:
         raw.close()
         listener.close()
+
+
+def test_late_binding_preserves_order(ipc_addr):
+    """Frames queued while the peer is down deliver IN ORDER on connect:
+    the writer's in-flight slot retries the dequeued frame FIRST (a naive
+    re-queue appended it behind newer frames and reordered)."""
+    dialer = PairDialer(ipc_addr, buffer_size=64)
+    try:
+        for i in range(6):
+            assert dialer.send(b"b%d" % i, block=False)
+        # let the writer dequeue b0 into the in-flight slot while down
+        time.sleep(0.5)
+        listener = PairListener(ipc_addr)
+        try:
+            got = []
+            deadline = time.monotonic() + 10
+            while len(got) < 6 and time.monotonic() < deadline:
+                try:
+                    got.extend(listener.recv_many(16, 300, 20))
+                except RecvTimeout:
+                    pass
+            assert got == [b"b%d" % i for i in range(6)]
+        finally:
+            listener.close()
+    finally:
+        dialer.close()
