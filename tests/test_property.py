@@ -1,0 +1,134 @@
+"""Property-based tests (hypothesis): codec roundtrips, matcher soundness,
+C++/Python codec agreement on arbitrary inputs."""
+import math
+
+import pytest
+from hypothesis import given, settings, strategies as st
+
+from detectmateservice_amd import ops
+from detectmateservice_amd.library.parsers.template_matcher import (
+    match_template,
+    split_template,
+)
+from detectmateservice_amd.schemas import DetectorSchema, LogSchema, ParserSchema
+
+_text = st.text(
+    alphabet=st.characters(blacklist_categories=("Cs",)), max_size=200
+)
+_i32 = st.integers(min_value=-(2**31), max_value=2**31 - 1)
+
+
+@settings(max_examples=200, deadline=None)
+@given(
+    log_id=_text, log=_text, source=_text, host=_text
+)
+def test_log_schema_roundtrip_property(log_id, log, source, host):
+    s = LogSchema(logID=log_id, log=log, logSource=source, hostname=host)
+    back = LogSchema.deserialize(s.serialize())
+    assert back.logID == log_id
+    assert back.log == log
+    assert back.logSource == source
+    assert back.hostname == host
+
+
+@settings(max_examples=200, deadline=None)
+@given(
+    event=_i32,
+    variables=st.lists(_text, max_size=8),
+    fmt_vars=st.dictionaries(
+        st.text(min_size=1, max_size=20), _text, max_size=6
+    ),
+    ts=_i32,
+)
+def test_parser_schema_roundtrip_property(event, variables, fmt_vars, ts):
+    s = ParserSchema(
+        EventID=event, variables=variables, logFormatVariables=fmt_vars,
+        receivedTimestamp=ts,
+    )
+    back = ParserSchema.deserialize(s.serialize())
+    assert back.EventID == event
+    assert back.variables == variables
+    assert back.logFormatVariables == fmt_vars
+    assert back.receivedTimestamp == ts
+
+
+@settings(max_examples=100, deadline=None)
+@given(
+    score=st.floats(allow_nan=False, allow_infinity=False, width=32),
+    stamps=st.lists(_i32, max_size=6),
+    alerts=st.dictionaries(st.text(min_size=1, max_size=16), _text, max_size=4),
+)
+def test_detector_schema_roundtrip_property(score, stamps, alerts):
+    s = DetectorSchema(score=score, extractedTimestamps=stamps, alertsObtain=alerts)
+    back = DetectorSchema.deserialize(s.serialize())
+    assert math.isclose(back.score, score, rel_tol=1e-6, abs_tol=1e-30) or (
+        score == 0.0 and back.score == 0.0
+    )
+    assert back.extractedTimestamps == stamps
+    assert back.alertsObtain == alerts
+
+
+@settings(max_examples=200, deadline=None)
+@given(
+    segments=st.lists(
+        st.text(alphabet="abcXY=_ ", min_size=1, max_size=6),
+        min_size=1, max_size=5,
+    ),
+    fillers=st.lists(st.text(alphabet="0123456789/", max_size=8), max_size=6),
+)
+def test_match_template_soundness(segments, fillers):
+    """If match_template returns captures, substituting them back into the
+    template reconstructs the content exactly. (Non-empty segments only:
+    adjacent wildcards — "<*><*>" — are inherently ambiguous and collapse,
+    in both the Python matcher and the HIP kernel.)"""
+    template = "<*>".join(segments)
+    segs = split_template(template)
+    n_wild = len(segs) - 1
+    fill = (fillers + [""] * n_wild)[:n_wild]
+    content = segs[0] + "".join(f + s for f, s in zip(fill, segs[1:]))
+    caps = match_template(content, segs)
+    if caps is None:
+        return  # matcher may reject ambiguous constructions; soundness only
+    rebuilt = segs[0]
+    ci = 0
+    for s_ in segs[1:]:
+        rebuilt += (caps[ci] if ci < len(caps) else "") + s_
+        ci += 1
+    assert rebuilt == content
+
+
+@settings(max_examples=100, deadline=None)
+@given(
+    lines=st.lists(
+        st.binary(max_size=100).map(lambda b: b.replace(b"\x00", b"x")),
+        min_size=1, max_size=10,
+    )
+)
+def test_cpp_decode_log_batch_matches_python(lines):
+    if not ops.have_extension():
+        pytest.skip("extension not built")
+    from detectmateservice_amd.ops import _dmx_C
+
+    frames = [
+        LogSchema(
+            logID=f"id{i}", log=l.decode("latin-1"), logSource="s"
+        ).serialize()
+        for i, l in enumerate(lines)
+    ]
+    blines, lens, ids, _s, _h = _dmx_C.decode_log_batch(frames, 128)
+    for i, l in enumerate(lines):
+        expect = l.decode("latin-1").encode("utf-8")[:128]
+        got = bytes(blines[i, : int(lens[i])].numpy().tobytes())
+        assert got == expect
+        assert bytes(ids[i]) == f"id{i}".encode()
+
+
+@settings(max_examples=100, deadline=None)
+@given(data=st.binary(max_size=300))
+def test_decoder_never_crashes_on_garbage(data):
+    """Arbitrary bytes must never raise unhandled errors beyond ValueError."""
+    for cls in (LogSchema, ParserSchema, DetectorSchema):
+        try:
+            cls.deserialize(data)
+        except ValueError:
+            pass
