@@ -400,3 +400,38 @@ def test_mfma_32x32x16_layout_probe():
     ref = A.float().cpu() @ B.float().cpu()
     err = (D - ref).abs().max()
     assert err < 0.1, f"32x32x16 layout mismatch: max err {err}"
+
+
+def test_bert_fused_four_layers():
+    """The fused kernel loops arbitrary layer counts (runtime param)."""
+    from detectmateservice_amd import ops
+    from detectmateservice_amd.models.bert_tiny import (
+        BertTinyConfig,
+        BertTinyDetectorModel,
+    )
+    from detectmateservice_amd.utils.synthetic import AuditLogGenerator
+
+    cfg = BertTinyConfig(layers=4)
+    gpu = BertTinyDetectorModel(cfg, device="cuda", seed=21)
+    cpu = BertTinyDetectorModel(cfg, device="cpu", seed=21)
+    assert gpu._fused_ok()
+    gen = AuditLogGenerator(seed=2)
+    raw = [gen.line()[0].encode() for _ in range(64)]
+    lines, lens = ops.pack_lines(raw, 256, device="cuda")
+    start = torch.zeros(64, dtype=torch.int32, device="cuda")
+    fused = gpu.score_spans(lines, start, lens.int()).cpu()
+    tokens = cpu.tokenize_spans(lines.cpu(), start.cpu(), lens.int().cpu())
+    ref = cpu.forward(tokens)
+    assert (fused - ref).abs().max() < 0.1
+
+
+def test_ops_fail_loudly_without_extension(monkeypatch):
+    """On a GPU box, a CUDA tensor hitting the ops layer with the
+    extension missing must raise (no silent eager fallback)."""
+    from detectmateservice_amd import ops
+
+    x = torch.randn(64, 128).bfloat16().cuda()
+    wt = torch.randn(128, 128).bfloat16().cuda()
+    monkeypatch.setattr(ops, "_C", None)
+    with pytest.raises(RuntimeError, match="not built"):
+        ops.fused_linear(x, wt)
