@@ -495,15 +495,31 @@ class PairListener:
             ).start()
 
     def _reader_loop(self, conn: socket.socket) -> None:
-        reader = (_WsFrameReader(conn, server_side=True) if self._ws
-                  else _FrameReader(conn, self._sp))
+        # Native C++ receive loop for plain fd sockets (GIL released around
+        # poll/recv/parse — frame_reader.cpp); Python readers for TLS/ws.
+        native = None
+        if not self._ws and self._ssl_ctx is None:
+            try:
+                from ..ops import _dmx_C  # type: ignore[attr-defined]
+
+                native = _dmx_C.FdFrameReader(conn.fileno(), self._sp)
+            except Exception:  # noqa: BLE001 - extension absent: python path
+                native = None
         try:
-            while not self._closed.is_set():
-                frames = reader.next_frames()
-                if frames is None:
-                    break
-                self._recv_q.put((conn, frames))
-        except (OSError, ValueError):
+            if native is not None:
+                while not self._closed.is_set():
+                    frames = native.read_batch(4096, 200)
+                    if frames:
+                        self._recv_q.put((conn, frames))
+            else:
+                reader = (_WsFrameReader(conn, server_side=True) if self._ws
+                          else _FrameReader(conn, self._sp))
+                while not self._closed.is_set():
+                    frames = reader.next_frames()
+                    if frames is None:
+                        break
+                    self._recv_q.put((conn, frames))
+        except (OSError, ValueError, EOFError, RuntimeError):
             pass
         finally:
             with self._peers_lock:

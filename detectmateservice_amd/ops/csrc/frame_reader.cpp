@@ -1,0 +1,94 @@
+// Native frame reader: the service data plane's hot receive loop in C++.
+//
+// The reference's data plane is the NNG C library; this
+// framework's equivalent native piece reads a connected socket fd, parses
+// length-prefixed frames (4-byte intra-node or 8-byte NNG-SP framing) and
+// returns a BATCH of frames per call — with the GIL released around
+// poll/read/parse so the Python engine thread never serializes on
+// per-frame syscalls. (The Python reader measured ~13k frames/s
+// pre-batching; the batched Python reader ~113k; this removes the
+// remaining per-chunk Python overhead for plain tcp/ipc peers. TLS and ws
+// connections keep the Python reader — ssl objects are not plain fds.)
+#include <torch/extension.h>
+
+#include <poll.h>
+#include <sys/socket.h>
+#include <unistd.h>
+
+#include <cstring>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+namespace {
+
+constexpr size_t kMaxFrame = 64ull * 1024 * 1024;
+
+class FdFrameReader {
+ public:
+  FdFrameReader(int fd, bool sp) : fd_(fd), sp_(sp) {}
+
+  // Read until at least one complete frame is available (or timeout/EOF).
+  // Returns: list of frame BYTES (py::bytes, not str); empty list on
+  // timeout; raises EOFError on peer close.
+  py::list read_batch(int max_frames, int timeout_ms) {
+    std::vector<std::string> out;
+    const size_t hdr = sp_ ? 8 : 4;
+    for (;;) {
+      // slice out complete frames already buffered
+      size_t pos = 0;
+      while ((int)out.size() < max_frames && buf_.size() - pos >= hdr) {
+        uint64_t len = 0;
+        for (size_t i = 0; i < hdr; ++i)
+          len = (len << 8) | (unsigned char)buf_[pos + i];
+        if (len > kMaxFrame) throw std::runtime_error("oversize frame");
+        if (buf_.size() - pos < hdr + len) break;
+        out.emplace_back(buf_.data() + pos + hdr, len);
+        pos += hdr + len;
+      }
+      if (pos) buf_.erase(buf_.begin(), buf_.begin() + pos);
+      if (!out.empty()) return to_bytes(out);
+
+      // blocking wait + read with the GIL released
+      int rc;
+      ssize_t n;
+      char tmp[262144];
+      {
+        py::gil_scoped_release release;
+        struct pollfd pfd{fd_, POLLIN, 0};
+        rc = ::poll(&pfd, 1, timeout_ms);
+        if (rc > 0) {
+          n = ::recv(fd_, tmp, sizeof(tmp), 0);
+        }
+      }
+      if (rc == 0) return to_bytes(out);     // timeout: empty
+      if (rc < 0) throw std::runtime_error("poll failed");
+      if (n == 0) {
+        PyErr_SetString(PyExc_EOFError, "peer closed");
+        throw py::error_already_set();
+      }
+      if (n < 0) throw std::runtime_error("recv failed");
+      buf_.insert(buf_.end(), tmp, tmp + n);
+    }
+  }
+
+ private:
+  static py::list to_bytes(const std::vector<std::string>& v) {
+    py::list out;
+    for (const auto& s : v) out.append(py::bytes(s));
+    return out;
+  }
+
+  int fd_;
+  bool sp_;
+  std::vector<char> buf_;
+};
+
+}  // namespace
+
+void register_frame_reader(py::module_& m) {
+  py::class_<FdFrameReader>(m, "FdFrameReader")
+      .def(py::init<int, bool>(), py::arg("fd"), py::arg("sp"))
+      .def("read_batch", &FdFrameReader::read_batch,
+           py::arg("max_frames") = 4096, py::arg("timeout_ms") = 200);
+}

@@ -19,6 +19,7 @@ CSRC = ROOT / "detectmateservice_amd" / "ops" / "csrc"
 sources = [
     str(CSRC / "bindings.cpp"),
     str(CSRC / "codec.cpp"),
+    str(CSRC / "frame_reader.cpp"),
     str(CSRC / "gemm_bf16.hip"),
     str(CSRC / "layernorm.hip"),
     str(CSRC / "attention.hip"),
