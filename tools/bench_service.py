@@ -107,10 +107,11 @@ def main():
 
         def pump(batch):
             sent = 0
-            for f in batch:
-                while not feeder.send(f, block=False):
+            while sent < len(batch):
+                n = feeder.send_many(batch[sent:sent + 4096], block=False)
+                if n == 0:
                     time.sleep(0.0005)
-                sent += 1
+                sent += n
             return sent
 
         pump(warmup)
