@@ -118,6 +118,17 @@ class Service:
         # passthrough mode for core* component types (reference core.py:204-206)
         return list(frames)
 
+    def source_batches(self, batch_size: int, stop_event):
+        """Source-mode delegation (engine_source_mode: reader services)."""
+        if self.library_component is None or not hasattr(
+            self.library_component, "stream_batches"
+        ):
+            raise RuntimeError(
+                f"component {self.settings.component_type!r} does not support "
+                "source mode (no stream_batches)"
+            )
+        return self.library_component.stream_batches(batch_size, stop_event)
+
     # ------------------------------------------------------------------
     # lifecycle (reference core.py:209-353)
     # ------------------------------------------------------------------

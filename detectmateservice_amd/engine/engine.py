@@ -146,6 +146,8 @@ class Engine:
 
     # ------------------------------------------------------------------
     def _run_loop(self) -> None:
+        if self.settings.engine_source_mode:
+            return self._run_source_loop()
         s = self.settings
         m = self.metrics
         self._log.info(
@@ -206,6 +208,31 @@ class Engine:
                         m.data_dropped_bytes_total.inc(len(out))
                         m.data_dropped_lines_total.inc(_count_lines(out))
         self._log.info("engine loop exited")
+
+    def _run_source_loop(self) -> None:
+        """Source mode: the component generates frames (reader services).
+
+        The input socket stays bound (admin/back-channel parity) but the
+        data flows component → outputs."""
+        s = self.settings
+        m = self.metrics
+        gen = self.processor.source_batches(s.engine_batch_size, self._stop_event)
+        self._log.info("engine source loop started (outputs=%d)", len(self._out_socks))
+        for batch in gen:
+            if self._stop_event.is_set():
+                break
+            t0 = time.perf_counter()
+            n_bytes = sum(len(f) for f in batch)
+            n_lines = sum(_count_lines(f) for f in batch)
+            m.data_read_bytes_total.inc(n_bytes)
+            m.data_read_lines_total.inc(n_lines)
+            m.data_processed_bytes_total.inc(n_bytes)
+            m.data_processed_lines_total.inc(n_lines)
+            m.engine_batch_size.observe(len(batch))
+            for out in batch:
+                self._send_to_outputs(out)
+            m.observe_batch(time.perf_counter() - t0, n_lines)
+        self._log.info("engine source loop exited")
 
     def _send_to_outputs(self, data: bytes) -> None:
         """Broadcast to all outputs with retry-then-drop per socket

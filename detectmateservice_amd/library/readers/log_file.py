@@ -22,6 +22,9 @@ class LogFileConfig(CoreConfig):
     log_source: str = "file"
     hostname: str = ""
     max_lines: Optional[int] = None
+    #: source-mode streaming: keep tailing the file for appended lines
+    follow: bool = False
+    poll_interval_s: float = 0.2
 
 
 class FileReaderConfig(LogFileConfig):
@@ -54,6 +57,43 @@ class FileReader(CoreComponent):
         if not line:
             return None
         return self._wrap(line).serialize()
+
+    def stream_batches(self, batch_size: int = 256, stop_event=None) -> Iterator[List[bytes]]:
+        """Source-mode generator: serialized LogSchema frames in batches.
+
+        With ``follow=True`` the reader tails the file (fluentd-style
+        ingestion without fluentd); otherwise it ends at EOF. Used by the
+        engine's source mode (``engine_source_mode: true``)."""
+        import time as _time
+
+        if not self.config.path:
+            raise ValueError("FileReader requires config.path")
+        limit = self.config.max_lines
+        emitted = 0
+        batch: List[bytes] = []
+        with open(self.config.path, "r", encoding="utf-8", errors="replace") as fh:
+            while stop_event is None or not stop_event.is_set():
+                line = fh.readline()
+                if not line:
+                    if batch:
+                        yield batch
+                        batch = []
+                    if not self.config.follow:
+                        return
+                    _time.sleep(self.config.poll_interval_s)
+                    continue
+                line = line.rstrip("\n")
+                if not line:
+                    continue
+                batch.append(self._wrap(line).serialize())
+                emitted += 1
+                if limit is not None and emitted >= limit:
+                    if batch:
+                        yield batch
+                    return
+                if len(batch) >= batch_size:
+                    yield batch
+                    batch = []
 
     def read(self) -> Iterator[LogSchema]:
         if not self.config.path:

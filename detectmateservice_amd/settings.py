@@ -45,10 +45,10 @@ from pydantic import BaseModel, ConfigDict, Field, field_validator, model_valida
 ENV_PREFIX = "DETECTMATE_"
 ENV_NESTED_DELIMITER = "__"
 
-#: Accepted engine address schemes (reference settings.py:30-37). ``ws`` is
-#: accepted for settings parity but carried over the same length-prefixed TCP
-#: framing as ``tcp`` (this framework defines its own wire framing; NNG is
-#: not part of the stack).
+#: Accepted engine address schemes (reference settings.py:30-37).
+#: Wire framings: tcp/tls+tcp speak the NNG SP mapping, ws speaks
+#: RFC 6455, ipc/inproc use the compact intra-node framing — see
+#: engine/sockets.py.
 _ADDR_SCHEMES = ("ipc", "tcp", "tls+tcp", "ws", "inproc")
 _ADDR_RE = re.compile(r"^(?P<scheme>[a-z+]+)://(?P<rest>.+)$")
 
@@ -128,6 +128,10 @@ class ServiceSettings(BaseModel):
     engine_recv_timeout: int = Field(default=100, ge=1, description="recv poll timeout, ms")
     engine_retry_count: int = Field(default=10, ge=1)
     engine_buffer_size: int = Field(default=128, ge=0, le=8192, description="per-socket queued frames")
+
+    #: source mode: the component GENERATES frames (readers tailing a
+    #: file) instead of receiving them on the engine socket
+    engine_source_mode: bool = False
 
     # --- batching (MI355X-native; no reference equivalent) ---
     engine_batch_size: int = Field(default=256, ge=1, le=1_048_576)

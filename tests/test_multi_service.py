@@ -98,3 +98,51 @@ def test_file_reader_streaming(tmp_path):
     frame = reader.process(b"delta\n")
     assert LogSchema.deserialize(frame).log == "delta"
     assert reader.process(b"\n") is None
+
+
+def test_reader_source_service(tmp_path):
+    """A FileReader SERVICE in source mode tails a file and pushes frames
+    to its outputs — a self-contained ingestion stage (no fluentd)."""
+    import yaml
+
+    log_file = tmp_path / "live.log"
+    log_file.write_text("first line\nsecond line\n")
+    out_addr = f"ipc://{tmp_path}/reader-out.ipc"
+    cfg_file = tmp_path / "reader.yaml"
+    cfg_file.write_text(yaml.safe_dump({
+        "readers": {"FileReader": {
+            "path": str(log_file), "follow": True, "poll_interval_s": 0.05,
+        }}
+    }))
+    settings = ServiceSettings(
+        component_type="FileReader",
+        engine_addr=f"ipc://{tmp_path}/reader-in.ipc",
+        out_addr=[out_addr],
+        engine_source_mode=True,
+        http_enabled=False,
+        config_file=cfg_file,
+        log_dir=tmp_path / "logs",
+    )
+    sink = PairListener(out_addr)
+    svc = Service(settings)
+    t = threading.Thread(target=svc.run, daemon=True)
+    t.start()
+    try:
+        logs = []
+        deadline = time.monotonic() + 10
+        while len(logs) < 2 and time.monotonic() < deadline:
+            try:
+                logs.append(LogSchema.deserialize(sink.recv(timeout_ms=500)))
+            except Exception:  # noqa: BLE001
+                pass
+        assert [l.log for l in logs] == ["first line", "second line"]
+        # follow mode: appended lines flow through
+        with open(log_file, "a") as fh:
+            fh.write("third line\n")
+        tail = LogSchema.deserialize(sink.recv(timeout_ms=10000))
+        assert tail.log == "third line"
+        assert tail.logSource == "file"
+    finally:
+        svc.shutdown()
+        t.join(timeout=5.0)
+        sink.close()
