@@ -94,14 +94,19 @@ class Engine:
     def _setup_output_sockets(self):
         socks = []
         for addr in self.settings.out_addr:
-            socks.append(
-                self._factory.create_dialer(
-                    addr,
-                    logger=self._log,
-                    tls_config=self.settings.tls_output,
-                    buffer_size=self.settings.engine_buffer_size,
-                )
+            kwargs = dict(
+                logger=self._log,
+                tls_config=self.settings.tls_output,
+                buffer_size=self.settings.engine_buffer_size,
             )
+            try:
+                socks.append(self._factory.create_dialer(
+                    addr, dial_timeout_s=self.settings.dial_timeout / 1000.0,
+                    **kwargs,
+                ))
+            except TypeError:
+                # injected test factories may not take dial_timeout_s
+                socks.append(self._factory.create_dialer(addr, **kwargs))
         return socks
 
     @property

@@ -697,9 +697,11 @@ class PairDialer:
         logger: Optional[logging.Logger] = None,
         tls_config: Optional[TlsOutputConfig] = None,
         buffer_size: int = 128,
+        dial_timeout_s: float = 1.0,
     ) -> None:
         self.addr = EngineAddr.validate(addr)
         self._log = logger or logging.getLogger(__name__)
+        self._dial_timeout_s = max(dial_timeout_s, 0.05)
         self._send_q: "queue.Queue[bytes]" = queue.Queue(maxsize=max(buffer_size, 1))
         self._recv_q: "queue.Queue[bytes]" = queue.Queue(maxsize=max(buffer_size, 1) * 4)
         self._closed = threading.Event()
@@ -728,12 +730,14 @@ class PairDialer:
         try:
             if scheme == "ipc":
                 s = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
-                s.settimeout(1.0)
+                s.settimeout(self._dial_timeout_s)
                 s.connect("/" + rest.lstrip("/"))
             else:
                 host, port = rest.rsplit(":", 1)
                 host = host.strip("[]")
-                s = socket.create_connection((host, int(port)), timeout=1.0)
+                s = socket.create_connection(
+                    (host, int(port)), timeout=self._dial_timeout_s
+                )
                 if scheme == "tls+tcp":
                     if self._tls_config is None:
                         raise ValueError("tls+tcp dialer requires tls_output config")
@@ -953,7 +957,9 @@ class PairSocketFactory:
         logger: Optional[logging.Logger] = None,
         tls_config: Optional[TlsOutputConfig] = None,
         buffer_size: int = 128,
+        dial_timeout_s: float = 1.0,
     ) -> PairDialer:
         return PairDialer(
-            addr, logger=logger, tls_config=tls_config, buffer_size=buffer_size
+            addr, logger=logger, tls_config=tls_config, buffer_size=buffer_size,
+            dial_timeout_s=dial_timeout_s,
         )
