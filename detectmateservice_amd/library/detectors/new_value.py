@@ -130,7 +130,17 @@ class NewValueDetector(CoreDetector):
     def process_batch(self, frames: List[bytes]) -> List[Optional[bytes]]:
         from ... import ops as _ops
 
-        if len(frames) < 8 or not _ops.have_extension() or not self.specs:
+        from ...components.base import BufferMode
+
+        mode = getattr(self.config, "buffer_mode", BufferMode.NO_BUF)
+        if isinstance(mode, str):
+            mode = BufferMode(mode)
+        if (
+            len(frames) < 8
+            or not _ops.have_extension()
+            or not self.specs
+            or mode != BufferMode.NO_BUF  # windowed modes need per-frame flow
+        ):
             return super().process_batch(frames)
         from ...ops import _dmx_C  # type: ignore[attr-defined]
 
