@@ -74,3 +74,18 @@ def test_client_cli_against_live_service(free_port, tmp_path, capsys):
     finally:
         svc.shutdown()
         t.join(timeout=10.0)
+
+
+def test_service_logger_file_handler(tmp_path):
+    from detectmateservice_amd.utils.logging import build_service_logger
+
+    log = build_service_logger("testtype", "tid-1", "DEBUG", tmp_path / "lg")
+    log.info("hello file")
+    for h in log.handlers:
+        h.flush()
+    f = tmp_path / "lg" / "testtype_tid-1.log"
+    assert f.exists() and "hello file" in f.read_text()
+    # duplicate-handler guard: building again adds no handlers
+    n = len(log.handlers)
+    build_service_logger("testtype", "tid-1", "DEBUG", tmp_path / "lg")
+    assert len(log.handlers) == n

@@ -160,3 +160,49 @@ def test_detector_service_pipeline(ipc_addr, tmp_path, free_port):
         client.close()
         service.shutdown()
         t.join(timeout=5.0)
+
+
+def test_status_includes_live_counters(running_service):
+    service, url = running_service
+    client = PairDialer(service.settings.engine_addr)
+    try:
+        assert client.wait_connected(5.0)
+        client.send(b"count-me")
+        assert client.recv(timeout_ms=3000) == b"count-me"
+        deadline = time.time() + 5
+        while time.time() < deadline:
+            m = httpx.get(f"{url}/admin/status", timeout=5.0).json()["metrics"]
+            if m["processed_lines"] and m["processed_lines"] >= 1:
+                break
+            time.sleep(0.1)
+        assert m["read_lines"] >= 1
+        assert m["processed_lines"] >= 1
+        assert m["processing_errors"] == 0
+    finally:
+        client.close()
+
+
+def test_service_over_websocket_engine(tmp_path, free_port):
+    """A Service listening on ws:// (real RFC6455 framing end-to-end)."""
+    settings = ServiceSettings(
+        component_type="core",
+        engine_addr="ws://127.0.0.1:0",
+        http_enabled=False,
+        log_dir=tmp_path / "logs",
+    )
+    service = Service(settings)
+    port = service.engine._pair_sock.bound_port
+    t = threading.Thread(target=service.run, daemon=True)
+    t.start()
+    try:
+        time.sleep(0.2)
+        client = PairDialer(f"ws://127.0.0.1:{port}")
+        try:
+            assert client.wait_connected(5.0)
+            client.send(b"ws-frame")
+            assert client.recv(timeout_ms=3000) == b"ws-frame"
+        finally:
+            client.close()
+    finally:
+        service.shutdown()
+        t.join(timeout=5.0)
