@@ -59,7 +59,7 @@ def fused_linear(
         return _require_ext().fused_linear_bf16(x, wt, bias, EPILOGUE[activation])
     y = torch.nn.functional.linear(x.float(), wt.float(), bias)
     if activation == "gelu":
-        y = torch.nn.functional.gelu(y)
+        y = torch.nn.functional.gelu(y, approximate="tanh")
     elif activation == "relu":
         y = torch.relu(y)
     return y.to(x.dtype)

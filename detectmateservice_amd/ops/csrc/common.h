@@ -49,9 +49,15 @@ static __device__ __forceinline__ short f32_to_bf16(float f) {
   return v.s;
 }
 
-// GELU (erf form, matches torch.nn.functional.gelu default)
+// GELU, tanh approximation (torch gelu(approximate="tanh"); the GPT-2 /
+// BERT-family standard approximation). The erf form cost ~2x the VALU
+// ops (libdevice erff is a long polynomial) and the fused kernel is
+// VALU-issue-bound; the framework uses the tanh form consistently
+// (kernels, CPU fallbacks, tests).
 static __device__ __forceinline__ float gelu_f32(float x) {
-  return 0.5f * x * (1.0f + erff(x * 0.70710678118654752440f));
+  const float c = 0.7978845608028654f;  // sqrt(2/pi)
+  const float t = tanhf(c * (x + 0.044715f * x * x * x));
+  return 0.5f * x * (1.0f + t);
 }
 
 static __device__ __forceinline__ float warp_reduce_sum_f32(float v) {
