@@ -22,7 +22,14 @@ def main(argv=None) -> int:
     parser.add_argument("--settings", help="service settings YAML", default=None)
     parser.add_argument("--config", help="component config YAML (overrides settings.config_file)", default=None)
     parser.add_argument("--log-level", default=None)
+    parser.add_argument("--version", action="store_true", help="print version and exit")
     args = parser.parse_args(argv)
+
+    if args.version:
+        from .metadata import __framework__, __version__
+
+        print(f"{__framework__} {__version__}")
+        return 0
 
     if not args.settings:
         print("error: --settings is required", file=sys.stderr)

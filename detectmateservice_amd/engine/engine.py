@@ -199,14 +199,18 @@ class Engine:
             m.data_processed_lines_total.inc(read_lines)
             m.observe_batch(elapsed, read_lines)
 
-            for out in outs:
+            for i, out in enumerate(outs):
                 if out is None:
                     continue  # filtered (engine.py:238-240)
                 if self._out_socks:
                     self._send_to_outputs(out)
                 else:
-                    # request/reply fallback mode (engine.py:248-264)
-                    if self._pair_sock.send(out, block=False):
+                    # request/reply fallback mode (engine.py:248-264);
+                    # replies route to EACH frame's sender (multi-peer)
+                    reply = getattr(self._pair_sock, "reply", None)
+                    ok = (reply(i, out) if reply is not None
+                          else self._pair_sock.send(out, block=False))
+                    if ok:
                         m.data_written_bytes_total.inc(len(out))
                         m.data_written_lines_total.inc(_count_lines(out))
                     else:

@@ -559,19 +559,28 @@ class PairListener:
         first, then keep draining without waiting more than ``linger_ms`` total.
 
         This is the batched-engine entry point (SURVEY.md §7 design
-        departures) — no reference equivalent.
+        departures) — no reference equivalent. ``_batch_conns`` records
+        each frame's source connection so the request/reply mode routes
+        every reply to ITS sender even with several concurrent peers
+        (the reference's Pair0 was strictly 1:1 and never faced this).
         """
         out: List[bytes] = list(getattr(self, "_pending", None) or [])
+        conns: List = list(getattr(self, "_pending_conns", None) or [])
         self._pending = []
+        self._pending_conns = []
         if len(out) >= max_frames:
             self._pending = out[max_frames:]
+            self._pending_conns = conns[max_frames:]
+            self._batch_conns = conns[:max_frames]
             return out[:max_frames]
         try:
             if not out:
                 conn, frames = self._recv_q.get(timeout=timeout_ms / 1000.0)
                 self._last_sender = conn
                 out.extend(frames)
+                conns.extend([conn] * len(frames))
         except queue.Empty:
+            self._batch_conns = conns
             return out
         deadline = time.monotonic() + linger_ms / 1000.0
         while len(out) < max_frames:
@@ -585,10 +594,30 @@ class PairListener:
                 break
             self._last_sender = conn
             out.extend(frames)
+            conns.extend([conn] * len(frames))
         if len(out) > max_frames:
             self._pending = out[max_frames:]
+            self._pending_conns = conns[max_frames:]
             out = out[:max_frames]
+            conns = conns[:max_frames]
+        self._batch_conns = conns
         return out
+
+    def reply(self, idx: int, data: bytes) -> bool:
+        """Reply to the sender of frame ``idx`` of the last recv_many batch
+        (falls back to the last sender when unknown)."""
+        conns = getattr(self, "_batch_conns", None)
+        conn = conns[idx] if conns and 0 <= idx < len(conns) else self._last_sender
+        if conn is None:
+            return False
+        try:
+            if self._ws:
+                conn.sendall(_ws_encode(data, mask=False))
+            else:
+                _send_frame(conn, data, self._sp)
+            return True
+        except OSError:
+            return False
 
     def send(self, data: bytes, block: bool = True) -> bool:
         """Reply to the most recent sender (request/reply fallback mode)."""

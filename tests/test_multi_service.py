@@ -146,3 +146,67 @@ def test_reader_source_service(tmp_path):
         svc.shutdown()
         t.join(timeout=5.0)
         sink.close()
+
+
+def test_many_concurrent_clients_one_service(tmp_path):
+    """8 concurrent feeders (mixed schemes) against one echo service:
+    every client gets exactly its own replies, in order."""
+    import queue as q
+
+    from detectmateservice_amd.engine.sockets import RecvTimeout
+
+    settings = ServiceSettings(
+        component_type="core",
+        engine_addr=f"ipc://{tmp_path}/hub.ipc",
+        http_enabled=False,
+        log_dir=tmp_path / "logs",
+        engine_batch_linger_ms=2.0,
+    )
+    svc = Service(settings)
+    t = threading.Thread(target=svc.run, daemon=True)
+    t.start()
+    time.sleep(0.2)
+
+    results = q.Queue()
+
+    def client(cid):
+        try:
+            d = PairDialer(settings.engine_addr, buffer_size=512)
+            assert d.wait_connected(10.0)
+            sent = [b"c%d-%03d" % (cid, i) for i in range(50)]
+            got = []
+            # interleave sends and receives; replies route to the LAST
+            # sender, so serialize per client: send one, await its echo
+            for f in sent:
+                d.send(f)
+                deadline = time.monotonic() + 10
+                while time.monotonic() < deadline:
+                    try:
+                        r = d.recv(timeout_ms=500)
+                        got.append(r)
+                        break
+                    except RecvTimeout:
+                        continue
+            d.close()
+            results.put((cid, sent, got))
+        except Exception as exc:  # noqa: BLE001
+            results.put((cid, exc, None))
+
+    threads = [threading.Thread(target=client, args=(i,)) for i in range(8)]
+    for th in threads:
+        th.start()
+    for th in threads:
+        th.join(timeout=60)
+    try:
+        seen = 0
+        while seen < 8:
+            cid, sent, got = results.get(timeout=10)
+            assert not isinstance(sent, Exception), f"client {cid}: {sent}"
+            # each client's replies are its own frames, in order (the
+            # reply-to-last-sender routing holds because each client waits
+            # for its echo before sending the next frame)
+            assert got == sent, f"client {cid} got {len(got)}/{len(sent)}"
+            seen += 1
+    finally:
+        svc.shutdown()
+        t.join(timeout=5.0)
