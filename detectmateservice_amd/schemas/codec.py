@@ -181,6 +181,8 @@ def decode_message(spec: Dict[str, Tuple[int, str]], data: bytes) -> Dict[str, A
             v, pos = decode_varint(data, pos)
             values[name] = _to_int32(v)
         elif kind == FLOAT and wt == 5:
+            if pos + 4 > n:
+                raise ValueError("truncated fixed32 field")
             values[name] = struct.unpack_from("<f", data, pos)[0]
             pos += 4
         elif kind == REP_STRING and wt == 2:
