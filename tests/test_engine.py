@@ -256,3 +256,24 @@ def test_engine_restart(ipc_addr):
     finally:
         eng.stop()
         eng.close()
+
+
+def test_batch_size_one_strict_per_message(ipc_addr, run_engine):
+    """engine_batch_size=1 degenerates to the reference's strict
+    per-message loop with identical semantics."""
+    run_engine(SimpleProcessor(), engine_batch_size=1, engine_batch_linger_ms=0.0)
+    client = PairDialer(ipc_addr)
+    try:
+        assert client.wait_connected(5.0)
+        for i in range(10):
+            client.send(b"s%d" % i)
+        got = []
+        deadline = time.monotonic() + 10
+        while len(got) < 10 and time.monotonic() < deadline:
+            try:
+                got.append(client.recv(timeout_ms=300))
+            except RecvTimeout:
+                pass
+        assert got == [b"out:s%d" % i for i in range(10)]
+    finally:
+        client.close()

@@ -164,6 +164,88 @@ def test_three_stage_pipeline(pipeline_procs):
         sink.close()
 
 
+def test_four_stage_with_output_aggregator(tmp_path, free_port):
+    """parser -> detector -> OutputAggregator services; OutputSchema frames
+    arrive at the final sink (the fluentout-equivalent egress)."""
+    import socket as s_mod
+    import uuid as uuid_mod
+
+    from detectmateservice_amd.schemas import OutputSchema
+
+    uid = uuid_mod.uuid4().hex[:6]
+    det_in = f"ipc://{tmp_path}/d4-{uid}.ipc"
+    out_in = f"ipc://{tmp_path}/o4-{uid}.ipc"
+    sink_addr = f"ipc://{tmp_path}/s4-{uid}.ipc"
+    sock = s_mod.socket(); sock.bind(("127.0.0.1", 0))
+    out_port = sock.getsockname()[1]; sock.close()
+
+    det_settings = tmp_path / "d4.yaml"
+    det_settings.write_text(yaml.safe_dump({
+        "component_type": "NewValueDetector",
+        "engine_addr": det_in,
+        "out_addr": [out_in],
+        "http_port": free_port,
+        "config_file": str(tmp_path / "d4c.yaml"),
+        "log_dir": str(tmp_path / "logs"),
+    }))
+    (tmp_path / "d4c.yaml").write_text(yaml.safe_dump({
+        "detectors": {"NewValueDetector": {
+            "data_use_training": 2,
+            "global": {"g": {"header_variables": [{"pos": "Type"}]}},
+        }}
+    }))
+    agg_settings = tmp_path / "a4.yaml"
+    agg_settings.write_text(yaml.safe_dump({
+        "component_type": "OutputAggregator",
+        "engine_addr": out_in,
+        "out_addr": [sink_addr],
+        "http_port": out_port,
+        "config_file": str(tmp_path / "a4c.yaml"),
+        "log_dir": str(tmp_path / "logs"),
+    }))
+    (tmp_path / "a4c.yaml").write_text(yaml.safe_dump({
+        "outputs": {"OutputAggregator": {"window_size": 1}}
+    }))
+
+    procs = [subprocess.Popen(
+        [sys.executable, "-m", "detectmateservice_amd.cli", "--settings", str(st)],
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+    ) for st in (det_settings, agg_settings)]
+    sink = PairListener(sink_addr)
+    feeder = PairDialer(det_in)
+    try:
+        assert _wait_running(free_port)
+        assert _wait_running(out_port)
+        assert feeder.wait_connected(10.0)
+
+        def pframe(t, lid):
+            return ParserSchema(
+                EventID=1, logID=lid, logFormatVariables={"Type": t}
+            ).serialize()
+
+        feeder.send(pframe("LOGIN", "t1"))
+        feeder.send(pframe("USER_ACCT", "t2"))
+        time.sleep(0.8)
+        feeder.send(pframe("WEIRD_TYPE", "bad"))
+        out = OutputSchema.deserialize(sink.recv(timeout_ms=15000))
+        assert out.detectorTypes == ["new_value_detector"]
+        assert out.logIDs == ["bad"]
+        assert "WEIRD_TYPE" in out.description
+    finally:
+        feeder.close()
+        sink.close()
+        for port in (free_port, out_port):
+            try:
+                httpx.post(f"http://127.0.0.1:{port}/admin/shutdown", timeout=2.0)
+            except Exception:  # noqa: BLE001
+                pass
+        for p in procs:
+            try:
+                p.wait(timeout=10)
+            except subprocess.TimeoutExpired:
+                p.kill()
+
+
 def test_pipeline_metrics_flow(pipeline_procs):
     """Prometheus counters advance along the pipeline."""
     feeder = PairDialer(pipeline_procs["parser_in"])
