@@ -190,7 +190,12 @@ static __device__ __forceinline__ void block_layernorm(
 #define PH_FFN 16
 #define PH_ALL 31
 
-template <int PHASES>
+// Per-wave phase timing (diagnostic builds): stamps[k] = s_memrealtime at
+// phase boundaries; slot pairs (work-done, barrier-crossed) separate each
+// wave's compute time from its barrier wait. 100 MHz constant clock.
+#define BF_NSTAMP 22
+
+template <int PHASES, bool TIMED = false>
 static __device__ __forceinline__ void bert_fused_body(
     const unsigned char* __restrict__ lines,  // [B, max_len]
     const int* __restrict__ start,            // [B] content span start
@@ -198,7 +203,14 @@ static __device__ __forceinline__ void bert_fused_body(
     const short* __restrict__ wb,             // bf16 weight blob
     const float* __restrict__ fb,             // f32 bias blob
     float* __restrict__ scores,               // [B]
-    int B, int max_len, int n_layers, float eps) {
+    int B, int max_len, int n_layers, float eps,
+    unsigned long long* __restrict__ stamps_out = nullptr) {
+  unsigned long long tstamp[TIMED ? BF_NSTAMP : 1];
+  int tidx = 0;
+#define BF_STAMP()                                                          \
+  if constexpr (TIMED) {                                                    \
+    if (tidx < BF_NSTAMP) tstamp[tidx++] = __builtin_amdgcn_s_memrealtime(); \
+  }
   const int line = blockIdx.x;
   if (line >= B) return;
   const int tid = threadIdx.x;
@@ -212,6 +224,7 @@ static __device__ __forceinline__ void bert_fused_body(
   lds_short* vt = buf + BUF_ELEMS;     // [128][VTS]
   lds_float* red = (lds_float*)(vt + 2 * BF_DH * VTS);  // [128] pooling
 
+  BF_STAMP();  // 0: kernel start
   // ---- embed: x[s][c] = tok_emb[byte+3 or 0][c] + pos_emb[s][c] ----
   {
     const int s0 = start[line], e0 = end[line];
@@ -230,7 +243,9 @@ static __device__ __forceinline__ void bert_fused_body(
       *(__attribute__((address_space(3))) short8v*)(x_lds + s * XS + c) = xv;
     }
   }
+  BF_STAMP();  // 1: embed work done
   __syncthreads();
+  BF_STAMP();  // 2: embed barrier crossed
 
   for (int layer = 0; layer < n_layers; ++layer) {
     const short* lw = wb + WB_LAYER0 + (long)layer * LW_SIZE;
@@ -241,7 +256,9 @@ static __device__ __forceinline__ void bert_fused_body(
       block_gemm<BF_H, 3 * BF_H, 1, 0, BF_H>(x_lds, XS, (glob_cshort*)(lw + LW_QKV),
                                              (glob_cfloat*)(lf + FB_BQKV), buf, QKS, x_lds, vt,
                                              wid, lane);
+    BF_STAMP();  // qkv work done
     __syncthreads();
+    BF_STAMP();  // qkv barrier crossed
 
     // ---- attention: wave = (head hh, 16 q-rows) ----
     if (PHASES & PH_ATTN) {
@@ -328,7 +345,9 @@ static __device__ __forceinline__ void bert_fused_body(
         }
       }
     }
+    BF_STAMP();  // attention work done
     __syncthreads();
+    BF_STAMP();  // attention barrier crossed
 
     // ---- proj: x += Wo(attn) ; LN1 ----
     if (PHASES & PH_PROJ)
@@ -338,7 +357,9 @@ static __device__ __forceinline__ void bert_fused_body(
     __syncthreads();
     if (PHASES & PH_LN)
       block_layernorm(x_lds, lw + LW_LN1G, lw + LW_LN1B, wid, lane, eps);
+    BF_STAMP();  // proj+LN1 work done
     __syncthreads();
+    BF_STAMP();  // proj+LN1 barrier crossed
 
     // ---- FFN in two K=256 halves: buf = gelu(x@W1_h); x += buf@W2_h ----
     if (PHASES & PH_FFN)
@@ -358,7 +379,9 @@ static __device__ __forceinline__ void bert_fused_body(
     }
     if (PHASES & PH_LN)
       block_layernorm(x_lds, lw + LW_LN2G, lw + LW_LN2B, wid, lane, eps);
+    BF_STAMP();  // ffn+LN2 work done
     __syncthreads();
+    BF_STAMP();  // ffn+LN2 barrier crossed
   }
 
   // ---- pool (mean over S) + score head ----
@@ -379,6 +402,16 @@ static __device__ __forceinline__ void bert_fused_body(
         scores[line] = v + fb[(long)n_layers * FB_SIZE];  // b_score
     }
   }
+  BF_STAMP();  // final
+  if constexpr (TIMED) {
+    if (lane == 0 && stamps_out != nullptr) {
+      unsigned long long* dst =
+          stamps_out + ((long)line * BF_WAVES + wid) * BF_NSTAMP;
+      for (int k = 0; k < BF_NSTAMP; ++k)
+        dst[k] = k < tidx ? tstamp[k] : 0ull;
+    }
+  }
+#undef BF_STAMP
 }
 
 extern "C" __global__ __launch_bounds__(BF_THREADS, 4)
@@ -402,6 +435,32 @@ __global__ __launch_bounds__(BF_THREADS, 4) void dmx_bert_fused_probe(
     int n_layers, float eps) {
   bert_fused_body<PHASES>(lines, start, end, wb, fb, scores, B, max_len,
                           n_layers, eps);
+}
+
+extern "C" __global__ __launch_bounds__(BF_THREADS, 4)
+void dmx_bert_fused_timed(const unsigned char* lines, const int* start,
+                          const int* end, const short* wb, const float* fb,
+                          float* scores, int B, int max_len, int n_layers,
+                          float eps, unsigned long long* stamps) {
+  bert_fused_body<PH_ALL, true>(lines, start, end, wb, fb, scores, B,
+                                max_len, n_layers, eps, stamps);
+}
+
+extern "C" void dmx_launch_bert_fused_timed(
+    const void* lines, const void* start, const void* end, const void* wb,
+    const void* fb, void* scores, void* stamps, int B, int max_len,
+    int n_layers, float eps, hipStream_t stream) {
+  const size_t lds =
+      ((size_t)BF_S * XS + BUF_ELEMS + (size_t)2 * BF_DH * VTS) *
+          sizeof(short) +
+      128 * sizeof(float);
+  hipFuncSetAttribute((const void*)dmx_bert_fused_timed,
+                      hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+  hipLaunchKernelGGL(dmx_bert_fused_timed, dim3(B), dim3(BF_THREADS), lds,
+                     stream, (const unsigned char*)lines, (const int*)start,
+                     (const int*)end, (const short*)wb, (const float*)fb,
+                     (float*)scores, B, max_len, n_layers, eps,
+                     (unsigned long long*)stamps);
 }
 
 extern "C" void dmx_launch_bert_fused_probe(

@@ -28,6 +28,9 @@ void dmx_launch_bert_fused_bf16(const void*, const void*, const void*,
 void dmx_launch_bert_fused_probe(const void*, const void*, const void*,
                                  const void*, const void*, void*, int, int,
                                  int, float, int, hipStream_t);
+void dmx_launch_bert_fused_timed(const void*, const void*, const void*,
+                                 const void*, const void*, void*, void*, int,
+                                 int, int, float, hipStream_t);
 void dmx_launch_template_match(const void*, const void*, int, int,
                                const void*, const void*, int, const void*,
                                int, const void*, const void*, int, int,
@@ -223,6 +226,26 @@ torch::Tensor bert_fused_probe(torch::Tensor lines, torch::Tensor start,
   return scores;
 }
 
+std::vector<torch::Tensor> bert_fused_timed(torch::Tensor lines,
+                                            torch::Tensor start,
+                                            torch::Tensor end,
+                                            torch::Tensor wb,
+                                            torch::Tensor fb,
+                                            int64_t n_layers, double eps) {
+  const auto B = lines.size(0), max_len = lines.size(1);
+  auto scores = torch::empty(
+      {B}, torch::TensorOptions().dtype(torch::kFloat32).device(lines.device()));
+  auto stamps = torch::zeros(
+      {B, 8, 22},
+      torch::TensorOptions().dtype(torch::kInt64).device(lines.device()));
+  dmx_launch_bert_fused_timed(lines.data_ptr(), start.data_ptr(),
+                              end.data_ptr(), wb.data_ptr(), fb.data_ptr(),
+                              scores.data_ptr(), stamps.data_ptr(), (int)B,
+                              (int)max_len, (int)n_layers, (float)eps,
+                              cur_stream());
+  return {scores, stamps};
+}
+
 torch::Tensor edit_distance(torch::Tensor A, torch::Tensor a_len,
                             torch::Tensor B, torch::Tensor b_len) {
   TORCH_CHECK(A.is_cuda() && A.dtype() == torch::kUInt8 && A.is_contiguous());
@@ -295,6 +318,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "whole-model BERT-tiny forward, one workgroup per line");
   m.def("bert_fused_probe", &bert_fused_probe,
         "phase-masked probe variant of the fused BERT kernel");
+  m.def("bert_fused_timed", &bert_fused_timed,
+        "fused kernel with per-wave phase timestamps (diagnostic)");
   m.def("template_match", &template_match, "batched wildcard template match");
   m.def("watch_hashes", &watch_hashes, "hash watched capture spans");
   m.def("hashset_insert", &hashset_insert, "insert hashes into GPU sets");

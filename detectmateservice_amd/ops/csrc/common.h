@@ -55,9 +55,14 @@ static __device__ __forceinline__ short f32_to_bf16(float f) {
 // VALU-issue-bound; the framework uses the tanh form consistently
 // (kernels, CPU fallbacks, tests).
 static __device__ __forceinline__ float gelu_f32(float x) {
-  const float c = 0.7978845608028654f;  // sqrt(2/pi)
-  const float t = tanhf(c * (x + 0.044715f * x * x * x));
-  return 0.5f * x * (1.0f + t);
+  // EXACT identity: 0.5*(1+tanh(a)) == sigmoid(2a), so
+  //   gelu_tanh(x) = x * sigmoid(2c*(x + 0.044715 x^3)),
+  // computable with ONE v_exp + one reciprocal instead of libm tanhf
+  // (branchy, multiple transcendentals — it made the FFN phase half the
+  // fused kernel's per-wave work, tools/probe_bert_timing.py).
+  const float two_c = 1.5957691216057308f;  // 2*sqrt(2/pi)
+  const float z = two_c * (x + 0.044715f * x * x * x);
+  return x / (1.0f + __expf(-z));
 }
 
 static __device__ __forceinline__ float warp_reduce_sum_f32(float v) {
