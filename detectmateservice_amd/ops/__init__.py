@@ -205,6 +205,11 @@ class TemplateMatcher:
     def match_packed(self, lines: torch.Tensor, line_len: torch.Tensor):
         """lines [B, max_len] u8 on self.device. Returns dict of tensors:
         event_id [B], fmt_caps [B,Fc,2], n_fmt_caps [B], caps [B,C,2], n_caps [B]."""
+        if lines.shape[0] == 0:  # zero-grid kernel launches are invalid
+            z = lambda *shape: torch.zeros(shape, dtype=torch.int32, device=lines.device)  # noqa: E731
+            return {"event_id": z(0), "fmt_caps": z(0, self.max_fmt_caps, 2),
+                    "n_fmt_caps": z(0), "caps": z(0, self.max_caps, 2),
+                    "n_caps": z(0)}
         if lines.is_cuda:
             ev, fc, nfc, caps, ncaps = _require_ext().template_match(
                 lines, line_len, self.fmt_bytes, self.fmt_seg_off,

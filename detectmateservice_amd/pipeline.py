@@ -91,6 +91,14 @@ class GpuPipeline:
         Returns event_id [B], anomaly [B] (bool), scores [B] (f32, 0 when
         transformer off), nv_unseen [B, W]."""
         B = lines.shape[0]
+        if B == 0:
+            return {
+                "event_id": torch.zeros(0, dtype=torch.int32, device=lines.device),
+                "anomaly": torch.zeros(0, dtype=torch.bool, device=lines.device),
+                "scores": torch.zeros(0, dtype=torch.float32, device=lines.device),
+                "nv_unseen": None,
+                "match": None,
+            }
         # roctx-visible stage ranges (torch.cuda.nvtx maps to rocTracer
         # markers on ROCm — SURVEY.md §5.1 tracing requirement)
         _rng = torch.cuda.nvtx.range if lines.is_cuda else None

@@ -177,3 +177,22 @@ def test_pipeline_end_to_end_cpu():
     ).encode()
     out = pipe.process_lines(normal[:3] + [bad])
     assert out["anomaly"].tolist() == [False, False, False, True]
+
+
+def test_empty_batch_paths():
+    """B=0 never reaches a kernel launch (zero grids are invalid)."""
+    m = ops.TemplateMatcher(["a=<*>"], device="cpu")
+    lines = torch.zeros((0, 64), dtype=torch.uint8)
+    lens = torch.zeros((0,), dtype=torch.int32)
+    out = m.match_packed(lines, lens)
+    assert out["event_id"].shape == (0,)
+
+    cfg = PipelineConfig(templates=["a=<*>"], use_transformer=False)
+    pipe = GpuPipeline(cfg, device="cpu")
+    res = pipe.process_lines([])
+    assert res["anomaly"].shape == (0,)
+
+    from detectmateservice_amd.library.parsers import MatcherParser
+
+    p = MatcherParser({"templates": ["a=<*>"]})
+    assert p.process_batch([]) == []
