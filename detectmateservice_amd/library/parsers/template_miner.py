@@ -46,6 +46,8 @@ def edit_distances(
     from .. import __name__ as _  # noqa: F401
     from ... import ops
 
+    if len(queries) == 0 or len(refs) == 0:
+        return torch.zeros((len(queries), len(refs)), dtype=torch.int32)
     dev = torch.device(device)
     if dev.type == "cuda" and ops.have_extension():
         from ...ops import _dmx_C  # type: ignore[attr-defined]

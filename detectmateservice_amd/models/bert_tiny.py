@@ -177,6 +177,8 @@ class BertTinyDetectorModel:
         layers → pool → score, all in LDS). Otherwise: tokenize + layered
         forward (same numerics at bf16 tolerance — tests/test_gpu_ops.py).
         """
+        if lines.shape[0] == 0:
+            return torch.zeros(0, dtype=torch.float32, device=lines.device)
         if self._fused_ok():
             from ..ops import _dmx_C  # type: ignore[attr-defined]
 

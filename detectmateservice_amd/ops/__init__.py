@@ -320,6 +320,8 @@ class GpuHashSets:
             self.sets: List[set] = [set() for _ in range(n_watch)]
 
     def insert(self, hashes: torch.Tensor) -> None:
+        if hashes.shape[0] == 0:
+            return
         if self.device.type == "cuda":
             _require_ext().hashset_insert(hashes, self.tables)
         else:
@@ -328,6 +330,8 @@ class GpuHashSets:
                 self.sets[w].update(int(x) for x in h[:, w] if x != 0)
 
     def probe(self, hashes: torch.Tensor) -> torch.Tensor:
+        if hashes.shape[0] == 0:
+            return torch.zeros((0, self.W), dtype=torch.int32, device=hashes.device)
         if self.device.type == "cuda":
             return _require_ext().hashset_probe(hashes, self.tables)
         h = hashes.cpu().numpy()
