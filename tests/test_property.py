@@ -132,3 +132,60 @@ def test_decoder_never_crashes_on_garbage(data):
             cls.deserialize(data)
         except ValueError:
             pass
+
+
+@settings(max_examples=100, deadline=None)
+@given(
+    payloads=st.lists(st.binary(max_size=300), min_size=1, max_size=8),
+    masked=st.booleans(),
+)
+def test_ws_frame_roundtrip_property(payloads, masked):
+    """RFC6455 encode -> _WsFrameReader parse over a real socketpair."""
+    import socket as s_mod
+
+    from detectmateservice_amd.engine.sockets import _ws_encode, _WsFrameReader
+
+    a, b = s_mod.socketpair()
+    try:
+        blob = b"".join(_ws_encode(p, mask=masked) for p in payloads)
+        a.sendall(blob)
+        a.shutdown(s_mod.SHUT_WR)
+        reader = _WsFrameReader(b, server_side=masked)
+        got = []
+        while True:
+            frames = reader.next_frames()
+            if frames is None:
+                break
+            got.extend(frames)
+            if len(got) >= len(payloads):
+                break
+        assert got == payloads
+    finally:
+        a.close()
+        b.close()
+
+
+@settings(max_examples=100, deadline=None)
+@given(payloads=st.lists(st.binary(min_size=0, max_size=100000), min_size=1, max_size=3))
+def test_sp_frame_roundtrip_property(payloads):
+    """SP 64-bit framing roundtrip through _FrameReader."""
+    import socket as s_mod
+    import struct as struct_mod
+
+    from detectmateservice_amd.engine.sockets import _FrameReader
+
+    a, b = s_mod.socketpair()
+    try:
+        blob = b"".join(struct_mod.pack(">Q", len(p)) + p for p in payloads)
+        a.sendall(blob)
+        a.shutdown(s_mod.SHUT_WR)
+        reader = _FrameReader(b, sp=True)
+        got = []
+        while len(got) < len(payloads):
+            frames = reader.next_frames()
+            assert frames is not None
+            got.extend(frames)
+        assert got == payloads
+    finally:
+        a.close()
+        b.close()
