@@ -91,6 +91,16 @@ class Service:
                 component_path, config=cfg, logger=self.logger
             )
 
+        # source-mode sanity: fail at construction, not in the engine thread
+        if settings.engine_source_mode and (
+            self.library_component is None
+            or not hasattr(self.library_component, "stream_batches")
+        ):
+            raise ValueError(
+                f"engine_source_mode requires a component with stream_batches "
+                f"(a reader); {settings.component_type!r} has none"
+            )
+
         # --- engine (reference core.py:155) ---
         self.engine = Engine(
             settings,

@@ -210,3 +210,22 @@ def test_many_concurrent_clients_one_service(tmp_path):
     finally:
         svc.shutdown()
         t.join(timeout=5.0)
+
+
+def test_source_mode_requires_reader(tmp_path):
+    with pytest.raises(ValueError, match="stream_batches"):
+        Service(ServiceSettings(
+            component_type="core",
+            engine_addr=f"ipc://{tmp_path}/bad-src.ipc",
+            engine_source_mode=True,
+            http_enabled=False,
+            log_dir=tmp_path / "logs",
+        ))
+
+
+def test_from_log_without_processing(tmp_path):
+    f = tmp_path / "r.log"
+    f.write_text("one\ntwo\n")
+    frames = From.log(None, f, do_process=False)
+    assert len(frames) == 2
+    assert LogSchema.deserialize(frames[0]).log == "one"

@@ -92,3 +92,22 @@ def test_bounds():
         ServiceSettings(engine_retry_count=0)
     with pytest.raises(Exception):
         ServiceSettings(engine_buffer_size=9000)
+
+
+def test_config_class_override(tmp_path, monkeypatch):
+    """settings.config_class overrides the resolver's convention-based
+    config class (reference settings.py:52)."""
+    from detectmateservice_amd import Service
+
+    svc = Service(ServiceSettings(
+        component_type="NewValueDetector",
+        config_class="detectors.new_value.NewValueDetectorConfig",
+        engine_addr=f"ipc://{tmp_path}/cc.ipc",
+        http_enabled=False,
+        log_dir=tmp_path / "logs",
+    ))
+    try:
+        assert svc.config_manager.schema is not None
+        assert svc.config_manager.schema.__name__ == "NewValueDetectorConfig"
+    finally:
+        svc.engine.close()
