@@ -171,7 +171,9 @@ def main() -> None:
     total_lines = args.batch * args.steps * world_size
     lines_per_sec = total_lines / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
-    p50_step = statistics.median(step_times)
+    step_sorted = sorted(step_times)
+    p50_step = step_sorted[len(step_sorted) // 2]
+    p99_step = step_sorted[min(len(step_sorted) - 1, int(len(step_sorted) * 0.99))]
     p50_line_us = p50_step / args.batch * 1e6
 
     if rank == 0:
@@ -196,6 +198,7 @@ def main() -> None:
                 "parallelism": f"dp{world_size}",
                 "p50_detect_latency_us_per_line": round(p50_line_us, 3),
                 "p50_batch_ms": round(p50_step * 1000.0, 3),
+                "p99_batch_ms": round(p99_step * 1000.0, 3),
                 "transformer": not args.no_transformer,
                 "hip_graph": use_graph[0],
                 "device": str(device),
