@@ -21,3 +21,6 @@ __all__ += ["TransformerDetector", "TransformerDetectorConfig"]
 from .embedding import EmbeddingDetector, EmbeddingDetectorConfig
 
 __all__ += ["EmbeddingDetector", "EmbeddingDetectorConfig"]
+from .fused_pipeline import FusedPipelineDetector, FusedPipelineDetectorConfig
+
+__all__ += ["FusedPipelineDetector", "FusedPipelineDetectorConfig"]

@@ -1,0 +1,133 @@
+"""FusedPipelineDetector: the whole GPU pipeline as ONE service component.
+
+Deployment collapse of parser→detector: LogSchema frames in, DetectorSchema
+alerts out, with everything between — template matching, NewValue hash
+probing, BERT-tiny scoring — running as the fused GPU pipeline
+(detectmateservice_amd.pipeline.GpuPipeline) on device tensors. The C++
+batched codec decodes frames straight into the packed SoA buffer; Python
+objects are built only for the (rare) alert frames.
+
+This is the highest-throughput single-service deployment (config:
+``detectors: {FusedPipelineDetector: {...}}`` mirrors PipelineConfig).
+"""
+from __future__ import annotations
+
+import time
+from typing import Any, Dict, List, Optional
+
+import torch
+
+from ...components.base import CoreComponent, CoreConfig
+from ...schemas import DetectorSchema, LogSchema
+from ... import ops
+from ...pipeline import GpuPipeline, PipelineConfig
+
+
+class FusedPipelineDetectorConfig(CoreConfig):
+    method_type: str = "fused_pipeline_detector"
+    templates: List[str] = []
+    path_templates: Optional[str] = None
+    log_format: Optional[str] = None
+    lowercase: bool = False
+    max_len: int = 256
+    #: [{kind: variable|header, pos: int, event: int}]
+    watches: List[Dict[str, Any]] = []
+    hashset_capacity: int = 1 << 16
+    use_transformer: bool = True
+    score_threshold: float = 3.0
+    data_use_training: int = 0
+    device: Optional[str] = None
+    seed: int = 1234
+
+
+class FusedPipelineDetector(CoreComponent):
+    CONFIG_CLASS = FusedPipelineDetectorConfig
+
+    def __init__(self, config=None) -> None:
+        super().__init__(config)
+        cfg = self.config
+        templates = list(cfg.templates or [])
+        if cfg.path_templates:
+            with open(cfg.path_templates, "r", encoding="utf-8") as fh:
+                templates.extend(l.rstrip("\n") for l in fh if l.strip())
+        device = cfg.device or ("cuda" if torch.cuda.is_available() else "cpu")
+        self.device = torch.device(device)
+        self.pipe = GpuPipeline(
+            PipelineConfig(
+                templates=templates,
+                log_format=cfg.log_format,
+                lowercase=cfg.lowercase,
+                max_len=cfg.max_len,
+                watches=list(cfg.watches or []),
+                hashset_capacity=cfg.hashset_capacity,
+                use_transformer=cfg.use_transformer,
+                score_threshold=cfg.score_threshold,
+                train_lines=cfg.data_use_training,
+                seed=cfg.seed,
+            ),
+            device=self.device,
+        )
+        self.detector_id = f"fused_pipeline-{id(self):x}"
+
+    def process(self, data: bytes) -> Optional[bytes]:
+        return self.process_batch([data])[0]
+
+    def process_batch(self, frames: List[bytes]) -> List[Optional[bytes]]:
+        if not frames:
+            return []
+        if ops.have_extension():
+            from ...ops import _dmx_C  # type: ignore[attr-defined]
+
+            lines, lens, log_ids, _s, _h = _dmx_C.decode_log_batch(
+                list(frames), self.config.max_len
+            )
+            log_ids = [bytes(i).decode("utf-8", "replace") for i in log_ids]
+        else:
+            logs = [LogSchema.deserialize(f) for f in frames]
+            lines, lens = ops.pack_lines(
+                [(l.log or "").encode() for l in logs], self.config.max_len
+            )
+            log_ids = [l.logID for l in logs]
+        out = self.pipe.process_packed(
+            lines.to(self.device), lens.to(self.device)
+        )
+        anomaly = out["anomaly"]
+        results: List[Optional[bytes]] = [None] * len(frames)
+        if not bool(anomaly.any()):
+            return results
+        idxs = torch.nonzero(anomaly, as_tuple=False).flatten().cpu().tolist()
+        scores = out["scores"].float().cpu()
+        nv = out["nv_unseen"].cpu() if out["nv_unseen"] is not None else None
+        now = int(time.time())
+        for i in idxs:
+            reasons = []
+            if nv is not None and int(nv[i].sum()) > 0:
+                reasons.append("unknown watched value")
+            if self.pipe.model is not None and float(scores[i]) > self.config.score_threshold:
+                reasons.append(f"score {float(scores[i]):.3f}")
+            results[i] = DetectorSchema(
+                detectorID=self.detector_id,
+                detectorType="fused_pipeline_detector",
+                alertID=f"fp-{log_ids[i]}",
+                detectionTimestamp=now,
+                logIDs=[log_ids[i]] if log_ids[i] else [],
+                score=float(scores[i]),
+                description="Anomaly: " + "; ".join(reasons),
+            ).serialize()
+        return results
+
+    # -- checkpoint -----------------------------------------------------
+    def state_dict(self) -> Dict[str, Any]:
+        state: Dict[str, Any] = {"seen_lines": self.pipe.seen_lines}
+        if self.pipe.hashsets is not None:
+            state["hashsets"] = self.pipe.hashsets.state_dict()
+        if self.pipe.model is not None:
+            state["model"] = self.pipe.model.state_dict()
+        return state
+
+    def load_state_dict(self, state: Dict[str, Any]) -> None:
+        self.pipe.seen_lines = int(state.get("seen_lines", 0))
+        if "hashsets" in state and self.pipe.hashsets is not None:
+            self.pipe.hashsets.load_state_dict(state["hashsets"])
+        if "model" in state and self.pipe.model is not None:
+            self.pipe.model.load_state_dict(state["model"])

@@ -149,3 +149,62 @@ def test_embedding_detector_resolvable():
 
     path, cfg = ComponentResolver().resolve("EmbeddingDetector")
     assert path.endswith("EmbeddingDetector")
+
+
+def test_fused_pipeline_detector_component():
+    """LogSchema in -> DetectorSchema alerts out through the fused
+    pipeline as ONE component (CPU fallback path here; GPU in -m gpu)."""
+    from detectmateservice_amd.library.detectors import FusedPipelineDetector
+    from detectmateservice_amd.schemas import DetectorSchema, LogSchema
+    from detectmateservice_amd.utils.synthetic import (
+        AUDIT_LOG_FORMAT,
+        AUDIT_TEMPLATES,
+        AuditLogGenerator,
+    )
+
+    det = FusedPipelineDetector({
+        "templates": list(AUDIT_TEMPLATES),
+        "log_format": AUDIT_LOG_FORMAT,
+        "watches": [{"kind": "variable", "pos": 5, "event": 1}],
+        "data_use_training": 64,
+        "use_transformer": False,
+        "device": "cpu",
+    })
+    gen = AuditLogGenerator(seed=5)
+    train = [LogSchema(logID=f"t{i}", log=gen.line()[0]).serialize() for i in range(64)]
+    assert all(o is None for o in det.process_batch(train))
+
+    normal = [LogSchema(logID=f"n{i}", log=gen.line()[0]).serialize() for i in range(16)]
+    assert sum(o is not None for o in det.process_batch(normal)) == 0
+
+    bad = LogSchema(logID="bad", log=(
+        "type=USER_ACCT msg=audit(1642723741.072:999): pid=1 uid=0 auid=1 ses=1 "
+        "msg='op=PAM:accounting acct=\"intruder\" exe=/usr/sbin/cron hostname=? "
+        "addr=? terminal=cron res=success'"
+    )).serialize()
+    out = det.process_batch(normal[:3] + [bad])
+    assert [o is not None for o in out] == [False, False, False, True]
+    alert = DetectorSchema.deserialize(out[3])
+    assert alert.logIDs == ["bad"]
+    assert "unknown watched value" in alert.description
+
+    # checkpoint roundtrip
+    state = det.state_dict()
+    det2 = FusedPipelineDetector({
+        "templates": list(AUDIT_TEMPLATES),
+        "log_format": AUDIT_LOG_FORMAT,
+        "watches": [{"kind": "variable", "pos": 5, "event": 1}],
+        "data_use_training": 0,
+        "use_transformer": False,
+        "device": "cpu",
+    })
+    det2.load_state_dict(state)
+    out2 = det2.process_batch(normal[:3] + [bad])
+    assert [o is not None for o in out2] == [False, False, False, True]
+
+
+def test_fused_pipeline_detector_resolvable():
+    from detectmateservice_amd.components.resolver import ComponentResolver
+
+    path, _ = ComponentResolver().resolve("FusedPipelineDetector")
+    assert path.endswith("FusedPipelineDetector")
