@@ -436,3 +436,44 @@ def test_ops_fail_loudly_without_extension(monkeypatch):
     monkeypatch.setattr(ops, "_C", None)
     with pytest.raises(RuntimeError, match="not built"):
         ops.fused_linear(x, wt)
+
+
+@pytest.mark.gpu
+def test_fused_pipeline_detector_gpu():
+    """FusedPipelineDetector component end-to-end on the GPU: train on
+    clean audit traffic, then a line with an unseen watched header value
+    must alert (and only that line)."""
+    from detectmateservice_amd.library.detectors import FusedPipelineDetector
+    from detectmateservice_amd.schemas import DetectorSchema, LogSchema
+    from detectmateservice_amd.utils.synthetic import (
+        AUDIT_LOG_FORMAT,
+        AUDIT_TEMPLATES,
+        AuditLogGenerator,
+    )
+
+    det = FusedPipelineDetector({
+        "templates": list(AUDIT_TEMPLATES),
+        "log_format": AUDIT_LOG_FORMAT,
+        "watches": [{"kind": "header", "pos": 0}],
+        "data_use_training": 64,
+        "use_transformer": True,
+        "score_threshold": 1.0e9,  # NV-only alerts (random-init model)
+        "device": "cuda",
+    })
+    gen = AuditLogGenerator(seed=17)
+    train = [LogSchema(logID=f"t{i}", log=gen.line()[0]).serialize()
+             for i in range(64)]
+    assert all(o is None for o in det.process_batch(train))
+
+    normal = [LogSchema(logID=f"n{i}", log=gen.line()[0]).serialize()
+              for i in range(8)]
+    bad = LogSchema(logID="bad", log=(
+        "type=ZZZ_NEVER_SEEN msg=audit(1.0:1): pid=1 uid=0 auid=1 ses=1 "
+        "msg='op=PAM:x acct=\"x\" exe=/bin/x hostname=? addr=? "
+        "terminal=x res=success'"
+    )).serialize()
+    out = det.process_batch(normal + [bad])
+    assert [o is not None for o in out] == [False] * 8 + [True]
+    alert = DetectorSchema.deserialize(out[8])
+    assert alert.logIDs == ["bad"]
+    assert "unknown watched value" in alert.description

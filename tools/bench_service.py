@@ -9,6 +9,11 @@ including protobuf, sockets and the admin plane.
 
 Usage: python tools/bench_service.py [--lines 200000] [--batch 4096]
 Prints one JSON line (same shape as bench.py, metric service_lines_per_sec).
+
+``--fused`` replaces the two-stage parser+detector chain with ONE
+FusedPipelineDetector service (the whole GPU pipeline behind a socket):
+feeder -> fused svc -> sink. That is the highest-throughput single-box
+service deployment.
 """
 import argparse
 import json
@@ -37,6 +42,11 @@ def main():
     ap.add_argument("--lines", type=int, default=200_000)
     ap.add_argument("--batch", type=int, default=4096, help="engine batch size")
     ap.add_argument("--warmup-lines", type=int, default=20_000)
+    ap.add_argument("--fused", action="store_true",
+                    help="single FusedPipelineDetector service instead of "
+                         "parser+detector chain")
+    ap.add_argument("--no-transformer", action="store_true",
+                    help="(--fused) skip BERT-tiny scoring, hash-only")
     args = ap.parse_args()
 
     tmp = tempfile.mkdtemp(prefix="dmx-bench-")
@@ -84,16 +94,44 @@ def main():
         "log_dir": os.path.join(tmp, "logs"),
     })
 
+    if args.fused:
+        fused_settings = write_yaml("fs.yaml", {
+            "component_type": "FusedPipelineDetector",
+            "engine_addr": parser_in,        # feeder dials the same addr
+            "out_addr": [sink_addr],
+            "http_enabled": False,
+            "engine_batch_size": args.batch,
+            "engine_batch_linger_ms": 3.0,
+            "engine_buffer_size": 8192,
+            "config_file": write_yaml("fc.yaml", {"detectors": {
+                "FusedPipelineDetector": {
+                    "log_format": AUDIT_LOG_FORMAT,
+                    "path_templates": tpl,
+                    "watches": [{"kind": "header", "pos": 0}],
+                    "use_transformer": not args.no_transformer,
+                    # full transformer compute, but only NV alerts fire —
+                    # a random-init model + fixed threshold would emit
+                    # false positives and break the sentinel protocol
+                    "score_threshold": 1.0e9,
+                    "data_use_training": args.warmup_lines,
+                }}}),
+            "log_dir": os.path.join(tmp, "logs"),
+        })
+        settings_files = (fused_settings,)
+    else:
+        settings_files = (parser_settings, detector_settings)
     procs = [
         subprocess.Popen([sys.executable, "-m", "detectmateservice_amd.cli",
                           "--settings", s],
                          stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
-        for s in (parser_settings, detector_settings)
+        for s in settings_files
     ]
     sink = PairListener(sink_addr)
     feeder = PairDialer(parser_in, buffer_size=8192)
     try:
-        assert feeder.wait_connected(20.0), "parser service did not come up"
+        # generous: on a fresh GPU box the first torch/hip init inside the
+        # service process can take over a minute
+        assert feeder.wait_connected(180.0), "ingest service did not come up"
         gen = AuditLogGenerator(seed=7, anomaly_rate=0.0)
         # pre-serialize frames so the feeder isn't the bottleneck
         def make_frames(n, tag):
@@ -153,8 +191,12 @@ def main():
             "dtype": "bf16",
             "data": "synthetic",
             "config": {
-                "mode": "3-stage service processes over ipc (SP-framed engine sockets)",
-                "pipeline": "feeder->MatcherParser svc->NewValueDetector svc->sink",
+                "mode": ("fused single-service over ipc"
+                         if args.fused else
+                         "3-stage service processes over ipc (SP-framed engine sockets)"),
+                "pipeline": ("feeder->FusedPipelineDetector svc->sink"
+                             if args.fused else
+                             "feeder->MatcherParser svc->NewValueDetector svc->sink"),
                 "engine_batch_size": args.batch,
                 "lines": total,
                 "elapsed_s": round(elapsed, 3),
