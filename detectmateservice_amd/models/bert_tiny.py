@@ -15,7 +15,6 @@ Random-init weights (no network for checkpoints — BASELINE.md).
 from __future__ import annotations
 
 import math
-import os
 from dataclasses import dataclass
 from typing import Dict, Optional
 
@@ -184,12 +183,7 @@ class BertTinyDetectorModel:
             from ..ops import _dmx_C  # type: ignore[attr-defined]
 
             wb, fb = self._fused_blobs()
-            # DMX_FUSED_X2=1 selects the two-lines-per-workgroup variant
-            # (shared weight-fragment loads; see bert_fused.hip x2 notes)
-            fn = (_dmx_C.bert_fused_bf16_x2
-                  if os.environ.get("DMX_FUSED_X2") == "1"
-                  else _dmx_C.bert_fused_bf16)
-            return fn(
+            return _dmx_C.bert_fused_bf16(
                 lines, start.int().contiguous(), end.int().contiguous(),
                 wb, fb, self.config.layers, 1e-5,
             )

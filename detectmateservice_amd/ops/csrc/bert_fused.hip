@@ -414,298 +414,6 @@ static __device__ __forceinline__ void bert_fused_body(
 #undef BF_STAMP
 }
 
-// ==== x2 variant: TWO lines per workgroup ================================
-// The 1-line kernel's GEMM phases are weight-L2-LATENCY bound (phase
-// probe: pure MFMA issue <10% of GEMM time). Amortization lever: ONE
-// B-fragment load feeds EIGHT MFMA chains (4 m-fragments x 2 lines)
-// instead of four. Costs: LDS doubles (~141 KiB -> 1 block/CU, losing
-// the 2-block barrier overlap, measured ~4%) and TLP halves (2 waves/
-// SIMD), traded for 2x fewer weight loads per line and a 256-VGPR
-// budget (deeper load pipelining). Same 8-wave geometry per line.
-#define BF2_ARENA (BF_S * XS + BUF_ELEMS + 2 * BF_DH * VTS)  // shorts/line
-
-template <int K, int N, int MODE, int ACT, int WTS>
-static __device__ __attribute__((noinline)) void block_gemm2(
-    const lds_short* in0, const lds_short* in1, int in_stride,
-    const glob_cshort* __restrict__ Wt, const glob_cfloat* __restrict__ bias,
-    lds_short* out0, lds_short* out1, int out_stride, lds_short* x0,
-    lds_short* x1, lds_short* vt0, lds_short* vt1, int wid, int lane) {
-  constexpr int N16 = N / 16;
-  constexpr int TOTAL = 4 * N16;
-  constexpr int KS = K / 32;
-  constexpr int FPW = TOTAL / BF_WAVES;
-  static_assert(FPW >= 4 && FPW % 4 == 0, "quad-chunked assignment");
-#pragma unroll 2
-  for (int ff = wid * FPW; ff < wid * FPW + FPW; ff += 4) {
-    const int fn = ff >> 2;
-    f32x4 acc0[4], acc1[4];
-#pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      acc0[i] = {0.f, 0.f, 0.f, 0.f};
-      acc1[i] = {0.f, 0.f, 0.f, 0.f};
-    }
-#pragma unroll 4
-    for (int ks = 0; ks < KS; ++ks) {
-      bf16x8 b = *(const __attribute__((address_space(1))) bf16x8*)(
-          Wt + (long)(fn * 16 + (lane & 15)) * WTS + ks * 32 + (lane >> 4) * 8);
-      const int aoff = ks * 32 + (lane >> 4) * 8;
-#pragma unroll
-      for (int fm = 0; fm < 4; ++fm) {
-        bf16x8 a0 = *(const __attribute__((address_space(3))) bf16x8*)(
-            in0 + (fm * 16 + (lane & 15)) * in_stride + aoff);
-        acc0[fm] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b, acc0[fm], 0, 0, 0);
-        bf16x8 a1 = *(const __attribute__((address_space(3))) bf16x8*)(
-            in1 + (fm * 16 + (lane & 15)) * in_stride + aoff);
-        acc1[fm] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b, acc1[fm], 0, 0, 0);
-      }
-    }
-    const int n = fn * 16 + (lane & 15);
-    const float bval = bias ? bias[n] : 0.f;
-#pragma unroll
-    for (int fm = 0; fm < 4; ++fm) {
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int m = fm * 16 + (lane >> 4) * 4 + r;
-        float v0 = acc0[fm][r] + bval;
-        float v1 = acc1[fm][r] + bval;
-        if (MODE == 0) {
-          if (ACT == 1) { v0 = gelu_f32(v0); v1 = gelu_f32(v1); }
-          out0[m * out_stride + n] = f32_to_bf16(v0);
-          out1[m * out_stride + n] = f32_to_bf16(v1);
-        } else if (MODE == 1) {
-          if (n < 2 * BF_H) {
-            out0[m * out_stride + n] = f32_to_bf16(v0);
-            out1[m * out_stride + n] = f32_to_bf16(v1);
-          } else {
-            vt0[(n - 2 * BF_H) * VTS + m] = f32_to_bf16(v0);
-            vt1[(n - 2 * BF_H) * VTS + m] = f32_to_bf16(v1);
-          }
-        } else {  // MODE 2: residual accumulate
-          const float xv0 = bf16_to_f32(x0[m * XS + n]);
-          x0[m * XS + n] = f32_to_bf16(v0 + xv0);
-          const float xv1 = bf16_to_f32(x1[m * XS + n]);
-          x1[m * XS + n] = f32_to_bf16(v1 + xv1);
-        }
-      }
-    }
-  }
-}
-
-// Per-line attention for the x2 body (identical math to the 1-line
-// kernel's PH_ATTN block; arena pointers select the line).
-static __device__ __forceinline__ void attn_one_line(
-    lds_short* buf, lds_short* vt, int wid, int lane) {
-  const int hh = wid >> 2;
-  const int q0 = (wid & 3) * 16;
-  const float scale = 0.125f;
-  f32x4 acc_p[4];
-#pragma unroll
-  for (int f = 0; f < 4; ++f) acc_p[f] = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-  for (int ks = 0; ks < BF_DH / 32; ++ks) {
-    bf16x8 a = *(const __attribute__((address_space(3))) bf16x8*)(
-        buf + (q0 + (lane & 15)) * QKS + hh * BF_DH + ks * 32 + (lane >> 4) * 8);
-#pragma unroll
-    for (int f = 0; f < 4; ++f) {
-      bf16x8 b = *(const __attribute__((address_space(3))) bf16x8*)(
-          buf + (f * 16 + (lane & 15)) * QKS + BF_H + hh * BF_DH + ks * 32 +
-          (lane >> 4) * 8);
-      acc_p[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc_p[f], 0, 0, 0);
-    }
-  }
-  float inv_sum[4];
-#pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    float m = -1e30f;
-#pragma unroll
-    for (int f = 0; f < 4; ++f) m = fmaxf(m, acc_p[f][r] * scale);
-#pragma unroll
-    for (int mask = 1; mask < 16; mask <<= 1) m = fmaxf(m, __shfl_xor(m, mask, 64));
-    float sum = 0.f;
-#pragma unroll
-    for (int f = 0; f < 4; ++f) {
-      const float e = __expf(acc_p[f][r] * scale - m);
-      acc_p[f][r] = e;
-      sum += e;
-    }
-#pragma unroll
-    for (int mask = 1; mask < 16; mask <<= 1) sum += __shfl_xor(sum, mask, 64);
-    inv_sum[r] = 1.f / sum;
-  }
-  __syncthreads();  // all waves done reading this line's Q/K
-  lds_short* my_p = buf + wid * 16 * VTS;
-#pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int row = (lane >> 4) * 4 + r;
-#pragma unroll
-    for (int f = 0; f < 4; ++f)
-      my_p[row * VTS + f * 16 + (lane & 15)] = f32_to_bf16(acc_p[f][r]);
-  }
-  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-  f32x4 acc_o[4];
-#pragma unroll
-  for (int f = 0; f < 4; ++f) acc_o[f] = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-  for (int ks = 0; ks < BF_S / 32; ++ks) {
-    bf16x8 a = *(const __attribute__((address_space(3))) bf16x8*)(
-        my_p + (lane & 15) * VTS + ks * 32 + (lane >> 4) * 8);
-#pragma unroll
-    for (int f = 0; f < 4; ++f) {
-      bf16x8 b = *(const __attribute__((address_space(3))) bf16x8*)(
-          vt + (hh * BF_DH + f * 16 + (lane & 15)) * VTS + ks * 32 +
-          (lane >> 4) * 8);
-      acc_o[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc_o[f], 0, 0, 0);
-    }
-  }
-#pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int q = q0 + (lane >> 4) * 4 + r;
-    const float inv = __shfl(inv_sum[r], (lane >> 4) * 4 + r, 64);
-#pragma unroll
-    for (int f = 0; f < 4; ++f) {
-      const int d = f * 16 + (lane & 15);
-      buf[O_OFF + q * XS + hh * BF_DH + d] = f32_to_bf16(acc_o[f][r] * inv);
-    }
-  }
-}
-
-extern "C" __global__ __launch_bounds__(BF_THREADS, 1)
-void dmx_bert_fused_bf16_x2(const unsigned char* __restrict__ lines,
-                            const int* __restrict__ start,
-                            const int* __restrict__ end,
-                            const short* __restrict__ wb,
-                            const float* __restrict__ fb,
-                            float* __restrict__ scores, int B, int max_len,
-                            int n_layers, float eps) {
-  const int line0 = blockIdx.x * 2;
-  const int line1 = line0 + 1;
-  if (line0 >= B) return;
-  const bool has1 = line1 < B;
-  const int tid = threadIdx.x;
-  const int wid = tid / DMX_WAVE;
-  const int lane = tid % DMX_WAVE;
-
-  extern __shared__ __attribute__((aligned(16))) short smem_raw[];
-  lds_short* smem = (lds_short*)smem_raw;
-  lds_short* x0 = smem;
-  lds_short* buf0 = x0 + BF_S * XS;
-  lds_short* vt0 = buf0 + BUF_ELEMS;
-  lds_short* x1 = smem + BF2_ARENA;
-  lds_short* buf1 = x1 + BF_S * XS;
-  lds_short* vt1 = buf1 + BUF_ELEMS;
-  lds_float* red = (lds_float*)(smem + 2 * BF2_ARENA);  // [2][128]
-
-  // ---- embed both lines ----
-  {
-    const int s00 = start[line0], e00 = end[line0];
-    const int s01 = has1 ? start[line1] : 0;
-    const int e01 = has1 ? end[line1] : 0;
-    for (int i = tid * 8; i < 2 * BF_S * BF_H; i += BF_THREADS * 8) {
-      const int sub = i >= BF_S * BF_H;
-      const int ii = sub ? i - BF_S * BF_H : i;
-      const int s = ii / BF_H, c = ii % BF_H;
-      const int ln = sub ? line1 : line0;
-      const int s0 = sub ? s01 : s00, e0 = sub ? e01 : e00;
-      int tok = 0;
-      const int idx = s0 + s;
-      if (idx < e0 && idx < max_len && (!sub || has1))
-        tok = (int)lines[(long)ln * max_len + idx] + 3;
-      short8v te = *(const short8v*)(wb + WB_TOK + (long)tok * BF_H + c);
-      short8v pe = *(const short8v*)(wb + WB_POS + (long)s * BF_H + c);
-      short8v xv;
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        xv[j] = f32_to_bf16(bf16_to_f32(te[j]) + bf16_to_f32(pe[j]));
-      lds_short* xd = sub ? x1 : x0;
-      *(__attribute__((address_space(3))) short8v*)(xd + s * XS + c) = xv;
-    }
-  }
-  __syncthreads();
-
-  for (int layer = 0; layer < n_layers; ++layer) {
-    const short* lw = wb + WB_LAYER0 + (long)layer * LW_SIZE;
-    const float* lf = fb + (long)layer * FB_SIZE;
-
-    block_gemm2<BF_H, 3 * BF_H, 1, 0, BF_H>(
-        x0, x1, XS, (glob_cshort*)(lw + LW_QKV), (glob_cfloat*)(lf + FB_BQKV),
-        buf0, buf1, QKS, x0, x1, vt0, vt1, wid, lane);
-    __syncthreads();
-
-    // attention per line; trailing barrier keeps every wave in the same
-    // sub-iteration (the in-attention barrier aliases P over Q/K, so
-    // waves MUST NOT drift across iterations at a barrier)
-    attn_one_line(buf0, vt0, wid, lane);
-    __syncthreads();
-    attn_one_line(buf1, vt1, wid, lane);
-    __syncthreads();
-
-    block_gemm2<BF_H, BF_H, 2, 0, BF_H>(
-        buf0 + O_OFF, buf1 + O_OFF, XS, (glob_cshort*)(lw + LW_WO),
-        (glob_cfloat*)(lf + FB_BO), nullptr, nullptr, 0, x0, x1, nullptr,
-        nullptr, wid, lane);
-    __syncthreads();
-    block_layernorm(x0, lw + LW_LN1G, lw + LW_LN1B, wid, lane, eps);
-    block_layernorm(x1, lw + LW_LN1G, lw + LW_LN1B, wid, lane, eps);
-    __syncthreads();
-
-#pragma unroll
-    for (int h = 0; h < 2; ++h) {
-      block_gemm2<BF_H, BF_FFN / 2, 0, 1, BF_H>(
-          x0, x1, XS, (glob_cshort*)(lw + LW_W1 + (long)h * (BF_FFN / 2) * BF_H),
-          (glob_cfloat*)(lf + FB_B1 + h * (BF_FFN / 2)), buf0, buf1, QKS,
-          nullptr, nullptr, nullptr, nullptr, wid, lane);
-      __syncthreads();
-      block_gemm2<BF_FFN / 2, BF_H, 2, 0, BF_FFN>(
-          buf0, buf1, QKS, (glob_cshort*)(lw + LW_W2 + (long)h * (BF_FFN / 2)),
-          h == 0 ? (glob_cfloat*)(lf + FB_B2) : nullptr, nullptr, nullptr, 0,
-          x0, x1, nullptr, nullptr, wid, lane);
-      __syncthreads();
-    }
-    block_layernorm(x0, lw + LW_LN2G, lw + LW_LN2B, wid, lane, eps);
-    block_layernorm(x1, lw + LW_LN2G, lw + LW_LN2B, wid, lane, eps);
-    __syncthreads();
-  }
-
-  // ---- pool + score, both lines ----
-  {
-    if (tid < 2 * BF_H) {
-      const int sub = tid >> 7;  // 0: line0, 1: line1
-      const int col = tid & (BF_H - 1);
-      const lds_short* xs = sub ? x1 : x0;
-      float s = 0.f;
-      for (int row = 0; row < BF_S; ++row) s += bf16_to_f32(xs[row * XS + col]);
-      const float w = bf16_to_f32(wb[WB_LAYER0 + (long)n_layers * LW_SIZE + col]);
-      red[sub * BF_H + col] = (s / BF_S) * w;
-    }
-    __syncthreads();
-    if (wid < 2) {
-      const int sub = wid;
-      float v = red[sub * BF_H + lane] + red[sub * BF_H + lane + 64];
-      v = warp_reduce_sum_f32(v);
-      if (lane == 0 && (sub == 0 || has1))
-        scores[sub ? line1 : line0] = v + fb[(long)n_layers * FB_SIZE];
-    }
-  }
-}
-
-extern "C" void dmx_launch_bert_fused_bf16_x2(
-    const void* lines, const void* start, const void* end, const void* wb,
-    const void* fb, void* scores, int B, int max_len, int n_layers, float eps,
-    hipStream_t stream) {
-  const size_t lds = (size_t)2 * BF2_ARENA * sizeof(short) + 256 * sizeof(float);
-  static bool attr_set = false;
-  if (!attr_set) {
-    hipFuncSetAttribute((const void*)dmx_bert_fused_bf16_x2,
-                        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
-    attr_set = true;
-  }
-  hipLaunchKernelGGL(dmx_bert_fused_bf16_x2, dim3((B + 1) / 2),
-                     dim3(BF_THREADS), lds, stream,
-                     (const unsigned char*)lines, (const int*)start,
-                     (const int*)end, (const short*)wb, (const float*)fb,
-                     (float*)scores, B, max_len, n_layers, eps);
-}
-
 extern "C" __global__ __launch_bounds__(BF_THREADS, 4)
 void dmx_bert_fused_bf16(const unsigned char* __restrict__ lines,
                          const int* __restrict__ start,

@@ -477,25 +477,3 @@ def test_fused_pipeline_detector_gpu():
     alert = DetectorSchema.deserialize(out[8])
     assert alert.logIDs == ["bad"]
     assert "unknown watched value" in alert.description
-
-
-@pytest.mark.gpu
-def test_bert_fused_x2_matches_x1():
-    """Two-lines-per-workgroup fused kernel == one-line kernel bit-for-bit
-    math (same MFMA fragments, same order), incl. odd batch sizes."""
-    from detectmateservice_amd.models.bert_tiny import BertTinyDetectorModel
-    from detectmateservice_amd.ops import _dmx_C
-    from detectmateservice_amd import ops
-
-    torch.manual_seed(0)
-    model = BertTinyDetectorModel(device="cuda")
-    wb, fb = model._fused_blobs()
-    for B in (1, 2, 7, 64, 513):
-        lines = torch.randint(32, 127, (B, 256), dtype=torch.uint8, device="cuda")
-        start = torch.randint(0, 8, (B,), dtype=torch.int32, device="cuda")
-        end = start + torch.randint(1, 200, (B,), dtype=torch.int32, device="cuda")
-        s1 = _dmx_C.bert_fused_bf16(lines, start, end, wb, fb, 2, 1e-5)
-        s2 = _dmx_C.bert_fused_bf16_x2(lines, start, end, wb, fb, 2, 1e-5)
-        torch.cuda.synchronize()
-        assert torch.allclose(s1, s2, atol=1e-4, rtol=1e-4), (
-            B, (s1 - s2).abs().max().item())
