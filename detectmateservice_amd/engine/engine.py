@@ -24,6 +24,7 @@ per-output drop accounting) are preserved exactly.
 from __future__ import annotations
 
 import logging
+import os
 import threading
 import time
 from typing import List, Optional, Protocol, runtime_checkable
@@ -160,11 +161,19 @@ class Engine:
             s.engine_addr, s.engine_batch_size, s.engine_batch_linger_ms,
             len(self._out_socks),
         )
+        # DMX_ENGINE_STATS=1: per-batch stage timing every ~2 s (perf triage)
+        stats_on = os.environ.get("DMX_ENGINE_STATS") == "1"
+        st = {"n": 0, "frames": 0, "recv": 0.0, "proc": 0.0, "other": 0.0,
+              "last": time.perf_counter()}
+        t_end = time.perf_counter()
         while not self._stop_event.is_set():
             try:
+                t_r0 = time.perf_counter()
                 frames = self._pair_sock.recv_many(
                     s.engine_batch_size, s.engine_recv_timeout, s.engine_batch_linger_ms
                 )
+                if stats_on:
+                    st["recv"] += time.perf_counter() - t_r0
             except RecvTimeout:
                 continue
             except SocketClosed:
@@ -216,6 +225,25 @@ class Engine:
                     else:
                         m.data_dropped_bytes_total.inc(len(out))
                         m.data_dropped_lines_total.inc(_count_lines(out))
+            if stats_on:
+                now = time.perf_counter()
+                st["n"] += 1
+                st["frames"] += len(frames)
+                st["proc"] += elapsed
+                st["other"] += (now - t_end) if t_end else 0.0
+                t_end = now
+                if now - st["last"] > 2.0 and st["n"]:
+                    tot = st["other"]
+                    self._log.info(
+                        "[stats] %d batches (%.0f fr/batch): recv-wait %.1fms/b, "
+                        "process %.1fms/b, other %.1fms/b, %.0f lines/s",
+                        st["n"], st["frames"] / st["n"],
+                        st["recv"] * 1e3 / st["n"], st["proc"] * 1e3 / st["n"],
+                        (tot - st["recv"] - st["proc"]) * 1e3 / st["n"],
+                        st["frames"] / tot if tot > 0 else 0.0,
+                    )
+                    st.update(n=0, frames=0, recv=0.0, proc=0.0, other=0.0,
+                              last=now)
         self._log.info("engine loop exited")
 
     def _run_source_loop(self) -> None:

@@ -120,10 +120,14 @@ def main():
         settings_files = (fused_settings,)
     else:
         settings_files = (parser_settings, detector_settings)
+    env = dict(os.environ)
+    if os.environ.get("DMX_ENGINE_STATS") == "1":
+        env["DMX_ENGINE_STATS"] = "1"
     procs = [
         subprocess.Popen([sys.executable, "-m", "detectmateservice_amd.cli",
                           "--settings", s],
-                         stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+                         stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL,
+                         env=env)
         for s in settings_files
     ]
     sink = PairListener(sink_addr)
@@ -162,6 +166,7 @@ def main():
         # frames AND the pipeline is idle (sink quiet + sockets drained).
         t0 = time.perf_counter()
         pump(frames)
+        pump_elapsed = time.perf_counter() - t0
         # wait until pipeline is idle: detector emits nothing for clean
         # traffic; send one marked anomalous line and wait for its alert
         # (it can only arrive after everything queued before it).
@@ -200,6 +205,7 @@ def main():
                 "engine_batch_size": args.batch,
                 "lines": total,
                 "elapsed_s": round(elapsed, 3),
+                "pump_s": round(pump_elapsed, 3),
             },
         }))
     finally:
