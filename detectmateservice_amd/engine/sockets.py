@@ -297,6 +297,7 @@ class _WsFrameReader:
         self.sock = sock
         self.server_side = server_side
         self.buf = bytearray()
+        self._frag: Optional[bytearray] = None  # in-progress fragmented msg
 
     def _fill(self, need: int) -> bool:
         while len(self.buf) < need:
@@ -351,7 +352,20 @@ class _WsFrameReader:
                 except OSError:
                     return out or None
                 continue
-            if opcode in (0x2, 0x1, 0x0):  # binary/text (no fragmentation)
+            if opcode in (0x2, 0x1, 0x0):  # binary/text incl. fragmentation
+                fin = bool(b0 & 0x80)
+                if opcode == 0x0:  # continuation
+                    if self._frag is None:
+                        raise ValueError("ws continuation without start frame")
+                    self._frag += payload
+                    if not fin:
+                        continue
+                    payload, self._frag = bytes(self._frag), None
+                elif not fin:  # fragmented message start
+                    if self._frag is not None:
+                        raise ValueError("ws nested fragmented message")
+                    self._frag = bytearray(payload)
+                    continue
                 out.append(payload)
                 if not self.buf:
                     return out

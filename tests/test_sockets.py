@@ -296,3 +296,73 @@ def test_late_binding_preserves_order(ipc_addr):
             listener.close()
     finally:
         dialer.close()
+
+
+def test_ws_fragmented_message_reassembly():
+    """RFC6455 §5.4: a message split into FIN=0 start + continuation
+    frames (with an interleaved ping) reassembles into one payload."""
+    import os as os_mod
+    import socket as s_mod
+
+    from detectmateservice_amd.engine.sockets import _WsFrameReader, _ws_encode
+
+    def frag_frames(data, parts, mask):
+        chunks = []
+        step = max(1, len(data) // parts)
+        pieces = [data[i:i + step] for i in range(0, len(data), step)]
+        for i, piece in enumerate(pieces):
+            first = i == 0
+            last = i == len(pieces) - 1
+            op = 0x2 if first else 0x0
+            b0 = (0x80 if last else 0) | op
+            hdr = bytearray([b0])
+            n = len(piece)
+            mask_bit = 0x80 if mask else 0
+            assert n < 126
+            hdr.append(mask_bit | n)
+            if mask:
+                mkey = os_mod.urandom(4)
+                hdr += mkey
+                piece = bytes(b ^ mkey[j % 4] for j, b in enumerate(piece))
+            chunks.append(bytes(hdr) + piece)
+        return chunks
+
+    a, b = s_mod.socketpair()
+    try:
+        msg = b"fragmented-payload-" + bytes(range(64))
+        frames = frag_frames(msg, 4, mask=True)
+        # interleave a ping between fragments (allowed by RFC6455)
+        ping = bytes([0x89, 0x80]) + b"\x00\x00\x00\x00"
+        blob = frames[0] + ping + b"".join(frames[1:])
+        blob += _ws_encode(b"whole", mask=True)
+        a.sendall(blob)
+        a.shutdown(s_mod.SHUT_WR)
+        reader = _WsFrameReader(b, server_side=True)
+        got = []
+        while len(got) < 2:
+            fr = reader.next_frames()
+            if fr is None:
+                break
+            got.extend(fr)
+        assert got == [msg, b"whole"]
+    finally:
+        a.close()
+        b.close()
+
+
+def test_ws_continuation_without_start_rejected():
+    import socket as s_mod
+
+    from detectmateservice_amd.engine.sockets import _WsFrameReader
+
+    a, b = s_mod.socketpair()
+    try:
+        # lone continuation frame (opcode 0, FIN=1, masked, empty)
+        a.sendall(bytes([0x80, 0x80]) + b"\x00\x00\x00\x00")
+        a.shutdown(s_mod.SHUT_WR)
+        reader = _WsFrameReader(b, server_side=True)
+        with pytest.raises(ValueError):
+            reader.next_frames()
+    finally:
+        a.close()
+        b.close()
