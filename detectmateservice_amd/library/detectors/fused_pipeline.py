@@ -75,23 +75,45 @@ class FusedPipelineDetector(CoreComponent):
     def process_batch(self, frames: List[bytes]) -> List[Optional[bytes]]:
         if not frames:
             return []
+        import os as _os
+        _stats = _os.environ.get("DMX_ENGINE_STATS") == "1"
+        _t0 = time.perf_counter() if _stats else 0.0
         if ops.have_extension():
             from ...ops import _dmx_C  # type: ignore[attr-defined]
 
-            lines, lens, log_ids, _s, _h = _dmx_C.decode_log_batch(
+            lines, lens, log_ids_raw, _s, _h = _dmx_C.decode_log_batch(
                 list(frames), self.config.max_len
             )
-            log_ids = [bytes(i).decode("utf-8", "replace") for i in log_ids]
+            log_ids = None  # decoded lazily, only for alert frames
         else:
             logs = [LogSchema.deserialize(f) for f in frames]
             lines, lens = ops.pack_lines(
                 [(l.log or "").encode() for l in logs], self.config.max_len
             )
+            log_ids_raw = None
             log_ids = [l.logID for l in logs]
+        _t1 = time.perf_counter() if _stats else 0.0
         out = self.pipe.process_packed(
-            lines.to(self.device), lens.to(self.device)
+            lines.to(self.device, non_blocking=True),
+            lens.to(self.device, non_blocking=True),
         )
         anomaly = out["anomaly"]
+        if _stats:
+            import torch as _torch
+            if anomaly.is_cuda:
+                _torch.cuda.synchronize()
+            _t2 = time.perf_counter()
+            self._stat_acc = getattr(self, "_stat_acc", [0, 0.0, 0.0, 0.0])
+            self._stat_acc[0] += 1
+            self._stat_acc[1] += _t1 - _t0
+            self._stat_acc[2] += _t2 - _t1
+            if self._stat_acc[0] % 16 == 0:
+                import logging as _logging
+                _logging.getLogger(__name__).info(
+                    "[fp-stats] decode %.2fms/b gpu %.2fms/b",
+                    self._stat_acc[1] * 1e3 / self._stat_acc[0],
+                    self._stat_acc[2] * 1e3 / self._stat_acc[0],
+                )
         results: List[Optional[bytes]] = [None] * len(frames)
         if not bool(anomaly.any()):
             return results
@@ -99,6 +121,10 @@ class FusedPipelineDetector(CoreComponent):
         scores = out["scores"].float().cpu()
         nv = out["nv_unseen"].cpu() if out["nv_unseen"] is not None else None
         now = int(time.time())
+        if log_ids is None:  # decode only the alert frames' ids
+            log_ids = {
+                i: bytes(log_ids_raw[i]).decode("utf-8", "replace") for i in idxs
+            }
         for i in idxs:
             reasons = []
             if nv is not None and int(nv[i].sum()) > 0:

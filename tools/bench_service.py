@@ -121,15 +121,18 @@ def main():
     else:
         settings_files = (parser_settings, detector_settings)
     env = dict(os.environ)
-    if os.environ.get("DMX_ENGINE_STATS") == "1":
-        env["DMX_ENGINE_STATS"] = "1"
+    stats_on = os.environ.get("DMX_ENGINE_STATS") == "1"
+    svc_out = (open(os.path.join(tmp, "svc_stdout.log"), "w")
+               if stats_on else subprocess.DEVNULL)
     procs = [
         subprocess.Popen([sys.executable, "-m", "detectmateservice_amd.cli",
                           "--settings", s],
-                         stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL,
-                         env=env)
+                         stdout=svc_out, stderr=subprocess.STDOUT
+                         if stats_on else subprocess.DEVNULL, env=env)
         for s in settings_files
     ]
+    if stats_on:
+        print(f"# service stdout: {tmp}/svc_stdout.log", file=sys.stderr)
     sink = PairListener(sink_addr)
     feeder = PairDialer(parser_in, buffer_size=8192)
     try:
