@@ -41,6 +41,11 @@ class MatcherParserConfig(CoreConfig):
     #: mining itself runs on the GPU edit-distance kernel — template_miner.py)
     auto_config: bool = False
     auto_config_lines: int = 1000
+    #: drift adaptation: when > 0, buffer UNMATCHED (EventID -1) content
+    #: lines and, once this many accumulate, mine them and APPEND any new
+    #: templates (existing EventIDs are stable — new templates get higher
+    #: ids). MI355X-first addition: the reference mines once at startup.
+    auto_config_refit_lines: int = 0
     log_format: Optional[str] = None
     time_format: Optional[str] = None
     params: Dict = {}
@@ -140,6 +145,7 @@ class MatcherParser(CoreComponent):
         self._auto_pending: Optional[List[str]] = (
             [] if (cfg.auto_config and not templates) else None
         )
+        self._drift_pending: List[str] = []
         # Which extracted header field carries the free-text content that
         # template matching applies to. Defaults to the LAST token of the
         # log_format (e.g. ``<Content>`` in the audit format); when the
@@ -174,6 +180,38 @@ class MatcherParser(CoreComponent):
             self._auto_pending.append(self._normalize(line))
         if len(self._auto_pending) >= self.config.auto_config_lines:
             self.mine_templates()
+
+    def _drift_observe(self, contents: List[str]) -> None:
+        """Buffer unmatched content lines; refit when the threshold hits."""
+        lim = self.config.auto_config_refit_lines
+        if lim <= 0 or self._auto_pending is not None:
+            return
+        room = 4 * lim - len(self._drift_pending)
+        if room > 0:
+            self._drift_pending.extend(contents[:room])
+        if len(self._drift_pending) >= lim:
+            self.refit_templates()
+
+    def refit_templates(self) -> List[str]:
+        """Mine the drift buffer and append templates not already known.
+        Existing templates keep their EventIDs (append-only)."""
+        from .template_miner import TemplateMiner
+        import torch
+
+        pending, self._drift_pending = self._drift_pending, []
+        if not pending:
+            return []
+        device = (self.config.params or {}).get("device")
+        if device is None:
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+        mined = TemplateMiner(device=device).fit(pending)
+        known = set(self.templates)
+        new = [m for m in mined if m not in known]
+        if new:
+            self.templates = self.templates + new
+            self._segments = [split_template(t) for t in self.templates]
+            self._batch_matcher = None  # rebuild packed tables
+        return new
 
     def mine_templates(self) -> List[str]:
         """Run the miner on buffered lines and install the templates."""
@@ -254,9 +292,17 @@ class MatcherParser(CoreComponent):
     def _process_python(self, frames: List[bytes]) -> List[Optional[bytes]]:
         now = int(time.time())
         out: List[Optional[bytes]] = []
+        drift: List[str] = []
         for raw in frames:
             log = LogSchema.deserialize(raw)
             header, event_id, template, variables = self.parse_line(log.log)
+            if event_id == -1 and self.config.auto_config_refit_lines > 0:
+                content = log.log
+                if self._format_re is not None:
+                    m = self._format_re.match(content)
+                    if m and self._content_field and self._content_field in m.groupdict():
+                        content = m.group(self._content_field)
+                drift.append(self._normalize(content))
             ts = self._extracted_timestamp(header)
             parsed = ParserSchema(
                 parserType="matcher_parser",
@@ -271,6 +317,8 @@ class MatcherParser(CoreComponent):
                 parsedTimestamp=now,
             )
             out.append(parsed.serialize())
+        if drift:
+            self._drift_observe(drift)
         return out
 
     # -- batched SoA path ----------------------------------------------
@@ -309,6 +357,20 @@ class MatcherParser(CoreComponent):
         dev_lens = lens.to(self._batch_device)
         match = matcher.match_packed(dev_lines, dev_lens)
         match_cpu = {k: v.cpu() for k, v in match.items()}
+        if self.config.auto_config_refit_lines > 0:
+            ev = match_cpu["event_id"]
+            idxs = (ev < 0).nonzero().flatten().tolist()[:256]
+            if idxs:
+                fc, nfc = match_cpu["fmt_caps"], match_cpu["n_fmt_caps"]
+                drift = []
+                for i in idxs:
+                    row = bytes(lines[i, : int(lens[i])].numpy().tobytes())
+                    nf = int(nfc[i])
+                    if nf > 0:  # content = last header capture
+                        s0, e0 = int(fc[i, nf - 1, 0]), int(fc[i, nf - 1, 1])
+                        row = row[s0:e0]
+                    drift.append(self._normalize(row.decode("utf-8", "replace")))
+                self._drift_observe(drift)
         encoded = _dmx_C.encode_parser_batch(
             lines,
             lens,

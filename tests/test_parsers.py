@@ -139,3 +139,58 @@ def test_time_format_extraction():
     expect = int(datetime.datetime(2000, 10, 10, 13, 55, 36,
                                    tzinfo=datetime.timezone.utc).timestamp())
     assert parsed.receivedTimestamp == expect
+
+
+def test_drift_refit_appends_templates():
+    """auto_config_refit_lines: unmatched lines accumulate and are mined
+    into NEW templates appended after the existing ones (stable EventIDs)."""
+    from detectmateservice_amd.library.parsers import MatcherParser
+    from detectmateservice_amd.schemas import LogSchema, ParserSchema
+
+    parser = MatcherParser({
+        "templates": ["session opened for user <*> by uid <*>"],
+        "auto_config_refit_lines": 24,
+    })
+    # known pattern parses as event 1
+    known = LogSchema(logID="k", log="session opened for user root by uid 0")
+    out = ParserSchema.deserialize(parser.process(known.serialize()))
+    assert out.EventID == 1
+
+    # a NEW recurring pattern the template set doesn't know
+    drift = [
+        LogSchema(logID=f"d{i}", log=f"connection from 10.0.0.{i} port {4000+i} accepted").serialize()
+        for i in range(30)
+    ]
+    outs = [ParserSchema.deserialize(o) for o in parser._process_python(drift)]
+    assert all(o.EventID == -1 for o in outs[:20])  # unmatched before refit
+    assert len(parser.templates) > 1  # refit appended a mined template
+
+    # the drifted pattern now matches with a NEW id; event 1 is unchanged
+    again = LogSchema(logID="d", log="connection from 10.0.0.99 port 4099 accepted")
+    o2 = ParserSchema.deserialize(parser.process(again.serialize()))
+    assert o2.EventID > 1
+    o1 = ParserSchema.deserialize(parser.process(known.serialize()))
+    assert o1.EventID == 1
+
+
+def test_drift_refit_batched_path():
+    """Same behavior through the batched C++/kernel path."""
+    from detectmateservice_amd import ops
+    from detectmateservice_amd.library.parsers import MatcherParser
+    from detectmateservice_amd.schemas import LogSchema, ParserSchema
+
+    if not ops.have_extension():
+        pytest.skip("extension not built")
+    parser = MatcherParser({
+        "templates": ["session opened for user <*> by uid <*>"],
+        "auto_config_refit_lines": 24,
+    })
+    drift = [
+        LogSchema(logID=f"d{i}", log=f"connection from 10.0.0.{i} port {4000+i} accepted").serialize()
+        for i in range(30)
+    ]
+    parser.process_batch(drift)
+    assert len(parser.templates) > 1
+    again = [LogSchema(logID="x", log="connection from 10.0.0.99 port 4099 accepted").serialize()] * 8
+    outs = [ParserSchema.deserialize(o) for o in parser.process_batch(again)]
+    assert all(o.EventID > 1 for o in outs)
