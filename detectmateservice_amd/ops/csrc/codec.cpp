@@ -74,24 +74,25 @@ bool skip_field(const uint8_t* p, size_t n, size_t& pos, int wt) {
 
 // ---- decode LogSchema batch ----------------------------------------------
 
-// returns (lines u8 [B, max_len], lens i32 [B], logIDs, logSources, hostnames)
-py::tuple decode_log_batch(const std::vector<py::bytes>& frames,
-                           int64_t max_len) {
-  const int64_t B = (int64_t)frames.size();
-  auto lines = torch::zeros({B, max_len}, torch::kUInt8);
-  auto lens = torch::zeros({B}, torch::kInt32);
-  uint8_t* lbuf = lines.data_ptr<uint8_t>();
-  int32_t* lenp = lens.data_ptr<int32_t>();
-  py::list log_ids, sources, hostnames;
+namespace {
+struct LogSpan {  // byte spans into the source frame (no copies)
+  const char* id = "";   size_t id_len = 0;
+  const char* src = "";  size_t src_len = 0;
+  const char* host = ""; size_t host_len = 0;
+};
 
+// varint walk + line copy for one batch; NO Python API — callable with
+// the GIL released (the service engine thread must not stall the C++
+// socket reader threads, which need the GIL for frame py::bytes).
+void decode_log_core(const std::vector<std::pair<const uint8_t*, size_t>>& raw,
+                     int64_t max_len, uint8_t* lbuf, int32_t* lenp,
+                     std::vector<LogSpan>& spans) {
+  const int64_t B = (int64_t)raw.size();
   for (int64_t i = 0; i < B; ++i) {
-    char* fptr;
-    Py_ssize_t flen;
-    PyBytes_AsStringAndSize(frames[i].ptr(), &fptr, &flen);
-    const uint8_t* p = (const uint8_t*)fptr;
-    const size_t n = (size_t)flen;
+    const uint8_t* p = raw[i].first;
+    const size_t n = raw[i].second;
     size_t pos = 0;
-    std::string log_id, source, hostname;
+    LogSpan& sp = spans[i];
     while (pos < n) {
       uint64_t key;
       if (!get_varint(p, n, pos, key)) break;
@@ -101,15 +102,15 @@ py::tuple decode_log_batch(const std::vector<py::bytes>& frames,
         if (!get_varint(p, n, pos, sl) || pos + sl > n) break;
         const char* s = (const char*)(p + pos);
         switch (field) {
-          case 2: log_id.assign(s, sl); break;
+          case 2: sp.id = s; sp.id_len = sl; break;
           case 3: {  // log line -> packed buffer
             const size_t copy = std::min<size_t>(sl, (size_t)max_len);
             std::memcpy(lbuf + i * max_len, s, copy);
             lenp[i] = (int32_t)copy;
             break;
           }
-          case 4: source.assign(s, sl); break;
-          case 5: hostname.assign(s, sl); break;
+          case 4: sp.src = s; sp.src_len = sl; break;
+          case 5: sp.host = s; sp.host_len = sl; break;
           default: break;
         }
         pos += sl;
@@ -117,11 +118,75 @@ py::tuple decode_log_batch(const std::vector<py::bytes>& frames,
         if (!skip_field(p, n, pos, wt)) break;
       }
     }
-    log_ids.append(py::bytes(log_id));
-    sources.append(py::bytes(source));
-    hostnames.append(py::bytes(hostname));
+  }
+}
+
+std::vector<std::pair<const uint8_t*, size_t>> frame_ptrs(
+    const std::vector<py::bytes>& frames) {
+  std::vector<std::pair<const uint8_t*, size_t>> raw(frames.size());
+  for (size_t i = 0; i < frames.size(); ++i) {
+    char* fptr;
+    Py_ssize_t flen;
+    PyBytes_AsStringAndSize(frames[i].ptr(), &fptr, &flen);
+    raw[i] = {(const uint8_t*)fptr, (size_t)flen};
+  }
+  return raw;
+}
+}  // namespace
+
+// returns (lines u8 [B, max_len], lens i32 [B], logIDs, logSources, hostnames)
+py::tuple decode_log_batch(const std::vector<py::bytes>& frames,
+                           int64_t max_len) {
+  const int64_t B = (int64_t)frames.size();
+  auto lines = torch::zeros({B, max_len}, torch::kUInt8);
+  auto lens = torch::zeros({B}, torch::kInt32);
+  auto raw = frame_ptrs(frames);
+  std::vector<LogSpan> spans(B);
+  {
+    py::gil_scoped_release release;  // frames held by the caller's list
+    decode_log_core(raw, max_len, lines.data_ptr<uint8_t>(),
+                    lens.data_ptr<int32_t>(), spans);
+  }
+  py::list log_ids, sources, hostnames;
+  for (int64_t i = 0; i < B; ++i) {
+    log_ids.append(py::bytes(spans[i].id, spans[i].id_len));
+    sources.append(py::bytes(spans[i].src, spans[i].src_len));
+    hostnames.append(py::bytes(spans[i].host, spans[i].host_len));
   }
   return py::make_tuple(lines, lens, log_ids, sources, hostnames);
+}
+
+// Hot-path variant: NO per-frame Python objects. Returns
+// (lines u8 [B,max_len] (pinned if pin), lens i32 [B],
+//  ids_blob bytes, ids_off i32 [B+1]) — logID i is
+// ids_blob[ids_off[i]:ids_off[i+1]]. Sources/hostnames are dropped
+// (the fused service does not use them; decode_log_batch does).
+py::tuple decode_log_batch_packed(const std::vector<py::bytes>& frames,
+                                  int64_t max_len, bool pin) {
+  const int64_t B = (int64_t)frames.size();
+  auto opts = torch::TensorOptions().dtype(torch::kUInt8);
+  if (pin) opts = opts.pinned_memory(true);
+  auto lines = torch::zeros({B, max_len}, opts);
+  auto lens = torch::zeros({B}, torch::TensorOptions().dtype(torch::kInt32));
+  auto ids_off = torch::zeros({B + 1}, torch::TensorOptions().dtype(torch::kInt32));
+  auto raw = frame_ptrs(frames);
+  std::vector<LogSpan> spans(B);
+  std::string blob;
+  {
+    py::gil_scoped_release release;
+    decode_log_core(raw, max_len, lines.data_ptr<uint8_t>(),
+                    lens.data_ptr<int32_t>(), spans);
+    int32_t* off = ids_off.data_ptr<int32_t>();
+    size_t total = 0;
+    for (int64_t i = 0; i < B; ++i) total += spans[i].id_len;
+    blob.reserve(total);
+    for (int64_t i = 0; i < B; ++i) {
+      off[i] = (int32_t)blob.size();
+      blob.append(spans[i].id, spans[i].id_len);
+    }
+    off[B] = (int32_t)blob.size();
+  }
+  return py::make_tuple(lines, lens, py::bytes(blob), ids_off);
 }
 
 // ---- encode ParserSchema batch -------------------------------------------
@@ -426,6 +491,10 @@ py::tuple parser_watch_hashes(
 }  // namespace
 
 void register_codec(py::module_& m) {
+  m.def("decode_log_batch_packed", &decode_log_batch_packed,
+        py::arg("frames"), py::arg("max_len"), py::arg("pin") = false,
+        "hot-path LogSchema batch decode: packed ids blob, optional pinned "
+        "lines buffer, GIL released during the parse");
   m.def("decode_log_batch", &decode_log_batch,
         "decode N LogSchema frames -> (lines u8 [B,max_len], lens, logIDs, "
         "sources, hostnames)");

@@ -208,3 +208,18 @@ def test_fused_pipeline_detector_resolvable():
 
     path, _ = ComponentResolver().resolve("FusedPipelineDetector")
     assert path.endswith("FusedPipelineDetector")
+
+
+@needs_ext
+def test_decode_log_batch_packed_matches_list_variant():
+    from detectmateservice_amd.ops import _dmx_C
+
+    frames = [
+        LogSchema(logID=f"id{i}", log=f"the line {i}", logSource="s").serialize()
+        for i in range(17)
+    ] + [LogSchema(logID="", log="no id").serialize()]
+    l1, n1, ids, _s, _h = _dmx_C.decode_log_batch(frames, 64)
+    l2, n2, blob, off = _dmx_C.decode_log_batch_packed(frames, 64, False)
+    assert torch.equal(l1, l2) and torch.equal(n1, n2)
+    for i in range(len(frames)):
+        assert blob[int(off[i]):int(off[i + 1])] == bytes(ids[i])
