@@ -38,6 +38,9 @@ class FusedPipelineDetectorConfig(CoreConfig):
     data_use_training: int = 0
     device: Optional[str] = None
     seed: int = 1234
+    #: capture the detect path as ONE hipGraph at this batch size (0 = off).
+    #: Smaller batches replay padded; set to the engine batch size.
+    graph_batch: int = 0
 
 
 class FusedPipelineDetector(CoreComponent):
@@ -98,10 +101,18 @@ class FusedPipelineDetector(CoreComponent):
             log_ids_raw = None
             log_ids = [l.logID for l in logs]
         _t1 = time.perf_counter() if _stats else 0.0
-        out = self.pipe.process_packed(
-            lines.to(self.device, non_blocking=True),
-            lens.to(self.device, non_blocking=True),
-        )
+        gb = self.config.graph_batch
+        if (gb > 0 and self.device.type == "cuda"
+                and self.pipe.seen_lines >= self.config.data_use_training
+                and lines.shape[0] <= gb):
+            if self.pipe._graph is None:
+                self.pipe.enable_graph(gb)
+            out = self.pipe.process_packed_graph_partial(lines, lens)
+        else:
+            out = self.pipe.process_packed(
+                lines.to(self.device, non_blocking=True),
+                lens.to(self.device, non_blocking=True),
+            )
         anomaly = out["anomaly"]
         if _stats:
             import torch as _torch

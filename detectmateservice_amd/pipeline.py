@@ -212,3 +212,30 @@ class GpuPipeline:
         self._graph.replay()
         self.seen_lines += lines.shape[0]
         return self._graph_out
+
+    def process_packed_graph_partial(
+        self, lines: torch.Tensor, lens: torch.Tensor
+    ) -> Dict[str, torch.Tensor]:
+        """Replay on a batch SMALLER than the captured size: pad rows get
+        len 0 (their kernels no-op on the span but the transformer still
+        runs them — acceptable when partial batches are rare stream
+        edges). Outputs are sliced back to the true batch size."""
+        assert self._graph is not None, "call enable_graph() first"
+        sl, sn = self._graph_in
+        B = lines.shape[0]
+        assert B <= sl.shape[0]
+        sl[:B].copy_(lines, non_blocking=True)
+        sn[:B].copy_(lens, non_blocking=True)
+        if B < sn.shape[0]:
+            sn[B:].zero_()
+        self._graph.replay()
+        self.seen_lines += B
+
+        def cut(v):
+            if torch.is_tensor(v):
+                return v[:B]
+            if isinstance(v, dict):
+                return {k: cut(x) for k, x in v.items()}
+            return v
+
+        return {k: cut(v) for k, v in self._graph_out.items()}

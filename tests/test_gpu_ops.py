@@ -477,3 +477,52 @@ def test_fused_pipeline_detector_gpu():
     alert = DetectorSchema.deserialize(out[8])
     assert alert.logIDs == ["bad"]
     assert "unknown watched value" in alert.description
+
+
+@pytest.mark.gpu
+def test_fused_pipeline_detector_graph_matches_eager():
+    """graph_batch replay (incl. padded partial batches) produces the same
+    alerts as the eager path."""
+    from detectmateservice_amd.library.detectors import FusedPipelineDetector
+    from detectmateservice_amd.schemas import LogSchema
+    from detectmateservice_amd.utils.synthetic import (
+        AUDIT_LOG_FORMAT,
+        AUDIT_TEMPLATES,
+        AuditLogGenerator,
+    )
+
+    def build(graph_batch):
+        return FusedPipelineDetector({
+            "templates": list(AUDIT_TEMPLATES),
+            "log_format": AUDIT_LOG_FORMAT,
+            "watches": [{"kind": "header", "pos": 0}],
+            "data_use_training": 64,
+            "use_transformer": True,
+            "score_threshold": 1.0e9,
+            "graph_batch": graph_batch,
+            "device": "cuda",
+            "seed": 99,
+        })
+
+    gen = AuditLogGenerator(seed=23)
+    train = [LogSchema(logID=f"t{i}", log=gen.line()[0]).serialize()
+             for i in range(64)]
+    lines = [gen.line()[0] for _ in range(40)]
+    bad = ("type=ZZZ_NEW msg=audit(1.0:1): pid=1 uid=0 auid=1 ses=1 "
+           "msg='op=PAM:x acct=\"x\" exe=/bin/x hostname=? addr=? "
+           "terminal=x res=success'")
+    batch = [LogSchema(logID=f"n{i}", log=l).serialize()
+             for i, l in enumerate(lines)] + [
+        LogSchema(logID="bad", log=bad).serialize()]
+
+    eager, graph = build(0), build(64)
+    eager.process_batch(train)
+    graph.process_batch(train)
+    out_e = eager.process_batch(batch)       # 41 frames, eager
+    out_g = graph.process_batch(batch)       # 41 <= graph_batch 64: padded
+    assert [o is not None for o in out_e] == [o is not None for o in out_g]
+    assert out_g[-1] is not None
+    # full-size batch through the graph
+    out_g2 = graph.process_batch(batch[:40] + batch[:23] + [batch[-1]])
+    assert out_g2[-1] is not None
+    assert sum(o is not None for o in out_g2) == 1

@@ -47,6 +47,9 @@ def main():
                          "parser+detector chain")
     ap.add_argument("--no-transformer", action="store_true",
                     help="(--fused) skip BERT-tiny scoring, hash-only")
+    ap.add_argument("--graph", action="store_true",
+                    help="(--fused) replay the detect path as one hipGraph "
+                         "per batch (graph_batch = engine batch size)")
     args = ap.parse_args()
 
     tmp = tempfile.mkdtemp(prefix="dmx-bench-")
@@ -113,6 +116,7 @@ def main():
                     # a random-init model + fixed threshold would emit
                     # false positives and break the sentinel protocol
                     "score_threshold": 1.0e9,
+                    "graph_batch": args.batch if args.graph else 0,
                     "data_use_training": args.warmup_lines,
                 }}}),
             "log_dir": os.path.join(tmp, "logs"),
