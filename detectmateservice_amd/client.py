@@ -48,12 +48,13 @@ class DetectMateClient:
             f"{self.url}/admin/restore", json={"path": path}, timeout=TIMEOUT_S
         ).json()
 
-    def reconfigure(self, config_file: str, persist: bool = False) -> dict:
+    def reconfigure(self, config_file: str, persist: bool = False,
+                    reload: bool = False) -> dict:
         with open(config_file, "r", encoding="utf-8") as fh:
             config = yaml.safe_load(fh) or {}
         return requests.post(
             f"{self.url}/admin/reconfigure",
-            json={"config": config, "persist": persist},
+            json={"config": config, "persist": persist, "reload": reload},
             timeout=TIMEOUT_S,
         ).json()
 
@@ -69,6 +70,8 @@ def main(argv=None) -> int:
     rec = sub.add_parser("reconfigure")
     rec.add_argument("config_file")
     rec.add_argument("--persist", action="store_true")
+    rec.add_argument("--reload", action="store_true",
+                     help="rebuild the live component from the new config")
     for cmd in ("checkpoint", "restore"):
         cp = sub.add_parser(cmd)
         cp.add_argument("path")
@@ -79,7 +82,8 @@ def main(argv=None) -> int:
         if args.command == "metrics":
             print(client.metrics())
         elif args.command == "reconfigure":
-            print(json.dumps(client.reconfigure(args.config_file, args.persist), indent=2))
+            print(json.dumps(client.reconfigure(
+                args.config_file, args.persist, args.reload), indent=2))
         elif args.command in ("checkpoint", "restore"):
             print(json.dumps(getattr(client, args.command)(args.path), indent=2))
         else:

@@ -83,6 +83,7 @@ class Service:
             settings.config_file, schema=schema, logger=self.logger
         )
 
+        self._component_path = component_path
         if component_path is not None:
             cfg = component_config
             if cfg is None and self._component_class_name:
@@ -178,13 +179,33 @@ class Service:
     def status(self) -> Dict[str, Any]:
         return self._create_status_report(self._running)
 
-    def reconfigure(self, config: Dict[str, Any], persist: bool = False) -> Dict[str, Any]:
+    def reconfigure(
+        self, config: Dict[str, Any], persist: bool = False, reload: bool = False
+    ) -> Dict[str, Any]:
         """Validate + swap config; optionally persist defaults-stripped YAML
-        (reference core.py:299-338)."""
+        (reference core.py:299-338).
+
+        ``reload`` (no reference equivalent — the reference's live
+        component keeps its constructor-time config, core.py:299-345
+        comment): rebuild the library component from the NEW config and
+        swap it in atomically. In-memory detector state (learned values,
+        calibration) restarts fresh; checkpoint/restore first to keep it.
+        """
         validated = self.config_manager.update(config)
         if persist:
             self.config_manager.save(self.config_manager.get())
-        self.logger.info("reconfigured (persist=%s)", persist)
+        if reload and self._component_path is not None:
+            cfg = None
+            if self._component_class_name:
+                cfg = self.config_manager.component_section(self._component_class_name)
+            new_component = ComponentLoader(logger=self.logger).load_component(
+                self._component_path, config=cfg, logger=self.logger
+            )
+            old, self.library_component = self.library_component, new_component
+            if old is not None:
+                old.teardown()
+            self.logger.info("component reloaded from new config")
+        self.logger.info("reconfigured (persist=%s reload=%s)", persist, reload)
         return self.config_manager.get()
 
     def shutdown(self) -> None:

@@ -22,6 +22,7 @@ def get_service():  # overridden by WebServer with the live Service
 class ReconfigPayload(BaseModel):
     config: Dict[str, Any]
     persist: bool = False
+    reload: bool = False  # rebuild the live component from the new config
 
 
 @router.post("/start")
@@ -43,7 +44,7 @@ def status(service=Depends(get_service)) -> Dict[str, Any]:
 
 @router.post("/reconfigure")
 def reconfigure(payload: ReconfigPayload, service=Depends(get_service)) -> Dict[str, Any]:
-    configs = service.reconfigure(payload.config, payload.persist)
+    configs = service.reconfigure(payload.config, payload.persist, payload.reload)
     return {"status": "reconfigured", "configs": configs}
 
 

@@ -92,3 +92,35 @@ def test_checkpoint_over_http(tmp_path, ipc_addr, free_port):
     finally:
         svc.shutdown()
         t.join(timeout=5.0)
+
+
+def test_reconfigure_reload_rebuilds_component(tmp_path):
+    """reconfigure(reload=True) swaps in a component built from the NEW
+    config (the default reference-parity behavior keeps the old one)."""
+    from detectmateservice_amd import Service, ServiceSettings
+
+    cfg_file = tmp_path / "cfg.yaml"
+    cfg_file.write_text(
+        "detectors:\n  RandomDetector:\n    params:\n      threshold: 0.5\n"
+    )
+    svc = Service(ServiceSettings(
+        component_type="RandomDetector",
+        engine_addr=f"ipc://{tmp_path}/rl.ipc",
+        http_enabled=False,
+        config_file=cfg_file,
+        log_dir=tmp_path / "logs",
+    ))
+    try:
+        first = svc.library_component
+        # plain reconfigure: live component unchanged (reference parity)
+        svc.reconfigure({"detectors": {"RandomDetector": {"params": {"threshold": 0.9}}}})
+        assert svc.library_component is first
+        # reload: a NEW component instance built from the new config
+        svc.reconfigure(
+            {"detectors": {"RandomDetector": {"params": {"threshold": 0.9}}}},
+            reload=True,
+        )
+        assert svc.library_component is not first
+        assert svc.library_component.config.params["threshold"] == 0.9
+    finally:
+        svc.engine.close()
