@@ -129,6 +129,23 @@ class Service:
         # passthrough mode for core* component types (reference core.py:204-206)
         return list(frames)
 
+    # -- packed data plane (engine_packed_mode) ------------------------
+    def supports_packed_frames(self) -> bool:
+        return hasattr(self.library_component, "process_packed_frames")
+
+    def packed_max_len(self) -> int:
+        fn = getattr(self.library_component, "packed_max_len", None)
+        return fn() if fn is not None else 256
+
+    def packed_pin_memory(self) -> bool:
+        fn = getattr(self.library_component, "packed_pin_memory", None)
+        return fn() if fn is not None else False
+
+    def process_packed_frames(self, lines, lens, ids_blob, ids_off):
+        return self.library_component.process_packed_frames(
+            lines, lens, ids_blob, ids_off
+        )
+
     def source_batches(self, batch_size: int, stop_event):
         """Source-mode delegation (engine_source_mode: reader services)."""
         if self.library_component is None or not hasattr(

@@ -154,6 +154,8 @@ class Engine:
     def _run_loop(self) -> None:
         if self.settings.engine_source_mode:
             return self._run_source_loop()
+        if self.settings.engine_packed_mode and self._try_packed_loop():
+            return
         s = self.settings
         m = self.metrics
         self._log.info(
@@ -245,6 +247,97 @@ class Engine:
                     st.update(n=0, frames=0, recv=0.0, proc=0.0, other=0.0,
                               last=now)
         self._log.info("engine loop exited")
+
+    def _try_packed_loop(self) -> bool:
+        """Native packed data plane: reader threads decode LogSchema frames
+        straight into tensors (frame_reader.cpp read_batch_packed); the
+        loop hands (lines, lens, ids) to the component with ZERO per-frame
+        Python objects. Requires a plain fd listener, the C++ extension
+        and a component exposing ``process_packed_frames``. Returns False
+        (caller falls back to the frame loop) when unavailable."""
+        s = self.settings
+        m = self.metrics
+        proc = getattr(self.processor, "process_packed_frames", None)
+        supports = getattr(self.processor, "supports_packed_frames", None)
+        if proc is None or (supports is not None and not supports()):
+            self._log.warning(
+                "engine_packed_mode: component has no process_packed_frames; "
+                "using the frame loop"
+            )
+            return False
+        enable = getattr(self._pair_sock, "enable_packed", None)
+        max_len = getattr(self.processor, "packed_max_len", lambda: 256)()
+        pin = getattr(self.processor, "packed_pin_memory", lambda: False)()
+        if enable is None or not enable(max_len, pin):
+            self._log.warning(
+                "engine_packed_mode: listener cannot enable the packed path "
+                "(TLS/ws/inproc or extension missing); using the frame loop"
+            )
+            return False
+        self._log.info("engine PACKED loop started on %s (max_len=%d pin=%s)",
+                       s.engine_addr, max_len, pin)
+        stats_on = os.environ.get("DMX_ENGINE_STATS") == "1"
+        st = {"n": 0, "frames": 0, "recv": 0.0, "proc": 0.0,
+              "last": time.perf_counter()}
+        while not self._stop_event.is_set():
+            try:
+                t_r0 = time.perf_counter()
+                conn, lines, lens, blob, off, nbytes = (
+                    self._pair_sock.recv_packed(s.engine_recv_timeout)
+                )
+                if stats_on:
+                    st["recv"] += time.perf_counter() - t_r0
+            except RecvTimeout:
+                continue
+            except SocketClosed:
+                break
+            except OSError as exc:
+                if self._stop_event.is_set():
+                    break
+                self._log.error("recv error: %s", exc)
+                continue
+            B = int(lines.shape[0])
+            if B == 0:
+                continue
+            m.data_read_bytes_total.inc(nbytes)
+            m.data_read_lines_total.inc(B)
+            m.engine_batch_size.observe(B)
+            t0 = time.perf_counter()
+            try:
+                alerts = proc(lines, lens, blob, off)
+            except Exception as exc:  # noqa: BLE001 - loop must survive
+                m.processing_errors_total.inc(B)
+                self._log.error("processing error on packed batch of %d: %s",
+                                B, exc)
+                continue
+            elapsed = time.perf_counter() - t0
+            m.data_processed_bytes_total.inc(nbytes)
+            m.data_processed_lines_total.inc(B)
+            m.observe_batch(elapsed, B)
+            for _idx, out in alerts:
+                if self._out_socks:
+                    self._send_to_outputs(out)
+                elif self._pair_sock.send(out, block=False):
+                    m.data_written_bytes_total.inc(len(out))
+                    m.data_written_lines_total.inc(1)
+                else:
+                    m.data_dropped_bytes_total.inc(len(out))
+                    m.data_dropped_lines_total.inc(1)
+            if stats_on:
+                st["n"] += 1
+                st["frames"] += B
+                st["proc"] += elapsed
+                now = time.perf_counter()
+                if now - st["last"] > 2.0 and st["n"]:
+                    self._log.info(
+                        "[packed-stats] %d batches (%.0f fr/b): recv-wait "
+                        "%.1fms/b process %.1fms/b",
+                        st["n"], st["frames"] / st["n"],
+                        st["recv"] * 1e3 / st["n"], st["proc"] * 1e3 / st["n"],
+                    )
+                    st.update(n=0, frames=0, recv=0.0, proc=0.0, last=now)
+        self._log.info("engine packed loop exited")
+        return True
 
     def _run_source_loop(self) -> None:
         """Source mode: the component generates frames (reader services).

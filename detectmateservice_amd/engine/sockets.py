@@ -432,6 +432,9 @@ class PairListener:
         self._last_sender: Optional[socket.socket] = None
         self._closed = threading.Event()
         self._ssl_ctx: Optional[ssl.SSLContext] = None
+        #: packed fast path (enable_packed): reader threads decode
+        #: LogSchema frames straight into tensors — no Python objects
+        self._packed: Optional[Tuple[int, bool]] = None
 
         scheme, rest = self.addr.scheme, self.addr.rest
         #: SP (NNG pair0) wire mapping on tcp/tls+tcp edges
@@ -522,6 +525,12 @@ class PairListener:
         try:
             if native is not None:
                 while not self._closed.is_set():
+                    packed = self._packed
+                    if packed is not None:
+                        chunk = native.read_batch_packed(4096, 200, *packed)
+                        if chunk[0].shape[0] > 0:
+                            self._recv_q.put((conn, chunk))
+                        continue
                     frames = native.read_batch(4096, 200)
                     if frames:
                         self._recv_q.put((conn, frames))
@@ -565,6 +574,45 @@ class PairListener:
         self._last_sender = conn
         self._pending = list(frames[1:])
         return frames[0]
+
+    # -- packed fast path ----------------------------------------------
+    def enable_packed(self, max_len: int, pin: bool) -> bool:
+        """Switch reader threads to the native socket→tensor decode path
+        (plain fd sockets only; TLS/ws readers are unaffected and keep
+        delivering byte frames)."""
+        if self._ws or self._ssl_ctx is not None:
+            return False
+        try:
+            from ..ops import _dmx_C  # noqa: F401
+        except Exception:  # noqa: BLE001
+            return False
+        self._packed = (max_len, pin)
+        return True
+
+    def recv_packed(self, timeout_ms: int):
+        """Pop ONE packed chunk: (conn, lines, lens, ids_blob, ids_off,
+        frame_bytes). Byte-frame items queued before enable_packed (or
+        from TLS/ws peers) are converted via the batch codec."""
+        try:
+            conn, item = self._recv_q.get(timeout=timeout_ms / 1000.0)
+        except queue.Empty:
+            raise RecvTimeout(self.addr) from None
+        self._last_sender = conn
+        import torch
+
+        if len(item) == 5 and torch.is_tensor(item[0]):
+            lines, lens, blob, off, nbytes = item
+            return conn, lines, lens, blob, off, int(nbytes)
+        # byte-frame item: convert (rare transition / mixed-peer case)
+        from ..ops import _dmx_C
+
+        max_len, pin = self._packed or (256, False)
+        frames = [f for f in item if f]
+        nbytes = sum(len(f) for f in frames)
+        lines, lens, blob, off = _dmx_C.decode_log_batch_packed(
+            list(frames), max_len, pin
+        )
+        return conn, lines, lens, blob, off, nbytes
 
     def recv_many(
         self, max_frames: int, timeout_ms: int, linger_ms: float = 0.0

@@ -75,6 +75,70 @@ class FusedPipelineDetector(CoreComponent):
     def process(self, data: bytes) -> Optional[bytes]:
         return self.process_batch([data])[0]
 
+    # -- packed data plane (engine_packed_mode) ------------------------
+    def packed_max_len(self) -> int:
+        return int(self.config.max_len)
+
+    def packed_pin_memory(self) -> bool:
+        return self.device.type == "cuda"
+
+    def _score_packed(self, lines: torch.Tensor, lens: torch.Tensor):
+        """Run the pipeline (graph replay when enabled) on a CPU-packed
+        batch; returns the output dict."""
+        gb = self.config.graph_batch
+        if (gb > 0 and self.device.type == "cuda"
+                and self.pipe.seen_lines >= self.config.data_use_training
+                and lines.shape[0] <= gb):
+            if self.pipe._graph is None:
+                self.pipe.enable_graph(gb)
+            return self.pipe.process_packed_graph_partial(lines, lens)
+        return self.pipe.process_packed(
+            lines.to(self.device, non_blocking=True),
+            lens.to(self.device, non_blocking=True),
+        )
+
+    def _alerts(self, out, id_of) -> List:
+        """[(idx, DetectorSchema bytes)] for anomalous rows; ``id_of(i)``
+        resolves a row's logID lazily (alerts are rare)."""
+        anomaly = out["anomaly"]
+        if not bool(anomaly.any()):
+            return []
+        idxs = torch.nonzero(anomaly, as_tuple=False).flatten().cpu().tolist()
+        scores = out["scores"].float().cpu()
+        nv = out["nv_unseen"].cpu() if out["nv_unseen"] is not None else None
+        now = int(time.time())
+        alerts = []
+        for i in idxs:
+            reasons = []
+            if nv is not None and int(nv[i].sum()) > 0:
+                reasons.append("unknown watched value")
+            if (self.pipe.model is not None
+                    and float(scores[i]) > self.config.score_threshold):
+                reasons.append(f"score {float(scores[i]):.3f}")
+            lid = id_of(i)
+            alerts.append((i, DetectorSchema(
+                detectorID=self.detector_id,
+                detectorType="fused_pipeline_detector",
+                alertID=f"fp-{lid}",
+                detectionTimestamp=now,
+                logIDs=[lid] if lid else [],
+                score=float(scores[i]),
+                description="Anomaly: " + "; ".join(reasons),
+            ).serialize()))
+        return alerts
+
+    def process_packed_frames(self, lines, lens, ids_blob, ids_off) -> List:
+        """Engine packed-loop entry: tensors in (already decoded by the
+        C++ socket reader), [(idx, alert bytes)] out."""
+        if lines.shape[0] == 0:
+            return []
+        out = self._score_packed(lines, lens)
+        return self._alerts(
+            out,
+            lambda i: ids_blob[int(ids_off[i]):int(ids_off[i + 1])].decode(
+                "utf-8", "replace"),
+        )
+
     def process_batch(self, frames: List[bytes]) -> List[Optional[bytes]]:
         if not frames:
             return []
@@ -101,25 +165,12 @@ class FusedPipelineDetector(CoreComponent):
             log_ids_raw = None
             log_ids = [l.logID for l in logs]
         _t1 = time.perf_counter() if _stats else 0.0
-        gb = self.config.graph_batch
-        if (gb > 0 and self.device.type == "cuda"
-                and self.pipe.seen_lines >= self.config.data_use_training
-                and lines.shape[0] <= gb):
-            if self.pipe._graph is None:
-                self.pipe.enable_graph(gb)
-            out = self.pipe.process_packed_graph_partial(lines, lens)
-        else:
-            out = self.pipe.process_packed(
-                lines.to(self.device, non_blocking=True),
-                lens.to(self.device, non_blocking=True),
-            )
-        anomaly = out["anomaly"]
+        out = self._score_packed(lines, lens)
         if _stats:
-            import torch as _torch
-            if anomaly.is_cuda:
-                _torch.cuda.synchronize()
+            if out["anomaly"].is_cuda:
+                torch.cuda.synchronize()
             _t2 = time.perf_counter()
-            self._stat_acc = getattr(self, "_stat_acc", [0, 0.0, 0.0, 0.0])
+            self._stat_acc = getattr(self, "_stat_acc", [0, 0.0, 0.0])
             self._stat_acc[0] += 1
             self._stat_acc[1] += _t1 - _t0
             self._stat_acc[2] += _t2 - _t1
@@ -130,34 +181,15 @@ class FusedPipelineDetector(CoreComponent):
                     self._stat_acc[1] * 1e3 / self._stat_acc[0],
                     self._stat_acc[2] * 1e3 / self._stat_acc[0],
                 )
-        results: List[Optional[bytes]] = [None] * len(frames)
-        if not bool(anomaly.any()):
-            return results
-        idxs = torch.nonzero(anomaly, as_tuple=False).flatten().cpu().tolist()
-        scores = out["scores"].float().cpu()
-        nv = out["nv_unseen"].cpu() if out["nv_unseen"] is not None else None
-        now = int(time.time())
-        if log_ids is None:  # decode only the alert frames' ids
+        if log_ids is None:
             blob, off = log_ids_raw
-            log_ids = {
-                i: blob[int(off[i]):int(off[i + 1])].decode("utf-8", "replace")
-                for i in idxs
-            }
-        for i in idxs:
-            reasons = []
-            if nv is not None and int(nv[i].sum()) > 0:
-                reasons.append("unknown watched value")
-            if self.pipe.model is not None and float(scores[i]) > self.config.score_threshold:
-                reasons.append(f"score {float(scores[i]):.3f}")
-            results[i] = DetectorSchema(
-                detectorID=self.detector_id,
-                detectorType="fused_pipeline_detector",
-                alertID=f"fp-{log_ids[i]}",
-                detectionTimestamp=now,
-                logIDs=[log_ids[i]] if log_ids[i] else [],
-                score=float(scores[i]),
-                description="Anomaly: " + "; ".join(reasons),
-            ).serialize()
+            id_of = lambda i: blob[int(off[i]):int(off[i + 1])].decode(  # noqa: E731
+                "utf-8", "replace")
+        else:
+            id_of = lambda i: log_ids[i]  # noqa: E731
+        results: List[Optional[bytes]] = [None] * len(frames)
+        for i, alert in self._alerts(out, id_of):
+            results[i] = alert
         return results
 
     # -- checkpoint -----------------------------------------------------

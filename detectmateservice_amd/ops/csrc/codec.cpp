@@ -12,6 +12,8 @@
 // verified against the official protobuf runtime in tests/test_schemas.py).
 #include <torch/extension.h>
 
+#include "proto_log.h"
+
 #include <cstdint>
 #include <cstring>
 #include <string>
@@ -75,51 +77,8 @@ bool skip_field(const uint8_t* p, size_t n, size_t& pos, int wt) {
 // ---- decode LogSchema batch ----------------------------------------------
 
 namespace {
-struct LogSpan {  // byte spans into the source frame (no copies)
-  const char* id = "";   size_t id_len = 0;
-  const char* src = "";  size_t src_len = 0;
-  const char* host = ""; size_t host_len = 0;
-};
-
-// varint walk + line copy for one batch; NO Python API — callable with
-// the GIL released (the service engine thread must not stall the C++
-// socket reader threads, which need the GIL for frame py::bytes).
-void decode_log_core(const std::vector<std::pair<const uint8_t*, size_t>>& raw,
-                     int64_t max_len, uint8_t* lbuf, int32_t* lenp,
-                     std::vector<LogSpan>& spans) {
-  const int64_t B = (int64_t)raw.size();
-  for (int64_t i = 0; i < B; ++i) {
-    const uint8_t* p = raw[i].first;
-    const size_t n = raw[i].second;
-    size_t pos = 0;
-    LogSpan& sp = spans[i];
-    while (pos < n) {
-      uint64_t key;
-      if (!get_varint(p, n, pos, key)) break;
-      const int field = (int)(key >> 3), wt = (int)(key & 7);
-      if (wt == 2) {
-        uint64_t sl;
-        if (!get_varint(p, n, pos, sl) || pos + sl > n) break;
-        const char* s = (const char*)(p + pos);
-        switch (field) {
-          case 2: sp.id = s; sp.id_len = sl; break;
-          case 3: {  // log line -> packed buffer
-            const size_t copy = std::min<size_t>(sl, (size_t)max_len);
-            std::memcpy(lbuf + i * max_len, s, copy);
-            lenp[i] = (int32_t)copy;
-            break;
-          }
-          case 4: sp.src = s; sp.src_len = sl; break;
-          case 5: sp.host = s; sp.host_len = sl; break;
-          default: break;
-        }
-        pos += sl;
-      } else {
-        if (!skip_field(p, n, pos, wt)) break;
-      }
-    }
-  }
-}
+using dmx_proto::LogSpan;
+using dmx_proto::decode_log_core;
 
 std::vector<std::pair<const uint8_t*, size_t>> frame_ptrs(
     const std::vector<py::bytes>& frames) {

@@ -11,6 +11,8 @@
 // connections keep the Python reader — ssl objects are not plain fds.)
 #include <torch/extension.h>
 
+#include "proto_log.h"
+
 #include <poll.h>
 #include <sys/socket.h>
 #include <unistd.h>
@@ -72,6 +74,83 @@ class FdFrameReader {
     }
   }
 
+  // Socket -> packed LogSchema tensors with ZERO Python objects per
+  // frame: SP/compact frame slicing AND proto decode happen here, GIL
+  // released, straight from the connection buffer into the [B, max_len]
+  // line tensor the GPU pipeline consumes. Returns (lines, lens,
+  // ids_blob, ids_off, frame_bytes); B == 0 on timeout.
+  py::tuple read_batch_packed(int max_frames, int timeout_ms, int max_len,
+                              bool pin) {
+    const size_t hdr = sp_ ? 8 : 4;
+    std::vector<std::pair<const uint8_t*, size_t>> raw;
+    size_t consumed = 0;
+    for (;;) {
+      // scan complete frames in the buffer (parse happens IN PLACE)
+      raw.clear();
+      size_t pos = 0;
+      while ((int)raw.size() < max_frames && buf_.size() - pos >= hdr) {
+        uint64_t len = 0;
+        for (size_t i = 0; i < hdr; ++i)
+          len = (len << 8) | (unsigned char)buf_[pos + i];
+        if (len > kMaxFrame) throw std::runtime_error("oversize frame");
+        if (buf_.size() - pos < hdr + len) break;
+        raw.emplace_back((const uint8_t*)buf_.data() + pos + hdr, (size_t)len);
+        pos += hdr + len;
+      }
+      consumed = pos;
+      if (!raw.empty()) break;
+
+      int rc;
+      ssize_t n;
+      char tmp[262144];
+      {
+        py::gil_scoped_release release;
+        struct pollfd pfd{fd_, POLLIN, 0};
+        rc = ::poll(&pfd, 1, timeout_ms);
+        if (rc > 0) n = ::recv(fd_, tmp, sizeof(tmp), 0);
+      }
+      if (rc == 0) {  // timeout: empty batch
+        auto opts = torch::TensorOptions().dtype(torch::kUInt8);
+        return py::make_tuple(torch::zeros({0, max_len}, opts),
+                              torch::zeros({0}, torch::kInt32), py::bytes(""),
+                              torch::zeros({1}, torch::kInt32), 0);
+      }
+      if (rc < 0) throw std::runtime_error("poll failed");
+      if (n == 0) {
+        PyErr_SetString(PyExc_EOFError, "peer closed");
+        throw py::error_already_set();
+      }
+      if (n < 0) throw std::runtime_error("recv failed");
+      buf_.insert(buf_.end(), tmp, tmp + n);
+    }
+
+    const int64_t B = (int64_t)raw.size();
+    auto lopts = torch::TensorOptions().dtype(torch::kUInt8);
+    if (pin) lopts = lopts.pinned_memory(true);
+    auto lines = torch::zeros({B, max_len}, lopts);
+    auto lens = torch::zeros({B}, torch::kInt32);
+    auto ids_off = torch::zeros({B + 1}, torch::kInt32);
+    std::vector<dmx_proto::LogSpan> spans(B);
+    std::string blob;
+    {
+      py::gil_scoped_release release;
+      dmx_proto::decode_log_core(raw, max_len, lines.data_ptr<uint8_t>(),
+                                 lens.data_ptr<int32_t>(), spans);
+      int32_t* off = ids_off.data_ptr<int32_t>();
+      size_t total = 0;
+      for (int64_t i = 0; i < B; ++i) total += spans[i].id_len;
+      blob.reserve(total);
+      for (int64_t i = 0; i < B; ++i) {
+        off[i] = (int32_t)blob.size();
+        blob.append(spans[i].id, spans[i].id_len);
+      }
+      off[B] = (int32_t)blob.size();
+    }
+    buf_.erase(buf_.begin(), buf_.begin() + consumed);
+    return py::make_tuple(lines, lens, py::bytes(blob), ids_off,
+                          (int64_t)consumed);
+  }
+
  private:
   static py::list to_bytes(const std::vector<std::string>& v) {
     py::list out;
@@ -90,5 +169,8 @@ void register_frame_reader(py::module_& m) {
   py::class_<FdFrameReader>(m, "FdFrameReader")
       .def(py::init<int, bool>(), py::arg("fd"), py::arg("sp"))
       .def("read_batch", &FdFrameReader::read_batch,
-           py::arg("max_frames") = 4096, py::arg("timeout_ms") = 200);
+           py::arg("max_frames") = 4096, py::arg("timeout_ms") = 200)
+      .def("read_batch_packed", &FdFrameReader::read_batch_packed,
+           py::arg("max_frames") = 4096, py::arg("timeout_ms") = 200,
+           py::arg("max_len") = 256, py::arg("pin") = false);
 }

@@ -137,6 +137,13 @@ class ServiceSettings(BaseModel):
     engine_batch_size: int = Field(default=256, ge=1, le=1_048_576)
     engine_batch_linger_ms: float = Field(default=2.0, ge=0.0)
 
+    #: packed data plane: socket reader threads decode LogSchema frames
+    #: straight into tensors in C++ (zero Python objects per frame).
+    #: Requires a plain ipc/tcp listener and a component exposing
+    #: ``process_packed_frames`` (e.g. FusedPipelineDetector); the engine
+    #: falls back to the frame loop otherwise.
+    engine_packed_mode: bool = False
+
     # --- outputs (reference settings.py:68-70) ---
     out_addr: List[str] = Field(default_factory=list)
     dial_timeout: int = Field(default=1000, ge=0, description="output dial timeout, ms")

@@ -47,6 +47,9 @@ def main():
                          "parser+detector chain")
     ap.add_argument("--no-transformer", action="store_true",
                     help="(--fused) skip BERT-tiny scoring, hash-only")
+    ap.add_argument("--packed", action="store_true",
+                    help="(--fused) native packed data plane: C++ socket "
+                         "reader decodes frames straight into tensors")
     ap.add_argument("--graph", action="store_true",
                     help="(--fused) replay the detect path as one hipGraph "
                          "per batch (graph_batch = engine batch size)")
@@ -106,6 +109,7 @@ def main():
             "engine_batch_size": args.batch,
             "engine_batch_linger_ms": 3.0,
             "engine_buffer_size": 8192,
+            "engine_packed_mode": args.packed,
             "config_file": write_yaml("fc.yaml", {"detectors": {
                 "FusedPipelineDetector": {
                     "log_format": AUDIT_LOG_FORMAT,
