@@ -268,7 +268,7 @@ class Engine:
         enable = getattr(self._pair_sock, "enable_packed", None)
         max_len = getattr(self.processor, "packed_max_len", lambda: 256)()
         pin = getattr(self.processor, "packed_pin_memory", lambda: False)()
-        if enable is None or not enable(max_len, pin):
+        if enable is None or not enable(max_len, pin, s.engine_batch_size):
             self._log.warning(
                 "engine_packed_mode: listener cannot enable the packed path "
                 "(TLS/ws/inproc or extension missing); using the frame loop"
@@ -283,7 +283,10 @@ class Engine:
             try:
                 t_r0 = time.perf_counter()
                 conn, lines, lens, blob, off, nbytes = (
-                    self._pair_sock.recv_packed(s.engine_recv_timeout)
+                    self._pair_sock.recv_packed(
+                        s.engine_recv_timeout, s.engine_batch_size,
+                        s.engine_batch_linger_ms,
+                    )
                 )
                 if stats_on:
                     st["recv"] += time.perf_counter() - t_r0

@@ -527,7 +527,9 @@ class PairListener:
                 while not self._closed.is_set():
                     packed = self._packed
                     if packed is not None:
-                        chunk = native.read_batch_packed(4096, 200, *packed)
+                        max_len, pin, max_frames = packed
+                        chunk = native.read_batch_packed(
+                            max_frames, 200, max_len, pin)
                         if chunk[0].shape[0] > 0:
                             self._recv_q.put((conn, chunk))
                         continue
@@ -576,7 +578,8 @@ class PairListener:
         return frames[0]
 
     # -- packed fast path ----------------------------------------------
-    def enable_packed(self, max_len: int, pin: bool) -> bool:
+    def enable_packed(self, max_len: int, pin: bool,
+                      max_frames: int = 4096) -> bool:
         """Switch reader threads to the native socket→tensor decode path
         (plain fd sockets only; TLS/ws readers are unaffected and keep
         delivering byte frames)."""
@@ -586,32 +589,71 @@ class PairListener:
             from ..ops import _dmx_C  # noqa: F401
         except Exception:  # noqa: BLE001
             return False
-        self._packed = (max_len, pin)
+        self._packed = (max_len, pin, max_frames)
         return True
 
-    def recv_packed(self, timeout_ms: int):
-        """Pop ONE packed chunk: (conn, lines, lens, ids_blob, ids_off,
-        frame_bytes). Byte-frame items queued before enable_packed (or
-        from TLS/ws peers) are converted via the batch codec."""
-        try:
-            conn, item = self._recv_q.get(timeout=timeout_ms / 1000.0)
-        except queue.Empty:
-            raise RecvTimeout(self.addr) from None
-        self._last_sender = conn
+    def _as_packed(self, item):
+        """Normalize a queue item to (lines, lens, blob, off, nbytes) —
+        byte-frame items (queued before enable_packed, or from TLS/ws
+        peers) are converted via the batch codec."""
         import torch
 
         if len(item) == 5 and torch.is_tensor(item[0]):
             lines, lens, blob, off, nbytes = item
-            return conn, lines, lens, blob, off, int(nbytes)
-        # byte-frame item: convert (rare transition / mixed-peer case)
+            return lines, lens, blob, off, int(nbytes)
         from ..ops import _dmx_C
 
-        max_len, pin = self._packed or (256, False)
+        max_len, pin = (self._packed or (256, False, 0))[:2]
         frames = [f for f in item if f]
         nbytes = sum(len(f) for f in frames)
         lines, lens, blob, off = _dmx_C.decode_log_batch_packed(
             list(frames), max_len, pin
         )
+        return lines, lens, blob, off, nbytes
+
+    def recv_packed(self, timeout_ms: int, max_frames: int = 0,
+                    linger_ms: float = 0.0):
+        """Pop packed chunks, merging up to ``max_frames`` rows within
+        ``linger_ms``: (conn, lines, lens, ids_blob, ids_off, frame_bytes).
+        ``conn`` is the FIRST chunk's sender (packed mode is fan-in via
+        out_addr; per-row reply routing is the frame loop's job)."""
+        import torch
+
+        try:
+            conn, item = self._recv_q.get(timeout=timeout_ms / 1000.0)
+        except queue.Empty:
+            raise RecvTimeout(self.addr) from None
+        self._last_sender = conn
+        lines, lens, blob, off, nbytes = self._as_packed(item)
+        if max_frames <= 0 or lines.shape[0] >= max_frames:
+            return conn, lines, lens, blob, off, nbytes
+        chunks = [(lines, lens, blob, off, nbytes)]
+        total = int(lines.shape[0])
+        deadline = time.monotonic() + linger_ms / 1000.0
+        while total < max_frames:
+            remaining = deadline - time.monotonic()
+            try:
+                _c, item = self._recv_q.get(
+                    timeout=max(remaining, 0) if remaining > 0 else None,
+                    block=remaining > 0,
+                )
+            except queue.Empty:
+                break
+            ch = self._as_packed(item)
+            chunks.append(ch)
+            total += int(ch[0].shape[0])
+        if len(chunks) == 1:
+            return conn, lines, lens, blob, off, nbytes
+        lines = torch.cat([c[0] for c in chunks])
+        lens = torch.cat([c[1] for c in chunks])
+        blob = b"".join(c[2] for c in chunks)
+        offs = [chunks[0][3]]
+        base = int(chunks[0][3][-1])
+        for c in chunks[1:]:
+            offs.append(c[3][1:] + base)
+            base += int(c[3][-1])
+        off = torch.cat(offs)
+        nbytes = sum(c[4] for c in chunks)
         return conn, lines, lens, blob, off, nbytes
 
     def recv_many(

@@ -98,18 +98,21 @@ class FdFrameReader {
         pos += hdr + len;
       }
       consumed = pos;
-      if (!raw.empty()) break;
+      if ((int)raw.size() >= max_frames) break;
 
+      // keep draining while the socket has data RIGHT NOW (batches grow
+      // to max_frames without waiting); block only when we have nothing
       int rc;
-      ssize_t n;
+      ssize_t n = -1;
       char tmp[262144];
       {
         py::gil_scoped_release release;
         struct pollfd pfd{fd_, POLLIN, 0};
-        rc = ::poll(&pfd, 1, timeout_ms);
+        rc = ::poll(&pfd, 1, raw.empty() ? timeout_ms : 0);
         if (rc > 0) n = ::recv(fd_, tmp, sizeof(tmp), 0);
       }
-      if (rc == 0) {  // timeout: empty batch
+      if (rc == 0) {
+        if (!raw.empty()) break;  // no more data now: ship what we have
         auto opts = torch::TensorOptions().dtype(torch::kUInt8);
         return py::make_tuple(torch::zeros({0, max_len}, opts),
                               torch::zeros({0}, torch::kInt32), py::bytes(""),
@@ -117,6 +120,7 @@ class FdFrameReader {
       }
       if (rc < 0) throw std::runtime_error("poll failed");
       if (n == 0) {
+        if (!raw.empty()) break;  // deliver the tail before raising EOF
         PyErr_SetString(PyExc_EOFError, "peer closed");
         throw py::error_already_set();
       }

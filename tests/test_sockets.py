@@ -366,3 +366,47 @@ def test_ws_continuation_without_start_rejected():
     finally:
         a.close()
         b.close()
+
+
+def test_recv_packed_merges_chunks(tmp_path):
+    """recv_packed merges queued packed chunks: lines concatenated, ids
+    blob offsets rebased."""
+    import torch
+
+    from detectmateservice_amd import ops
+    from detectmateservice_amd.engine.sockets import PairDialer, PairListener
+    from detectmateservice_amd.schemas import LogSchema
+
+    if not ops.have_extension():
+        pytest.skip("extension not built")
+    addr = f"ipc://{tmp_path}/pk.ipc"
+    listener = PairListener(addr, buffer_size=8192)
+    assert listener.enable_packed(64, False, max_frames=4096)
+    dialer = PairDialer(addr)
+    try:
+        assert dialer.wait_connected(10.0)
+        frames = [LogSchema(logID=f"id{i}", log=f"line {i}").serialize()
+                  for i in range(50)]
+        # two separated sends -> likely two packed chunks in the queue
+        assert dialer.send_many(frames[:20], block=True) == 20
+        import time as t_mod
+        t_mod.sleep(0.3)
+        assert dialer.send_many(frames[20:], block=True) == 30
+        got = 0
+        ids = []
+        while got < 50:
+            conn, lines, lens, blob, off, nbytes = listener.recv_packed(
+                5000, max_frames=4096, linger_ms=200.0)
+            B = int(lines.shape[0])
+            assert nbytes > 0
+            for i in range(B):
+                text = bytes(lines[i, : int(lens[i])].numpy().tobytes())
+                lid = blob[int(off[i]):int(off[i + 1])].decode()
+                ids.append(lid)
+                assert text.decode() == f"line {lid[2:]}"
+            got += B
+        assert ids == [f"id{i}" for i in range(50)]
+        assert isinstance(off, torch.Tensor)
+    finally:
+        dialer.close()
+        listener.close()
