@@ -434,7 +434,18 @@ class PairListener:
         self._ssl_ctx: Optional[ssl.SSLContext] = None
         #: packed fast path (enable_packed): reader threads decode
         #: LogSchema frames straight into tensors — no Python objects
-        self._packed: Optional[Tuple[int, bool]] = None
+        self._packed: Optional[Tuple[int, bool, int]] = None
+        self._reader_threads: List[threading.Thread] = []
+        # Resolve the native reader ONCE here (synchronously): a lazy
+        # import inside the reader thread costs a full torch import on a
+        # cold process (~2 s) and delays the first frames past recv
+        # deadlines.
+        try:
+            from ..ops import _dmx_C as _native  # type: ignore[attr-defined]
+
+            self._native_mod = _native
+        except Exception:  # noqa: BLE001 - extension absent: python readers
+            self._native_mod = None
 
         scheme, rest = self.addr.scheme, self.addr.rest
         #: SP (NNG pair0) wire mapping on tcp/tls+tcp edges
@@ -506,21 +517,21 @@ class PairListener:
                 continue
             with self._peers_lock:
                 self._peers.append(conn)
-            threading.Thread(
+            rt = threading.Thread(
                 target=self._reader_loop, args=(conn,),
                 name="PairListenerReader", daemon=True,
-            ).start()
+            )
+            self._reader_threads.append(rt)
+            rt.start()
 
     def _reader_loop(self, conn: socket.socket) -> None:
         # Native C++ receive loop for plain fd sockets (GIL released around
         # poll/recv/parse — frame_reader.cpp); Python readers for TLS/ws.
         native = None
-        if not self._ws and self._ssl_ctx is None:
+        if not self._ws and self._ssl_ctx is None and self._native_mod is not None:
             try:
-                from ..ops import _dmx_C  # type: ignore[attr-defined]
-
-                native = _dmx_C.FdFrameReader(conn.fileno(), self._sp)
-            except Exception:  # noqa: BLE001 - extension absent: python path
+                native = self._native_mod.FdFrameReader(conn.fileno(), self._sp)
+            except Exception:  # noqa: BLE001 - bad fd etc: python path
                 native = None
         try:
             if native is not None:
@@ -583,11 +594,7 @@ class PairListener:
         """Switch reader threads to the native socket→tensor decode path
         (plain fd sockets only; TLS/ws readers are unaffected and keep
         delivering byte frames)."""
-        if self._ws or self._ssl_ctx is not None:
-            return False
-        try:
-            from ..ops import _dmx_C  # noqa: F401
-        except Exception:  # noqa: BLE001
+        if self._ws or self._ssl_ctx is not None or self._native_mod is None:
             return False
         self._packed = (max_len, pin, max_frames)
         return True
@@ -601,12 +608,10 @@ class PairListener:
         if len(item) == 5 and torch.is_tensor(item[0]):
             lines, lens, blob, off, nbytes = item
             return lines, lens, blob, off, int(nbytes)
-        from ..ops import _dmx_C
-
         max_len, pin = (self._packed or (256, False, 0))[:2]
         frames = [f for f in item if f]
         nbytes = sum(len(f) for f in frames)
-        lines, lens, blob, off = _dmx_C.decode_log_batch_packed(
+        lines, lens, blob, off = self._native_mod.decode_log_batch_packed(
             list(frames), max_len, pin
         )
         return lines, lens, blob, off, nbytes
@@ -749,10 +754,20 @@ class PairListener:
         with self._peers_lock:
             for conn in self._peers:
                 try:
+                    conn.shutdown(socket.SHUT_RDWR)
+                except OSError:
+                    pass
+                try:
                     conn.close()
                 except OSError:
                     pass
             self._peers.clear()
+        # join reader threads: a daemon thread still inside the C++
+        # reader when the interpreter finalizes would be killed at GIL
+        # reacquisition (pthread_exit unwinding C++ frames -> terminate)
+        for rt in self._reader_threads:
+            rt.join(timeout=1.0)
+        self._reader_threads.clear()
         if self.addr.scheme == "ipc":
             try:
                 os.unlink("/" + self.addr.rest.lstrip("/"))
