@@ -597,6 +597,11 @@ class PairListener:
         if self._ws or self._ssl_ctx is not None or self._native_mod is None:
             return False
         self._packed = (max_len, pin, max_frames)
+        # Each queued chunk can hold a full [max_frames, max_len] (pinned)
+        # buffer (~2 MB at 8192x256): bound the queue so a fast producer
+        # backpressures through the socket instead of ballooning pinned
+        # host memory.
+        self._recv_q.maxsize = 64
         return True
 
     def _as_packed(self, item):

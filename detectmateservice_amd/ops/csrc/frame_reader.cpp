@@ -130,8 +130,18 @@ class FdFrameReader {
 
     const int64_t B = (int64_t)raw.size();
     auto lopts = torch::TensorOptions().dtype(torch::kUInt8);
-    if (pin) lopts = lopts.pinned_memory(true);
-    auto lines = torch::zeros({B, max_len}, lopts);
+    torch::Tensor lines;
+    if (pin) {
+      // CONSTANT-size pinned alloc (narrowed to B): the pinned caching
+      // allocator reuses same-size blocks, while per-chunk exact-size
+      // pinned allocs (hipHostMalloc ~ms) collapsed small-batch
+      // throughput ~20x when the service outpaced the feeder.
+      lopts = lopts.pinned_memory(true);
+      lines = torch::empty({(int64_t)max_frames, max_len}, lopts)
+                  .narrow(0, 0, B);
+    } else {
+      lines = torch::empty({B, max_len}, lopts);
+    }
     auto lens = torch::zeros({B}, torch::kInt32);
     auto ids_off = torch::zeros({B + 1}, torch::kInt32);
     std::vector<dmx_proto::LogSpan> spans(B);
