@@ -996,12 +996,16 @@ class PairDialer:
                 continue
             # coalesce everything queued into ONE sendall (per-frame
             # sendall measured ~13k frames/s; batching removes the
-            # syscall + GIL ping-pong per frame)
-            while len(batch) < 512:
-                try:
-                    batch = batch + self._send_q.get_nowait()
-                except queue.Empty:
-                    break
+            # syscall + GIL ping-pong per frame). 4096-frame cap ≈ 1 MB
+            # per sendall; only backlog is drained, so idle-path latency
+            # is unaffected.
+            if len(batch) < 4096:
+                batch = list(batch)
+                while len(batch) < 4096:
+                    try:
+                        batch.extend(self._send_q.get_nowait())
+                    except queue.Empty:
+                        break
             if self._ws:
                 payload = b"".join(_ws_encode(f, mask=True) for f in batch)
             elif self._native_mod is not None:
