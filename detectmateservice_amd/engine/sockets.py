@@ -542,7 +542,14 @@ class PairListener:
                         chunk = native.read_batch_packed(
                             max_frames, 200, max_len, pin)
                         if chunk[0].shape[0] > 0:
-                            self._recv_q.put((conn, chunk))
+                            # bounded put: never deadlock a close() join
+                            while not self._closed.is_set():
+                                try:
+                                    self._recv_q.put((conn, chunk),
+                                                     timeout=0.5)
+                                    break
+                                except queue.Full:
+                                    continue
                         continue
                     frames = native.read_batch(4096, 200)
                     if frames:
