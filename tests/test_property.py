@@ -189,3 +189,36 @@ def test_sp_frame_roundtrip_property(payloads):
     finally:
         a.close()
         b.close()
+
+
+@settings(max_examples=50, deadline=None)
+@given(
+    payloads=st.lists(st.binary(min_size=0, max_size=2000), min_size=1,
+                      max_size=20),
+    sp=st.booleans(),
+)
+def test_pack_frames_roundtrip_through_reader(payloads, sp):
+    """C++ pack_frames blob parsed back by the C++ FdFrameReader."""
+    import socket as s_mod
+
+    from detectmateservice_amd import ops
+
+    if not ops.have_extension():
+        pytest.skip("extension not built")
+    from detectmateservice_amd.ops import _dmx_C
+
+    a, b = s_mod.socketpair()
+    try:
+        a.sendall(_dmx_C.pack_frames(payloads, sp))
+        a.shutdown(s_mod.SHUT_WR)
+        reader = _dmx_C.FdFrameReader(b.fileno(), sp)
+        got = []
+        while len(got) < len(payloads):
+            frames = reader.read_batch(4096, 2000)
+            if not frames:
+                break
+            got.extend(frames)
+        assert got == payloads
+    finally:
+        a.close()
+        b.close()
