@@ -872,6 +872,8 @@ class PairDialer:
         self._sp = self.addr.scheme in ("tcp", "tls+tcp")
         self._ws = self.addr.scheme == "ws"
         self._inflight: Optional[bytes] = None
+        self._send_stats = ([0, 0.0, 0.0, 0] if os.environ.get(
+            "DMX_ENGINE_STATS") == "1" else None)
         try:
             from ..ops import _dmx_C as _native  # type: ignore[attr-defined]
 
@@ -1013,6 +1015,8 @@ class PairDialer:
                         batch.extend(self._send_q.get_nowait())
                     except queue.Empty:
                         break
+            _st = self._send_stats
+            _t0 = time.perf_counter() if _st is not None else 0.0
             if self._ws:
                 payload = b"".join(_ws_encode(f, mask=True) for f in batch)
             elif self._native_mod is not None:
@@ -1026,8 +1030,21 @@ class PairDialer:
                     parts.append(hdr.pack(len(f)))
                     parts.append(f)
                 payload = b"".join(parts)
+            if _st is not None:
+                _t1 = time.perf_counter()
             try:
                 conn.sendall(payload)
+                if _st is not None:
+                    _st[0] += len(batch)
+                    _st[1] += _t1 - _t0
+                    _st[2] += time.perf_counter() - _t1
+                    _st[3] += 1
+                    if _st[3] % 64 == 0:
+                        self._log.warning(
+                            "[send-stats] %d frames in %d sendalls: pack "
+                            "%.2fms/batch send %.2fms/batch (%.0f fr/batch)",
+                            _st[0], _st[3], _st[1] * 1e3 / _st[3],
+                            _st[2] * 1e3 / _st[3], _st[0] / _st[3])
             except OSError:
                 self._log.debug("send to %s failed; reconnecting", self.addr)
                 try:
