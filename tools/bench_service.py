@@ -47,6 +47,8 @@ def main():
                          "parser+detector chain")
     ap.add_argument("--no-transformer", action="store_true",
                     help="(--fused) skip BERT-tiny scoring, hash-only")
+    ap.add_argument("--feeders", type=int, default=1,
+                    help="parallel feeder connections (fan-in sources)")
     ap.add_argument("--packed", action="store_true",
                     help="(--fused) native packed data plane: C++ socket "
                          "reader decodes frames straight into tensors")
@@ -142,11 +144,15 @@ def main():
     if stats_on:
         print(f"# service stdout: {tmp}/svc_stdout.log", file=sys.stderr)
     sink = PairListener(sink_addr)
-    feeder = PairDialer(parser_in, buffer_size=8192)
+    feeders = [PairDialer(parser_in, buffer_size=8192)
+               for _ in range(max(1, args.feeders))]
+    feeder = feeders[0]
     try:
         # generous: on a fresh GPU box the first torch/hip init inside the
         # service process can take over a minute
         assert feeder.wait_connected(180.0), "ingest service did not come up"
+        for f in feeders[1:]:
+            assert f.wait_connected(30.0)
         gen = AuditLogGenerator(seed=7, anomaly_rate=0.0)
         # pre-serialize frames so the feeder isn't the bottleneck
         def make_frames(n, tag):
@@ -158,14 +164,29 @@ def main():
         warmup = make_frames(args.warmup_lines, "w")
         frames = make_frames(args.lines, "m")
 
-        def pump(batch):
+        def pump_one(f, batch):
             sent = 0
             while sent < len(batch):
-                n = feeder.send_many(batch[sent:sent + 4096], block=False)
+                n = f.send_many(batch[sent:sent + 4096], block=False)
                 if n == 0:
                     time.sleep(0.0005)
                 sent += n
             return sent
+
+        def pump(batch):
+            if len(feeders) == 1:
+                return pump_one(feeder, batch)
+            import threading as _th
+            k = len(feeders)
+            per = (len(batch) + k - 1) // k
+            ts = [_th.Thread(target=pump_one,
+                             args=(f, batch[i * per:(i + 1) * per]))
+                  for i, f in enumerate(feeders)]
+            for th in ts:
+                th.start()
+            for th in ts:
+                th.join()
+            return len(batch)
 
         pump(warmup)
         time.sleep(2.0)  # drain training frames through both stages
@@ -186,15 +207,24 @@ def main():
             "msg='op=PAM:x acct=\"x\" exe=/bin/x hostname=? addr=? "
             "terminal=x res=success'"
         )
-        while not feeder.send(LogSchema(logID="sentinel", log=bad).serialize(),
-                              block=False):
-            time.sleep(0.0005)
-        while True:
+        # one sentinel PER feeder connection: the pipeline is only idle
+        # once every connection's queued frames have drained
+        for k, f in enumerate(feeders):
+            while not f.send(
+                LogSchema(logID=f"sentinel{k}", log=bad).serialize(),
+                block=False,
+            ):
+                time.sleep(0.0005)
+        seen = 0
+        while seen < len(feeders):
             try:
-                sink.recv(timeout_ms=30000)
-                break
+                frame = sink.recv(timeout_ms=30000)
             except RecvTimeout:
                 raise SystemExit("sentinel alert never arrived")
+            from detectmateservice_amd.schemas import DetectorSchema
+            alert = DetectorSchema.deserialize(frame)
+            if any(i.startswith("sentinel") for i in alert.logIDs):
+                seen += 1
         elapsed = time.perf_counter() - t0
 
         total = args.lines + 1
@@ -214,13 +244,15 @@ def main():
                              if args.fused else
                              "feeder->MatcherParser svc->NewValueDetector svc->sink"),
                 "engine_batch_size": args.batch,
+                "feeders": args.feeders,
                 "lines": total,
                 "elapsed_s": round(elapsed, 3),
                 "pump_s": round(pump_elapsed, 3),
             },
         }))
     finally:
-        feeder.close()
+        for f in feeders:
+            f.close()
         sink.close()
         for p in procs:
             p.terminate()
