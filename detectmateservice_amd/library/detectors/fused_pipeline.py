@@ -80,6 +80,12 @@ class FusedPipelineDetector(CoreComponent):
         return int(self.config.max_len)
 
     def packed_pin_memory(self) -> bool:
+        import os as _os
+
+        # DMX_PACKED_PIN=0 disables pinned staging (A/B: pinned-cache
+        # event syncs vs pageable H2D)
+        if _os.environ.get("DMX_PACKED_PIN") == "0":
+            return False
         return self.device.type == "cuda"
 
     def _score_packed(self, lines: torch.Tensor, lens: torch.Tensor):
