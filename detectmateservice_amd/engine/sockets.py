@@ -58,6 +58,16 @@ SP_PAIR0_PROTO = 0x10
 _LEN64 = struct.Struct(">Q")
 
 
+def _tune_buffers(sock: socket.socket) -> None:
+    """Best-effort 4 MB kernel buffers: batched senders write ~1 MB blobs
+    per sendall; the 208 KB default forces sender/reader lockstep."""
+    for opt in (socket.SO_SNDBUF, socket.SO_RCVBUF):
+        try:
+            sock.setsockopt(socket.SOL_SOCKET, opt, 4 << 20)
+        except OSError:
+            pass
+
+
 def _sp_header_bytes(proto: int = SP_PAIR0_PROTO) -> bytes:
     return b"\x00SP\x00" + struct.pack(">H", proto) + b"\x00\x00"
 
@@ -502,6 +512,7 @@ class PairListener:
                 continue
             except OSError:
                 return
+            _tune_buffers(conn)
             if self._ssl_ctx is not None:
                 try:
                     conn = self._ssl_ctx.wrap_socket(conn, server_side=True)
@@ -921,6 +932,7 @@ class PairDialer:
                     s.close()
                     return False
             s.settimeout(None)
+            _tune_buffers(s)
             self._conn = s
             threading.Thread(
                 target=self._reader_loop, args=(s,),
