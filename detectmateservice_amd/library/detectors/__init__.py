@@ -24,3 +24,12 @@ __all__ += ["EmbeddingDetector", "EmbeddingDetectorConfig"]
 from .fused_pipeline import FusedPipelineDetector, FusedPipelineDetectorConfig
 
 __all__ += ["FusedPipelineDetector", "FusedPipelineDetectorConfig"]
+from .sklearn_detector import (
+    FrequencyDetector,
+    FrequencyDetectorConfig,
+    SklearnDetector,
+    SklearnDetectorConfig,
+)
+
+__all__ += ["SklearnDetector", "SklearnDetectorConfig",
+            "FrequencyDetector", "FrequencyDetectorConfig"]
