@@ -25,9 +25,6 @@ void dmx_launch_attention_mfma_bf16(const void*, void*, int, int, int, int,
 void dmx_launch_bert_fused_bf16(const void*, const void*, const void*,
                                 const void*, const void*, void*, int, int,
                                 int, float, hipStream_t);
-void dmx_launch_bert_fused_bf16_v4(const void*, const void*, const void*,
-                                   const void*, const void*, void*, int, int,
-                                   int, float, hipStream_t);
 void dmx_launch_bert_fused_probe(const void*, const void*, const void*,
                                  const void*, const void*, void*, int, int,
                                  int, float, int, hipStream_t);
@@ -185,26 +182,6 @@ torch::Tensor bert_fused_bf16(torch::Tensor lines, torch::Tensor start,
   return scores;
 }
 
-torch::Tensor bert_fused_bf16_v4(torch::Tensor lines, torch::Tensor start,
-                                 torch::Tensor end, torch::Tensor wb,
-                                 torch::Tensor fb, int64_t n_layers,
-                                 double eps) {
-  TORCH_CHECK(lines.is_cuda() && lines.dtype() == torch::kUInt8 &&
-                  lines.is_contiguous(),
-              "lines must be contiguous u8 on GPU");
-  TORCH_CHECK(wb.dtype() == torch::kBFloat16 && wb.is_contiguous());
-  TORCH_CHECK(fb.dtype() == torch::kFloat32 && fb.is_contiguous());
-  TORCH_CHECK(start.dtype() == torch::kInt32 && end.dtype() == torch::kInt32);
-  const auto B = lines.size(0), max_len = lines.size(1);
-  auto scores = torch::empty(
-      {B}, torch::TensorOptions().dtype(torch::kFloat32).device(lines.device()));
-  dmx_launch_bert_fused_bf16_v4(lines.data_ptr(), start.data_ptr(),
-                                end.data_ptr(), wb.data_ptr(), fb.data_ptr(),
-                                scores.data_ptr(), (int)B, (int)max_len,
-                                (int)n_layers, (float)eps, cur_stream());
-  return scores;
-}
-
 std::vector<torch::Tensor> template_match(
     torch::Tensor lines, torch::Tensor line_len, torch::Tensor fmt_bytes,
     torch::Tensor fmt_seg_off, torch::Tensor seg_bytes, torch::Tensor seg_off,
@@ -339,8 +316,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "MFMA MHA reading fused QKV layout [B,S,3*H*Dh] -> [B,S,H*Dh]");
   m.def("bert_fused_bf16", &bert_fused_bf16,
         "whole-model BERT-tiny forward, one workgroup per line");
-  m.def("bert_fused_bf16_v4", &bert_fused_bf16_v4,
-        "fused BERT-tiny forward, 3-blocks/CU occupancy variant");
   m.def("bert_fused_probe", &bert_fused_probe,
         "phase-masked probe variant of the fused BERT kernel");
   m.def("bert_fused_timed", &bert_fused_timed,

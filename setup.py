@@ -25,7 +25,6 @@ sources = [
     str(CSRC / "attention.hip"),
     str(CSRC / "attention_mfma.hip"),
     str(CSRC / "bert_fused.hip"),
-    str(CSRC / "bert_fused_v4.hip"),
     str(CSRC / "template_match.hip"),
     str(CSRC / "hashset.hip"),
     str(CSRC / "edit_distance.hip"),

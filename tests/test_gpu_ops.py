@@ -526,24 +526,3 @@ def test_fused_pipeline_detector_graph_matches_eager():
     out_g2 = graph.process_batch(batch[:40] + batch[:23] + [batch[-1]])
     assert out_g2[-1] is not None
     assert sum(o is not None for o in out_g2) == 1
-
-
-@pytest.mark.gpu
-def test_bert_fused_v4_matches_v3():
-    """3-blocks/CU occupancy variant == production kernel numerics
-    (same MFMA fragment math; softmax split across waves)."""
-    from detectmateservice_amd.models.bert_tiny import BertTinyDetectorModel
-    from detectmateservice_amd.ops import _dmx_C
-
-    torch.manual_seed(3)
-    model = BertTinyDetectorModel(device="cuda")
-    wb, fb = model._fused_blobs()
-    for B in (1, 5, 64, 1024):
-        lines = torch.randint(32, 127, (B, 256), dtype=torch.uint8, device="cuda")
-        start = torch.randint(0, 8, (B,), dtype=torch.int32, device="cuda")
-        end = start + torch.randint(1, 200, (B,), dtype=torch.int32, device="cuda")
-        s3 = _dmx_C.bert_fused_bf16(lines, start, end, wb, fb, 2, 1e-5)
-        s4 = _dmx_C.bert_fused_bf16_v4(lines, start, end, wb, fb, 2, 1e-5)
-        torch.cuda.synchronize()
-        assert torch.allclose(s3, s4, atol=2e-4, rtol=2e-4), (
-            B, (s3 - s4).abs().max().item())
