@@ -20,7 +20,6 @@
 #include <cstring>
 #include <stdexcept>
 #include <string>
-#include <thread>
 #include <vector>
 
 namespace {
@@ -149,33 +148,8 @@ class FdFrameReader {
     std::string blob;
     {
       py::gil_scoped_release release;
-      // parallel proto decode: rows are disjoint, spans slots are
-      // per-frame — the reader thread was the service's last bound
-      // (~4M lines/s single-threaded incl. recv+alloc+queue)
-      const int nt = B >= 2048 ? 4 : 1;
-      if (nt == 1) {
-        dmx_proto::decode_log_core(raw, max_len, lines.data_ptr<uint8_t>(),
-                                   lens.data_ptr<int32_t>(), spans);
-      } else {
-        uint8_t* lbuf = lines.data_ptr<uint8_t>();
-        int32_t* lenp = lens.data_ptr<int32_t>();
-        const int64_t per = (B + nt - 1) / nt;
-        std::vector<std::thread> ts;
-        for (int w = 0; w < nt; ++w) {
-          const int64_t lo = w * per;
-          const int64_t hi = std::min<int64_t>(B, lo + per);
-          if (lo >= hi) break;
-          ts.emplace_back([&, lo, hi]() {
-            std::vector<std::pair<const uint8_t*, size_t>> sub(
-                raw.begin() + lo, raw.begin() + hi);
-            std::vector<dmx_proto::LogSpan> sub_spans(hi - lo);
-            dmx_proto::decode_log_core(sub, max_len, lbuf + lo * max_len,
-                                       lenp + lo, sub_spans);
-            std::copy(sub_spans.begin(), sub_spans.end(), spans.begin() + lo);
-          });
-        }
-        for (auto& th : ts) th.join();
-      }
+      dmx_proto::decode_log_core(raw, max_len, lines.data_ptr<uint8_t>(),
+                                 lens.data_ptr<int32_t>(), spans);
       int32_t* off = ids_off.data_ptr<int32_t>();
       size_t total = 0;
       for (int64_t i = 0; i < B; ++i) total += spans[i].id_len;
