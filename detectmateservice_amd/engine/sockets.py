@@ -1136,6 +1136,218 @@ class PairDialer:
 # ---------------------------------------------------------------------------
 
 
+# ---------------------------------------------------------------------------
+# shm:// — shared-memory ring transport for co-located services.
+# The ipc:// socket path is bound by kernel copies (~4M lines/s at 258 B
+# frames — BASELINE.md); the ring maps one /dev/shm file into both
+# processes: the producer writes each frame once, the consumer's packed
+# path proto-decodes IN PLACE into tensors (ops/csrc/shm_ring.cpp).
+# Lifecycle: the LISTENER owns the files (unlinks on close); dialers
+# attach (creating the ring if they start first, so late binding keeps
+# buffered frames — reference engine.py:173-179 semantics) and re-attach
+# when the listener recreated the ring.
+# ---------------------------------------------------------------------------
+
+
+def _shm_paths(addr: "EngineAddr"):
+    rest = addr.rest
+    base = rest if rest.startswith("/") else "/" + rest
+    if not base.startswith("/dev/shm/"):
+        base = "/dev/shm" + base
+    return base + ".c2s", base + ".s2c"
+
+
+def _shm_ring_bytes() -> int:
+    return int(os.environ.get("DMX_SHM_RING_BYTES", str(32 << 20)))
+
+
+class ShmListener:
+    """Bound shm input: c2s ring for data, s2c ring for replies."""
+
+    def __init__(self, addr: str, logger=None, buffer_size: int = 128,
+                 tls_config=None) -> None:
+        if tls_config is not None:
+            raise ValueError("shm:// does not support TLS")
+        self.addr = EngineAddr.validate(addr)
+        self._log = logger or logging.getLogger(__name__)
+        from ..ops import _dmx_C  # extension required for shm
+
+        self._c2s_path, self._s2c_path = _shm_paths(self.addr)
+        self._c2s = _dmx_C.ShmRing(self._c2s_path, _shm_ring_bytes(), True)
+        self._s2c = _dmx_C.ShmRing(self._s2c_path, _shm_ring_bytes(), True)
+        self._pending: List[bytes] = []
+        self._packed: Optional[Tuple[int, bool, int]] = None
+        self._closed = threading.Event()
+
+    # -- frame path ----------------------------------------------------
+    def recv(self, timeout_ms: Optional[int] = None) -> bytes:
+        if self._closed.is_set():
+            raise SocketClosed(self.addr)
+        if self._pending:
+            return self._pending.pop(0)
+        frames = self._c2s.read_batch(
+            4096, 100 if timeout_ms is None else timeout_ms)
+        while not frames and timeout_ms is None:
+            if self._closed.is_set():
+                raise SocketClosed(self.addr)
+            frames = self._c2s.read_batch(4096, 100)
+        if not frames:
+            raise RecvTimeout(self.addr)
+        self._pending = [bytes(f) for f in frames[1:]]
+        return bytes(frames[0])
+
+    def recv_many(self, max_frames: int, timeout_ms: int,
+                  linger_ms: float = 0.0) -> List[bytes]:
+        if self._closed.is_set():
+            raise SocketClosed(self.addr)
+        out = self._pending[:max_frames]
+        self._pending = self._pending[max_frames:]
+        deadline = time.monotonic() + linger_ms / 1000.0
+        first = True
+        while len(out) < max_frames:
+            if out and time.monotonic() >= deadline:
+                break
+            tmo = timeout_ms if first and not out else max(
+                1, int((deadline - time.monotonic()) * 1000))
+            frames = self._c2s.read_batch(max_frames - len(out), tmo)
+            first = False
+            if not frames:
+                break
+            out.extend(bytes(f) for f in frames)
+        return out
+
+    # -- packed fast path ----------------------------------------------
+    def enable_packed(self, max_len: int, pin: bool,
+                      max_frames: int = 4096) -> bool:
+        self._packed = (max_len, pin, max_frames)
+        return True
+
+    def recv_packed(self, timeout_ms: int, max_frames: int = 0,
+                    linger_ms: float = 0.0):
+        if self._closed.is_set():
+            raise SocketClosed(self.addr)
+        max_len, pin, mf = self._packed or (256, False, 4096)
+        if max_frames > 0:
+            mf = max_frames
+        lines, lens, blob, off, nbytes = self._c2s.read_batch_packed(
+            mf, timeout_ms, max_len, pin)
+        if lines.shape[0] == 0:
+            raise RecvTimeout(self.addr)
+        return None, lines, lens, blob, off, int(nbytes)
+
+    # -- reply path (request/reply compatibility mode) -----------------
+    def send(self, data: bytes, block: bool = True) -> bool:
+        return self._s2c.write_frames([data]) == 1
+
+    def reply(self, idx: int, data: bytes) -> bool:
+        return self.send(data)
+
+    def close(self) -> None:
+        self._closed.set()
+        for p in (self._c2s_path, self._s2c_path):
+            try:
+                os.unlink(p)
+            except OSError:
+                pass
+
+
+class ShmDialer:
+    """Output side of a shm ring (stage output / feeder)."""
+
+    REATTACH_AFTER_FULL = 50
+
+    def __init__(self, addr: str, logger=None, buffer_size: int = 128,
+                 tls_config=None, dial_timeout_s: float = 1.0) -> None:
+        if tls_config is not None:
+            raise ValueError("shm:// does not support TLS")
+        self.addr = EngineAddr.validate(addr)
+        self._log = logger or logging.getLogger(__name__)
+        self._c2s_path, self._s2c_path = _shm_paths(self.addr)
+        self._closed = threading.Event()
+        self._full_streak = 0
+        self._attach()
+
+    def _attach(self) -> None:
+        from ..ops import _dmx_C
+
+        self._c2s = _dmx_C.ShmRing(self._c2s_path, _shm_ring_bytes(), False)
+        self._s2c = _dmx_C.ShmRing(self._s2c_path, _shm_ring_bytes(), False)
+        try:
+            self._ino = os.stat(self._c2s_path).st_ino
+        except OSError:
+            self._ino = None
+        self._pending_in: List[bytes] = []
+
+    def _maybe_reattach(self) -> None:
+        """The listener owns the files: if it was restarted (new inode),
+        remap — mirrors the socket dialer's background reconnect."""
+        self._full_streak += 1
+        if self._full_streak < self.REATTACH_AFTER_FULL:
+            return
+        self._full_streak = 0
+        try:
+            ino = os.stat(self._c2s_path).st_ino
+        except OSError:
+            return  # listener gone entirely: keep dropping
+        if ino != self._ino:
+            self._log.debug("shm ring recreated; re-attaching %s", self.addr)
+            self._attach()
+
+    def wait_connected(self, timeout_s: float) -> bool:
+        return True  # the ring buffers; drop-when-full supplies elasticity
+
+    def send(self, data: bytes, block: bool = True) -> bool:
+        if self._closed.is_set():
+            raise SocketClosed(self.addr)
+        deadline = time.monotonic() + 5.0
+        while True:
+            if self._c2s.write_frames([data]) == 1:
+                self._full_streak = 0
+                return True
+            self._maybe_reattach()
+            if not block or time.monotonic() >= deadline:
+                return False
+            time.sleep(0.0005)
+
+    def send_many(self, frames, block: bool = True, chunk: int = 4096) -> int:
+        if self._closed.is_set():
+            raise SocketClosed(self.addr)
+        frames = list(frames)
+        accepted = 0
+        while accepted < len(frames):
+            n = self._c2s.write_frames(frames[accepted:accepted + chunk])
+            accepted += n
+            if n == 0:
+                self._maybe_reattach()
+                if not block:
+                    break
+                time.sleep(0.0005)
+            else:
+                self._full_streak = 0
+        return accepted
+
+    def recv(self, timeout_ms: Optional[int] = None) -> bytes:
+        if self._pending_in:
+            return self._pending_in.pop(0)
+        frames = self._s2c.read_batch(
+            4096, 100 if timeout_ms is None else timeout_ms)
+        while not frames and timeout_ms is None:
+            if self._closed.is_set():
+                raise SocketClosed(self.addr)
+            frames = self._s2c.read_batch(4096, 100)
+        if not frames:
+            raise RecvTimeout(self.addr)
+        self._pending_in = [bytes(f) for f in frames[1:]]
+        return bytes(frames[0])
+
+    @property
+    def pending(self) -> int:
+        return 0
+
+    def close(self) -> None:
+        self._closed.set()
+
+
 class PairSocketFactory:
     """Creates bound input sockets; injectable seam for tests/alt transports
     (reference engine.py:111-113)."""
@@ -1150,6 +1362,9 @@ class PairSocketFactory:
         parsed = EngineAddr.validate(addr)
         if parsed.scheme == "inproc":
             return InprocListener(addr, buffer_size=buffer_size)
+        if parsed.scheme == "shm":
+            return ShmListener(addr, logger=logger, buffer_size=buffer_size,
+                               tls_config=tls_config)
         return PairListener(
             addr, logger=logger, tls_config=tls_config, buffer_size=buffer_size
         )
@@ -1161,7 +1376,12 @@ class PairSocketFactory:
         tls_config: Optional[TlsOutputConfig] = None,
         buffer_size: int = 128,
         dial_timeout_s: float = 1.0,
-    ) -> PairDialer:
+    ):
+        parsed = EngineAddr.validate(addr)
+        if parsed.scheme == "shm":
+            return ShmDialer(addr, logger=logger, buffer_size=buffer_size,
+                             tls_config=tls_config,
+                             dial_timeout_s=dial_timeout_s)
         return PairDialer(
             addr, logger=logger, tls_config=tls_config, buffer_size=buffer_size,
             dial_timeout_s=dial_timeout_s,

@@ -302,10 +302,12 @@ torch::Tensor hashset_probe(torch::Tensor hashes, torch::Tensor tables) {
 
 void register_codec(py::module_& m);   // codec.cpp
 void register_frame_reader(py::module_& m);  // frame_reader.cpp
+void register_shm_ring(py::module_& m);      // shm_ring.cpp
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   register_codec(m);
   register_frame_reader(m);
+  register_shm_ring(m);
   m.def("fused_linear_bf16", &fused_linear_bf16,
         "C = act(x @ wt^T + bias), bf16 MFMA (epilogue: 0 none, 1 gelu, 2 relu)");
   m.def("probe_mfma", &probe_mfma, "MFMA 16x16x32 bf16 layout probe");

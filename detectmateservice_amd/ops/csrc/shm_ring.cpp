@@ -1,0 +1,286 @@
+// Shared-memory ring transport for co-located services (shm:// scheme).
+//
+// The ipc:// socket path tops out at ~4M lines/s for 258 B frames: every
+// frame is copied into the kernel by sendall and back out by recv
+// (BASELINE.md "packed plane journey"). Co-located pipeline stages on an
+// MI355X node don't need a kernel hop at all: this ring maps one
+// /dev/shm file into both processes — the producer writes each frame
+// ONCE, the consumer parses IN PLACE (the packed path proto-decodes
+// straight from the ring into the [B, max_len] tensor the GPU pipeline
+// consumes). Single-producer/single-consumer cursors are seqcst atomics;
+// multiple producers serialize on a bounded spinlock. Flow control is
+// drop-don't-block, matching the engine's retry-then-drop semantics
+// (reference engine.py:281-301): a full ring (consumer dead or slow)
+// makes write_frames return the accepted count.
+//
+// Layout: 4 KiB header + power-of-two data area.
+//   [u64 magic][u64 size][u64 head][u64 tail][u32 wlock]
+// Frames: [u32 len][payload], byte-wrapped at the data boundary.
+#include <torch/extension.h>
+
+#include "proto_log.h"
+
+#include <fcntl.h>
+#include <sys/mman.h>
+#include <sys/stat.h>
+#include <unistd.h>
+
+#include <atomic>
+#include <chrono>
+#include <cerrno>
+#include <cstring>
+#include <stdexcept>
+#include <string>
+#include <thread>
+#include <vector>
+
+namespace {
+
+constexpr uint64_t kMagic = 0x444d5853484d3031ull;  // "DMXSHM01"
+constexpr size_t kHdrBytes = 4096;
+constexpr size_t kMaxFrame = 64ull * 1024 * 1024;
+
+struct RingHdr {
+  std::atomic<uint64_t> magic;
+  uint64_t size;
+  std::atomic<uint64_t> head;  // producer cursor (monotonic byte count)
+  std::atomic<uint64_t> tail;  // consumer cursor
+  std::atomic<uint32_t> wlock;
+};
+
+class ShmRing {
+ public:
+  ShmRing(const std::string& path, int64_t size_bytes, bool create)
+      : path_(path) {
+    // power-of-two data size
+    uint64_t sz = 1;
+    while ((int64_t)sz < size_bytes) sz <<= 1;
+    int fd = ::open(path.c_str(), O_RDWR | O_CREAT, 0600);
+    if (fd < 0) throw std::runtime_error("shm open failed: " + path);
+    struct stat st{};
+    if (fstat(fd, &st) != 0) {
+      ::close(fd);
+      throw std::runtime_error("shm fstat failed");
+    }
+    bool init = false;
+    if ((size_t)st.st_size < kHdrBytes) {
+      if (ftruncate(fd, kHdrBytes + sz) != 0) {
+        ::close(fd);
+        throw std::runtime_error("shm ftruncate failed");
+      }
+      init = true;
+    } else {
+      sz = (uint64_t)st.st_size - kHdrBytes;
+    }
+    void* m = mmap(nullptr, kHdrBytes + sz, PROT_READ | PROT_WRITE,
+                   MAP_SHARED, fd, 0);
+    ::close(fd);
+    if (m == MAP_FAILED) throw std::runtime_error("shm mmap failed");
+    map_ = (uint8_t*)m;
+    map_bytes_ = kHdrBytes + sz;
+    hdr_ = (RingHdr*)map_;
+    data_ = map_ + kHdrBytes;
+    if (init) {
+      hdr_->size = sz;
+      hdr_->head.store(0);
+      hdr_->tail.store(0);
+      hdr_->wlock.store(0);
+      hdr_->magic.store(kMagic);  // last: attachers wait on it
+    } else {
+      for (int i = 0; i < 20000; ++i) {  // ~2 s for a racing creator
+        if (hdr_->magic.load() == kMagic) break;
+        std::this_thread::sleep_for(std::chrono::microseconds(100));
+      }
+      if (hdr_->magic.load() != kMagic)
+        throw std::runtime_error("shm ring never initialized: " + path);
+    }
+    size_ = hdr_->size;
+    mask_ = size_ - 1;
+    (void)create;
+  }
+
+  ~ShmRing() {
+    if (map_) munmap(map_, map_bytes_);
+  }
+
+  int64_t write_frames(const std::vector<py::bytes>& frames) {
+    std::vector<std::pair<const uint8_t*, size_t>> raw(frames.size());
+    for (size_t i = 0; i < frames.size(); ++i) {
+      char* p;
+      Py_ssize_t n;
+      PyBytes_AsStringAndSize(frames[i].ptr(), &p, &n);
+      raw[i] = {(const uint8_t*)p, (size_t)n};
+    }
+    int64_t accepted = 0;
+    {
+      py::gil_scoped_release release;
+      // bounded spinlock (multi-producer fan-in)
+      uint32_t expect = 0;
+      int spins = 0;
+      while (!hdr_->wlock.compare_exchange_weak(expect, 1)) {
+        expect = 0;
+        if (++spins > 200000) return 0;  // holder died: drop, don't hang
+        if ((spins & 1023) == 0) std::this_thread::yield();
+      }
+      uint64_t head = hdr_->head.load(std::memory_order_relaxed);
+      const uint64_t tail = hdr_->tail.load(std::memory_order_acquire);
+      for (auto& f : raw) {
+        const uint64_t need = 4 + f.second;
+        if (need > size_) break;  // oversize for this ring: drop
+        if (size_ - (head - tail) < need) break;  // full: drop the rest
+        const uint32_t len32 = (uint32_t)f.second;
+        put_bytes(head, (const uint8_t*)&len32, 4);  // native-endian u32
+        put_bytes(head + 4, f.first, f.second);
+        head += need;
+        ++accepted;
+      }
+      hdr_->head.store(head, std::memory_order_release);
+      hdr_->wlock.store(0, std::memory_order_release);
+    }
+    return accepted;
+  }
+
+  // frames currently readable (diagnostics)
+  int64_t pending() const {
+    return (int64_t)(hdr_->head.load() - hdr_->tail.load());
+  }
+
+  py::list read_batch(int max_frames, int timeout_ms) {
+    std::vector<std::string> out;
+    {
+      py::gil_scoped_release release;
+      wait_data(timeout_ms);
+      uint64_t tail = hdr_->tail.load(std::memory_order_relaxed);
+      const uint64_t head = hdr_->head.load(std::memory_order_acquire);
+      while ((int)out.size() < max_frames && head - tail >= 4) {
+        uint32_t len = 0;
+        get_bytes(tail, (uint8_t*)&len, 4);
+        if (len > kMaxFrame || head - tail < 4 + (uint64_t)len) break;
+        out.emplace_back();
+        out.back().resize(len);
+        get_bytes(tail + 4, (uint8_t*)out.back().data(), len);
+        tail += 4 + len;
+      }
+      hdr_->tail.store(tail, std::memory_order_release);
+    }
+    py::list res;
+    for (auto& s : out) res.append(py::bytes(s));
+    return res;
+  }
+
+  // ring -> packed LogSchema tensors (same contract as
+  // FdFrameReader.read_batch_packed): proto decode IN PLACE from the
+  // mapped ring; no kernel copies, no Python objects per frame.
+  py::tuple read_batch_packed(int max_frames, int timeout_ms, int max_len,
+                              bool pin) {
+    std::vector<std::pair<const uint8_t*, size_t>> raw;
+    std::vector<std::string> wrapped;  // frames crossing the boundary
+    uint64_t tail0, tail;
+    {
+      py::gil_scoped_release release;
+      wait_data(timeout_ms);
+      tail0 = tail = hdr_->tail.load(std::memory_order_relaxed);
+      const uint64_t head = hdr_->head.load(std::memory_order_acquire);
+      while ((int)raw.size() < max_frames && head - tail >= 4) {
+        uint32_t len = 0;
+        get_bytes(tail, (uint8_t*)&len, 4);
+        if (len > kMaxFrame || head - tail < 4 + (uint64_t)len) break;
+        const uint64_t off = (tail + 4) & mask_;
+        if (off + len <= size_) {
+          raw.emplace_back(data_ + off, (size_t)len);
+        } else {  // wraps: copy to scratch (rare: ~1 per ring lap)
+          wrapped.emplace_back();
+          wrapped.back().resize(len);
+          get_bytes(tail + 4, (uint8_t*)wrapped.back().data(), len);
+          raw.emplace_back((const uint8_t*)wrapped.back().data(), (size_t)len);
+        }
+        tail += 4 + len;
+      }
+    }
+    const int64_t B = (int64_t)raw.size();
+    if (B == 0) {
+      auto opts = torch::TensorOptions().dtype(torch::kUInt8);
+      return py::make_tuple(torch::zeros({0, max_len}, opts),
+                            torch::zeros({0}, torch::kInt32), py::bytes(""),
+                            torch::zeros({1}, torch::kInt32), 0);
+    }
+    auto lopts = torch::TensorOptions().dtype(torch::kUInt8);
+    torch::Tensor lines;
+    if (pin) {  // constant-size pinned alloc (see frame_reader.cpp)
+      lopts = lopts.pinned_memory(true);
+      lines = torch::empty({(int64_t)max_frames, max_len}, lopts)
+                  .narrow(0, 0, B);
+    } else {
+      lines = torch::empty({B, max_len}, lopts);
+    }
+    auto lens = torch::zeros({B}, torch::kInt32);
+    auto ids_off = torch::zeros({B + 1}, torch::kInt32);
+    std::vector<dmx_proto::LogSpan> spans(B);
+    std::string blob;
+    {
+      py::gil_scoped_release release;
+      dmx_proto::decode_log_core(raw, max_len, lines.data_ptr<uint8_t>(),
+                                 lens.data_ptr<int32_t>(), spans);
+      int32_t* off = ids_off.data_ptr<int32_t>();
+      size_t total = 0;
+      for (int64_t i = 0; i < B; ++i) total += spans[i].id_len;
+      blob.reserve(total);
+      for (int64_t i = 0; i < B; ++i) {
+        off[i] = (int32_t)blob.size();
+        blob.append(spans[i].id, spans[i].id_len);
+      }
+      off[B] = (int32_t)blob.size();
+      // release the ring space only AFTER the decode read everything
+      hdr_->tail.store(tail, std::memory_order_release);
+    }
+    return py::make_tuple(lines, lens, py::bytes(blob), ids_off,
+                          (int64_t)(tail - tail0));
+  }
+
+ private:
+  void wait_data(int timeout_ms) {  // GIL must be released by caller
+    const auto deadline = std::chrono::steady_clock::now() +
+                          std::chrono::milliseconds(timeout_ms);
+    while (hdr_->head.load(std::memory_order_acquire) ==
+           hdr_->tail.load(std::memory_order_relaxed)) {
+      if (std::chrono::steady_clock::now() >= deadline) return;
+      std::this_thread::sleep_for(std::chrono::microseconds(50));
+    }
+  }
+
+  void put_bytes(uint64_t pos, const uint8_t* src, size_t n) {
+    const uint64_t off = pos & mask_;
+    const size_t first = std::min<size_t>(n, size_ - off);
+    std::memcpy(data_ + off, src, first);
+    if (first < n) std::memcpy(data_, src + first, n - first);
+  }
+
+  void get_bytes(uint64_t pos, uint8_t* dst, size_t n) const {
+    const uint64_t off = pos & mask_;
+    const size_t first = std::min<size_t>(n, size_ - off);
+    std::memcpy(dst, data_ + off, first);
+    if (first < n) std::memcpy(dst, data_, n - first);
+  }
+
+  std::string path_;
+  uint8_t* map_ = nullptr;
+  size_t map_bytes_ = 0;
+  RingHdr* hdr_ = nullptr;
+  uint8_t* data_ = nullptr;
+  uint64_t size_ = 0, mask_ = 0;
+};
+
+}  // namespace
+
+void register_shm_ring(py::module_& m) {
+  py::class_<ShmRing>(m, "ShmRing")
+      .def(py::init<const std::string&, int64_t, bool>(), py::arg("path"),
+           py::arg("size_bytes") = 16 << 20, py::arg("create") = false)
+      .def("write_frames", &ShmRing::write_frames)
+      .def("read_batch", &ShmRing::read_batch, py::arg("max_frames") = 4096,
+           py::arg("timeout_ms") = 100)
+      .def("read_batch_packed", &ShmRing::read_batch_packed,
+           py::arg("max_frames") = 4096, py::arg("timeout_ms") = 100,
+           py::arg("max_len") = 256, py::arg("pin") = false)
+      .def("pending", &ShmRing::pending);
+}

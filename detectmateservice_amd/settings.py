@@ -49,7 +49,7 @@ ENV_NESTED_DELIMITER = "__"
 #: Wire framings: tcp/tls+tcp speak the NNG SP mapping, ws speaks
 #: RFC 6455, ipc/inproc use the compact intra-node framing — see
 #: engine/sockets.py.
-_ADDR_SCHEMES = ("ipc", "tcp", "tls+tcp", "ws", "inproc")
+_ADDR_SCHEMES = ("ipc", "tcp", "tls+tcp", "ws", "inproc", "shm")
 _ADDR_RE = re.compile(r"^(?P<scheme>[a-z+]+)://(?P<rest>.+)$")
 
 

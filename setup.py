@@ -20,6 +20,7 @@ sources = [
     str(CSRC / "bindings.cpp"),
     str(CSRC / "codec.cpp"),
     str(CSRC / "frame_reader.cpp"),
+    str(CSRC / "shm_ring.cpp"),
     str(CSRC / "gemm_bf16.hip"),
     str(CSRC / "layernorm.hip"),
     str(CSRC / "attention.hip"),

@@ -307,3 +307,73 @@ def test_packed_engine_loop_end_to_end(tmp_path):
         sink.close()
         proc.terminate()
         proc.wait(timeout=10)
+
+
+def test_packed_engine_loop_over_shm(tmp_path):
+    """Full service over shm:// rings: zero-copy frames in, alerts out."""
+    import subprocess
+    import sys
+    import time as time_mod
+    import uuid as uuid_mod
+
+    import yaml as yaml_mod
+
+    from detectmateservice_amd import ops
+    from detectmateservice_amd.engine.sockets import ShmDialer, ShmListener
+    from detectmateservice_amd.schemas import DetectorSchema, LogSchema
+    from detectmateservice_amd.utils.synthetic import (
+        AUDIT_LOG_FORMAT,
+        AUDIT_TEMPLATES,
+        AuditLogGenerator,
+    )
+
+    if not ops.have_extension():
+        pytest.skip("extension not built")
+    uid = uuid_mod.uuid4().hex[:8]
+    fused_in = f"shm:///dmx-svc-{uid}"
+    sink_addr = f"shm:///dmx-snk-{uid}"
+    cfg = tmp_path / "fc.yaml"
+    cfg.write_text(yaml_mod.safe_dump({"detectors": {"FusedPipelineDetector": {
+        "templates": list(AUDIT_TEMPLATES),
+        "log_format": AUDIT_LOG_FORMAT,
+        "watches": [{"kind": "header", "pos": 0}],
+        "use_transformer": False,
+        "data_use_training": 64,
+        "device": "cpu",
+    }}}))
+    settings = tmp_path / "fs.yaml"
+    settings.write_text(yaml_mod.safe_dump({
+        "component_type": "FusedPipelineDetector",
+        "engine_addr": fused_in,
+        "out_addr": [sink_addr],
+        "http_enabled": False,
+        "engine_packed_mode": True,
+        "config_file": str(cfg),
+        "log_dir": str(tmp_path / "logs"),
+    }))
+    sink = ShmListener(sink_addr)
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "detectmateservice_amd.cli", "--settings",
+         str(settings)],
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL,
+    )
+    feeder = ShmDialer(fused_in)
+    try:
+        gen = AuditLogGenerator(seed=9, anomaly_rate=0.0)
+        train = [LogSchema(logID=f"t{i}", log=gen.line()[0]).serialize()
+                 for i in range(64)]
+        assert feeder.send_many(train, block=True) == 64
+        time_mod.sleep(1.0)
+        bad = ("type=ZZZ_SHM msg=audit(1.0:1): pid=1 uid=0 auid=1 ses=1 "
+               "msg='op=PAM:x acct=\"x\" exe=/bin/x hostname=? addr=? "
+               "terminal=x res=success'")
+        assert feeder.send(LogSchema(logID="shm-bad", log=bad).serialize(),
+                           block=True)
+        alert = DetectorSchema.deserialize(sink.recv(timeout_ms=60000))
+        assert alert.logIDs == ["shm-bad"]
+        assert "unknown watched value" in alert.description
+    finally:
+        feeder.close()
+        sink.close()
+        proc.terminate()
+        proc.wait(timeout=10)

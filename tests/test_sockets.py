@@ -410,3 +410,100 @@ def test_recv_packed_merges_chunks(tmp_path):
     finally:
         dialer.close()
         listener.close()
+
+
+def test_shm_ring_roundtrip(tmp_path, monkeypatch):
+    """shm:// transport: frames + replies through the mapped ring pair."""
+    import uuid as uuid_mod
+
+    from detectmateservice_amd import ops
+
+    if not ops.have_extension():
+        pytest.skip("extension not built")
+    from detectmateservice_amd.engine.sockets import ShmDialer, ShmListener
+
+    monkeypatch.setenv("DMX_SHM_RING_BYTES", str(1 << 20))
+    addr = f"shm:///dmx-test-{uuid_mod.uuid4().hex[:8]}"
+    listener = ShmListener(addr)
+    dialer = ShmDialer(addr)
+    try:
+        assert dialer.wait_connected(1.0)
+        assert dialer.send(b"hello")
+        assert listener.recv(timeout_ms=2000) == b"hello"
+        assert listener.send(b"world")
+        assert dialer.recv(timeout_ms=2000) == b"world"
+        # batch + recv_many
+        frames = [f"f{i}".encode() for i in range(500)]
+        assert dialer.send_many(frames, block=True) == 500
+        got = []
+        while len(got) < 500:
+            got.extend(listener.recv_many(4096, 2000, linger_ms=5.0))
+        assert got == frames
+    finally:
+        dialer.close()
+        listener.close()
+    import os as os_mod
+
+    assert not os_mod.path.exists(f"/dev/shm/dmx-test-{addr[-8:]}.c2s".replace(addr[-8:], addr.rsplit('-', 1)[1]))
+
+
+def test_shm_late_binding_keeps_buffered_frames(monkeypatch):
+    """Dialer starts FIRST; its buffered frames survive until the
+    listener attaches (socket late-binding parity)."""
+    import uuid as uuid_mod
+
+    from detectmateservice_amd import ops
+
+    if not ops.have_extension():
+        pytest.skip("extension not built")
+    from detectmateservice_amd.engine.sockets import ShmDialer, ShmListener
+
+    monkeypatch.setenv("DMX_SHM_RING_BYTES", str(1 << 20))
+    addr = f"shm:///dmx-late-{uuid_mod.uuid4().hex[:8]}"
+    dialer = ShmDialer(addr)
+    listener = None
+    try:
+        for i in range(10):
+            assert dialer.send(f"early{i}".encode())
+        listener = ShmListener(addr)
+        got = listener.recv_many(64, 2000, linger_ms=5.0)
+        assert got == [f"early{i}".encode() for i in range(10)]
+    finally:
+        dialer.close()
+        if listener is not None:
+            listener.close()
+
+
+def test_shm_packed_recv(monkeypatch):
+    """Packed path: LogSchema frames decode in place from the ring."""
+    import uuid as uuid_mod
+
+    from detectmateservice_amd import ops
+
+    if not ops.have_extension():
+        pytest.skip("extension not built")
+    from detectmateservice_amd.engine.sockets import ShmDialer, ShmListener
+    from detectmateservice_amd.schemas import LogSchema
+
+    monkeypatch.setenv("DMX_SHM_RING_BYTES", str(1 << 20))
+    addr = f"shm:///dmx-pk-{uuid_mod.uuid4().hex[:8]}"
+    listener = ShmListener(addr)
+    dialer = ShmDialer(addr)
+    try:
+        assert listener.enable_packed(64, False)
+        frames = [LogSchema(logID=f"id{i}", log=f"line {i}").serialize()
+                  for i in range(200)]
+        assert dialer.send_many(frames, block=True) == 200
+        got = 0
+        while got < 200:
+            _c, lines, lens, blob, off, nb = listener.recv_packed(
+                2000, max_frames=4096)
+            for i in range(lines.shape[0]):
+                lid = blob[int(off[i]):int(off[i + 1])].decode()
+                text = bytes(lines[i, : int(lens[i])].numpy().tobytes()).decode()
+                assert text == f"line {lid[2:]}"
+            got += lines.shape[0]
+            assert nb > 0
+    finally:
+        dialer.close()
+        listener.close()

@@ -47,6 +47,10 @@ def main():
                          "parser+detector chain")
     ap.add_argument("--no-transformer", action="store_true",
                     help="(--fused) skip BERT-tiny scoring, hash-only")
+    ap.add_argument("--shm", action="store_true",
+                    help="shared-memory ring transport (shm://) instead of "
+                         "ipc sockets — zero kernel copies between "
+                         "co-located stages")
     ap.add_argument("--prepack", action="store_true",
                     help="feeder pre-packs all frames into wire blobs and "
                          "sendalls them from a raw socket (removes the "
@@ -63,9 +67,14 @@ def main():
 
     tmp = tempfile.mkdtemp(prefix="dmx-bench-")
     uid = uuid.uuid4().hex[:6]
-    parser_in = f"ipc://{tmp}/parser-{uid}.ipc"
-    detector_in = f"ipc://{tmp}/det-{uid}.ipc"
-    sink_addr = f"ipc://{tmp}/sink-{uid}.ipc"
+    if args.shm:
+        parser_in = f"shm:///dmx-b-{uid}-in"
+        detector_in = f"shm:///dmx-b-{uid}-det"
+        sink_addr = f"shm:///dmx-b-{uid}-out"
+    else:
+        parser_in = f"ipc://{tmp}/parser-{uid}.ipc"
+        detector_in = f"ipc://{tmp}/det-{uid}.ipc"
+        sink_addr = f"ipc://{tmp}/sink-{uid}.ipc"
 
     tpl = os.path.join(tmp, "templates.txt")
     with open(tpl, "w") as fh:
@@ -147,9 +156,15 @@ def main():
     ]
     if stats_on:
         print(f"# service stdout: {tmp}/svc_stdout.log", file=sys.stderr)
-    sink = PairListener(sink_addr)
-    feeders = [PairDialer(parser_in, buffer_size=8192)
-               for _ in range(max(1, args.feeders))]
+    if args.shm:
+        from detectmateservice_amd.engine.sockets import ShmDialer, ShmListener
+
+        sink = ShmListener(sink_addr)
+        feeders = [ShmDialer(parser_in)]
+    else:
+        sink = PairListener(sink_addr)
+        feeders = [PairDialer(parser_in, buffer_size=8192)
+                   for _ in range(max(1, args.feeders))]
     feeder = feeders[0]
     try:
         # generous: on a fresh GPU box the first torch/hip init inside the
@@ -179,6 +194,8 @@ def main():
 
         prepacked = None
         raw_sock = None
+        if args.prepack and args.shm:
+            raise SystemExit("--prepack applies to the socket transport")
         if args.prepack:
             import socket as s_mod
 
@@ -273,7 +290,8 @@ def main():
             "dtype": "bf16",
             "data": "synthetic",
             "config": {
-                "mode": ("fused single-service over ipc"
+                "mode": (("fused single-service over shm" if args.shm else
+                          "fused single-service over ipc")
                          if args.fused else
                          "3-stage service processes over ipc (SP-framed engine sockets)"),
                 "pipeline": ("feeder->FusedPipelineDetector svc->sink"
