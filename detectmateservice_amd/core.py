@@ -146,6 +146,14 @@ class Service:
             lines, lens, ids_blob, ids_off
         )
 
+    @property
+    def submit_packed_frames(self):
+        return getattr(self.library_component, "submit_packed_frames", None)
+
+    @property
+    def collect_packed_frames(self):
+        return getattr(self.library_component, "collect_packed_frames", None)
+
     def source_batches(self, batch_size: int, stop_event):
         """Source-mode delegation (engine_source_mode: reader services)."""
         if self.library_component is None or not hasattr(

@@ -274,11 +274,55 @@ class Engine:
                 "(TLS/ws/inproc or extension missing); using the frame loop"
             )
             return False
-        self._log.info("engine PACKED loop started on %s (max_len=%d pin=%s)",
-                       s.engine_addr, max_len, pin)
+        # pipelined mode: overlap batch N's async GPU work with batch
+        # N+1's recv+decode (submit launches without syncing; collect
+        # does the one readback). Falls back to the synchronous call
+        # when the component lacks the split API.
+        submit = getattr(self.processor, "submit_packed_frames", None)
+        collect = getattr(self.processor, "collect_packed_frames", None)
+        pipelined = submit is not None and collect is not None
+        self._log.info(
+            "engine PACKED loop started on %s (max_len=%d pin=%s pipelined=%s)",
+            s.engine_addr, max_len, pin, pipelined)
         stats_on = os.environ.get("DMX_ENGINE_STATS") == "1"
         st = {"n": 0, "frames": 0, "recv": 0.0, "proc": 0.0,
               "last": time.perf_counter()}
+        prev = None          # in-flight token
+        prev_meta = None     # (B, nbytes, t_submit)
+
+        def emit(alerts):
+            for _idx, out in alerts:
+                if self._out_socks:
+                    self._send_to_outputs(out)
+                elif self._pair_sock.send(out, block=False):
+                    m.data_written_bytes_total.inc(len(out))
+                    m.data_written_lines_total.inc(1)
+                else:
+                    m.data_dropped_bytes_total.inc(len(out))
+                    m.data_dropped_lines_total.inc(1)
+
+        def drain_prev():
+            nonlocal prev, prev_meta
+            if prev is None:
+                return
+            pB, pbytes, t_sub = prev_meta
+            try:
+                emit(collect(prev))
+            except Exception as exc:  # noqa: BLE001
+                m.processing_errors_total.inc(pB)
+                self._log.error("processing error on packed batch of %d: %s",
+                                pB, exc)
+                prev, prev_meta = None, None
+                return
+            m.data_processed_bytes_total.inc(pbytes)
+            m.data_processed_lines_total.inc(pB)
+            m.observe_batch(time.perf_counter() - t_sub, pB)
+            if stats_on:
+                st["n"] += 1
+                st["frames"] += pB
+                st["proc"] += time.perf_counter() - t_sub
+            prev, prev_meta = None, None
+
         while not self._stop_event.is_set():
             try:
                 t_r0 = time.perf_counter()
@@ -291,6 +335,7 @@ class Engine:
                 if stats_on:
                     st["recv"] += time.perf_counter() - t_r0
             except RecvTimeout:
+                drain_prev()  # idle: finish the in-flight batch
                 continue
             except SocketClosed:
                 break
@@ -301,35 +346,40 @@ class Engine:
                 continue
             B = int(lines.shape[0])
             if B == 0:
+                drain_prev()
                 continue
             m.data_read_bytes_total.inc(nbytes)
             m.data_read_lines_total.inc(B)
             m.engine_batch_size.observe(B)
             t0 = time.perf_counter()
-            try:
-                alerts = proc(lines, lens, blob, off)
-            except Exception as exc:  # noqa: BLE001 - loop must survive
-                m.processing_errors_total.inc(B)
-                self._log.error("processing error on packed batch of %d: %s",
-                                B, exc)
-                continue
-            elapsed = time.perf_counter() - t0
-            m.data_processed_bytes_total.inc(nbytes)
-            m.data_processed_lines_total.inc(B)
-            m.observe_batch(elapsed, B)
-            for _idx, out in alerts:
-                if self._out_socks:
-                    self._send_to_outputs(out)
-                elif self._pair_sock.send(out, block=False):
-                    m.data_written_bytes_total.inc(len(out))
-                    m.data_written_lines_total.inc(1)
-                else:
-                    m.data_dropped_bytes_total.inc(len(out))
-                    m.data_dropped_lines_total.inc(1)
+            if pipelined:
+                drain_prev()  # collect batch N (GPU overlapped our recv)
+                try:
+                    prev = submit(lines, lens, blob, off)
+                    prev_meta = (B, nbytes, t0)
+                except Exception as exc:  # noqa: BLE001
+                    m.processing_errors_total.inc(B)
+                    self._log.error(
+                        "processing error on packed batch of %d: %s", B, exc)
+                    prev, prev_meta = None, None
+            else:
+                try:
+                    alerts = proc(lines, lens, blob, off)
+                except Exception as exc:  # noqa: BLE001 - loop must survive
+                    m.processing_errors_total.inc(B)
+                    self._log.error(
+                        "processing error on packed batch of %d: %s", B, exc)
+                    continue
+                elapsed = time.perf_counter() - t0
+                m.data_processed_bytes_total.inc(nbytes)
+                m.data_processed_lines_total.inc(B)
+                m.observe_batch(elapsed, B)
+                emit(alerts)
+                if stats_on:
+                    st["n"] += 1
+                    st["frames"] += B
+                    st["proc"] += elapsed
             if stats_on:
-                st["n"] += 1
-                st["frames"] += B
-                st["proc"] += elapsed
                 now = time.perf_counter()
                 if now - st["last"] > 2.0 and st["n"]:
                     self._log.info(
@@ -339,6 +389,7 @@ class Engine:
                         st["recv"] * 1e3 / st["n"], st["proc"] * 1e3 / st["n"],
                     )
                     st.update(n=0, frames=0, recv=0.0, proc=0.0, last=now)
+        drain_prev()
         self._log.info("engine packed loop exited")
         return True
 

@@ -145,6 +145,35 @@ class FusedPipelineDetector(CoreComponent):
                 "utf-8", "replace"),
         )
 
+    # -- pipelined packed path (submit/collect) ------------------------
+    # The engine overlaps batch N's GPU work with batch N+1's recv+decode:
+    # submit launches the pipeline WITHOUT synchronizing; collect performs
+    # the one device sync (anomaly readback) and builds the alerts.
+    def submit_packed_frames(self, lines, lens, ids_blob, ids_off):
+        if lines.shape[0] == 0:
+            return None
+        out = self._score_packed(lines, lens)  # async: kernels queued
+        if self.pipe._graph is not None:
+            # graph replay writes into STATIC buffers: clone (async) so
+            # the next submit's replay can't clobber this batch's outputs
+            out = {
+                k: (v.clone() if torch.is_tensor(v) else v)
+                for k, v in out.items() if k != "match"
+            }
+        # keep `lines` alive until the async H2D copy completes (the
+        # pinned block would otherwise return to the cache mid-copy)
+        return (out, ids_blob, ids_off, lines)
+
+    def collect_packed_frames(self, token) -> List:
+        if token is None:
+            return []
+        out, ids_blob, ids_off, _lines = token
+        return self._alerts(
+            out,
+            lambda i: ids_blob[int(ids_off[i]):int(ids_off[i + 1])].decode(
+                "utf-8", "replace"),
+        )
+
     def process_batch(self, frames: List[bytes]) -> List[Optional[bytes]]:
         if not frames:
             return []
