@@ -181,7 +181,11 @@ def main():
             ]
 
         warmup = make_frames(args.warmup_lines, "w")
-        frames = make_frames(args.lines, "m")
+        # bounded pool cycled to --lines (pre-building tens of millions of
+        # Python frames costs minutes and GBs; logID uniqueness only
+        # matters for the sentinel)
+        pool_n = min(args.lines, 1_000_000)
+        frames = make_frames(pool_n, "m")
 
         def pump_one(f, batch):
             sent = 0
@@ -242,7 +246,11 @@ def main():
         # detector input: we time until the parser has accepted all
         # frames AND the pipeline is idle (sink quiet + sockets drained).
         t0 = time.perf_counter()
-        pump(frames)
+        sent_total = 0
+        while sent_total < args.lines:
+            todo = min(args.lines - sent_total, pool_n)
+            pump(frames[:todo])
+            sent_total += todo
         pump_elapsed = time.perf_counter() - t0
         # wait until pipeline is idle: detector emits nothing for clean
         # traffic; send one marked anomalous line and wait for its alert
