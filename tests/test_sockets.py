@@ -544,3 +544,49 @@ def test_shm_ring_multi_producer(monkeypatch):
         assert a == list(range(n)) and b == list(range(n))
     finally:
         d1.close(); d2.close(); listener.close()
+
+
+def test_shm_dialer_reattaches_after_listener_restart(monkeypatch):
+    """The listener owns the ring files: when it restarts (new inode),
+    a dialer detects the recreated ring on persistent-full and remaps
+    (socket background-reconnect parity)."""
+    import uuid as uuid_mod
+
+    from detectmateservice_amd import ops
+
+    if not ops.have_extension():
+        pytest.skip("extension not built")
+    from detectmateservice_amd.engine.sockets import ShmDialer, ShmListener
+
+    monkeypatch.setenv("DMX_SHM_RING_BYTES", str(64 << 10))  # tiny ring
+    addr = f"shm:///dmx-re-{uuid_mod.uuid4().hex[:8]}"
+    l1 = ShmListener(addr)
+    dialer = ShmDialer(addr)
+    dialer.REATTACH_AFTER_FULL = 5
+    try:
+        assert dialer.send(b"one")
+        assert l1.recv(timeout_ms=2000) == b"one"
+        l1.close()  # unlinks the ring files
+
+        # fill the (orphaned) mapping until the dialer gives up...
+        filler = [b"x" * 1000] * 200
+        for _ in range(20):
+            dialer.send_many(filler, block=False)
+        # ...new listener creates a FRESH ring at the same address
+        l2 = ShmListener(addr)
+        try:
+            delivered = False
+            for i in range(200):
+                if dialer.send(f"after-{i}".encode(), block=False):
+                    try:
+                        got = l2.recv_many(4096, 200, linger_ms=5.0)
+                    except Exception:
+                        got = []
+                    if any(g.startswith(b"after-") for g in got):
+                        delivered = True
+                        break
+            assert delivered, "dialer never re-attached to the new ring"
+        finally:
+            l2.close()
+    finally:
+        dialer.close()
