@@ -277,3 +277,74 @@ def test_batch_size_one_strict_per_message(ipc_addr, run_engine):
         assert got == [b"out:s%d" % i for i in range(10)]
     finally:
         client.close()
+
+
+def test_packed_loop_survives_component_errors(tmp_path):
+    """A component whose submit raises must not kill the packed loop;
+    processing_errors_total counts and later batches still process
+    (reference engine.py:233-236 error policy on the native plane)."""
+    import time as time_mod
+
+    from detectmateservice_amd import ops
+
+    if not ops.have_extension():
+        pytest.skip("extension not built")
+    from detectmateservice_amd.engine.engine import Engine
+    from detectmateservice_amd.engine.sockets import PairDialer
+    from detectmateservice_amd.schemas import LogSchema
+    from detectmateservice_amd.settings import ServiceSettings
+
+    calls = {"n": 0}
+    got = []
+
+    class Flaky:
+        def process(self, d):
+            return None
+
+        def process_batch(self, frames):
+            return [None] * len(frames)
+
+        def supports_packed_frames(self):
+            return True
+
+        def packed_max_len(self):
+            return 64
+
+        def packed_pin_memory(self):
+            return False
+
+        def process_packed_frames(self, lines, lens, blob, off):
+            calls["n"] += 1
+            if calls["n"] == 1:
+                raise RuntimeError("boom")
+            got.append(int(lines.shape[0]))
+            return []
+
+    addr = f"ipc://{tmp_path}/pkerr.ipc"
+    settings = ServiceSettings(
+        component_type="core", engine_addr=addr, http_enabled=False,
+        engine_packed_mode=True, engine_batch_size=64,
+        log_dir=tmp_path / "logs",
+    )
+    engine = Engine(settings, processor=Flaky())
+    engine.start()
+    dialer = PairDialer(addr)
+    try:
+        assert dialer.wait_connected(10.0)
+        frames = [LogSchema(logID=str(i), log="x").serialize() for i in range(8)]
+        for f in frames:
+            assert dialer.send(f, block=True)
+        deadline = time_mod.monotonic() + 10.0
+        while calls["n"] < 1 and time_mod.monotonic() < deadline:
+            time_mod.sleep(0.05)
+        # second batch after the error must be processed
+        for f in frames:
+            assert dialer.send(f, block=True)
+        while not got and time_mod.monotonic() < deadline:
+            time_mod.sleep(0.05)
+        assert calls["n"] >= 2 and got, (calls, got)
+        assert engine.metrics.processing_errors_total._value.get() >= 8
+    finally:
+        dialer.close()
+        engine.stop()
+        engine.close()

@@ -111,3 +111,13 @@ def test_config_class_override(tmp_path, monkeypatch):
         assert svc.config_manager.schema.__name__ == "NewValueDetectorConfig"
     finally:
         svc.engine.close()
+
+
+def test_shm_scheme_accepted_tls_rejected():
+    s = ServiceSettings(engine_addr="shm:///dmx-x", out_addr=["shm:///dmx-y"])
+    assert s.engine_addr == "shm:///dmx-x"
+    # shm listener refuses TLS config at socket construction
+    from detectmateservice_amd.engine.sockets import ShmListener
+
+    with pytest.raises(ValueError):
+        ShmListener("shm:///dmx-x", tls_config=object())
