@@ -362,6 +362,13 @@ class Engine:
                     self._log.error(
                         "processing error on packed batch of %d: %s", B, exc)
                     prev, prev_meta = None, None
+                # sparse traffic: if nothing else is already queued,
+                # finish this batch NOW instead of waiting out the recv
+                # timeout (keeps single-frame round-trip latency low;
+                # under load the queue is non-empty and pipelining holds)
+                has_pending = getattr(self._pair_sock, "has_pending", None)
+                if has_pending is None or not has_pending():
+                    drain_prev()
             else:
                 try:
                     alerts = proc(lines, lens, blob, off)

@@ -606,6 +606,9 @@ class PairListener:
         self._pending = list(frames[1:])
         return frames[0]
 
+    def has_pending(self) -> bool:
+        return bool(getattr(self, "_pending", None)) or not self._recv_q.empty()
+
     # -- packed fast path ----------------------------------------------
     def enable_packed(self, max_len: int, pin: bool,
                       max_frames: int = 4096) -> bool:
@@ -1216,6 +1219,9 @@ class ShmListener:
             out.extend(bytes(f) for f in frames)
         return out
 
+    def has_pending(self) -> bool:
+        return bool(getattr(self, "_pending", None)) or not self._recv_q.empty()
+
     # -- packed fast path ----------------------------------------------
     def enable_packed(self, max_len: int, pin: bool,
                       max_frames: int = 4096) -> bool:
@@ -1234,6 +1240,9 @@ class ShmListener:
         if lines.shape[0] == 0:
             raise RecvTimeout(self.addr)
         return None, lines, lens, blob, off, int(nbytes)
+
+    def has_pending(self) -> bool:
+        return bool(self._pending) or self._c2s.pending() > 0
 
     # -- reply path (request/reply compatibility mode) -----------------
     def send(self, data: bytes, block: bool = True) -> bool:
