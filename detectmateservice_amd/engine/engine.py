@@ -288,26 +288,31 @@ class Engine:
         st = {"n": 0, "frames": 0, "recv": 0.0, "proc": 0.0,
               "last": time.perf_counter()}
         prev = None          # in-flight token
-        prev_meta = None     # (B, nbytes, t_submit)
+        prev_meta = None     # (B, nbytes, t_submit, conn)
+        reply_conn = getattr(self._pair_sock, "reply_conn", None)
 
-        def emit(alerts):
+        def emit(alerts, conn=None):
             for _idx, out in alerts:
                 if self._out_socks:
                     self._send_to_outputs(out)
-                elif self._pair_sock.send(out, block=False):
-                    m.data_written_bytes_total.inc(len(out))
-                    m.data_written_lines_total.inc(1)
                 else:
-                    m.data_dropped_bytes_total.inc(len(out))
-                    m.data_dropped_lines_total.inc(1)
+                    # request/reply mode: route to THE BATCH'S sender
+                    ok = (reply_conn(conn, out) if reply_conn is not None
+                          else self._pair_sock.send(out, block=False))
+                    if ok:
+                        m.data_written_bytes_total.inc(len(out))
+                        m.data_written_lines_total.inc(1)
+                    else:
+                        m.data_dropped_bytes_total.inc(len(out))
+                        m.data_dropped_lines_total.inc(1)
 
         def drain_prev():
             nonlocal prev, prev_meta
             if prev is None:
                 return
-            pB, pbytes, t_sub = prev_meta
+            pB, pbytes, t_sub, pconn = prev_meta
             try:
-                emit(collect(prev))
+                emit(collect(prev), pconn)
             except Exception as exc:  # noqa: BLE001
                 m.processing_errors_total.inc(pB)
                 self._log.error("processing error on packed batch of %d: %s",
@@ -356,7 +361,7 @@ class Engine:
                 drain_prev()  # collect batch N (GPU overlapped our recv)
                 try:
                     prev = submit(lines, lens, blob, off)
-                    prev_meta = (B, nbytes, t0)
+                    prev_meta = (B, nbytes, t0, conn)
                 except Exception as exc:  # noqa: BLE001
                     m.processing_errors_total.inc(B)
                     self._log.error(
@@ -381,7 +386,7 @@ class Engine:
                 m.data_processed_bytes_total.inc(nbytes)
                 m.data_processed_lines_total.inc(B)
                 m.observe_batch(elapsed, B)
-                emit(alerts)
+                emit(alerts, conn)
                 if stats_on:
                     st["n"] += 1
                     st["frames"] += B

@@ -756,7 +756,12 @@ class PairListener:
 
     def send(self, data: bytes, block: bool = True) -> bool:
         """Reply to the most recent sender (request/reply fallback mode)."""
-        conn = self._last_sender
+        return self.reply_conn(self._last_sender, data)
+
+    def reply_conn(self, conn, data: bytes) -> bool:
+        """Reply to a SPECIFIC peer (the packed loop routes each batch's
+        replies to that batch's sender; merged multi-peer chunks reply
+        to the first chunk's sender)."""
         if conn is None:
             with self._peers_lock:
                 conn = self._peers[-1] if self._peers else None
