@@ -1253,6 +1253,10 @@ class ShmListener:
     def send(self, data: bytes, block: bool = True) -> bool:
         return self._s2c.write_frames([data]) == 1
 
+    def reply_conn(self, conn, data: bytes) -> bool:
+        """Ring replies are single-channel (s2c); conn is advisory."""
+        return self.send(data)
+
     def reply(self, idx: int, data: bytes) -> bool:
         return self.send(data)
 
