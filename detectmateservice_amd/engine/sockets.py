@@ -10,8 +10,10 @@ same capability surface as the reference's socket layer without NNG:
   ``tls+tcp://host:port`` (stdlib ``ssl``; server cert+key from one PEM as
   the reference's ``cert_key_file``, client CA + SNI as ``ca_file`` /
   ``server_name`` — engine_socket.py:60-73, engine.py:156-170),
-  ``ws://`` (real RFC 6455 WebSocket framing), and ``inproc://``
-  (process-local queue pair).
+  ``ws://`` (real RFC 6455 WebSocket framing), ``inproc://``
+  (process-local queue pair), and ``shm://`` (shared-memory ring under
+  /dev/shm for co-located services — zero kernel copies, in-place
+  packed decode; ops/csrc/shm_ring.cpp).
 * a listener (stage input) that accepts peers and can reply to the sender
   of the last received frame (the reference's request/reply fallback mode,
   engine.py:248-264),
