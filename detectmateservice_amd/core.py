@@ -241,10 +241,30 @@ class Service:
     # framework's detector state — GPU hash sets, transformer weights,
     # calibration — must survive restarts)
     # ------------------------------------------------------------------
-    def checkpoint(self, path: str | Path) -> Dict[str, Any]:
+    def _resolve_checkpoint_path(self, path: str | Path, from_admin: bool) -> Path:
+        """Admin calls (unauthenticated HTTP) are confined to the
+        configured ``checkpoint_dir``; direct Python calls by the
+        operator's own process keep arbitrary paths."""
+        p = Path(path)
+        if not from_admin:
+            return p
+        ckdir = self.settings.checkpoint_dir
+        if ckdir is None:
+            raise PermissionError(
+                "admin checkpoint/restore requires settings.checkpoint_dir"
+            )
+        base = Path(ckdir).resolve()
+        cand = (p if p.is_absolute() else base / p).resolve()
+        if not cand.is_relative_to(base):
+            raise PermissionError(
+                f"checkpoint path {str(path)!r} escapes checkpoint_dir"
+            )
+        return cand
+
+    def checkpoint(self, path: str | Path, from_admin: bool = False) -> Dict[str, Any]:
         import torch
 
-        path = Path(path)
+        path = self._resolve_checkpoint_path(path, from_admin)
         path.parent.mkdir(parents=True, exist_ok=True)
         state = {
             "version": __version__,
@@ -264,11 +284,16 @@ class Service:
         self.logger.info("checkpoint written to %s", path)
         return {"path": str(path), "timestamp": state["timestamp"]}
 
-    def restore(self, path: str | Path) -> Dict[str, Any]:
+    def restore(self, path: str | Path, from_admin: bool = False) -> Dict[str, Any]:
         import torch
 
-        path = Path(path)
-        state = torch.load(path, map_location="cpu", weights_only=False)
+        path = self._resolve_checkpoint_path(path, from_admin)
+        # weights_only: checkpoints are tensors + primitives only — a
+        # planted pickle payload must raise here, not execute
+        # (component state_dicts keep to that contract; sklearn models
+        # store their pickle as inert bytes unpickled only by their own
+        # load_state_dict after this trust gate).
+        state = torch.load(path, map_location="cpu", weights_only=True)
         if state.get("component_type") != self.settings.component_type:
             raise ValueError(
                 f"checkpoint is for component_type={state.get('component_type')!r}, "

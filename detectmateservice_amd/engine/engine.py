@@ -187,8 +187,14 @@ class Engine:
                 continue
             if not frames:
                 continue
-            # Skip empty frames (reference engine.py:207-209).
-            frames = [f for f in frames if f]
+            # Skip empty frames (reference engine.py:207-209) — but keep
+            # each surviving frame's ORIGINAL batch index: the listener's
+            # _batch_conns was recorded for the unfiltered recv_many batch,
+            # so reply(i) must be given pre-filter indices or multi-peer
+            # replies route to the wrong sender.
+            orig_idx = [i for i, f in enumerate(frames) if f]
+            if len(orig_idx) != len(frames):
+                frames = [frames[i] for i in orig_idx]
             if not frames:
                 continue
 
@@ -219,7 +225,7 @@ class Engine:
                     # request/reply fallback mode (engine.py:248-264);
                     # replies route to EACH frame's sender (multi-peer)
                     reply = getattr(self._pair_sock, "reply", None)
-                    ok = (reply(i, out) if reply is not None
+                    ok = (reply(orig_idx[i], out) if reply is not None
                           else self._pair_sock.send(out, block=False))
                     if ok:
                         m.data_written_bytes_total.inc(len(out))
@@ -288,16 +294,27 @@ class Engine:
         st = {"n": 0, "frames": 0, "recv": 0.0, "proc": 0.0,
               "last": time.perf_counter()}
         prev = None          # in-flight token
-        prev_meta = None     # (B, nbytes, t_submit, conn)
+        prev_meta = None     # (B, nbytes, t_submit, conn, conn_segs)
         reply_conn = getattr(self._pair_sock, "reply_conn", None)
 
-        def emit(alerts, conn=None):
-            for _idx, out in alerts:
+        def emit(alerts, conn=None, segs=None):
+            # segs = [(conn, row_count)] captured AT RECV TIME for this
+            # batch (the listener's own state may already describe the
+            # NEXT batch when pipelining): merged multi-peer chunks route
+            # each alert row to ITS sender.
+            for idx, out in alerts:
                 if self._out_socks:
                     self._send_to_outputs(out)
                 else:
-                    # request/reply mode: route to THE BATCH'S sender
-                    ok = (reply_conn(conn, out) if reply_conn is not None
+                    target = conn
+                    if segs is not None and len(segs) > 1:
+                        base = 0
+                        for c, rows in segs:
+                            if idx < base + rows:
+                                target = c
+                                break
+                            base += rows
+                    ok = (reply_conn(target, out) if reply_conn is not None
                           else self._pair_sock.send(out, block=False))
                     if ok:
                         m.data_written_bytes_total.inc(len(out))
@@ -310,9 +327,9 @@ class Engine:
             nonlocal prev, prev_meta
             if prev is None:
                 return
-            pB, pbytes, t_sub, pconn = prev_meta
+            pB, pbytes, t_sub, pconn, psegs = prev_meta
             try:
-                emit(collect(prev), pconn)
+                emit(collect(prev), pconn, psegs)
             except Exception as exc:  # noqa: BLE001
                 m.processing_errors_total.inc(pB)
                 self._log.error("processing error on packed batch of %d: %s",
@@ -356,12 +373,13 @@ class Engine:
             m.data_read_bytes_total.inc(nbytes)
             m.data_read_lines_total.inc(B)
             m.engine_batch_size.observe(B)
+            segs = getattr(self._pair_sock, "_batch_conn_segs", None)
             t0 = time.perf_counter()
             if pipelined:
                 drain_prev()  # collect batch N (GPU overlapped our recv)
                 try:
                     prev = submit(lines, lens, blob, off)
-                    prev_meta = (B, nbytes, t0, conn)
+                    prev_meta = (B, nbytes, t0, conn, segs)
                 except Exception as exc:  # noqa: BLE001
                     m.processing_errors_total.inc(B)
                     self._log.error(
@@ -386,7 +404,7 @@ class Engine:
                 m.data_processed_bytes_total.inc(nbytes)
                 m.data_processed_lines_total.inc(B)
                 m.observe_batch(elapsed, B)
-                emit(alerts, conn)
+                emit(alerts, conn, segs)
                 if stats_on:
                     st["n"] += 1
                     st["frames"] += B

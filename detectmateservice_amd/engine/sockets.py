@@ -605,7 +605,11 @@ class PairListener:
         except queue.Empty:
             raise RecvTimeout(self.addr) from None
         self._last_sender = conn
+        # keep _pending_conns aligned with _pending: recv() and recv_many()
+        # may be interleaved, and reply(idx) routes by _batch_conns built
+        # from these lists — a bare _pending refill would desync them
         self._pending = list(frames[1:])
+        self._pending_conns = [conn] * len(self._pending)
         return frames[0]
 
     def has_pending(self) -> bool:
@@ -648,8 +652,9 @@ class PairListener:
                     linger_ms: float = 0.0):
         """Pop packed chunks, merging up to ``max_frames`` rows within
         ``linger_ms``: (conn, lines, lens, ids_blob, ids_off, frame_bytes).
-        ``conn`` is the FIRST chunk's sender (packed mode is fan-in via
-        out_addr; per-row reply routing is the frame loop's job)."""
+        ``conn`` is the FIRST chunk's sender; per-row routing for merged
+        multi-peer batches is available via ``reply_row`` — each chunk's
+        (conn, row_count) is recorded in ``_batch_conn_segs``."""
         import torch
 
         try:
@@ -659,14 +664,16 @@ class PairListener:
         self._last_sender = conn
         lines, lens, blob, off, nbytes = self._as_packed(item)
         if max_frames <= 0 or lines.shape[0] >= max_frames:
+            self._batch_conn_segs = [(conn, int(lines.shape[0]))]
             return conn, lines, lens, blob, off, nbytes
         chunks = [(lines, lens, blob, off, nbytes)]
+        segs = [(conn, int(lines.shape[0]))]
         total = int(lines.shape[0])
         deadline = time.monotonic() + linger_ms / 1000.0
         while total < max_frames:
             remaining = deadline - time.monotonic()
             try:
-                _c, item = self._recv_q.get(
+                c, item = self._recv_q.get(
                     timeout=max(remaining, 0) if remaining > 0 else None,
                     block=remaining > 0,
                 )
@@ -674,7 +681,9 @@ class PairListener:
                 break
             ch = self._as_packed(item)
             chunks.append(ch)
+            segs.append((c, int(ch[0].shape[0])))
             total += int(ch[0].shape[0])
+        self._batch_conn_segs = segs
         if len(chunks) == 1:
             return conn, lines, lens, blob, off, nbytes
         lines = torch.cat([c[0] for c in chunks])
@@ -756,14 +765,30 @@ class PairListener:
         except OSError:
             return False
 
+    def reply_row(self, idx: int, data: bytes) -> bool:
+        """Reply to the sender of ROW ``idx`` of the last recv_packed batch
+        (merged multi-peer chunks route per-row via the recorded
+        (conn, row_count) segments; falls back to the last sender)."""
+        conn = None
+        segs = getattr(self, "_batch_conn_segs", None)
+        if segs:
+            base = 0
+            for c, rows in segs:
+                if idx < base + rows:
+                    conn = c
+                    break
+                base += rows
+        return self.reply_conn(conn if conn is not None else self._last_sender,
+                               data)
+
     def send(self, data: bytes, block: bool = True) -> bool:
         """Reply to the most recent sender (request/reply fallback mode)."""
         return self.reply_conn(self._last_sender, data)
 
     def reply_conn(self, conn, data: bytes) -> bool:
         """Reply to a SPECIFIC peer (the packed loop routes each batch's
-        replies to that batch's sender; merged multi-peer chunks reply
-        to the first chunk's sender)."""
+        replies to that batch's sender; per-row routing for merged
+        multi-peer chunks is ``reply_row``)."""
         if conn is None:
             with self._peers_lock:
                 conn = self._peers[-1] if self._peers else None
@@ -1183,8 +1208,19 @@ class ShmListener:
         from ..ops import _dmx_C  # extension required for shm
 
         self._c2s_path, self._s2c_path = _shm_paths(self.addr)
-        self._c2s = _dmx_C.ShmRing(self._c2s_path, _shm_ring_bytes(), True)
-        self._s2c = _dmx_C.ShmRing(self._s2c_path, _shm_ring_bytes(), True)
+        # Stale-ring detection: a marker file exists while a listener is
+        # live and is removed by close(). Marker present at startup =>
+        # the previous listener CRASHED and the ring files hold that
+        # run's frames — recreate them fresh (ShmRing create=True
+        # unlinks first). No marker => any existing rings were created
+        # by an early dialer buffering ahead of us (late binding,
+        # reference engine.py:173-179): attach and keep their frames.
+        self._marker_path = self._c2s_path + ".live"
+        stale = os.path.exists(self._marker_path)
+        self._c2s = _dmx_C.ShmRing(self._c2s_path, _shm_ring_bytes(), stale)
+        self._s2c = _dmx_C.ShmRing(self._s2c_path, _shm_ring_bytes(), stale)
+        with open(self._marker_path, "w") as fh:
+            fh.write(str(os.getpid()))
         self._pending: List[bytes] = []
         self._packed: Optional[Tuple[int, bool, int]] = None
         self._closed = threading.Event()
@@ -1226,9 +1262,6 @@ class ShmListener:
             out.extend(bytes(f) for f in frames)
         return out
 
-    def has_pending(self) -> bool:
-        return bool(getattr(self, "_pending", None)) or not self._recv_q.empty()
-
     # -- packed fast path ----------------------------------------------
     def enable_packed(self, max_len: int, pin: bool,
                       max_frames: int = 4096) -> bool:
@@ -1262,9 +1295,13 @@ class ShmListener:
     def reply(self, idx: int, data: bytes) -> bool:
         return self.send(data)
 
+    def reply_row(self, idx: int, data: bytes) -> bool:
+        """Ring replies are single-channel (s2c)."""
+        return self.send(data)
+
     def close(self) -> None:
         self._closed.set()
-        for p in (self._c2s_path, self._s2c_path):
+        for p in (self._c2s_path, self._s2c_path, self._marker_path):
             try:
                 os.unlink(p)
             except OSError:

@@ -60,16 +60,25 @@ inline void put_int(std::string& out, int field, int64_t v) {
   put_varint(out, (uint64_t)v);  // sign-extended two's complement
 }
 
+// Overflow-safe: get_varint leaves pos <= n, so compare lengths against
+// n - pos instead of computing pos + len (which a crafted ~2^64 varint
+// length could wrap below n, defeating the bounds check).
 bool skip_field(const uint8_t* p, size_t n, size_t& pos, int wt) {
   uint64_t tmp;
   switch (wt) {
     case 0: return get_varint(p, n, pos, tmp);
-    case 1: pos += 8; return pos <= n;
+    case 1:
+      if (n - pos < 8) return false;
+      pos += 8;
+      return true;
     case 2:
-      if (!get_varint(p, n, pos, tmp)) return false;
-      pos += tmp;
-      return pos <= n;
-    case 5: pos += 4; return pos <= n;
+      if (!get_varint(p, n, pos, tmp) || tmp > (uint64_t)(n - pos)) return false;
+      pos += (size_t)tmp;
+      return true;
+    case 5:
+      if (n - pos < 4) return false;
+      pos += 4;
+      return true;
     default: return false;
   }
 }
@@ -421,7 +430,7 @@ py::tuple parser_watch_hashes(
         event_id = (int32_t)(uint32_t)v;
       } else if (wt == 2) {
         uint64_t sl;
-        if (!get_varint(p, n, pos, sl) || pos + sl > n) break;
+        if (!get_varint(p, n, pos, sl) || sl > (uint64_t)(n - pos)) break;
         const char* s = (const char*)(p + pos);
         if (field == 6) {
           variables.emplace_back(s, (size_t)sl);
@@ -437,7 +446,7 @@ py::tuple parser_watch_hashes(
             if (!get_varint(e, sl, epos, ekey)) break;
             if ((ekey & 7) == 2) {
               uint64_t el;
-              if (!get_varint(e, sl, epos, el) || epos + el > sl) break;
+              if (!get_varint(e, sl, epos, el) || el > (uint64_t)(sl - epos)) break;
               if ((ekey >> 3) == 1) k = {(const char*)e + epos, (size_t)el};
               else if ((ekey >> 3) == 2) v = {(const char*)e + epos, (size_t)el};
               epos += el;

@@ -22,16 +22,26 @@ inline bool get_varint(const uint8_t* p, size_t n, size_t& pos, uint64_t& v) {
   return false;
 }
 
+// All length arithmetic below is overflow-safe: get_varint leaves pos <= n,
+// so `len > n - pos` rejects any oversized length without computing pos+len
+// (a 10-byte crafted varint near 2^64 would wrap pos+len below n and turn
+// the bounds check into an out-of-bounds read).
 inline bool skip_field(const uint8_t* p, size_t n, size_t& pos, int wt) {
   uint64_t tmp;
   switch (wt) {
     case 0: return get_varint(p, n, pos, tmp);
-    case 1: pos += 8; return pos <= n;
+    case 1:
+      if (n - pos < 8) return false;
+      pos += 8;
+      return true;
     case 2:
-      if (!get_varint(p, n, pos, tmp)) return false;
-      pos += tmp;
-      return pos <= n;
-    case 5: pos += 4; return pos <= n;
+      if (!get_varint(p, n, pos, tmp) || tmp > (uint64_t)(n - pos)) return false;
+      pos += (size_t)tmp;
+      return true;
+    case 5:
+      if (n - pos < 4) return false;
+      pos += 4;
+      return true;
     default: return false;
   }
 }
@@ -59,7 +69,7 @@ inline void decode_log_core(
       const int field = (int)(key >> 3), wt = (int)(key & 7);
       if (wt == 2) {
         uint64_t sl;
-        if (!get_varint(p, n, pos, sl) || pos + sl > n) break;
+        if (!get_varint(p, n, pos, sl) || sl > (uint64_t)(n - pos)) break;
         const char* s = (const char*)(p + pos);
         switch (field) {
           case 2: sp.id = s; sp.id_len = sl; break;

@@ -29,6 +29,7 @@
 #include <chrono>
 #include <cerrno>
 #include <cstring>
+#include <deque>
 #include <stdexcept>
 #include <string>
 #include <thread>
@@ -55,6 +56,10 @@ class ShmRing {
     // power-of-two data size
     uint64_t sz = 1;
     while ((int64_t)sz < size_bytes) sz <<= 1;
+    // create=true means "this endpoint owns a FRESH ring": drop any stale
+    // file so a restarted listener neither replays a crashed run's frames
+    // nor silently keeps the old size when DMX_SHM_RING_BYTES changed.
+    if (create) ::unlink(path.c_str());
     int fd = ::open(path.c_str(), O_RDWR | O_CREAT, 0600);
     if (fd < 0) throw std::runtime_error("shm open failed: " + path);
     struct stat st{};
@@ -96,7 +101,6 @@ class ShmRing {
     }
     size_ = hdr_->size;
     mask_ = size_ - 1;
-    (void)create;
   }
 
   ~ShmRing() {
@@ -174,7 +178,12 @@ class ShmRing {
   py::tuple read_batch_packed(int max_frames, int timeout_ms, int max_len,
                               bool pin) {
     std::vector<std::pair<const uint8_t*, size_t>> raw;
-    std::vector<std::string> wrapped;  // frames crossing the boundary
+    // Frames crossing the ring boundary are copied to scratch. A deque is
+    // required (NOT vector<string>): raw holds .data() pointers into these
+    // strings, and short strings store their bytes inline (SSO) — a
+    // vector reallocation would move the string objects and dangle every
+    // SSO pointer already recorded. Deque push_back never moves elements.
+    std::deque<std::string> wrapped;
     uint64_t tail0, tail;
     {
       py::gil_scoped_release release;

@@ -160,6 +160,12 @@ class ServiceSettings(BaseModel):
     # --- component config file (reference settings.py:53) ---
     config_file: Optional[Path] = None
 
+    #: directory that /admin/checkpoint + /admin/restore may read/write.
+    #: Unset => the admin endpoints refuse (the Python API is unaffected):
+    #: the admin HTTP surface is unauthenticated, so letting callers name
+    #: arbitrary filesystem paths would allow checkpoint planting/probing.
+    checkpoint_dir: Optional[Path] = None
+
     # --- MI355X-native compute placement ---
     device: Optional[str] = None
     dist_backend: Optional[str] = None

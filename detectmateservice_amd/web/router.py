@@ -9,7 +9,7 @@ from __future__ import annotations
 
 from typing import Any, Dict, Optional
 
-from fastapi import APIRouter, Depends
+from fastapi import APIRouter, Depends, HTTPException
 from pydantic import BaseModel
 
 router = APIRouter(prefix="/admin")
@@ -55,14 +55,22 @@ def shutdown(service=Depends(get_service)) -> Dict[str, Any]:
 
 
 class CheckpointPayload(BaseModel):
+    #: name relative to settings.checkpoint_dir (absolute paths must
+    #: resolve inside it); refused with 403 when no checkpoint_dir is set
     path: str
 
 
 @router.post("/checkpoint")
 def checkpoint(payload: CheckpointPayload, service=Depends(get_service)) -> Dict[str, Any]:
-    return service.checkpoint(payload.path)
+    try:
+        return service.checkpoint(payload.path, from_admin=True)
+    except PermissionError as exc:
+        raise HTTPException(status_code=403, detail=str(exc)) from None
 
 
 @router.post("/restore")
 def restore(payload: CheckpointPayload, service=Depends(get_service)) -> Dict[str, Any]:
-    return service.restore(payload.path)
+    try:
+        return service.restore(payload.path, from_admin=True)
+    except PermissionError as exc:
+        raise HTTPException(status_code=403, detail=str(exc)) from None

@@ -223,3 +223,72 @@ def test_decode_log_batch_packed_matches_list_variant():
     assert torch.equal(l1, l2) and torch.equal(n1, n2)
     for i in range(len(frames)):
         assert blob[int(off[i]):int(off[i + 1])] == bytes(ids[i])
+
+
+def _varint(v: int) -> bytes:
+    out = bytearray()
+    while True:
+        b = v & 0x7F
+        v >>= 7
+        if v:
+            out.append(b | 0x80)
+        else:
+            out.append(b)
+            return bytes(out)
+
+
+@pytest.mark.skipif(not ops.have_extension(), reason="extension not built")
+def test_packed_decode_rejects_overflowing_varint_lengths():
+    """Crafted 10-byte varint string lengths near 2^64 must not wrap the
+    bounds check (ADVICE high: heap over-read). The decoder stops at the
+    malformed field; adjacent valid frames still decode."""
+    from detectmateservice_amd.ops import _dmx_C
+
+    evil_frames = []
+    for length in (2**64 - 8, 2**64 - 1, 2**63, 2**32):
+        # field 3 (log), wire type 2, huge length, tiny payload
+        evil_frames.append(bytes([0x1A]) + _varint(length) + b"AB")
+        # field 2 (logID) with huge length — span fields
+        evil_frames.append(bytes([0x12]) + _varint(length) + b"CD")
+    good = LogSchema(logID="ok", log="fine line").serialize()
+    frames = evil_frames + [good]
+    lines, lens, blob, off = _dmx_C.decode_log_batch_packed(frames, 64, False)
+    assert lines.shape == (len(frames), 64)
+    # malformed frames contribute empty/zero rows, never OOB copies
+    for i in range(len(evil_frames)):
+        assert int(lens[i]) == 0
+        assert blob[int(off[i]):int(off[i + 1])] == b""
+    assert bytes(lines[-1, : int(lens[-1])].numpy().tobytes()) == b"fine line"
+    assert blob[int(off[len(frames) - 1]):int(off[len(frames)])] == b"ok"
+
+
+@pytest.mark.skipif(not ops.have_extension(), reason="extension not built")
+def test_packed_decode_rejects_overflowing_skip_lengths():
+    """skip_field wire-type 2 with a wrapping length (unknown field) must
+    terminate the walk instead of advancing pos past the buffer."""
+    from detectmateservice_amd.ops import _dmx_C
+
+    # unknown field 9, wt 2, huge length, then a valid log field that the
+    # decoder must NOT reach (the frame is malformed from field 9 on)
+    f = bytes([0x4A]) + _varint(2**64 - 2) + bytes([0x1A, 0x03]) + b"abc"
+    lines, lens, blob, off = _dmx_C.decode_log_batch_packed([f], 64, False)
+    assert int(lens[0]) == 0
+
+    # fixed-width skips at the buffer edge: wt 1 with <8 bytes left,
+    # wt 5 with <4 bytes left
+    for frame in (bytes([0x49]) + b"\x01\x02", bytes([0x4D]) + b"\x01"):
+        lines, lens, blob, off = _dmx_C.decode_log_batch_packed([frame], 64, False)
+        assert int(lens[0]) == 0
+
+
+@pytest.mark.skipif(not ops.have_extension(), reason="extension not built")
+def test_parser_hash_decode_rejects_overflowing_varints():
+    """The ParserSchema watch-hash decoder shares the same varint walk;
+    crafted lengths in variables / map entries must not over-read."""
+    from detectmateservice_amd.ops import _dmx_C
+
+    evil = bytes([0x32]) + _varint(2**64 - 4) + b"XY"  # field 6 repeated var
+    good = ParserSchema(EventID=1, logID="g", variables=["v0"]).serialize()
+    hashes, event_ids, log_ids = _dmx_C.parser_watch_hashes(
+        [evil, good], [(-1, 0)], [], False)
+    assert int(event_ids[1]) == 1

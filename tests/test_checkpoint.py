@@ -10,7 +10,7 @@ from detectmateservice_amd import Service, ServiceSettings
 from detectmateservice_amd.schemas import ParserSchema
 
 
-def _detector_settings(tmp_path, ipc_addr, port=0, http=False):
+def _detector_settings(tmp_path, ipc_addr, port=0, http=False, ck_dir=None):
     cfg_file = tmp_path / "det.yaml"
     cfg_file.write_text(yaml.safe_dump({
         "detectors": {"NewValueDetector": {
@@ -26,6 +26,7 @@ def _detector_settings(tmp_path, ipc_addr, port=0, http=False):
         http_enabled=http,
         http_port=port,
         log_dir=tmp_path / "logs",
+        checkpoint_dir=ck_dir,
     )
 
 
@@ -76,7 +77,11 @@ def test_restore_rejects_wrong_component_type(tmp_path, ipc_addr):
 
 
 def test_checkpoint_over_http(tmp_path, ipc_addr, free_port):
-    settings = _detector_settings(tmp_path, ipc_addr, port=free_port, http=True)
+    """Admin checkpoint/restore operate on names INSIDE the configured
+    checkpoint_dir; traversal and absolute escapes are refused with 403."""
+    ck_dir = tmp_path / "ckpts"
+    settings = _detector_settings(
+        tmp_path, ipc_addr, port=free_port, http=True, ck_dir=ck_dir)
     svc = Service(settings)
     t = threading.Thread(target=svc.run, daemon=True)
     t.start()
@@ -84,14 +89,62 @@ def test_checkpoint_over_http(tmp_path, ipc_addr, free_port):
         assert svc.web_server.wait_started(10.0)
         svc.process_batch([_frame("/a"), _frame("/b")])
         url = f"http://127.0.0.1:{free_port}"
-        ck = str(tmp_path / "http_ck.pt")
-        r = httpx.post(f"{url}/admin/checkpoint", json={"path": ck}, timeout=10.0)
-        assert r.status_code == 200 and r.json()["path"] == ck
-        r = httpx.post(f"{url}/admin/restore", json={"path": ck}, timeout=10.0)
+        r = httpx.post(f"{url}/admin/checkpoint", json={"path": "http_ck.pt"},
+                       timeout=10.0)
         assert r.status_code == 200
+        assert (ck_dir / "http_ck.pt").exists()
+        r = httpx.post(f"{url}/admin/restore", json={"path": "http_ck.pt"},
+                       timeout=10.0)
+        assert r.status_code == 200
+        # traversal out of checkpoint_dir: refused
+        r = httpx.post(f"{url}/admin/restore",
+                       json={"path": "../det.yaml"}, timeout=10.0)
+        assert r.status_code == 403
+        # absolute path outside checkpoint_dir: refused
+        r = httpx.post(f"{url}/admin/checkpoint",
+                       json={"path": "/etc/cron.d/x.pt"}, timeout=10.0)
+        assert r.status_code == 403
     finally:
         svc.shutdown()
         t.join(timeout=5.0)
+
+
+def test_checkpoint_over_http_refused_without_dir(tmp_path, ipc_addr, free_port):
+    """No checkpoint_dir configured => the unauthenticated admin surface
+    must not touch the filesystem at all."""
+    settings = _detector_settings(tmp_path, ipc_addr, port=free_port, http=True)
+    svc = Service(settings)
+    t = threading.Thread(target=svc.run, daemon=True)
+    t.start()
+    try:
+        assert svc.web_server.wait_started(10.0)
+        url = f"http://127.0.0.1:{free_port}"
+        for ep in ("checkpoint", "restore"):
+            r = httpx.post(f"{url}/admin/{ep}", json={"path": "x.pt"},
+                           timeout=10.0)
+            assert r.status_code == 403
+    finally:
+        svc.shutdown()
+        t.join(timeout=5.0)
+
+
+def test_restore_refuses_pickle_payload(tmp_path, ipc_addr):
+    """A checkpoint file carrying an arbitrary pickle object (the RCE
+    vector ADVICE flagged) must raise on load, not execute."""
+    import torch
+
+    class Evil:  # torch.save will pickle this; weights_only load must balk
+        def __reduce__(self):
+            return (print, ("pwned",))
+
+    ck = tmp_path / "evil.pt"
+    torch.save({"component_type": "NewValueDetector", "component_state": Evil()}, ck)
+    svc = Service(_detector_settings(tmp_path, ipc_addr))
+    try:
+        with pytest.raises(Exception):
+            svc.restore(ck)
+    finally:
+        svc.engine.close()
 
 
 def test_reconfigure_reload_rebuilds_component(tmp_path):

@@ -348,3 +348,56 @@ def test_packed_loop_survives_component_errors(tmp_path):
         dialer.close()
         engine.stop()
         engine.close()
+
+
+class RecordingReplySock:
+    """Fake pair socket recording reply(idx) calls (seam: reference
+    engine_socket.py:23-32 factory injection)."""
+
+    def __init__(self, batches):
+        self.addr = "inproc://fake"
+        self._batches = list(batches)
+        self.replies = []
+
+    def recv_many(self, max_frames, timeout_ms, linger_ms=0.0):
+        if self._batches:
+            return self._batches.pop(0)
+        raise RecvTimeout(self.addr)
+
+    def reply(self, idx, data):
+        self.replies.append((idx, data))
+        return True
+
+    def send(self, data, block=True):
+        self.replies.append((None, data))
+        return True
+
+    def close(self):
+        pass
+
+
+class RecordingFactory:
+    def __init__(self, sock):
+        self.sock = sock
+
+    def create(self, addr, **kw):
+        return self.sock
+
+    def create_dialer(self, addr, **kw):
+        raise AssertionError("no outputs expected")
+
+
+def test_reply_indices_survive_empty_frame_filtering(ipc_addr):
+    """Empty frames are skipped (reference engine.py:207-209) but reply
+    routing must use each surviving frame's ORIGINAL batch index —
+    _batch_conns was recorded for the unfiltered batch (ADVICE low #1)."""
+    sock = RecordingReplySock([[b"", b"x", b"", b"y"]])
+    eng = Engine(_settings(ipc_addr), SimpleProcessor(),
+                 socket_factory=RecordingFactory(sock))
+    eng.start()
+    deadline = time.time() + 5.0
+    while len(sock.replies) < 2 and time.time() < deadline:
+        time.sleep(0.01)
+    eng.stop()
+    # frames at original indices 1 ("x") and 3 ("y") get the replies
+    assert sock.replies == [(1, b"out:x"), (3, b"out:y")]
