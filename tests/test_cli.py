@@ -49,6 +49,7 @@ def test_client_cli_against_live_service(free_port, tmp_path, capsys):
     settings = ServiceSettings(
         component_type="core", engine_addr=f"ipc://{tmp_path}/cli.ipc",
         http_port=free_port, log_dir=tmp_path / "logs",
+        checkpoint_dir=tmp_path / "ckpts",
     )
     svc = Service(settings)
     t = threading.Thread(target=svc.run, daemon=True)
@@ -66,10 +67,10 @@ def test_client_cli_against_live_service(free_port, tmp_path, capsys):
         cfg = tmp_path / "re.yaml"
         cfg.write_text("detectors: {}\n")
         assert client_main(["--url", url, "reconfigure", str(cfg)]) == 0
-        ck = tmp_path / "c.pt"
-        assert client_main(["--url", url, "checkpoint", str(ck)]) == 0
-        assert ck.exists()
-        assert client_main(["--url", url, "restore", str(ck)]) == 0
+        # checkpoint names resolve under settings.checkpoint_dir
+        assert client_main(["--url", url, "checkpoint", "c.pt"]) == 0
+        assert (tmp_path / "ckpts" / "c.pt").exists()
+        assert client_main(["--url", url, "restore", "c.pt"]) == 0
         assert client_main(["--url", url, "shutdown"]) == 0
     finally:
         svc.shutdown()
