@@ -69,35 +69,91 @@ typedef __attribute__((address_space(3))) float lds_float;
 #define FB_B2 (FB_B1 + BF_FFN)
 #define FB_SIZE (FB_B2 + BF_H)
 
-// ---- in-block GEMM: out = act(in_lds[64][K] @ Wt[N][WTS] + bias) ---------
-// MODE 0: write out_lds[m][n]; MODE 1 (qkv): n<2H -> out (Q|K), else vt
-// transposed; MODE 2: x[m][n] += v (residual-accumulate). ACT 1 = GELU.
-template <int K, int N, int MODE, int ACT, int WTS>
-static __device__ __attribute__((noinline)) void block_gemm(
-    const lds_short* in_lds, int in_stride, const glob_cshort* __restrict__ Wt,
-    const glob_cfloat* __restrict__ bias, lds_short* out_lds, int out_stride,
-    lds_short* x_lds, lds_short* vt_lds, int wid, int lane) {
-  constexpr int N16 = N / 16;
-  constexpr int TOTAL = 4 * N16;
+// ---- cross-barrier weight prefetch (v5) ----------------------------------
+// Round-1 PMC evidence (profiles/r08): SQ_WAIT_ANY ~ 18x SQ_BUSY — waves
+// spend ~95% of residency parked, and five issue-side ablations were
+// neutral. The wall is structural: every GEMM phase opened with a
+// DEPENDENT chain of L2 weight-fragment loads issued AFTER the barrier
+// that every wave crosses together, so the whole CU idles ~KS x 200
+// cycles per phase before the first MFMA can retire. Weights are
+// read-only, so those loads have NO hazard against the barrier — but the
+// compiler cannot hoist loads across __syncthreads(). v5 does it by
+// hand: each GEMM's full B-fragment set is issued into registers during
+// the PREVIOUS phase and consumed after the barrier, so phases open
+// MFMA-ready (the s_waitcnt lands ~a whole phase after issue). Costs
+// VGPRs (fragments held across a phase): launch bounds drop to
+// 2 waves/SIMD (<=256 VGPR, 1 block/CU) — round-1 measurements showed
+// throughput FLAT in occupancy 2-6 waves/SIMD, so the co-residency this
+// gives up was not buying anything the prefetch doesn't replace.
+//
+// Per wave a GEMM owns N/128 "quads" (quad = one fn column x 4
+// m-fragments); fragments per wave = (N/128) * (K/32).
+// PKS = fragments per quad prefetched across the barrier (the rest
+// stream in-phase under MFMA cover): full prefetch of every GEMM spilled
+// 17 VGPRs in-loop; phase-START latency only needs the first fragments
+// resident, later ks-steps hide behind ~136 cycles of MFMA per step.
+template <int K, int N, int PKS, int WTS>
+static __device__ __forceinline__ void load_wfrags(
+    const glob_cshort* __restrict__ Wt, int wid, int lane, bf16x8* out) {
+  constexpr int NQ = N / 128;
+#pragma unroll
+  for (int q = 0; q < NQ; ++q)
+#pragma unroll
+    for (int ks = 0; ks < PKS; ++ks)
+      out[q * PKS + ks] = *(const __attribute__((address_space(1))) bf16x8*)(
+          Wt + (long)((wid * NQ + q) * 16 + (lane & 15)) * WTS + ks * 32 +
+          (lane >> 4) * 8);
+}
+
+// Biases ride along with the fragments: an in-phase bias load would make
+// the compiler emit a vmcnt(0) drain in the GEMM epilogue, flushing the
+// NEXT phase's just-issued prefetch loads with it.
+template <int N>
+static __device__ __forceinline__ void load_bias(
+    const glob_cfloat* __restrict__ bias, int wid, int lane, float* out) {
+  constexpr int NQ = N / 128;
+#pragma unroll
+  for (int q = 0; q < NQ; ++q)
+    out[q] = bias ? bias[(wid * NQ + q) * 16 + (lane & 15)] : 0.f;
+}
+
+// ---- in-block GEMM: out = act(in_lds[64][K] @ W + bias) ------------------
+// Weight fragments arrive PRELOADED in registers (load_wfrags, issued
+// before the preceding barrier). MODE 0: write out_lds[m][n]; MODE 1
+// (qkv): n<2H -> out (Q|K), else vt transposed; MODE 2: x[m][n] += v
+// (residual-accumulate). ACT 1 = GELU.
+template <int K, int N, int PKS, int MODE, int ACT, int WTS>
+static __device__ __forceinline__ void block_gemm_pre(
+    const lds_short* in_lds, int in_stride, const bf16x8* w,
+    const glob_cshort* __restrict__ Wt, const float* bias_pre,
+    lds_short* out_lds, int out_stride, lds_short* x_lds, lds_short* vt_lds,
+    int wid, int lane) {
   constexpr int KS = K / 32;
-  constexpr int FPW = TOTAL / BF_WAVES;
-  static_assert(FPW >= 4 && FPW % 4 == 0, "quad-chunked assignment");
-  // quad-chunk: each iteration owns a FULL fn column (all 4 m-fragments):
-  // ONE L2 weight-fragment load feeds FOUR independent MFMA chains.
-  // unroll 2 quads: the second quad's (independent) weight loads issue
-  // under the first quad's MFMA chains, hiding the L2 latency that
-  // dominated the per-quad cost (phase probe: GEMMs ~3.9 of 4.5 ms while
-  // pure MFMA issue accounts for <10% of that)
-#pragma unroll 2
-  for (int ff = wid * FPW; ff < wid * FPW + FPW; ff += 4) {
-    const int fn = ff >> 2;
+  constexpr int NQ = N / 128;
+#pragma unroll
+  for (int q = 0; q < NQ; ++q) {
+    const int fn = wid * NQ + q;
     f32x4 acc[4];
 #pragma unroll
     for (int i = 0; i < 4; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll 4
-    for (int ks = 0; ks < KS; ++ks) {
-      bf16x8 b = *(const __attribute__((address_space(1))) bf16x8*)(
-          Wt + (long)(fn * 16 + (lane & 15)) * WTS + ks * 32 + (lane >> 4) * 8);
+    // prefetched head fragments
+#pragma unroll
+    for (int ks = 0; ks < PKS; ++ks) {
+      const bf16x8 b = w[q * PKS + ks];
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm) {
+        bf16x8 a = *(const __attribute__((address_space(3))) bf16x8*)(
+            in_lds + (fm * 16 + (lane & 15)) * in_stride + ks * 32 +
+            (lane >> 4) * 8);
+        acc[fm] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[fm], 0, 0, 0);
+      }
+    }
+    // in-phase tail: loads issue under the head's MFMA chains
+#pragma unroll
+    for (int ks = PKS; ks < KS; ++ks) {
+      const bf16x8 b = *(const __attribute__((address_space(1))) bf16x8*)(
+          Wt + (long)(fn * 16 + (lane & 15)) * WTS + ks * 32 +
+          (lane >> 4) * 8);
 #pragma unroll
       for (int fm = 0; fm < 4; ++fm) {
         bf16x8 a = *(const __attribute__((address_space(3))) bf16x8*)(
@@ -107,7 +163,7 @@ static __device__ __attribute__((noinline)) void block_gemm(
       }
     }
     const int n = fn * 16 + (lane & 15);
-    const float bval = bias ? bias[n] : 0.f;
+    const float bval = bias_pre[q];
 #pragma unroll
     for (int fm = 0; fm < 4; ++fm) {
 #pragma unroll
@@ -244,6 +300,13 @@ static __device__ __forceinline__ void bert_fused_body(
     }
   }
   BF_STAMP();  // 1: embed work done
+  // layer-0 qkv weight fragments: issued HERE so they are in flight
+  // across the embed barrier (load_wfrags comment above)
+  bf16x8 wq[6];
+  float bq[3];
+  load_wfrags<BF_H, 3 * BF_H, 2, BF_H>(
+      (glob_cshort*)(wb + WB_LAYER0 + LW_QKV), wid, lane, wq);
+  load_bias<3 * BF_H>((glob_cfloat*)(fb + FB_BQKV), wid, lane, bq);
   __syncthreads();
   BF_STAMP();  // 2: embed barrier crossed
 
@@ -253,10 +316,17 @@ static __device__ __forceinline__ void bert_fused_body(
 
     // ---- qkv: Q|K -> buf[64][QKS], V -> vt transposed ----
     if (PHASES & PH_QKV)
-      block_gemm<BF_H, 3 * BF_H, 1, 0, BF_H>(x_lds, XS, (glob_cshort*)(lw + LW_QKV),
-                                             (glob_cfloat*)(lf + FB_BQKV), buf, QKS, x_lds, vt,
-                                             wid, lane);
+      block_gemm_pre<BF_H, 3 * BF_H, 2, 1, 0, BF_H>(
+          x_lds, XS, wq, (glob_cshort*)(lw + LW_QKV), bq, buf, QKS, x_lds,
+          vt, wid, lane);
     BF_STAMP();  // qkv work done
+    // proj weights ride across the qkv barrier + the whole attention
+    // phase (attention reads no global weights)
+    bf16x8 wpr[4];
+    float bpr[1];
+    load_wfrags<BF_H, BF_H, 4, BF_H>((glob_cshort*)(lw + LW_WO), wid, lane,
+                                     wpr);
+    load_bias<BF_H>((glob_cfloat*)(lf + FB_BO), wid, lane, bpr);
     __syncthreads();
     BF_STAMP();  // qkv barrier crossed
 
@@ -346,14 +416,26 @@ static __device__ __forceinline__ void bert_fused_body(
       }
     }
     BF_STAMP();  // attention work done
+    // FFN W1 half-0 fragments cross the attn barrier + proj + LN1
+    bf16x8 w1[4];
+    float b1[2];
+    load_wfrags<BF_H, BF_FFN / 2, 2, BF_H>((glob_cshort*)(lw + LW_W1), wid,
+                                           lane, w1);
+    load_bias<BF_FFN / 2>((glob_cfloat*)(lf + FB_B1), wid, lane, b1);
     __syncthreads();
     BF_STAMP();  // attention barrier crossed
 
     // ---- proj: x += Wo(attn) ; LN1 ----
     if (PHASES & PH_PROJ)
-      block_gemm<BF_H, BF_H, 2, 0, BF_H>(buf + O_OFF, XS, (glob_cshort*)(lw + LW_WO),
-                                         (glob_cfloat*)(lf + FB_BO), nullptr, 0, x_lds,
-                                         nullptr, wid, lane);
+      block_gemm_pre<BF_H, BF_H, 4, 2, 0, BF_H>(
+          buf + O_OFF, XS, wpr, (glob_cshort*)(lw + LW_WO), bpr, nullptr, 0,
+          x_lds, nullptr, wid, lane);
+    // W2 half-0 fragments cross the proj barrier + LN1 + W1h0
+    bf16x8 w2[4];
+    float b2[1];
+    load_wfrags<BF_FFN / 2, BF_H, 4, BF_FFN>((glob_cshort*)(lw + LW_W2), wid,
+                                             lane, w2);
+    load_bias<BF_H>((glob_cfloat*)(lf + FB_B2), wid, lane, b2);
     __syncthreads();
     if (PHASES & PH_LN)
       block_layernorm(x_lds, lw + LW_LN1G, lw + LW_LN1B, wid, lane, eps);
@@ -362,21 +444,42 @@ static __device__ __forceinline__ void bert_fused_body(
     BF_STAMP();  // proj+LN1 barrier crossed
 
     // ---- FFN in two K=256 halves: buf = gelu(x@W1_h); x += buf@W2_h ----
-    if (PHASES & PH_FFN)
-#pragma unroll
-    for (int h = 0; h < 2; ++h) {
-      block_gemm<BF_H, BF_FFN / 2, 0, 1, BF_H>(
-          x_lds, XS, (glob_cshort*)(lw + LW_W1 + (long)h * (BF_FFN / 2) * BF_H),
-          (glob_cfloat*)(lf + FB_B1 + h * (BF_FFN / 2)), buf, QKS, nullptr,
+    if (PHASES & PH_FFN) {
+      block_gemm_pre<BF_H, BF_FFN / 2, 2, 0, 1, BF_H>(
+          x_lds, XS, w1, (glob_cshort*)(lw + LW_W1), b1, buf, QKS, nullptr,
           nullptr, wid, lane);
+      // W1 half-1 fragments reuse w1's registers (h0 consumed above)
+      load_wfrags<BF_H, BF_FFN / 2, 2, BF_H>(
+          (glob_cshort*)(lw + LW_W1 + (long)(BF_FFN / 2) * BF_H), wid, lane,
+          w1);
+      load_bias<BF_FFN / 2>((glob_cfloat*)(lf + FB_B1 + BF_FFN / 2), wid,
+                            lane, b1);
       __syncthreads();
       // bias b2 added once (half 0); half 1 adds only the partial product
-      block_gemm<BF_FFN / 2, BF_H, 2, 0, BF_FFN>(
-          buf, QKS, (glob_cshort*)(lw + LW_W2 + (long)h * (BF_FFN / 2)),
-          h == 0 ? (glob_cfloat*)(lf + FB_B2) : nullptr, nullptr, 0, x_lds,
+      block_gemm_pre<BF_FFN / 2, BF_H, 4, 2, 0, BF_FFN>(
+          buf, QKS, w2, (glob_cshort*)(lw + LW_W2), b2, nullptr, 0, x_lds,
           nullptr, wid, lane);
+      load_wfrags<BF_FFN / 2, BF_H, 4, BF_FFN>(
+          (glob_cshort*)(lw + LW_W2 + (BF_FFN / 2)), wid, lane, w2);
+      b2[0] = 0.f;  // half 1 adds only the partial product
       __syncthreads();
+      block_gemm_pre<BF_H, BF_FFN / 2, 2, 0, 1, BF_H>(
+          x_lds, XS, w1,
+          (glob_cshort*)(lw + LW_W1 + (long)(BF_FFN / 2) * BF_H), b1, buf,
+          QKS, nullptr, nullptr, wid, lane);
+      __syncthreads();
+      block_gemm_pre<BF_FFN / 2, BF_H, 4, 2, 0, BF_FFN>(
+          buf, QKS, w2, (glob_cshort*)(lw + LW_W2 + (BF_FFN / 2)), b2,
+          nullptr, 0, x_lds, nullptr, wid, lane);
     }
+    // next layer's qkv fragments cross the FFN-tail barrier + LN2
+    if (layer + 1 < n_layers) {
+      load_wfrags<BF_H, 3 * BF_H, 2, BF_H>(
+          (glob_cshort*)(lw + LW_SIZE + LW_QKV), wid, lane, wq);
+      load_bias<3 * BF_H>((glob_cfloat*)(lf + FB_SIZE + FB_BQKV), wid, lane,
+                          bq);
+    }
+    __syncthreads();
     if (PHASES & PH_LN)
       block_layernorm(x_lds, lw + LW_LN2G, lw + LW_LN2B, wid, lane, eps);
     BF_STAMP();  // ffn+LN2 work done
@@ -414,7 +517,7 @@ static __device__ __forceinline__ void bert_fused_body(
 #undef BF_STAMP
 }
 
-extern "C" __global__ __launch_bounds__(BF_THREADS, 4)
+extern "C" __global__ __launch_bounds__(BF_THREADS, 2)
 void dmx_bert_fused_bf16(const unsigned char* __restrict__ lines,
                          const int* __restrict__ start,
                          const int* __restrict__ end,
@@ -429,7 +532,7 @@ void dmx_bert_fused_bf16(const unsigned char* __restrict__ lines,
 // probe variants (in-kernel phase ablation; guide §5.4 rule 19: co-compiled
 // variants can perturb codegen by a few % — read the deltas, not absolutes)
 template <int PHASES>
-__global__ __launch_bounds__(BF_THREADS, 4) void dmx_bert_fused_probe(
+__global__ __launch_bounds__(BF_THREADS, 2) void dmx_bert_fused_probe(
     const unsigned char* lines, const int* start, const int* end,
     const short* wb, const float* fb, float* scores, int B, int max_len,
     int n_layers, float eps) {
@@ -437,7 +540,7 @@ __global__ __launch_bounds__(BF_THREADS, 4) void dmx_bert_fused_probe(
                           n_layers, eps);
 }
 
-extern "C" __global__ __launch_bounds__(BF_THREADS, 4)
+extern "C" __global__ __launch_bounds__(BF_THREADS, 2)
 void dmx_bert_fused_timed(const unsigned char* lines, const int* start,
                           const int* end, const short* wb, const float* fb,
                           float* scores, int B, int max_len, int n_layers,
