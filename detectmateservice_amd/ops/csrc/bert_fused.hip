@@ -35,7 +35,7 @@ typedef __attribute__((address_space(1))) const float glob_cfloat;
 typedef __attribute__((address_space(3))) const short lds_cshort;
 typedef __attribute__((address_space(3))) float lds_float;
 
-#define BF_WAVES 8
+#define BF_WAVES 4
 #define BF_THREADS (BF_WAVES * DMX_WAVE)
 #define BF_S 64
 #define BF_H 128
@@ -46,7 +46,7 @@ typedef __attribute__((address_space(3))) float lds_float;
 #define XS (BF_H + 8)       // 136, x rows
 #define QKS (2 * BF_H + 8)  // 264, Q|K rows and FFN-half rows
 #define VTS (BF_S + 8)      // 72, transposed-V and P rows
-#define O_OFF (BF_WAVES * 16 * VTS)    // 9216: attn-out after the P tiles
+#define O_OFF (BF_WAVES * 32 * VTS)    // 9216: attn-out after the P tiles
 #define BUF_ELEMS (BF_S * QKS + 1024)  // 17920 = max(QK, P+O, FFN-half)
 
 // bf16 weight-blob element offsets (host packs identically: bert_tiny.py)
@@ -95,14 +95,20 @@ typedef __attribute__((address_space(3))) float lds_float;
 template <int K, int N, int PKS, int WTS>
 static __device__ __forceinline__ void load_wfrags(
     const glob_cshort* __restrict__ Wt, int wid, int lane, bf16x8* out) {
-  constexpr int NQ = N / 128;
+  constexpr int NQ = N / (16 * BF_WAVES);
+  // opaque lane: the per-lane address offsets are layer-invariant and
+  // LICM otherwise hoists ~25 of them out of the layer loop, spilling
+  // them across every phase (the guide's "recompute per block" pitfall);
+  // recomputing costs a couple of VALU per phase.
+  int ln = lane;
+  asm volatile("" : "+v"(ln));
 #pragma unroll
   for (int q = 0; q < NQ; ++q)
 #pragma unroll
     for (int ks = 0; ks < PKS; ++ks)
       out[q * PKS + ks] = *(const __attribute__((address_space(1))) bf16x8*)(
-          Wt + (long)((wid * NQ + q) * 16 + (lane & 15)) * WTS + ks * 32 +
-          (lane >> 4) * 8);
+          Wt + (long)((wid * NQ + q) * 16 + (ln & 15)) * WTS + ks * 32 +
+          (ln >> 4) * 8);
 }
 
 // Biases ride along with the fragments: an in-phase bias load would make
@@ -111,10 +117,12 @@ static __device__ __forceinline__ void load_wfrags(
 template <int N>
 static __device__ __forceinline__ void load_bias(
     const glob_cfloat* __restrict__ bias, int wid, int lane, float* out) {
-  constexpr int NQ = N / 128;
+  constexpr int NQ = N / (16 * BF_WAVES);
+  int ln = lane;
+  asm volatile("" : "+v"(ln));
 #pragma unroll
   for (int q = 0; q < NQ; ++q)
-    out[q] = bias ? bias[(wid * NQ + q) * 16 + (lane & 15)] : 0.f;
+    out[q] = bias ? bias[(wid * NQ + q) * 16 + (ln & 15)] : 0.f;
 }
 
 // ---- in-block GEMM: out = act(in_lds[64][K] @ W + bias) ------------------
@@ -129,9 +137,14 @@ static __device__ __forceinline__ void block_gemm_pre(
     lds_short* out_lds, int out_stride, lds_short* x_lds, lds_short* vt_lds,
     int wid, int lane) {
   constexpr int KS = K / 32;
-  constexpr int NQ = N / 128;
+  constexpr int NQ = N / (16 * BF_WAVES);
 #pragma unroll
   for (int q = 0; q < NQ; ++q) {
+    // cap register pressure: without this fence the fully-unrolled quad
+    // loop interleaves every quad's 16 accumulators (192 VGPR spills at
+    // NQ=6). Mask 0x120 (VMEM_READ|DS_READ) still lets the NEXT quad's
+    // weight/activation loads hoist under THIS quad's MFMAs.
+    if (q > 0) __builtin_amdgcn_sched_barrier(0x120);
     const int fn = wid * NQ + q;
     f32x4 acc[4];
 #pragma unroll
@@ -149,11 +162,13 @@ static __device__ __forceinline__ void block_gemm_pre(
       }
     }
     // in-phase tail: loads issue under the head's MFMA chains
+    int ln = lane;
+    asm volatile("" : "+v"(ln));
 #pragma unroll
     for (int ks = PKS; ks < KS; ++ks) {
       const bf16x8 b = *(const __attribute__((address_space(1))) bf16x8*)(
-          Wt + (long)(fn * 16 + (lane & 15)) * WTS + ks * 32 +
-          (lane >> 4) * 8);
+          Wt + (long)(fn * 16 + (ln & 15)) * WTS + ks * 32 +
+          (ln >> 4) * 8);
 #pragma unroll
       for (int fm = 0; fm < 4; ++fm) {
         bf16x8 a = *(const __attribute__((address_space(3))) bf16x8*)(
@@ -196,7 +211,11 @@ static __device__ __forceinline__ void block_gemm_pre(
 static __device__ __forceinline__ void block_layernorm(
     lds_short* x_lds, const short* __restrict__ gamma,
     const short* __restrict__ beta, int wid, int lane, float eps) {
-  const int row = wid * 8 + (lane >> 3);
+  // 4 waves x 2 passes x 8 rows = the 64 rows (8 lanes per row, 16
+  // elements per lane, 3-step shfl_xor reduction per 8-lane row group)
+#pragma unroll
+  for (int rr = 0; rr < BF_S / (BF_WAVES * 8); ++rr) {
+  const int row = wid * (BF_S / BF_WAVES) + rr * 8 + (lane >> 3);
   const int c0 = (lane & 7) * 16;
   short8v va = *(const __attribute__((address_space(3))) short8v*)(x_lds + row * XS + c0);
   short8v vb = *(const __attribute__((address_space(3))) short8v*)(x_lds + row * XS + c0 + 8);
@@ -234,6 +253,7 @@ static __device__ __forceinline__ void block_layernorm(
   }
   *(__attribute__((address_space(3))) short8v*)(x_lds + row * XS + c0) = oa;
   *(__attribute__((address_space(3))) short8v*)(x_lds + row * XS + c0 + 8) = ob;
+  }
 }
 
 // Phase bits for the profiling probe (production uses PH_ALL; skipped
@@ -303,8 +323,8 @@ static __device__ __forceinline__ void bert_fused_body(
   // layer-0 qkv weight fragments: issued HERE so they are in flight
   // across the embed barrier (load_wfrags comment above)
   bf16x8 wq[6];
-  float bq[3];
-  load_wfrags<BF_H, 3 * BF_H, 2, BF_H>(
+  float bq[6];
+  load_wfrags<BF_H, 3 * BF_H, 1, BF_H>(
       (glob_cshort*)(wb + WB_LAYER0 + LW_QKV), wid, lane, wq);
   load_bias<3 * BF_H>((glob_cfloat*)(fb + FB_BQKV), wid, lane, bq);
   __syncthreads();
@@ -316,109 +336,118 @@ static __device__ __forceinline__ void bert_fused_body(
 
     // ---- qkv: Q|K -> buf[64][QKS], V -> vt transposed ----
     if (PHASES & PH_QKV)
-      block_gemm_pre<BF_H, 3 * BF_H, 2, 1, 0, BF_H>(
+      block_gemm_pre<BF_H, 3 * BF_H, 1, 1, 0, BF_H>(
           x_lds, XS, wq, (glob_cshort*)(lw + LW_QKV), bq, buf, QKS, x_lds,
           vt, wid, lane);
     BF_STAMP();  // qkv work done
     // proj weights ride across the qkv barrier + the whole attention
     // phase (attention reads no global weights)
-    bf16x8 wpr[4];
-    float bpr[1];
+    bf16x8 wpr[8];
+    float bpr[2];
     load_wfrags<BF_H, BF_H, 4, BF_H>((glob_cshort*)(lw + LW_WO), wid, lane,
                                      wpr);
     load_bias<BF_H>((glob_cfloat*)(lf + FB_BO), wid, lane, bpr);
     __syncthreads();
     BF_STAMP();  // qkv barrier crossed
 
-    // ---- attention: wave = (head hh, 16 q-rows) ----
+    // ---- attention: wave = (head hh, 32 q-rows as two 16-row groups) ----
     if (PHASES & PH_ATTN) {
-      const int hh = wid >> 2;
-      const int q0 = (wid & 3) * 16;
+      const int hh = wid >> 1;
+      const int q0 = (wid & 1) * 32;
       const float scale = 0.125f;  // 1/sqrt(64)
 
-      f32x4 acc_p[4];
+      // both groups' QK^T + softmax BEFORE the P-alias barrier (every
+      // wave must be done reading Q/K before any P store)
+      f32x4 acc_p[2][4];
+      float inv_sum[2][4];
 #pragma unroll
-      for (int f = 0; f < 4; ++f) acc_p[f] = {0.f, 0.f, 0.f, 0.f};
+      for (int g = 0; g < 2; ++g) {
 #pragma unroll
-      for (int ks = 0; ks < BF_DH / 32; ++ks) {
-        bf16x8 a = *(const __attribute__((address_space(3))) bf16x8*)(
-            buf + (q0 + (lane & 15)) * QKS + hh * BF_DH + ks * 32 +
-            (lane >> 4) * 8);
+        for (int f = 0; f < 4; ++f) acc_p[g][f] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-        for (int f = 0; f < 4; ++f) {
-          bf16x8 b = *(const __attribute__((address_space(3))) bf16x8*)(
-              buf + (f * 16 + (lane & 15)) * QKS + BF_H + hh * BF_DH +
-              ks * 32 + (lane >> 4) * 8);
-          acc_p[f] =
-              __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc_p[f], 0, 0, 0);
-        }
-      }
-      float inv_sum[4];
+        for (int ks = 0; ks < BF_DH / 32; ++ks) {
+          bf16x8 a = *(const __attribute__((address_space(3))) bf16x8*)(
+              buf + (q0 + g * 16 + (lane & 15)) * QKS + hh * BF_DH + ks * 32 +
+              (lane >> 4) * 8);
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        float m = -1e30f;
-#pragma unroll
-        for (int f = 0; f < 4; ++f) m = fmaxf(m, acc_p[f][r] * scale);
-#pragma unroll
-        for (int mask = 1; mask < 16; mask <<= 1)
-          m = fmaxf(m, __shfl_xor(m, mask, 64));
-        float sum = 0.f;
-#pragma unroll
-        for (int f = 0; f < 4; ++f) {
-          const float e = __expf(acc_p[f][r] * scale - m);
-          acc_p[f][r] = e;
-          sum += e;
+          for (int f = 0; f < 4; ++f) {
+            bf16x8 b = *(const __attribute__((address_space(3))) bf16x8*)(
+                buf + (f * 16 + (lane & 15)) * QKS + BF_H + hh * BF_DH +
+                ks * 32 + (lane >> 4) * 8);
+            acc_p[g][f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a, b, acc_p[g][f], 0, 0, 0);
+          }
         }
 #pragma unroll
-        for (int mask = 1; mask < 16; mask <<= 1)
-          sum += __shfl_xor(sum, mask, 64);
-        inv_sum[r] = 1.f / sum;
+        for (int r = 0; r < 4; ++r) {
+          float m = -1e30f;
+#pragma unroll
+          for (int f = 0; f < 4; ++f) m = fmaxf(m, acc_p[g][f][r] * scale);
+#pragma unroll
+          for (int mask = 1; mask < 16; mask <<= 1)
+            m = fmaxf(m, __shfl_xor(m, mask, 64));
+          float sum = 0.f;
+#pragma unroll
+          for (int f = 0; f < 4; ++f) {
+            const float e = __expf(acc_p[g][f][r] * scale - m);
+            acc_p[g][f][r] = e;
+            sum += e;
+          }
+#pragma unroll
+          for (int mask = 1; mask < 16; mask <<= 1)
+            sum += __shfl_xor(sum, mask, 64);
+          inv_sum[g][r] = 1.f / sum;
+        }
       }
       // P tiles alias the Q|K area: every wave must be done reading Q/K
       __syncthreads();
-      lds_short* my_p = buf + wid * 16 * VTS;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int row = (lane >> 4) * 4 + r;
+      for (int g = 0; g < 2; ++g) {
+        lds_short* my_p = buf + (wid * 32 + g * 16) * VTS;
 #pragma unroll
-        for (int f = 0; f < 4; ++f)
-          my_p[row * VTS + f * 16 + (lane & 15)] = f32_to_bf16(acc_p[f][r]);
-      }
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // wave-local P
-
-      f32x4 acc_o[4];
+        for (int r = 0; r < 4; ++r) {
+          const int row = (lane >> 4) * 4 + r;
 #pragma unroll
-      for (int f = 0; f < 4; ++f) acc_o[f] = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-      for (int ks = 0; ks < BF_S / 32; ++ks) {
-        bf16x8 a = *(const __attribute__((address_space(3))) bf16x8*)(
-            my_p + (lane & 15) * VTS + ks * 32 + (lane >> 4) * 8);
-#pragma unroll
-        for (int f = 0; f < 4; ++f) {
-          bf16x8 b = *(const __attribute__((address_space(3))) bf16x8*)(
-              vt + (hh * BF_DH + f * 16 + (lane & 15)) * VTS + ks * 32 +
-              (lane >> 4) * 8);
-          acc_o[f] =
-              __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc_o[f], 0, 0, 0);
+          for (int f = 0; f < 4; ++f)
+            my_p[row * VTS + f * 16 + (lane & 15)] =
+                f32_to_bf16(acc_p[g][f][r]);
         }
-      }
-      // attn out -> buf[O_OFF + q*XS + hh*64 + d] (disjoint from P tiles)
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // wave-local P
+
+        f32x4 acc_o[4];
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int q = q0 + (lane >> 4) * 4 + r;
-        const float inv = __shfl(inv_sum[r], (lane >> 4) * 4 + r, 64);
+        for (int f = 0; f < 4; ++f) acc_o[f] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-        for (int f = 0; f < 4; ++f) {
-          const int d = f * 16 + (lane & 15);
-          buf[O_OFF + q * XS + hh * BF_DH + d] =
-              f32_to_bf16(acc_o[f][r] * inv);
+        for (int ks = 0; ks < BF_S / 32; ++ks) {
+          bf16x8 a = *(const __attribute__((address_space(3))) bf16x8*)(
+              my_p + (lane & 15) * VTS + ks * 32 + (lane >> 4) * 8);
+#pragma unroll
+          for (int f = 0; f < 4; ++f) {
+            bf16x8 b = *(const __attribute__((address_space(3))) bf16x8*)(
+                vt + (hh * BF_DH + f * 16 + (lane & 15)) * VTS + ks * 32 +
+                (lane >> 4) * 8);
+            acc_o[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a, b, acc_o[f], 0, 0, 0);
+          }
+        }
+        // attn out -> buf[O_OFF + q*XS + hh*64 + d] (disjoint from P tiles)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int q = q0 + g * 16 + (lane >> 4) * 4 + r;
+          const float inv = __shfl(inv_sum[g][r], (lane >> 4) * 4 + r, 64);
+#pragma unroll
+          for (int f = 0; f < 4; ++f) {
+            const int d = f * 16 + (lane & 15);
+            buf[O_OFF + q * XS + hh * BF_DH + d] =
+                f32_to_bf16(acc_o[f][r] * inv);
+          }
         }
       }
     }
     BF_STAMP();  // attention work done
     // FFN W1 half-0 fragments cross the attn barrier + proj + LN1
-    bf16x8 w1[4];
-    float b1[2];
+    bf16x8 w1[8];
+    float b1[4];
     load_wfrags<BF_H, BF_FFN / 2, 2, BF_H>((glob_cshort*)(lw + LW_W1), wid,
                                            lane, w1);
     load_bias<BF_FFN / 2>((glob_cfloat*)(lf + FB_B1), wid, lane, b1);
@@ -431,8 +460,8 @@ static __device__ __forceinline__ void bert_fused_body(
           buf + O_OFF, XS, wpr, (glob_cshort*)(lw + LW_WO), bpr, nullptr, 0,
           x_lds, nullptr, wid, lane);
     // W2 half-0 fragments cross the proj barrier + LN1 + W1h0
-    bf16x8 w2[4];
-    float b2[1];
+    bf16x8 w2[8];
+    float b2[2];
     load_wfrags<BF_FFN / 2, BF_H, 4, BF_FFN>((glob_cshort*)(lw + LW_W2), wid,
                                              lane, w2);
     load_bias<BF_H>((glob_cfloat*)(lf + FB_B2), wid, lane, b2);
@@ -461,7 +490,7 @@ static __device__ __forceinline__ void bert_fused_body(
           nullptr, wid, lane);
       load_wfrags<BF_FFN / 2, BF_H, 4, BF_FFN>(
           (glob_cshort*)(lw + LW_W2 + (BF_FFN / 2)), wid, lane, w2);
-      b2[0] = 0.f;  // half 1 adds only the partial product
+      b2[0] = b2[1] = 0.f;  // half 1 adds only the partial product
       __syncthreads();
       block_gemm_pre<BF_H, BF_FFN / 2, 2, 0, 1, BF_H>(
           x_lds, XS, w1,
@@ -474,7 +503,7 @@ static __device__ __forceinline__ void bert_fused_body(
     }
     // next layer's qkv fragments cross the FFN-tail barrier + LN2
     if (layer + 1 < n_layers) {
-      load_wfrags<BF_H, 3 * BF_H, 2, BF_H>(
+      load_wfrags<BF_H, 3 * BF_H, 1, BF_H>(
           (glob_cshort*)(lw + LW_SIZE + LW_QKV), wid, lane, wq);
       load_bias<3 * BF_H>((glob_cfloat*)(lf + FB_SIZE + FB_BQKV), wid, lane,
                           bq);
