@@ -69,135 +69,71 @@ typedef __attribute__((address_space(3))) float lds_float;
 #define FB_B2 (FB_B1 + BF_FFN)
 #define FB_SIZE (FB_B2 + BF_H)
 
-// ---- cross-barrier weight prefetch (v5) ----------------------------------
-// Round-1 PMC evidence (profiles/r08): SQ_WAIT_ANY ~ 18x SQ_BUSY — waves
-// spend ~95% of residency parked, and five issue-side ablations were
-// neutral. The wall is structural: every GEMM phase opened with a
-// DEPENDENT chain of L2 weight-fragment loads issued AFTER the barrier
-// that every wave crosses together, so the whole CU idles ~KS x 200
-// cycles per phase before the first MFMA can retire. Weights are
-// read-only, so those loads have NO hazard against the barrier — but the
-// compiler cannot hoist loads across __syncthreads(). v5 does it by
-// hand: each GEMM's full B-fragment set is issued into registers during
-// the PREVIOUS phase and consumed after the barrier, so phases open
-// MFMA-ready (the s_waitcnt lands ~a whole phase after issue). Costs
-// VGPRs (fragments held across a phase): launch bounds drop to
-// 2 waves/SIMD (<=256 VGPR, 1 block/CU) — round-1 measurements showed
-// throughput FLAT in occupancy 2-6 waves/SIMD, so the co-residency this
-// gives up was not buying anything the prefetch doesn't replace.
+// ---- v7: ROW-partitioned waves (the structural fix) ----------------------
+// Rounds 1-2 history: the N-partitioned design (every wave computes a
+// column stripe of every GEMM, so every GEMM output feeds every wave's
+// next input) forced ~10 block-wide barriers per layer; PMC showed waves
+// parked ~95% of residency (SQ_WAIT_ANY ~ 18x SQ_BUSY, profiles/r08) and
+// cross-barrier register prefetch variants (v5/v6) measured 5.97-6.27M
+// lines/s vs 7.09M for the round-1 kernel — the barrier structure, not
+// the loads, was the wall.
 //
-// Per wave a GEMM owns N/128 "quads" (quad = one fn column x 4
-// m-fragments); fragments per wave = (N/128) * (K/32).
-// PKS = fragments per quad prefetched across the barrier (the rest
-// stream in-phase under MFMA cover): full prefetch of every GEMM spilled
-// 17 VGPRs in-loop; phase-START latency only needs the first fragments
-// resident, later ks-steps hide behind ~136 cycles of MFMA per step.
-template <int K, int N, int PKS, int WTS>
-static __device__ __forceinline__ void load_wfrags(
-    const glob_cshort* __restrict__ Wt, int wid, int lane, bf16x8* out) {
-  constexpr int NQ = N / (16 * BF_WAVES);
-  // opaque lane: the per-lane address offsets are layer-invariant and
-  // LICM otherwise hoists ~25 of them out of the layer loop, spilling
-  // them across every phase (the guide's "recompute per block" pitfall);
-  // recomputing costs a couple of VALU per phase.
-  int ln = lane;
-  asm volatile("" : "+v"(ln));
-#pragma unroll
-  for (int q = 0; q < NQ; ++q)
-#pragma unroll
-    for (int ks = 0; ks < PKS; ++ks)
-      out[q * PKS + ks] = *(const __attribute__((address_space(1))) bf16x8*)(
-          Wt + (long)((wid * NQ + q) * 16 + (ln & 15)) * WTS + ks * 32 +
-          (ln >> 4) * 8);
-}
-
-// Biases ride along with the fragments: an in-phase bias load would make
-// the compiler emit a vmcnt(0) drain in the GEMM epilogue, flushing the
-// NEXT phase's just-issued prefetch loads with it.
-template <int N>
-static __device__ __forceinline__ void load_bias(
-    const glob_cfloat* __restrict__ bias, int wid, int lane, float* out) {
-  constexpr int NQ = N / (16 * BF_WAVES);
-  int ln = lane;
-  asm volatile("" : "+v"(ln));
-#pragma unroll
-  for (int q = 0; q < NQ; ++q)
-    out[q] = bias ? bias[(wid * NQ + q) * 16 + (ln & 15)] : 0.f;
-}
-
-// ---- in-block GEMM: out = act(in_lds[64][K] @ W + bias) ------------------
-// Weight fragments arrive PRELOADED in registers (load_wfrags, issued
-// before the preceding barrier). MODE 0: write out_lds[m][n]; MODE 1
-// (qkv): n<2H -> out (Q|K), else vt transposed; MODE 2: x[m][n] += v
-// (residual-accumulate). ACT 1 = GELU.
-template <int K, int N, int PKS, int MODE, int ACT, int WTS>
-static __device__ __forceinline__ void block_gemm_pre(
-    const lds_short* in_lds, int in_stride, const bf16x8* w,
-    const glob_cshort* __restrict__ Wt, const float* bias_pre,
+// v7 partitions by ROWS: wave g owns token rows [16g, 16g+16). Since
+// out[m][:] of every GEMM depends only on in[m][:], the whole chain
+// qkv -> attn-out -> proj -> LN1 -> FFN -> LN2 -> next qkv is WAVE-LOCAL
+// except attention itself (queries attend to all keys). Exactly TWO
+// __syncthreads per layer remain: after qkv (K|V visible to all waves)
+// and after attention (buf/vt reusable). A-fragments load once per GEMM
+// and stay in registers for every n-fragment; B-fragments stream from
+// the XCD L2 with no barrier anywhere to trap the pipeline.
+template <int K, int N, int MODE, int ACT, int WTS>
+static __device__ __forceinline__ void row_gemm(
+    const lds_short* in_lds, int in_stride, int m0,
+    const glob_cshort* __restrict__ Wt, const glob_cfloat* __restrict__ bias,
     lds_short* out_lds, int out_stride, lds_short* x_lds, lds_short* vt_lds,
-    int wid, int lane) {
+    int lane) {
   constexpr int KS = K / 32;
-  constexpr int NQ = N / (16 * BF_WAVES);
+  constexpr int NF = N / 16;
+  // opaque lane: per-lane address offsets are layer-invariant; without
+  // this LICM hoists them out of the layer loop and spills them
+  int ln = lane;
+  asm volatile("" : "+v"(ln));
+  // the wave's A-fragments: row m0+(ln&15), all K — loaded ONCE
+  bf16x8 a[KS];
 #pragma unroll
-  for (int q = 0; q < NQ; ++q) {
-    // cap register pressure: without this fence the fully-unrolled quad
-    // loop interleaves every quad's 16 accumulators (192 VGPR spills at
-    // NQ=6). Mask 0x120 (VMEM_READ|DS_READ) still lets the NEXT quad's
-    // weight/activation loads hoist under THIS quad's MFMAs.
-    if (q > 0) __builtin_amdgcn_sched_barrier(0x120);
-    const int fn = wid * NQ + q;
-    f32x4 acc[4];
+  for (int ks = 0; ks < KS; ++ks)
+    a[ks] = *(const __attribute__((address_space(3))) bf16x8*)(
+        in_lds + (m0 + (ln & 15)) * in_stride + ks * 32 + (ln >> 4) * 8);
 #pragma unroll
-    for (int i = 0; i < 4; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
-    // prefetched head fragments
+  for (int nf = 0; nf < NF; ++nf) {
+    // pressure cap: keep MFMA/VALU of n-fragments ordered, let the NEXT
+    // fragment's weight loads hoist under THIS fragment's MFMA chain
+    if ((nf & 1) == 0 && nf > 0) __builtin_amdgcn_sched_barrier(0x120);
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-    for (int ks = 0; ks < PKS; ++ks) {
-      const bf16x8 b = w[q * PKS + ks];
-#pragma unroll
-      for (int fm = 0; fm < 4; ++fm) {
-        bf16x8 a = *(const __attribute__((address_space(3))) bf16x8*)(
-            in_lds + (fm * 16 + (lane & 15)) * in_stride + ks * 32 +
-            (lane >> 4) * 8);
-        acc[fm] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[fm], 0, 0, 0);
-      }
-    }
-    // in-phase tail: loads issue under the head's MFMA chains
-    int ln = lane;
-    asm volatile("" : "+v"(ln));
-#pragma unroll
-    for (int ks = PKS; ks < KS; ++ks) {
+    for (int ks = 0; ks < KS; ++ks) {
       const bf16x8 b = *(const __attribute__((address_space(1))) bf16x8*)(
-          Wt + (long)(fn * 16 + (ln & 15)) * WTS + ks * 32 +
-          (ln >> 4) * 8);
-#pragma unroll
-      for (int fm = 0; fm < 4; ++fm) {
-        bf16x8 a = *(const __attribute__((address_space(3))) bf16x8*)(
-            in_lds + (fm * 16 + (lane & 15)) * in_stride + ks * 32 +
-            (lane >> 4) * 8);
-        acc[fm] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[fm], 0, 0, 0);
-      }
+          Wt + (long)(nf * 16 + (ln & 15)) * WTS + ks * 32 + (ln >> 4) * 8);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[ks], b, acc, 0, 0, 0);
     }
-    const int n = fn * 16 + (lane & 15);
-    const float bval = bias_pre[q];
+    const int n = nf * 16 + (ln & 15);
+    const float bval = bias ? bias[n] : 0.f;
 #pragma unroll
-    for (int fm = 0; fm < 4; ++fm) {
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int m = fm * 16 + (lane >> 4) * 4 + r;
-        float v = acc[fm][r] + bval;
-        if (MODE == 0) {
-          if (ACT == 1) v = gelu_f32(v);
+    for (int r = 0; r < 4; ++r) {
+      const int m = m0 + (ln >> 4) * 4 + r;
+      float v = acc[r] + bval;
+      if (MODE == 0) {
+        if (ACT == 1) v = gelu_f32(v);
+        out_lds[m * out_stride + n] = f32_to_bf16(v);
+      } else if (MODE == 1) {
+        if (n < 2 * BF_H) {
           out_lds[m * out_stride + n] = f32_to_bf16(v);
-        } else if (MODE == 1) {
-          if (n < 2 * BF_H) {
-            out_lds[m * out_stride + n] = f32_to_bf16(v);
-          } else {
-            vt_lds[(n - 2 * BF_H) * VTS + m] = f32_to_bf16(v);  // V transposed
-          }
-        } else {  // MODE 2
-          const float xv = bf16_to_f32(x_lds[m * XS + n]);
-          x_lds[m * XS + n] = f32_to_bf16(v + xv);
+        } else {
+          vt_lds[(n - 2 * BF_H) * VTS + m] = f32_to_bf16(v);  // V transposed
         }
+      } else {  // MODE 2
+        const float xv = bf16_to_f32(x_lds[m * XS + n]);
+        x_lds[m * XS + n] = f32_to_bf16(v + xv);
       }
     }
   }
@@ -301,116 +237,100 @@ static __device__ __forceinline__ void bert_fused_body(
   lds_float* red = (lds_float*)(vt + 2 * BF_DH * VTS);  // [128] pooling
 
   BF_STAMP();  // 0: kernel start
-  // ---- embed: x[s][c] = tok_emb[byte+3 or 0][c] + pos_emb[s][c] ----
+  const int m0 = wid * 16;  // the wave's own token rows [m0, m0+16)
+
+  // ---- embed (own rows): x[s][c] = tok_emb[byte+3 or 0][c] + pos[s][c] ----
   {
     const int s0 = start[line], e0 = end[line];
-    for (int i = tid * 8; i < BF_S * BF_H; i += BF_THREADS * 8) {
-      const int s = i / BF_H, c = i % BF_H;
+#pragma unroll
+    for (int it = 0; it < 16 * BF_H / (DMX_WAVE * 8); ++it) {
+      const int i = m0 * BF_H + it * DMX_WAVE * 8 + lane * 8;
+      const int sr = i / BF_H, c = i % BF_H;
       int tok = 0;
-      const int idx = s0 + s;
+      const int idx = s0 + sr;
       if (idx < e0 && idx < max_len)
         tok = (int)lines[(long)line * max_len + idx] + 3;
       short8v te = *(const short8v*)(wb + WB_TOK + (long)tok * BF_H + c);
-      short8v pe = *(const short8v*)(wb + WB_POS + (long)s * BF_H + c);
+      short8v pe = *(const short8v*)(wb + WB_POS + (long)sr * BF_H + c);
       short8v xv;
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         xv[j] = f32_to_bf16(bf16_to_f32(te[j]) + bf16_to_f32(pe[j]));
-      *(__attribute__((address_space(3))) short8v*)(x_lds + s * XS + c) = xv;
+      *(__attribute__((address_space(3))) short8v*)(x_lds + sr * XS + c) = xv;
     }
   }
-  BF_STAMP();  // 1: embed work done
-  // layer-0 qkv weight fragments: issued HERE so they are in flight
-  // across the embed barrier (load_wfrags comment above)
-  bf16x8 wq[6];
-  float bq[6];
-  load_wfrags<BF_H, 3 * BF_H, 1, BF_H>(
-      (glob_cshort*)(wb + WB_LAYER0 + LW_QKV), wid, lane, wq);
-  load_bias<3 * BF_H>((glob_cfloat*)(fb + FB_BQKV), wid, lane, bq);
-  __syncthreads();
-  BF_STAMP();  // 2: embed barrier crossed
+  BF_STAMP();  // 1: embed done (row-local: NO barrier before qkv)
+  BF_STAMP();  // 2: (slot kept for stamp-layout compatibility)
 
   for (int layer = 0; layer < n_layers; ++layer) {
     const short* lw = wb + WB_LAYER0 + (long)layer * LW_SIZE;
     const float* lf = fb + (long)layer * FB_SIZE;
 
-    // ---- qkv: Q|K -> buf[64][QKS], V -> vt transposed ----
+    // ---- qkv (row-local): Q|K -> buf[own rows], V -> vt own columns ----
     if (PHASES & PH_QKV)
-      block_gemm_pre<BF_H, 3 * BF_H, 1, 1, 0, BF_H>(
-          x_lds, XS, wq, (glob_cshort*)(lw + LW_QKV), bq, buf, QKS, x_lds,
-          vt, wid, lane);
+      row_gemm<BF_H, 3 * BF_H, 1, 0, BF_H>(
+          x_lds, XS, m0, (glob_cshort*)(lw + LW_QKV),
+          (glob_cfloat*)(lf + FB_BQKV), buf, QKS, x_lds, vt, lane);
     BF_STAMP();  // qkv work done
-    // proj weights ride across the qkv barrier + the whole attention
-    // phase (attention reads no global weights)
-    bf16x8 wpr[8];
-    float bpr[2];
-    load_wfrags<BF_H, BF_H, 4, BF_H>((glob_cshort*)(lw + LW_WO), wid, lane,
-                                     wpr);
-    load_bias<BF_H>((glob_cfloat*)(lf + FB_BO), wid, lane, bpr);
-    __syncthreads();
+    __syncthreads();  // K|V of ALL rows visible (barrier 1 of 2)
     BF_STAMP();  // qkv barrier crossed
 
-    // ---- attention: wave = (head hh, 32 q-rows as two 16-row groups) ----
+    // ---- attention (own 16 q-rows, both heads sequentially) ----
     if (PHASES & PH_ATTN) {
-      const int hh = wid >> 1;
-      const int q0 = (wid & 1) * 32;
       const float scale = 0.125f;  // 1/sqrt(64)
-
-      // both groups' QK^T + softmax BEFORE the P-alias barrier (every
-      // wave must be done reading Q/K before any P store)
-      f32x4 acc_p[2][4];
-      float inv_sum[2][4];
 #pragma unroll
-      for (int g = 0; g < 2; ++g) {
+      for (int hh = 0; hh < 2; ++hh) {
+        f32x4 acc_p[4];
 #pragma unroll
-        for (int f = 0; f < 4; ++f) acc_p[g][f] = {0.f, 0.f, 0.f, 0.f};
+        for (int f = 0; f < 4; ++f) acc_p[f] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int ks = 0; ks < BF_DH / 32; ++ks) {
           bf16x8 a = *(const __attribute__((address_space(3))) bf16x8*)(
-              buf + (q0 + g * 16 + (lane & 15)) * QKS + hh * BF_DH + ks * 32 +
+              buf + (m0 + (lane & 15)) * QKS + hh * BF_DH + ks * 32 +
               (lane >> 4) * 8);
 #pragma unroll
           for (int f = 0; f < 4; ++f) {
             bf16x8 b = *(const __attribute__((address_space(3))) bf16x8*)(
                 buf + (f * 16 + (lane & 15)) * QKS + BF_H + hh * BF_DH +
                 ks * 32 + (lane >> 4) * 8);
-            acc_p[g][f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                a, b, acc_p[g][f], 0, 0, 0);
+            acc_p[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a, b, acc_p[f], 0, 0, 0);
           }
         }
+        float inv_sum[4];
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           float m = -1e30f;
 #pragma unroll
-          for (int f = 0; f < 4; ++f) m = fmaxf(m, acc_p[g][f][r] * scale);
+          for (int f = 0; f < 4; ++f) m = fmaxf(m, acc_p[f][r] * scale);
 #pragma unroll
           for (int mask = 1; mask < 16; mask <<= 1)
             m = fmaxf(m, __shfl_xor(m, mask, 64));
           float sum = 0.f;
 #pragma unroll
           for (int f = 0; f < 4; ++f) {
-            const float e = __expf(acc_p[g][f][r] * scale - m);
-            acc_p[g][f][r] = e;
+            const float e = __expf(acc_p[f][r] * scale - m);
+            acc_p[f][r] = e;
             sum += e;
           }
 #pragma unroll
           for (int mask = 1; mask < 16; mask <<= 1)
             sum += __shfl_xor(sum, mask, 64);
-          inv_sum[g][r] = 1.f / sum;
+          inv_sum[r] = 1.f / sum;
         }
-      }
-      // P tiles alias the Q|K area: every wave must be done reading Q/K
-      __syncthreads();
-#pragma unroll
-      for (int g = 0; g < 2; ++g) {
-        lds_short* my_p = buf + (wid * 32 + g * 16) * VTS;
+        // P tile overwrites the wave's OWN Q slice, head hh's 64 dims:
+        // Q[own rows][hh*64..hh*64+64) is dead once THIS head's QK^T is
+        // done (head 1 reads cols 64..128, P(h0) goes to cols 0..64) —
+        // wave-local, no barrier. NOTE head 1's P overwrites cols 64..128
+        // AFTER its own QK^T consumed them.
+        lds_short* my_p = buf + m0 * QKS + hh * BF_DH;
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int row = (lane >> 4) * 4 + r;
 #pragma unroll
           for (int f = 0; f < 4; ++f)
-            my_p[row * VTS + f * 16 + (lane & 15)] =
-                f32_to_bf16(acc_p[g][f][r]);
+            my_p[row * QKS + f * 16 + (lane & 15)] =
+                f32_to_bf16(acc_p[f][r]);
         }
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // wave-local P
 
@@ -420,7 +340,7 @@ static __device__ __forceinline__ void bert_fused_body(
 #pragma unroll
         for (int ks = 0; ks < BF_S / 32; ++ks) {
           bf16x8 a = *(const __attribute__((address_space(3))) bf16x8*)(
-              my_p + (lane & 15) * VTS + ks * 32 + (lane >> 4) * 8);
+              my_p + (lane & 15) * QKS + ks * 32 + (lane >> 4) * 8);
 #pragma unroll
           for (int f = 0; f < 4; ++f) {
             bf16x8 b = *(const __attribute__((address_space(3))) bf16x8*)(
@@ -430,91 +350,58 @@ static __device__ __forceinline__ void bert_fused_body(
                 a, b, acc_o[f], 0, 0, 0);
           }
         }
-        // attn out -> buf[O_OFF + q*XS + hh*64 + d] (disjoint from P tiles)
+        // attn out overwrites the wave's OWN Q columns for this head
+        // (P(hh) there was fully consumed by the PV reads above; DS ops
+        // of one wave execute in order, so the write-after-read is safe
+        // without a wait). Other waves never read MY Q slice.
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
-          const int q = q0 + g * 16 + (lane >> 4) * 4 + r;
-          const float inv = __shfl(inv_sum[g][r], (lane >> 4) * 4 + r, 64);
+          const int q = m0 + (lane >> 4) * 4 + r;
+          const float inv = __shfl(inv_sum[r], (lane >> 4) * 4 + r, 64);
 #pragma unroll
           for (int f = 0; f < 4; ++f) {
             const int d = f * 16 + (lane & 15);
-            buf[O_OFF + q * XS + hh * BF_DH + d] =
-                f32_to_bf16(acc_o[f][r] * inv);
+            buf[q * QKS + hh * BF_DH + d] = f32_to_bf16(acc_o[f][r] * inv);
           }
         }
       }
     }
     BF_STAMP();  // attention work done
-    // FFN W1 half-0 fragments cross the attn barrier + proj + LN1
-    bf16x8 w1[8];
-    float b1[4];
-    load_wfrags<BF_H, BF_FFN / 2, 2, BF_H>((glob_cshort*)(lw + LW_W1), wid,
-                                           lane, w1);
-    load_bias<BF_FFN / 2>((glob_cfloat*)(lf + FB_B1), wid, lane, b1);
-    __syncthreads();
+    __syncthreads();  // buf/vt reusable by row-local phases (barrier 2 of 2)
     BF_STAMP();  // attention barrier crossed
 
-    // ---- proj: x += Wo(attn) ; LN1 ----
+    // ---- proj (row-local): x += Wo(attn-out in own Q cols) ; LN1 ----
     if (PHASES & PH_PROJ)
-      block_gemm_pre<BF_H, BF_H, 4, 2, 0, BF_H>(
-          buf + O_OFF, XS, wpr, (glob_cshort*)(lw + LW_WO), bpr, nullptr, 0,
-          x_lds, nullptr, wid, lane);
-    // W2 half-0 fragments cross the proj barrier + LN1 + W1h0
-    bf16x8 w2[8];
-    float b2[2];
-    load_wfrags<BF_FFN / 2, BF_H, 4, BF_FFN>((glob_cshort*)(lw + LW_W2), wid,
-                                             lane, w2);
-    load_bias<BF_H>((glob_cfloat*)(lf + FB_B2), wid, lane, b2);
-    __syncthreads();
+      row_gemm<BF_H, BF_H, 2, 0, BF_H>(
+          buf, QKS, m0, (glob_cshort*)(lw + LW_WO),
+          (glob_cfloat*)(lf + FB_BO), nullptr, 0, x_lds, nullptr, lane);
     if (PHASES & PH_LN)
       block_layernorm(x_lds, lw + LW_LN1G, lw + LW_LN1B, wid, lane, eps);
     BF_STAMP();  // proj+LN1 work done
-    __syncthreads();
-    BF_STAMP();  // proj+LN1 barrier crossed
+    BF_STAMP();  // (stamp slot)
 
-    // ---- FFN in two K=256 halves: buf = gelu(x@W1_h); x += buf@W2_h ----
+    // ---- FFN (row-local), two K=256 halves sharing buf[own rows] ----
     if (PHASES & PH_FFN) {
-      block_gemm_pre<BF_H, BF_FFN / 2, 2, 0, 1, BF_H>(
-          x_lds, XS, w1, (glob_cshort*)(lw + LW_W1), b1, buf, QKS, nullptr,
-          nullptr, wid, lane);
-      // W1 half-1 fragments reuse w1's registers (h0 consumed above)
-      load_wfrags<BF_H, BF_FFN / 2, 2, BF_H>(
-          (glob_cshort*)(lw + LW_W1 + (long)(BF_FFN / 2) * BF_H), wid, lane,
-          w1);
-      load_bias<BF_FFN / 2>((glob_cfloat*)(lf + FB_B1 + BF_FFN / 2), wid,
-                            lane, b1);
-      __syncthreads();
-      // bias b2 added once (half 0); half 1 adds only the partial product
-      block_gemm_pre<BF_FFN / 2, BF_H, 4, 2, 0, BF_FFN>(
-          buf, QKS, w2, (glob_cshort*)(lw + LW_W2), b2, nullptr, 0, x_lds,
-          nullptr, wid, lane);
-      load_wfrags<BF_FFN / 2, BF_H, 4, BF_FFN>(
-          (glob_cshort*)(lw + LW_W2 + (BF_FFN / 2)), wid, lane, w2);
-      b2[0] = b2[1] = 0.f;  // half 1 adds only the partial product
-      __syncthreads();
-      block_gemm_pre<BF_H, BF_FFN / 2, 2, 0, 1, BF_H>(
-          x_lds, XS, w1,
-          (glob_cshort*)(lw + LW_W1 + (long)(BF_FFN / 2) * BF_H), b1, buf,
-          QKS, nullptr, nullptr, wid, lane);
-      __syncthreads();
-      block_gemm_pre<BF_FFN / 2, BF_H, 4, 2, 0, BF_FFN>(
-          buf, QKS, w2, (glob_cshort*)(lw + LW_W2 + (BF_FFN / 2)), b2,
-          nullptr, 0, x_lds, nullptr, wid, lane);
+#pragma unroll
+      for (int h = 0; h < 2; ++h) {
+        row_gemm<BF_H, BF_FFN / 2, 0, 1, BF_H>(
+            x_lds, XS, m0,
+            (glob_cshort*)(lw + LW_W1 + (long)h * (BF_FFN / 2) * BF_H),
+            (glob_cfloat*)(lf + FB_B1 + h * (BF_FFN / 2)), buf, QKS, nullptr,
+            nullptr, lane);
+        // W2 half h reads buf[own rows] written just above: wave-local
+        row_gemm<BF_FFN / 2, BF_H, 2, 0, BF_FFN>(
+            buf, QKS, m0, (glob_cshort*)(lw + LW_W2 + (long)h * (BF_FFN / 2)),
+            h == 0 ? (glob_cfloat*)(lf + FB_B2) : nullptr, nullptr, 0, x_lds,
+            nullptr, lane);
+      }
     }
-    // next layer's qkv fragments cross the FFN-tail barrier + LN2
-    if (layer + 1 < n_layers) {
-      load_wfrags<BF_H, 3 * BF_H, 1, BF_H>(
-          (glob_cshort*)(lw + LW_SIZE + LW_QKV), wid, lane, wq);
-      load_bias<3 * BF_H>((glob_cfloat*)(lf + FB_SIZE + FB_BQKV), wid, lane,
-                          bq);
-    }
-    __syncthreads();
     if (PHASES & PH_LN)
       block_layernorm(x_lds, lw + LW_LN2G, lw + LW_LN2B, wid, lane, eps);
     BF_STAMP();  // ffn+LN2 work done
-    __syncthreads();
-    BF_STAMP();  // ffn+LN2 barrier crossed
+    BF_STAMP();  // (stamp slot)
   }
+  __syncthreads();  // pool reads every row
 
   // ---- pool (mean over S) + score head ----
   {
