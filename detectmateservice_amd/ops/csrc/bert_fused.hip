@@ -35,7 +35,7 @@ typedef __attribute__((address_space(1))) const float glob_cfloat;
 typedef __attribute__((address_space(3))) const short lds_cshort;
 typedef __attribute__((address_space(3))) float lds_float;
 
-#define BF_WAVES 4
+#define BF_WAVES 8
 #define BF_THREADS (BF_WAVES * DMX_WAVE)
 #define BF_S 64
 #define BF_H 128
@@ -46,7 +46,7 @@ typedef __attribute__((address_space(3))) float lds_float;
 #define XS (BF_H + 8)       // 136, x rows
 #define QKS (2 * BF_H + 8)  // 264, Q|K rows and FFN-half rows
 #define VTS (BF_S + 8)      // 72, transposed-V and P rows
-#define O_OFF (BF_WAVES * 32 * VTS)    // 9216: attn-out after the P tiles
+#define O_OFF (BF_WAVES * 16 * VTS)    // 9216: attn-out after the P tiles
 #define BUF_ELEMS (BF_S * QKS + 1024)  // 17920 = max(QK, P+O, FFN-half)
 
 // bf16 weight-blob element offsets (host packs identically: bert_tiny.py)
@@ -69,71 +69,64 @@ typedef __attribute__((address_space(3))) float lds_float;
 #define FB_B2 (FB_B1 + BF_FFN)
 #define FB_SIZE (FB_B2 + BF_H)
 
-// ---- v7: ROW-partitioned waves (the structural fix) ----------------------
-// Rounds 1-2 history: the N-partitioned design (every wave computes a
-// column stripe of every GEMM, so every GEMM output feeds every wave's
-// next input) forced ~10 block-wide barriers per layer; PMC showed waves
-// parked ~95% of residency (SQ_WAIT_ANY ~ 18x SQ_BUSY, profiles/r08) and
-// cross-barrier register prefetch variants (v5/v6) measured 5.97-6.27M
-// lines/s vs 7.09M for the round-1 kernel — the barrier structure, not
-// the loads, was the wall.
-//
-// v7 partitions by ROWS: wave g owns token rows [16g, 16g+16). Since
-// out[m][:] of every GEMM depends only on in[m][:], the whole chain
-// qkv -> attn-out -> proj -> LN1 -> FFN -> LN2 -> next qkv is WAVE-LOCAL
-// except attention itself (queries attend to all keys). Exactly TWO
-// __syncthreads per layer remain: after qkv (K|V visible to all waves)
-// and after attention (buf/vt reusable). A-fragments load once per GEMM
-// and stay in registers for every n-fragment; B-fragments stream from
-// the XCD L2 with no barrier anywhere to trap the pipeline.
+// ---- in-block GEMM: out = act(in_lds[64][K] @ Wt[N][WTS] + bias) ---------
+// MODE 0: write out_lds[m][n]; MODE 1 (qkv): n<2H -> out (Q|K), else vt
+// transposed; MODE 2: x[m][n] += v (residual-accumulate). ACT 1 = GELU.
 template <int K, int N, int MODE, int ACT, int WTS>
-static __device__ __forceinline__ void row_gemm(
-    const lds_short* in_lds, int in_stride, int m0,
-    const glob_cshort* __restrict__ Wt, const glob_cfloat* __restrict__ bias,
-    lds_short* out_lds, int out_stride, lds_short* x_lds, lds_short* vt_lds,
-    int lane) {
+static __device__ __attribute__((noinline)) void block_gemm(
+    const lds_short* in_lds, int in_stride, const glob_cshort* __restrict__ Wt,
+    const glob_cfloat* __restrict__ bias, lds_short* out_lds, int out_stride,
+    lds_short* x_lds, lds_short* vt_lds, int wid, int lane) {
+  constexpr int N16 = N / 16;
+  constexpr int TOTAL = 4 * N16;
   constexpr int KS = K / 32;
-  constexpr int NF = N / 16;
-  // opaque lane: per-lane address offsets are layer-invariant; without
-  // this LICM hoists them out of the layer loop and spills them
-  int ln = lane;
-  asm volatile("" : "+v"(ln));
-  // the wave's A-fragments: row m0+(ln&15), all K — loaded ONCE
-  bf16x8 a[KS];
+  constexpr int FPW = TOTAL / BF_WAVES;
+  static_assert(FPW >= 4 && FPW % 4 == 0, "quad-chunked assignment");
+  // quad-chunk: each iteration owns a FULL fn column (all 4 m-fragments):
+  // ONE L2 weight-fragment load feeds FOUR independent MFMA chains.
+  // unroll 2 quads: the second quad's (independent) weight loads issue
+  // under the first quad's MFMA chains, hiding the L2 latency that
+  // dominated the per-quad cost (phase probe: GEMMs ~3.9 of 4.5 ms while
+  // pure MFMA issue accounts for <10% of that)
+#pragma unroll 2
+  for (int ff = wid * FPW; ff < wid * FPW + FPW; ff += 4) {
+    const int fn = ff >> 2;
+    f32x4 acc[4];
 #pragma unroll
-  for (int ks = 0; ks < KS; ++ks)
-    a[ks] = *(const __attribute__((address_space(3))) bf16x8*)(
-        in_lds + (m0 + (ln & 15)) * in_stride + ks * 32 + (ln >> 4) * 8);
-#pragma unroll
-  for (int nf = 0; nf < NF; ++nf) {
-    // pressure cap: keep MFMA/VALU of n-fragments ordered, let the NEXT
-    // fragment's weight loads hoist under THIS fragment's MFMA chain
-    if ((nf & 1) == 0 && nf > 0) __builtin_amdgcn_sched_barrier(0x120);
-    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
+    for (int i = 0; i < 4; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll 4
     for (int ks = 0; ks < KS; ++ks) {
-      const bf16x8 b = *(const __attribute__((address_space(1))) bf16x8*)(
-          Wt + (long)(nf * 16 + (ln & 15)) * WTS + ks * 32 + (ln >> 4) * 8);
-      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[ks], b, acc, 0, 0, 0);
+      bf16x8 b = *(const __attribute__((address_space(1))) bf16x8*)(
+          Wt + (long)(fn * 16 + (lane & 15)) * WTS + ks * 32 + (lane >> 4) * 8);
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm) {
+        bf16x8 a = *(const __attribute__((address_space(3))) bf16x8*)(
+            in_lds + (fm * 16 + (lane & 15)) * in_stride + ks * 32 +
+            (lane >> 4) * 8);
+        acc[fm] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[fm], 0, 0, 0);
+      }
     }
-    const int n = nf * 16 + (ln & 15);
+    const int n = fn * 16 + (lane & 15);
     const float bval = bias ? bias[n] : 0.f;
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int m = m0 + (ln >> 4) * 4 + r;
-      float v = acc[r] + bval;
-      if (MODE == 0) {
-        if (ACT == 1) v = gelu_f32(v);
-        out_lds[m * out_stride + n] = f32_to_bf16(v);
-      } else if (MODE == 1) {
-        if (n < 2 * BF_H) {
+    for (int fm = 0; fm < 4; ++fm) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = fm * 16 + (lane >> 4) * 4 + r;
+        float v = acc[fm][r] + bval;
+        if (MODE == 0) {
+          if (ACT == 1) v = gelu_f32(v);
           out_lds[m * out_stride + n] = f32_to_bf16(v);
-        } else {
-          vt_lds[(n - 2 * BF_H) * VTS + m] = f32_to_bf16(v);  // V transposed
+        } else if (MODE == 1) {
+          if (n < 2 * BF_H) {
+            out_lds[m * out_stride + n] = f32_to_bf16(v);
+          } else {
+            vt_lds[(n - 2 * BF_H) * VTS + m] = f32_to_bf16(v);  // V transposed
+          }
+        } else {  // MODE 2
+          const float xv = bf16_to_f32(x_lds[m * XS + n]);
+          x_lds[m * XS + n] = f32_to_bf16(v + xv);
         }
-      } else {  // MODE 2
-        const float xv = bf16_to_f32(x_lds[m * XS + n]);
-        x_lds[m * XS + n] = f32_to_bf16(v + xv);
       }
     }
   }
@@ -147,11 +140,7 @@ static __device__ __forceinline__ void row_gemm(
 static __device__ __forceinline__ void block_layernorm(
     lds_short* x_lds, const short* __restrict__ gamma,
     const short* __restrict__ beta, int wid, int lane, float eps) {
-  // 4 waves x 2 passes x 8 rows = the 64 rows (8 lanes per row, 16
-  // elements per lane, 3-step shfl_xor reduction per 8-lane row group)
-#pragma unroll
-  for (int rr = 0; rr < BF_S / (BF_WAVES * 8); ++rr) {
-  const int row = wid * (BF_S / BF_WAVES) + rr * 8 + (lane >> 3);
+  const int row = wid * 8 + (lane >> 3);
   const int c0 = (lane & 7) * 16;
   short8v va = *(const __attribute__((address_space(3))) short8v*)(x_lds + row * XS + c0);
   short8v vb = *(const __attribute__((address_space(3))) short8v*)(x_lds + row * XS + c0 + 8);
@@ -189,7 +178,6 @@ static __device__ __forceinline__ void block_layernorm(
   }
   *(__attribute__((address_space(3))) short8v*)(x_lds + row * XS + c0) = oa;
   *(__attribute__((address_space(3))) short8v*)(x_lds + row * XS + c0 + 8) = ob;
-  }
 }
 
 // Phase bits for the profiling probe (production uses PH_ALL; skipped
@@ -237,171 +225,164 @@ static __device__ __forceinline__ void bert_fused_body(
   lds_float* red = (lds_float*)(vt + 2 * BF_DH * VTS);  // [128] pooling
 
   BF_STAMP();  // 0: kernel start
-  const int m0 = wid * 16;  // the wave's own token rows [m0, m0+16)
-
-  // ---- embed (own rows): x[s][c] = tok_emb[byte+3 or 0][c] + pos[s][c] ----
+  // ---- embed: x[s][c] = tok_emb[byte+3 or 0][c] + pos_emb[s][c] ----
   {
     const int s0 = start[line], e0 = end[line];
-#pragma unroll
-    for (int it = 0; it < 16 * BF_H / (DMX_WAVE * 8); ++it) {
-      const int i = m0 * BF_H + it * DMX_WAVE * 8 + lane * 8;
-      const int sr = i / BF_H, c = i % BF_H;
+    for (int i = tid * 8; i < BF_S * BF_H; i += BF_THREADS * 8) {
+      const int s = i / BF_H, c = i % BF_H;
       int tok = 0;
-      const int idx = s0 + sr;
+      const int idx = s0 + s;
       if (idx < e0 && idx < max_len)
         tok = (int)lines[(long)line * max_len + idx] + 3;
       short8v te = *(const short8v*)(wb + WB_TOK + (long)tok * BF_H + c);
-      short8v pe = *(const short8v*)(wb + WB_POS + (long)sr * BF_H + c);
+      short8v pe = *(const short8v*)(wb + WB_POS + (long)s * BF_H + c);
       short8v xv;
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         xv[j] = f32_to_bf16(bf16_to_f32(te[j]) + bf16_to_f32(pe[j]));
-      *(__attribute__((address_space(3))) short8v*)(x_lds + sr * XS + c) = xv;
+      *(__attribute__((address_space(3))) short8v*)(x_lds + s * XS + c) = xv;
     }
   }
-  BF_STAMP();  // 1: embed done (row-local: NO barrier before qkv)
-  BF_STAMP();  // 2: (slot kept for stamp-layout compatibility)
+  BF_STAMP();  // 1: embed work done
+  __syncthreads();
+  BF_STAMP();  // 2: embed barrier crossed
 
   for (int layer = 0; layer < n_layers; ++layer) {
     const short* lw = wb + WB_LAYER0 + (long)layer * LW_SIZE;
     const float* lf = fb + (long)layer * FB_SIZE;
 
-    // ---- qkv (row-local): Q|K -> buf[own rows], V -> vt own columns ----
+    // ---- qkv: Q|K -> buf[64][QKS], V -> vt transposed ----
     if (PHASES & PH_QKV)
-      row_gemm<BF_H, 3 * BF_H, 1, 0, BF_H>(
-          x_lds, XS, m0, (glob_cshort*)(lw + LW_QKV),
-          (glob_cfloat*)(lf + FB_BQKV), buf, QKS, x_lds, vt, lane);
+      block_gemm<BF_H, 3 * BF_H, 1, 0, BF_H>(x_lds, XS, (glob_cshort*)(lw + LW_QKV),
+                                             (glob_cfloat*)(lf + FB_BQKV), buf, QKS, x_lds, vt,
+                                             wid, lane);
     BF_STAMP();  // qkv work done
-    __syncthreads();  // K|V of ALL rows visible (barrier 1 of 2)
+    __syncthreads();
     BF_STAMP();  // qkv barrier crossed
 
-    // ---- attention (own 16 q-rows, both heads sequentially) ----
+    // ---- attention: wave = (head hh, 16 q-rows) ----
     if (PHASES & PH_ATTN) {
+      const int hh = wid >> 2;
+      const int q0 = (wid & 3) * 16;
       const float scale = 0.125f;  // 1/sqrt(64)
-#pragma unroll
-      for (int hh = 0; hh < 2; ++hh) {
-        f32x4 acc_p[4];
-#pragma unroll
-        for (int f = 0; f < 4; ++f) acc_p[f] = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-        for (int ks = 0; ks < BF_DH / 32; ++ks) {
-          bf16x8 a = *(const __attribute__((address_space(3))) bf16x8*)(
-              buf + (m0 + (lane & 15)) * QKS + hh * BF_DH + ks * 32 +
-              (lane >> 4) * 8);
-#pragma unroll
-          for (int f = 0; f < 4; ++f) {
-            bf16x8 b = *(const __attribute__((address_space(3))) bf16x8*)(
-                buf + (f * 16 + (lane & 15)) * QKS + BF_H + hh * BF_DH +
-                ks * 32 + (lane >> 4) * 8);
-            acc_p[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                a, b, acc_p[f], 0, 0, 0);
-          }
-        }
-        float inv_sum[4];
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          float m = -1e30f;
-#pragma unroll
-          for (int f = 0; f < 4; ++f) m = fmaxf(m, acc_p[f][r] * scale);
-#pragma unroll
-          for (int mask = 1; mask < 16; mask <<= 1)
-            m = fmaxf(m, __shfl_xor(m, mask, 64));
-          float sum = 0.f;
-#pragma unroll
-          for (int f = 0; f < 4; ++f) {
-            const float e = __expf(acc_p[f][r] * scale - m);
-            acc_p[f][r] = e;
-            sum += e;
-          }
-#pragma unroll
-          for (int mask = 1; mask < 16; mask <<= 1)
-            sum += __shfl_xor(sum, mask, 64);
-          inv_sum[r] = 1.f / sum;
-        }
-        // P tile overwrites the wave's OWN Q slice, head hh's 64 dims:
-        // Q[own rows][hh*64..hh*64+64) is dead once THIS head's QK^T is
-        // done (head 1 reads cols 64..128, P(h0) goes to cols 0..64) —
-        // wave-local, no barrier. NOTE head 1's P overwrites cols 64..128
-        // AFTER its own QK^T consumed them.
-        lds_short* my_p = buf + m0 * QKS + hh * BF_DH;
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int row = (lane >> 4) * 4 + r;
-#pragma unroll
-          for (int f = 0; f < 4; ++f)
-            my_p[row * QKS + f * 16 + (lane & 15)] =
-                f32_to_bf16(acc_p[f][r]);
-        }
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // wave-local P
 
-        f32x4 acc_o[4];
+      f32x4 acc_p[4];
 #pragma unroll
-        for (int f = 0; f < 4; ++f) acc_o[f] = {0.f, 0.f, 0.f, 0.f};
+      for (int f = 0; f < 4; ++f) acc_p[f] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-        for (int ks = 0; ks < BF_S / 32; ++ks) {
-          bf16x8 a = *(const __attribute__((address_space(3))) bf16x8*)(
-              my_p + (lane & 15) * QKS + ks * 32 + (lane >> 4) * 8);
+      for (int ks = 0; ks < BF_DH / 32; ++ks) {
+        bf16x8 a = *(const __attribute__((address_space(3))) bf16x8*)(
+            buf + (q0 + (lane & 15)) * QKS + hh * BF_DH + ks * 32 +
+            (lane >> 4) * 8);
 #pragma unroll
-          for (int f = 0; f < 4; ++f) {
-            bf16x8 b = *(const __attribute__((address_space(3))) bf16x8*)(
-                vt + (hh * BF_DH + f * 16 + (lane & 15)) * VTS + ks * 32 +
-                (lane >> 4) * 8);
-            acc_o[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                a, b, acc_o[f], 0, 0, 0);
-          }
+        for (int f = 0; f < 4; ++f) {
+          bf16x8 b = *(const __attribute__((address_space(3))) bf16x8*)(
+              buf + (f * 16 + (lane & 15)) * QKS + BF_H + hh * BF_DH +
+              ks * 32 + (lane >> 4) * 8);
+          acc_p[f] =
+              __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc_p[f], 0, 0, 0);
         }
-        // attn out overwrites the wave's OWN Q columns for this head
-        // (P(hh) there was fully consumed by the PV reads above; DS ops
-        // of one wave execute in order, so the write-after-read is safe
-        // without a wait). Other waves never read MY Q slice.
+      }
+      float inv_sum[4];
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int q = m0 + (lane >> 4) * 4 + r;
-          const float inv = __shfl(inv_sum[r], (lane >> 4) * 4 + r, 64);
+      for (int r = 0; r < 4; ++r) {
+        float m = -1e30f;
 #pragma unroll
-          for (int f = 0; f < 4; ++f) {
-            const int d = f * 16 + (lane & 15);
-            buf[q * QKS + hh * BF_DH + d] = f32_to_bf16(acc_o[f][r] * inv);
-          }
+        for (int f = 0; f < 4; ++f) m = fmaxf(m, acc_p[f][r] * scale);
+#pragma unroll
+        for (int mask = 1; mask < 16; mask <<= 1)
+          m = fmaxf(m, __shfl_xor(m, mask, 64));
+        float sum = 0.f;
+#pragma unroll
+        for (int f = 0; f < 4; ++f) {
+          const float e = __expf(acc_p[f][r] * scale - m);
+          acc_p[f][r] = e;
+          sum += e;
+        }
+#pragma unroll
+        for (int mask = 1; mask < 16; mask <<= 1)
+          sum += __shfl_xor(sum, mask, 64);
+        inv_sum[r] = 1.f / sum;
+      }
+      // P tiles alias the Q|K area: every wave must be done reading Q/K
+      __syncthreads();
+      lds_short* my_p = buf + wid * 16 * VTS;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = (lane >> 4) * 4 + r;
+#pragma unroll
+        for (int f = 0; f < 4; ++f)
+          my_p[row * VTS + f * 16 + (lane & 15)] = f32_to_bf16(acc_p[f][r]);
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // wave-local P
+
+      f32x4 acc_o[4];
+#pragma unroll
+      for (int f = 0; f < 4; ++f) acc_o[f] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ks = 0; ks < BF_S / 32; ++ks) {
+        bf16x8 a = *(const __attribute__((address_space(3))) bf16x8*)(
+            my_p + (lane & 15) * VTS + ks * 32 + (lane >> 4) * 8);
+#pragma unroll
+        for (int f = 0; f < 4; ++f) {
+          bf16x8 b = *(const __attribute__((address_space(3))) bf16x8*)(
+              vt + (hh * BF_DH + f * 16 + (lane & 15)) * VTS + ks * 32 +
+              (lane >> 4) * 8);
+          acc_o[f] =
+              __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc_o[f], 0, 0, 0);
+        }
+      }
+      // attn out -> buf[O_OFF + q*XS + hh*64 + d] (disjoint from P tiles)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int q = q0 + (lane >> 4) * 4 + r;
+        const float inv = __shfl(inv_sum[r], (lane >> 4) * 4 + r, 64);
+#pragma unroll
+        for (int f = 0; f < 4; ++f) {
+          const int d = f * 16 + (lane & 15);
+          buf[O_OFF + q * XS + hh * BF_DH + d] =
+              f32_to_bf16(acc_o[f][r] * inv);
         }
       }
     }
     BF_STAMP();  // attention work done
-    __syncthreads();  // buf/vt reusable by row-local phases (barrier 2 of 2)
+    __syncthreads();
     BF_STAMP();  // attention barrier crossed
 
-    // ---- proj (row-local): x += Wo(attn-out in own Q cols) ; LN1 ----
+    // ---- proj: x += Wo(attn) ; LN1 ----
     if (PHASES & PH_PROJ)
-      row_gemm<BF_H, BF_H, 2, 0, BF_H>(
-          buf, QKS, m0, (glob_cshort*)(lw + LW_WO),
-          (glob_cfloat*)(lf + FB_BO), nullptr, 0, x_lds, nullptr, lane);
+      block_gemm<BF_H, BF_H, 2, 0, BF_H>(buf + O_OFF, XS, (glob_cshort*)(lw + LW_WO),
+                                         (glob_cfloat*)(lf + FB_BO), nullptr, 0, x_lds,
+                                         nullptr, wid, lane);
+    __syncthreads();
     if (PHASES & PH_LN)
       block_layernorm(x_lds, lw + LW_LN1G, lw + LW_LN1B, wid, lane, eps);
     BF_STAMP();  // proj+LN1 work done
-    BF_STAMP();  // (stamp slot)
+    __syncthreads();
+    BF_STAMP();  // proj+LN1 barrier crossed
 
-    // ---- FFN (row-local), two K=256 halves sharing buf[own rows] ----
-    if (PHASES & PH_FFN) {
+    // ---- FFN in two K=256 halves: buf = gelu(x@W1_h); x += buf@W2_h ----
+    if (PHASES & PH_FFN)
 #pragma unroll
-      for (int h = 0; h < 2; ++h) {
-        row_gemm<BF_H, BF_FFN / 2, 0, 1, BF_H>(
-            x_lds, XS, m0,
-            (glob_cshort*)(lw + LW_W1 + (long)h * (BF_FFN / 2) * BF_H),
-            (glob_cfloat*)(lf + FB_B1 + h * (BF_FFN / 2)), buf, QKS, nullptr,
-            nullptr, lane);
-        // W2 half h reads buf[own rows] written just above: wave-local
-        row_gemm<BF_FFN / 2, BF_H, 2, 0, BF_FFN>(
-            buf, QKS, m0, (glob_cshort*)(lw + LW_W2 + (long)h * (BF_FFN / 2)),
-            h == 0 ? (glob_cfloat*)(lf + FB_B2) : nullptr, nullptr, 0, x_lds,
-            nullptr, lane);
-      }
+    for (int h = 0; h < 2; ++h) {
+      block_gemm<BF_H, BF_FFN / 2, 0, 1, BF_H>(
+          x_lds, XS, (glob_cshort*)(lw + LW_W1 + (long)h * (BF_FFN / 2) * BF_H),
+          (glob_cfloat*)(lf + FB_B1 + h * (BF_FFN / 2)), buf, QKS, nullptr,
+          nullptr, wid, lane);
+      __syncthreads();
+      // bias b2 added once (half 0); half 1 adds only the partial product
+      block_gemm<BF_FFN / 2, BF_H, 2, 0, BF_FFN>(
+          buf, QKS, (glob_cshort*)(lw + LW_W2 + (long)h * (BF_FFN / 2)),
+          h == 0 ? (glob_cfloat*)(lf + FB_B2) : nullptr, nullptr, 0, x_lds,
+          nullptr, wid, lane);
+      __syncthreads();
     }
     if (PHASES & PH_LN)
       block_layernorm(x_lds, lw + LW_LN2G, lw + LW_LN2B, wid, lane, eps);
     BF_STAMP();  // ffn+LN2 work done
-    BF_STAMP();  // (stamp slot)
+    __syncthreads();
+    BF_STAMP();  // ffn+LN2 barrier crossed
   }
-  __syncthreads();  // pool reads every row
 
   // ---- pool (mean over S) + score head ----
   {
@@ -433,7 +414,7 @@ static __device__ __forceinline__ void bert_fused_body(
 #undef BF_STAMP
 }
 
-extern "C" __global__ __launch_bounds__(BF_THREADS, 2)
+extern "C" __global__ __launch_bounds__(BF_THREADS, 4)
 void dmx_bert_fused_bf16(const unsigned char* __restrict__ lines,
                          const int* __restrict__ start,
                          const int* __restrict__ end,
@@ -448,7 +429,7 @@ void dmx_bert_fused_bf16(const unsigned char* __restrict__ lines,
 // probe variants (in-kernel phase ablation; guide §5.4 rule 19: co-compiled
 // variants can perturb codegen by a few % — read the deltas, not absolutes)
 template <int PHASES>
-__global__ __launch_bounds__(BF_THREADS, 2) void dmx_bert_fused_probe(
+__global__ __launch_bounds__(BF_THREADS, 4) void dmx_bert_fused_probe(
     const unsigned char* lines, const int* start, const int* end,
     const short* wb, const float* fb, float* scores, int B, int max_len,
     int n_layers, float eps) {
@@ -456,7 +437,7 @@ __global__ __launch_bounds__(BF_THREADS, 2) void dmx_bert_fused_probe(
                           n_layers, eps);
 }
 
-extern "C" __global__ __launch_bounds__(BF_THREADS, 2)
+extern "C" __global__ __launch_bounds__(BF_THREADS, 4)
 void dmx_bert_fused_timed(const unsigned char* lines, const int* start,
                           const int* end, const short* wb, const float* fb,
                           float* scores, int B, int max_len, int n_layers,
