@@ -72,11 +72,22 @@ typedef __attribute__((address_space(3))) float lds_float;
 // ---- in-block GEMM: out = act(in_lds[64][K] @ Wt[N][WTS] + bias) ---------
 // MODE 0: write out_lds[m][n]; MODE 1 (qkv): n<2H -> out (Q|K), else vt
 // transposed; MODE 2: x[m][n] += v (residual-accumulate). ACT 1 = GELU.
-template <int K, int N, int MODE, int ACT, int WTS>
+// ABL (perf-ablation diagnostics, numerically wrong on purpose):
+// bit 1 = B operand from an opaque register instead of the L2 weight
+// load (removes vmcnt parks); bit 2 = A operand from a register instead
+// of the LDS read (removes lgkmcnt parks). Epilogue stores stay, so the
+// MFMA chains cannot be dead-code-eliminated (guide §5.4 rule 17).
+template <int K, int N, int MODE, int ACT, int WTS, int ABL = 0>
 static __device__ __attribute__((noinline)) void block_gemm(
     const lds_short* in_lds, int in_stride, const glob_cshort* __restrict__ Wt,
     const glob_cfloat* __restrict__ bias, lds_short* out_lds, int out_stride,
     lds_short* x_lds, lds_short* vt_lds, int wid, int lane) {
+  bf16x8 zb;
+  if constexpr (ABL != 0) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) zb[j] = (short)0x3f01;  // ~1.008 bf16
+    asm volatile("" : "+v"(zb));  // opaque: keeps realistic data/clocks
+  }
   constexpr int N16 = N / 16;
   constexpr int TOTAL = 4 * N16;
   constexpr int KS = K / 32;
@@ -96,13 +107,24 @@ static __device__ __attribute__((noinline)) void block_gemm(
     for (int i = 0; i < 4; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll 4
     for (int ks = 0; ks < KS; ++ks) {
-      bf16x8 b = *(const __attribute__((address_space(1))) bf16x8*)(
-          Wt + (long)(fn * 16 + (lane & 15)) * WTS + ks * 32 + (lane >> 4) * 8);
+      bf16x8 b;
+      if constexpr (ABL & 1) {
+        b = zb;
+      } else {
+        b = *(const __attribute__((address_space(1))) bf16x8*)(
+            Wt + (long)(fn * 16 + (lane & 15)) * WTS + ks * 32 +
+            (lane >> 4) * 8);
+      }
 #pragma unroll
       for (int fm = 0; fm < 4; ++fm) {
-        bf16x8 a = *(const __attribute__((address_space(3))) bf16x8*)(
-            in_lds + (fm * 16 + (lane & 15)) * in_stride + ks * 32 +
-            (lane >> 4) * 8);
+        bf16x8 a;
+        if constexpr (ABL & 2) {
+          a = zb;
+        } else {
+          a = *(const __attribute__((address_space(3))) bf16x8*)(
+              in_lds + (fm * 16 + (lane & 15)) * in_stride + ks * 32 +
+              (lane >> 4) * 8);
+        }
         acc[fm] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[fm], 0, 0, 0);
       }
     }
@@ -195,7 +217,7 @@ static __device__ __forceinline__ void block_layernorm(
 // wave's compute time from its barrier wait. 100 MHz constant clock.
 #define BF_NSTAMP 22
 
-template <int PHASES, bool TIMED = false>
+template <int PHASES, bool TIMED = false, int ABL = 0>
 static __device__ __forceinline__ void bert_fused_body(
     const unsigned char* __restrict__ lines,  // [B, max_len]
     const int* __restrict__ start,            // [B] content span start
@@ -253,7 +275,7 @@ static __device__ __forceinline__ void bert_fused_body(
 
     // ---- qkv: Q|K -> buf[64][QKS], V -> vt transposed ----
     if (PHASES & PH_QKV)
-      block_gemm<BF_H, 3 * BF_H, 1, 0, BF_H>(x_lds, XS, (glob_cshort*)(lw + LW_QKV),
+      block_gemm<BF_H, 3 * BF_H, 1, 0, BF_H, ABL>(x_lds, XS, (glob_cshort*)(lw + LW_QKV),
                                              (glob_cfloat*)(lf + FB_BQKV), buf, QKS, x_lds, vt,
                                              wid, lane);
     BF_STAMP();  // qkv work done
@@ -351,7 +373,7 @@ static __device__ __forceinline__ void bert_fused_body(
 
     // ---- proj: x += Wo(attn) ; LN1 ----
     if (PHASES & PH_PROJ)
-      block_gemm<BF_H, BF_H, 2, 0, BF_H>(buf + O_OFF, XS, (glob_cshort*)(lw + LW_WO),
+      block_gemm<BF_H, BF_H, 2, 0, BF_H, ABL>(buf + O_OFF, XS, (glob_cshort*)(lw + LW_WO),
                                          (glob_cfloat*)(lf + FB_BO), nullptr, 0, x_lds,
                                          nullptr, wid, lane);
     __syncthreads();
@@ -365,13 +387,13 @@ static __device__ __forceinline__ void bert_fused_body(
     if (PHASES & PH_FFN)
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
-      block_gemm<BF_H, BF_FFN / 2, 0, 1, BF_H>(
+      block_gemm<BF_H, BF_FFN / 2, 0, 1, BF_H, ABL>(
           x_lds, XS, (glob_cshort*)(lw + LW_W1 + (long)h * (BF_FFN / 2) * BF_H),
           (glob_cfloat*)(lf + FB_B1 + h * (BF_FFN / 2)), buf, QKS, nullptr,
           nullptr, wid, lane);
       __syncthreads();
       // bias b2 added once (half 0); half 1 adds only the partial product
-      block_gemm<BF_FFN / 2, BF_H, 2, 0, BF_FFN>(
+      block_gemm<BF_FFN / 2, BF_H, 2, 0, BF_FFN, ABL>(
           buf, QKS, (glob_cshort*)(lw + LW_W2 + (long)h * (BF_FFN / 2)),
           h == 0 ? (glob_cfloat*)(lf + FB_B2) : nullptr, nullptr, 0, x_lds,
           nullptr, wid, lane);
@@ -428,13 +450,13 @@ void dmx_bert_fused_bf16(const unsigned char* __restrict__ lines,
 
 // probe variants (in-kernel phase ablation; guide §5.4 rule 19: co-compiled
 // variants can perturb codegen by a few % — read the deltas, not absolutes)
-template <int PHASES>
+template <int PHASES, int ABL = 0>
 __global__ __launch_bounds__(BF_THREADS, 4) void dmx_bert_fused_probe(
     const unsigned char* lines, const int* start, const int* end,
     const short* wb, const float* fb, float* scores, int B, int max_len,
     int n_layers, float eps) {
-  bert_fused_body<PHASES>(lines, start, end, wb, fb, scores, B, max_len,
-                          n_layers, eps);
+  bert_fused_body<PHASES, false, ABL>(lines, start, end, wb, fb, scores, B,
+                                      max_len, n_layers, eps);
 }
 
 extern "C" __global__ __launch_bounds__(BF_THREADS, 4)
@@ -471,26 +493,33 @@ extern "C" void dmx_launch_bert_fused_probe(
       ((size_t)BF_S * XS + BUF_ELEMS + (size_t)2 * BF_DH * VTS) *
           sizeof(short) +
       128 * sizeof(float);
-#define LAUNCH_PROBE(MASK)                                                   \
-  case MASK:                                                                 \
-    hipFuncSetAttribute((const void*)dmx_bert_fused_probe<MASK>,             \
+#define LAUNCH_PROBE(PH, AB)                                                 \
+  case ((PH) | ((AB) << 8)):                                                 \
+    hipFuncSetAttribute((const void*)dmx_bert_fused_probe<PH, AB>,           \
                         hipFuncAttributeMaxDynamicSharedMemorySize,          \
                         (int)lds);                                           \
-    hipLaunchKernelGGL(dmx_bert_fused_probe<MASK>, dim3(B),                  \
+    hipLaunchKernelGGL((dmx_bert_fused_probe<PH, AB>), dim3(B),              \
                        dim3(BF_THREADS), lds, stream,                        \
                        (const unsigned char*)lines, (const int*)start,       \
                        (const int*)end, (const short*)wb, (const float*)fb,  \
                        (float*)scores, B, max_len, n_layers, eps);           \
     break;
   switch (phase_mask) {
-    LAUNCH_PROBE(0)
-    LAUNCH_PROBE(PH_QKV)
-    LAUNCH_PROBE(PH_QKV | PH_ATTN)
-    LAUNCH_PROBE(PH_QKV | PH_ATTN | PH_PROJ)
-    LAUNCH_PROBE(PH_QKV | PH_ATTN | PH_PROJ | PH_LN)
-    LAUNCH_PROBE(PH_ALL)
-    LAUNCH_PROBE(PH_FFN)
-    LAUNCH_PROBE(PH_LN)
+    LAUNCH_PROBE(0, 0)
+    LAUNCH_PROBE(PH_QKV, 0)
+    LAUNCH_PROBE(PH_QKV | PH_ATTN, 0)
+    LAUNCH_PROBE(PH_QKV | PH_ATTN | PH_PROJ, 0)
+    LAUNCH_PROBE(PH_QKV | PH_ATTN | PH_PROJ | PH_LN, 0)
+    LAUNCH_PROBE(PH_ALL, 0)
+    LAUNCH_PROBE(PH_FFN, 0)
+    LAUNCH_PROBE(PH_LN, 0)
+    // perf-ablation variants (numerically wrong by design):
+    LAUNCH_PROBE(PH_ALL, 1)   // B operand constant (no L2 weight loads)
+    LAUNCH_PROBE(PH_ALL, 2)   // A operand constant (no LDS activation reads)
+    LAUNCH_PROBE(PH_ALL, 3)   // both constant (pure MFMA + epilogue)
+    LAUNCH_PROBE(PH_FFN, 1)
+    LAUNCH_PROBE(PH_FFN, 2)
+    LAUNCH_PROBE(PH_FFN, 3)
     default:
       break;
   }
