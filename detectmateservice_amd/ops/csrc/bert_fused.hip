@@ -77,11 +77,19 @@ typedef __attribute__((address_space(3))) float lds_float;
 // load (removes vmcnt parks); bit 2 = A operand from a register instead
 // of the LDS read (removes lgkmcnt parks). Epilogue stores stay, so the
 // MFMA chains cannot be dead-code-eliminated (guide §5.4 rule 17).
+// All pointers __restrict__: without it the epilogue's LDS stores
+// between quads alias the activation reads for the analyzer (the
+// pointers arrive as opaque noinline args), so the compiler re-reads
+// every A fragment per quad (32 ds_reads for 32 MFMAs in the .s) and
+// cannot hoist reads across the quad boundary. The caller guarantees
+// in_lds never overlaps this GEMM's outputs.
 template <int K, int N, int MODE, int ACT, int WTS, int ABL = 0>
 static __device__ __attribute__((noinline)) void block_gemm(
-    const lds_short* in_lds, int in_stride, const glob_cshort* __restrict__ Wt,
-    const glob_cfloat* __restrict__ bias, lds_short* out_lds, int out_stride,
-    lds_short* x_lds, lds_short* vt_lds, int wid, int lane) {
+    const lds_short* __restrict__ in_lds, int in_stride,
+    const glob_cshort* __restrict__ Wt, const glob_cfloat* __restrict__ bias,
+    lds_short* __restrict__ out_lds, int out_stride,
+    lds_short* __restrict__ x_lds, lds_short* __restrict__ vt_lds, int wid,
+    int lane) {
   bf16x8 zb;
   if constexpr (ABL != 0) {
 #pragma unroll
@@ -102,6 +110,10 @@ static __device__ __attribute__((noinline)) void block_gemm(
 #pragma unroll 2
   for (int ff = wid * FPW; ff < wid * FPW + FPW; ff += 4) {
     const int fn = ff >> 2;
+    // bias up front: a bias load in the epilogue made the compiler wait
+    // vmcnt(0) mid-GEMM, draining the weight-load pipeline with it
+    const int n = fn * 16 + (lane & 15);
+    const float bval = bias ? bias[n] : 0.f;
     f32x4 acc[4];
 #pragma unroll
     for (int i = 0; i < 4; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
@@ -128,8 +140,6 @@ static __device__ __attribute__((noinline)) void block_gemm(
         acc[fm] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[fm], 0, 0, 0);
       }
     }
-    const int n = fn * 16 + (lane & 15);
-    const float bval = bias ? bias[n] : 0.f;
 #pragma unroll
     for (int fm = 0; fm < 4; ++fm) {
 #pragma unroll
