@@ -61,7 +61,7 @@ def parse_args():
                    help="shm: protobuf frames through the shm ring with C++ "
                         "in-place decode (full reader path, default); "
                         "off: device-resident line pool (GPU pipeline only)")
-    p.add_argument("--shards", type=int, default=2,
+    p.add_argument("--shards", type=int, default=4,
                    help="ingest shards (ring + feeder + reader threads) per rank")
     p.add_argument("--chunk", type=int, default=8192,
                    help="frames per ring read (constant size so pinned "
@@ -115,10 +115,11 @@ class ShardReader(threading.Thread):
         self.pin = pin
 
     def run(self):
+        # per-CHUNK queueing: the GPU starts on a step's first decoded
+        # chunk while this thread is still decoding the rest of it
         try:
             for _ in range(self.n_steps):
                 got = 0
-                chunks = []
                 while got < self.lines_per_step:
                     want = min(self.chunk, self.lines_per_step - got)
                     lines, lens, blob, off, nb = self.ring.read_batch_packed(
@@ -126,9 +127,9 @@ class ShardReader(threading.Thread):
                     b = int(lines.shape[0])
                     if b == 0:
                         continue
-                    chunks.append((lines, lens))
+                    self.out_q.put((lines, lens))
                     got += b
-                self.out_q.put(chunks)
+                self.out_q.put("step_end")
         except Exception as exc:  # noqa: BLE001 - surface in the main loop
             self.out_q.put(exc)
 
@@ -216,7 +217,7 @@ def main() -> None:
                     LogSchema(logID=f"l{j}", log=gen.line()[0]).serialize()
                     for j in range(nlines)
                 ])
-            q: "queue.Queue" = queue.Queue(maxsize=3)
+            q: "queue.Queue" = queue.Queue(maxsize=24)
             feeders.append(ShardFeeder(ring, pool, nlines * n_total_steps))
             readers.append(ShardReader(ring, nlines, n_total_steps,
                                         args.chunk, args.max_len, q, use_gpu))
@@ -230,10 +231,13 @@ def main() -> None:
         def step(i: int) -> None:
             outs = []
             for q in queues:
-                item = q.get(timeout=120)
-                if isinstance(item, Exception):
-                    raise item
-                for lines, lens in item:
+                while True:
+                    item = q.get(timeout=120)
+                    if isinstance(item, Exception):
+                        raise item
+                    if item == "step_end":
+                        break
+                    lines, lens = item
                     dl = lines.to(device, non_blocking=True)
                     dn = lens.to(device, non_blocking=True)
                     outs.append(pipe.process_packed(dl, dn))

@@ -22,22 +22,29 @@ import torch
 import torch.distributed as dist
 
 
-def init_from_env(backend: Optional[str] = None) -> Tuple[int, int]:
+def init_from_env(backend: Optional[str] = None,
+                  timeout_s: float = 120.0) -> Tuple[int, int]:
     """Initialize the default process group from torchrun env vars.
 
     Returns (rank, world_size); no-op (0, 1) when WORLD_SIZE is unset.
     Backend defaults to nccl (=RCCL) when a GPU is visible, else gloo.
-    """
+    The timeout bounds collective waits so a dead peer surfaces as an
+    error the engine loops can handle (drop-don't-block posture)."""
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     if world_size <= 1:
         return 0, 1
     rank = int(os.environ.get("RANK", "0"))
     if not dist.is_initialized():
+        import datetime
+
         if backend is None:
             backend = "nccl" if torch.cuda.is_available() else "gloo"
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29500")
-        dist.init_process_group(backend=backend, rank=rank, world_size=world_size)
+        dist.init_process_group(
+            backend=backend, rank=rank, world_size=world_size,
+            timeout=datetime.timedelta(seconds=timeout_s),
+        )
     return rank, world_size
 
 
@@ -108,27 +115,111 @@ def all_reduce_hashsets(tables: torch.Tensor, group=None) -> None:
     """Merge data-parallel NewValue hash sets across ranks.
 
     Open-addressing tables cannot be unioned by elementwise max directly
-    (slots differ per insertion order), so gather all tables and re-insert
-    locally. Intended for low-frequency sync (end of training phase)."""
+    (slots differ per insertion order), so gather all tables and
+    re-insert every peer table's keys with ONE batched insert-kernel
+    call per peer: the transposed table [capacity, W] IS a valid hash
+    matrix for dmx_hashset_insert (column w holds watch w's keys; empty
+    slots are 0, which the kernel skips). Intended for low-frequency
+    sync (end of training phase / explicit dp_sync)."""
     world = dist.get_world_size(group)
+    if world <= 1:
+        return
     gathered = [torch.empty_like(tables) for _ in range(world)]
     dist.all_gather(gathered, tables, group=group)
-    from .. import ops
+    if not tables.is_cuda:  # pragma: no cover - CPU tables are python sets
+        raise RuntimeError("all_reduce_hashsets expects device tensor tables")
+    from ..ops import _dmx_C
 
-    W, cap = tables.shape
-    for other in gathered:
-        if other.data_ptr() == tables.data_ptr():
+    me = dist.get_rank(group)
+    for r, other in enumerate(gathered):
+        if r == me:
             continue
-        # re-insert non-empty keys from `other`
-        for w in range(W):
-            keys = other[w][other[w] != 0]
-            if keys.numel() == 0:
-                continue
-            h = torch.zeros((keys.numel(), W), dtype=torch.int64, device=tables.device)
-            h[:, w] = keys
-            if tables.is_cuda:
-                from ..ops import _dmx_C
+        _dmx_C.hashset_insert(other.t().contiguous(), tables)
 
-                _dmx_C.hashset_insert(h, tables)
-            else:  # pragma: no cover - CPU tables are python sets
-                raise RuntimeError("all_reduce_hashsets expects tensor tables")
+
+# ---------------------------------------------------------------------------
+# frame transport for settings-driven service placement (dist_mode):
+# serialized protobuf frames as one u8 blob + offsets, moved by
+# broadcast (fanout) or P2P (stage). A 3-word header precedes each
+# batch; flag 1 = heartbeat (no frames, keeps sinks responsive to
+# stop), flag 2 = shutdown sentinel.
+# ---------------------------------------------------------------------------
+
+FRAME_DATA = 0
+FRAME_HEARTBEAT = 1
+FRAME_SHUTDOWN = 2
+
+
+def _frames_to_tensors(frames: Sequence[bytes]):
+    offs = torch.zeros(len(frames) + 1, dtype=torch.int64)
+    total = 0
+    for i, f in enumerate(frames):
+        total += len(f)
+        offs[i + 1] = total
+    blob = b"".join(frames)
+    data = (
+        torch.frombuffer(bytearray(blob), dtype=torch.uint8)
+        if blob
+        else torch.zeros(0, dtype=torch.uint8)
+    )
+    return data, offs
+
+
+def _tensors_to_frames(data: torch.Tensor, offs: torch.Tensor) -> List[bytes]:
+    raw = bytes(data.cpu().numpy().tobytes())
+    o = offs.tolist()
+    return [raw[o[i]:o[i + 1]] for i in range(len(o) - 1)]
+
+
+def broadcast_frames_src(frames: Sequence[bytes], src: int,
+                         device: torch.device, flag: int = FRAME_DATA,
+                         group=None) -> None:
+    """Source side of the fanout hop (reference multi_output broadcast,
+    engine.py:266-302, as ONE collective over xGMI/gloo)."""
+    data, offs = _frames_to_tensors(frames)
+    hdr = torch.tensor([len(frames), data.numel(), flag], dtype=torch.int64,
+                       device=device)
+    dist.broadcast(hdr, src=src, group=group)
+    if flag == FRAME_DATA and len(frames) > 0:
+        dist.broadcast(offs.to(device), src=src, group=group)
+        dist.broadcast(data.to(device), src=src, group=group)
+
+
+def broadcast_frames_sink(src: int, device: torch.device,
+                          group=None) -> Tuple[List[bytes], int]:
+    hdr = torch.zeros(3, dtype=torch.int64, device=device)
+    dist.broadcast(hdr, src=src, group=group)
+    n, nbytes, flag = int(hdr[0]), int(hdr[1]), int(hdr[2])
+    if flag != FRAME_DATA or n == 0:
+        return [], flag
+    offs = torch.zeros(n + 1, dtype=torch.int64, device=device)
+    dist.broadcast(offs, src=src, group=group)
+    data = torch.zeros(nbytes, dtype=torch.uint8, device=device)
+    dist.broadcast(data, src=src, group=group)
+    return _tensors_to_frames(data, offs), flag
+
+
+def send_frames(frames: Sequence[bytes], dst: int, device: torch.device,
+                flag: int = FRAME_DATA, tag: int = 100) -> None:
+    """P2P stage hop (one xGMI link on GPU — SURVEY.md §5.8)."""
+    data, offs = _frames_to_tensors(frames)
+    hdr = torch.tensor([len(frames), data.numel(), flag], dtype=torch.int64,
+                       device=device)
+    dist.send(hdr, dst=dst, tag=tag)
+    if flag == FRAME_DATA and len(frames) > 0:
+        dist.send(offs.to(device), dst=dst, tag=tag + 1)
+        dist.send(data.to(device), dst=dst, tag=tag + 2)
+
+
+def recv_frames(src: int, device: torch.device,
+                tag: int = 100) -> Tuple[List[bytes], int]:
+    hdr = torch.zeros(3, dtype=torch.int64, device=device)
+    dist.recv(hdr, src=src, tag=tag)
+    n, nbytes, flag = int(hdr[0]), int(hdr[1]), int(hdr[2])
+    if flag != FRAME_DATA or n == 0:
+        return [], flag
+    offs = torch.zeros(n + 1, dtype=torch.int64, device=device)
+    dist.recv(offs, src=src, tag=tag + 1)
+    data = torch.zeros(nbytes, dtype=torch.uint8, device=device)
+    dist.recv(data, src=src, tag=tag + 2)
+    return _tensors_to_frames(data, offs), flag
