@@ -242,12 +242,15 @@ def main() -> None:
                     dn = lens.to(device, non_blocking=True)
                     outs.append(pipe.process_packed(dl, dn))
             if dist is not None:
+                # DP aggregation over RCCL/xGMI — the same helper the
+                # Service dp path uses (parallel/dist.py)
+                from detectmateservice_amd.parallel import dist as dmx_dist
+
                 summary = torch.stack([
                     sum(o["anomaly"].sum() for o in outs).float(),
                     sum(o["scores"].sum() for o in outs),
                 ]).to(device)
-                gathered = [torch.empty_like(summary) for _ in range(world_size)]
-                dist.all_gather(gathered, summary)
+                dmx_dist.all_gather_summaries(summary)
 
         def cleanup():
             for _, path in rings:
@@ -270,11 +273,12 @@ def main() -> None:
             else:
                 out = pipe.process_packed(lines, lens)
             if dist is not None:
+                from detectmateservice_amd.parallel import dist as dmx_dist
+
                 summary = torch.stack(
                     [out["anomaly"].sum().float(), out["scores"].sum()]
                 ).to(device)
-                gathered = [torch.empty_like(summary) for _ in range(world_size)]
-                dist.all_gather(gathered, summary)
+                dmx_dist.all_gather_summaries(summary)
 
         def cleanup():
             pass

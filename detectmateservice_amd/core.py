@@ -44,6 +44,25 @@ class Service:
         component_config: Optional[Dict[str, Any]] = None,
         socket_factory=None,
     ) -> None:
+        # --- settings-driven distributed placement (dist_mode) ---------
+        # Initialize the process group from torchrun env vars, then
+        # substitute {rank}/{world} placeholders in addresses/identity so
+        # ONE YAML launched as `torchrun --nproc-per-node N detectmate
+        # --settings x.yaml` places N ranks (reference principle: the
+        # topology lives in config, parser_settings.yaml out_addr).
+        self.dist_ctx = None
+        if settings.dist_mode is not None:
+            from types import SimpleNamespace
+
+            from .parallel import dist as dmx_dist
+
+            rank, world = dmx_dist.init_from_env(settings.dist_backend)
+            settings = settings.resolve_dist_placeholders(rank, world)
+            if world > 1:
+                self.dist_ctx = SimpleNamespace(
+                    mode=settings.dist_mode, rank=rank, world=world,
+                    src=settings.dist_src_rank,
+                )
         self.settings = settings
         self.logger = build_service_logger(
             settings.component_type,
@@ -109,6 +128,7 @@ class Service:
             socket_factory=socket_factory,
             logger=self.logger,
             metrics=self.metrics,
+            dist_ctx=self.dist_ctx,
         )
 
         # --- admin web server (constructed unconditionally, core.py:81) ---
@@ -235,6 +255,27 @@ class Service:
 
     def shutdown(self) -> None:
         self._service_exit_event.set()
+
+    # ------------------------------------------------------------------
+    def dp_sync(self) -> Dict[str, Any]:
+        """Merge data-parallel detector state across ranks (dist_mode
+        "dp"): a COLLECTIVE — every rank must call it (operators hit
+        POST /admin/dp-sync on all ranks, or a cadence task does).
+        Components implement ``dist_sync()``; NewValue detectors merge
+        learned value sets, the fused GPU detector all-reduces its hash
+        tables (parallel/dist.py::all_reduce_hashsets)."""
+        import torch.distributed as tdist
+
+        if self.dist_ctx is None or not tdist.is_initialized():
+            return {"synced": False, "reason": "no process group"}
+        fn = getattr(self.library_component, "dist_sync", None)
+        if fn is None:
+            return {"synced": False,
+                    "reason": f"component {self.settings.component_type!r} "
+                              "has no dist_sync"}
+        fn()
+        self.logger.info("dp_sync complete (world=%d)", self.dist_ctx.world)
+        return {"synced": True, "world": self.dist_ctx.world}
 
     # ------------------------------------------------------------------
     # checkpoint / resume (SURVEY.md §5.4: the reference has none; this

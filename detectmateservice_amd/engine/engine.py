@@ -63,9 +63,14 @@ class Engine:
         socket_factory: Optional[PairSocketFactory] = None,
         logger: Optional[logging.Logger] = None,
         metrics: Optional[ServiceMetrics] = None,
+        dist_ctx=None,
     ) -> None:
         self.settings = settings
         self.processor = processor
+        #: settings-driven placement (core.py builds it from dist_mode):
+        #: an object with .mode/.rank/.world/.src — selects the fanout /
+        #: stage loops below instead of the socket loop
+        self._dist_ctx = dist_ctx
         self._log = logger or logging.getLogger(__name__)
         self._factory = socket_factory or PairSocketFactory()
         self.metrics = metrics or ServiceMetrics(
@@ -152,6 +157,9 @@ class Engine:
 
     # ------------------------------------------------------------------
     def _run_loop(self) -> None:
+        ctx = self._dist_ctx
+        if ctx is not None and ctx.world > 1 and ctx.mode in ("fanout", "stage"):
+            return self._run_dist_loop()
         if self.settings.engine_source_mode:
             return self._run_source_loop()
         if self.settings.engine_packed_mode and self._try_packed_loop():
@@ -422,6 +430,141 @@ class Engine:
         drain_prev()
         self._log.info("engine packed loop exited")
         return True
+
+
+    # ------------------------------------------------------------------
+    # settings-driven distributed placement (dist_mode fanout/stage):
+    # the reference wires its parallelism entirely through config
+    # (container/config/parser_settings.yaml out_addr); here torchrun +
+    # one YAML places the same topology across ranks/GPUs with RCCL
+    # collectives (gloo on CPU) instead of socket hops.
+    # ------------------------------------------------------------------
+    def _run_dist_loop(self) -> None:
+        import torch
+        import torch.distributed as tdist
+
+        from ..parallel import dist as dmx_dist
+
+        ctx = self._dist_ctx
+        s = self.settings
+        m = self.metrics
+        device = torch.device("cpu")
+        if tdist.get_backend() == "nccl":
+            device = torch.device("cuda", torch.cuda.current_device())
+        mode, rank, world = ctx.mode, ctx.rank, ctx.world
+        src = ctx.src if mode == "fanout" else 0
+        head = rank == src
+        self._log.info(
+            "engine DIST loop started: mode=%s rank=%d/%d src=%d head=%s",
+            mode, rank, world, src, head)
+        if head:
+            self._dist_head_loop(dmx_dist, mode, rank, world, device)
+        else:
+            self._dist_sink_loop(dmx_dist, mode, rank, world, src, device)
+        self._log.info("engine dist loop exited")
+
+    def _dist_head_loop(self, dmx_dist, mode, rank, world, device) -> None:
+        s, m = self.settings, self.metrics
+        while not self._stop_event.is_set():
+            try:
+                frames = self._pair_sock.recv_many(
+                    s.engine_batch_size, s.engine_recv_timeout,
+                    s.engine_batch_linger_ms)
+            except RecvTimeout:
+                frames = []
+            except SocketClosed:
+                break
+            frames = [f for f in frames if f]
+            outs: List[bytes] = []
+            if frames:
+                n_bytes = sum(len(f) for f in frames)
+                m.data_read_bytes_total.inc(n_bytes)
+                m.data_read_lines_total.inc(len(frames))
+                t0 = time.perf_counter()
+                try:
+                    outs = [o for o in self.processor.process_batch(frames)
+                            if o is not None]
+                except Exception as exc:  # noqa: BLE001
+                    m.processing_errors_total.inc(len(frames))
+                    self._log.error("dist processing error: %s", exc)
+                    outs = []
+                m.data_processed_bytes_total.inc(n_bytes)
+                m.data_processed_lines_total.inc(len(frames))
+                m.observe_batch(time.perf_counter() - t0, max(len(frames), 1))
+            flag = dmx_dist.FRAME_DATA if outs else dmx_dist.FRAME_HEARTBEAT
+            try:
+                if mode == "fanout":
+                    dmx_dist.broadcast_frames_src(outs, rank, device, flag)
+                else:
+                    dmx_dist.send_frames(outs, rank + 1, device, flag)
+            except RuntimeError as exc:
+                self._log.error("dist send failed (peer lost?): %s", exc)
+                break
+            if outs:
+                nb = sum(len(o) for o in outs)
+                m.data_written_bytes_total.inc(nb)
+                m.data_written_lines_total.inc(len(outs))
+        # wake every peer out of its collective before exiting
+        try:
+            if mode == "fanout":
+                dmx_dist.broadcast_frames_src([], rank, device,
+                                              dmx_dist.FRAME_SHUTDOWN)
+            else:
+                dmx_dist.send_frames([], rank + 1, device,
+                                     dmx_dist.FRAME_SHUTDOWN)
+        except RuntimeError:
+            pass
+
+    def _dist_sink_loop(self, dmx_dist, mode, rank, world, src, device) -> None:
+        m = self.metrics
+        last = mode == "fanout" or rank == world - 1
+        while not self._stop_event.is_set():
+            try:
+                if mode == "fanout":
+                    frames, flag = dmx_dist.broadcast_frames_sink(src, device)
+                else:
+                    frames, flag = dmx_dist.recv_frames(rank - 1, device)
+            except RuntimeError as exc:
+                self._log.error("dist recv failed (peer lost?): %s", exc)
+                break
+            if flag == dmx_dist.FRAME_SHUTDOWN:
+                if mode == "stage" and rank + 1 < world:
+                    try:
+                        dmx_dist.send_frames([], rank + 1, device,
+                                             dmx_dist.FRAME_SHUTDOWN)
+                    except RuntimeError:
+                        pass
+                break
+            outs: List[bytes] = []
+            if frames:
+                n_bytes = sum(len(f) for f in frames)
+                m.data_read_bytes_total.inc(n_bytes)
+                m.data_read_lines_total.inc(len(frames))
+                t0 = time.perf_counter()
+                try:
+                    outs = [o for o in self.processor.process_batch(frames)
+                            if o is not None]
+                except Exception as exc:  # noqa: BLE001
+                    m.processing_errors_total.inc(len(frames))
+                    self._log.error("dist processing error: %s", exc)
+                    outs = []
+                m.data_processed_bytes_total.inc(n_bytes)
+                m.data_processed_lines_total.inc(len(frames))
+                m.observe_batch(time.perf_counter() - t0, max(len(frames), 1))
+            if mode == "stage" and not last:
+                flag = (dmx_dist.FRAME_DATA if outs
+                        else dmx_dist.FRAME_HEARTBEAT)
+                try:
+                    dmx_dist.send_frames(outs, rank + 1, device, flag)
+                except RuntimeError as exc:
+                    self._log.error("dist forward failed: %s", exc)
+                    break
+                if outs:
+                    m.data_written_bytes_total.inc(sum(len(o) for o in outs))
+                    m.data_written_lines_total.inc(len(outs))
+            else:
+                for out in outs:
+                    self._send_to_outputs(out)
 
     def _run_source_loop(self) -> None:
         """Source mode: the component generates frames (reader services).

@@ -228,6 +228,33 @@ class FusedPipelineDetector(CoreComponent):
         return results
 
     # -- checkpoint -----------------------------------------------------
+    def dist_sync(self, group=None) -> None:
+        """COLLECTIVE merge of the data-parallel GPU hash sets (one
+        batched insert-kernel call per peer table —
+        parallel/dist.py::all_reduce_hashsets)."""
+        import torch.distributed as tdist
+
+        if not tdist.is_initialized() or tdist.get_world_size(group) <= 1:
+            return
+        if self.pipe.hashsets is None:
+            return
+        hs = self.pipe.hashsets
+        if hasattr(hs, "tables"):  # GPU open-addressing tables
+            from ...parallel import dist as dmx_dist
+
+            dmx_dist.all_reduce_hashsets(hs.tables, group=group)
+        else:  # CPU fallback keeps Python sets
+            world = tdist.get_world_size(group)
+            local = [sorted(s) for s in hs.sets]
+            gathered: list = [None] * world
+            tdist.all_gather_object(gathered, local, group=group)
+            me = tdist.get_rank(group)
+            for r, other in enumerate(gathered):
+                if r == me or not other:
+                    continue
+                for w, values in enumerate(other):
+                    hs.sets[w].update(values)
+
     def state_dict(self) -> Dict[str, Any]:
         state: Dict[str, Any] = {"seen_lines": self.pipe.seen_lines}
         if self.pipe.hashsets is not None:

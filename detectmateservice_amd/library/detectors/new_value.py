@@ -225,6 +225,32 @@ class NewValueDetector(CoreDetector):
         alert.alertsObtain = {k: f"Unknown value: '{v}'" for k, v in anomalies.items()}
         return True
 
+    # -- data-parallel state merge (dist_mode "dp") --------------------
+    def dist_sync(self, group=None) -> None:
+        """COLLECTIVE union of every rank's learned value sets: N
+        data-parallel detector ranks each see a shard of the stream;
+        after sync each knows every rank's values (so a value learned on
+        rank 0 does not alert on rank 1). Low-frequency — end of the
+        training phase or an operator's /admin/dp-sync."""
+        import torch.distributed as tdist
+
+        if not tdist.is_initialized() or tdist.get_world_size(group) <= 1:
+            return
+        world = tdist.get_world_size(group)
+        local = {k: sorted(v) for k, v in self.known.items()}
+        gathered: list = [None] * world
+        tdist.all_gather_object(gathered, local, group=group)
+        me = tdist.get_rank(group)
+        for r, other in enumerate(gathered):
+            if r == me or not other:
+                continue
+            for key, values in other.items():
+                if key not in self.known:
+                    self.known[key] = set()
+                    self.known_h[key] = set()
+                for v in values:
+                    self._learn(key, v)
+
     # -- checkpoint/resume (SURVEY.md §5.4) ----------------------------
     def state_dict(self) -> Dict[str, Any]:
         return {

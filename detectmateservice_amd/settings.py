@@ -172,6 +172,53 @@ class ServiceSettings(BaseModel):
     dist_world_size: int = Field(default=1, ge=1)
     dist_rank: int = Field(default=0, ge=0)
 
+    #: settings-driven distributed placement (reference principle:
+    #: topology is config — container/config/parser_settings.yaml wires
+    #: stages by address; here `torchrun ... detectmate --settings x.yaml`
+    #: wires them by rank):
+    #:   "dp"     N independent services; NewValue state merges on
+    #:            dp_sync (admin POST /admin/dp-sync or collective cadence)
+    #:   "fanout" rank dist_src_rank ingests + processes, broadcasts its
+    #:            outputs to every other rank (the reference multi_output
+    #:            1->N as ONE collective over xGMI)
+    #:   "stage"  chain: rank 0 ingests from its socket, each rank
+    #:            processes and forwards to rank+1, the last rank emits
+    #:            to its out_addr (pipeline placement, P2P hops)
+    #: Addresses and component_name may contain "{rank}"/"{world}"
+    #: placeholders, substituted after the process group initializes.
+    dist_mode: Optional[str] = None
+    dist_src_rank: int = Field(default=0, ge=0)
+
+    @field_validator("dist_mode")
+    @classmethod
+    def _check_dist_mode(cls, v):
+        if v is not None and v not in ("dp", "fanout", "stage"):
+            raise ValueError(f"dist_mode must be dp|fanout|stage, got {v!r}")
+        return v
+
+    def resolve_dist_placeholders(self, rank: int, world: int) -> "ServiceSettings":
+        """Substitute {rank}/{world} in addresses and identity; the
+        component_id regenerates from the substituted identity so each
+        rank gets a stable, distinct id."""
+        d = self.model_dump()
+        def sub(v):
+            return (v.replace("{rank}", str(rank)).replace("{world}", str(world))
+                    if isinstance(v, str) else v)
+        changed = False
+        for key in ("engine_addr", "component_name"):
+            nv = sub(d.get(key))
+            if nv != d.get(key):
+                d[key] = nv
+                changed = True
+        new_out = [sub(a) for a in d.get("out_addr", [])]
+        if new_out != d.get("out_addr"):
+            d["out_addr"] = new_out
+            changed = True
+        d["dist_rank"], d["dist_world_size"] = rank, world
+        if changed:
+            d["component_id"] = None  # regenerate per-rank identity
+        return ServiceSettings.model_validate(d)
+
     # ------------------------------------------------------------------
     @field_validator("engine_addr")
     @classmethod

@@ -54,6 +54,13 @@ def shutdown(service=Depends(get_service)) -> Dict[str, Any]:
     return {"status": "shutting down"}
 
 
+@router.post("/dp-sync")
+def dp_sync(service=Depends(get_service)) -> Dict[str, Any]:
+    """Collective merge of data-parallel detector state (dist_mode
+    "dp"); call on EVERY rank."""
+    return service.dp_sync()
+
+
 class CheckpointPayload(BaseModel):
     #: name relative to settings.checkpoint_dir (absolute paths must
     #: resolve inside it); refused with 403 when no checkpoint_dir is set

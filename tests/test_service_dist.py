@@ -1,0 +1,232 @@
+"""End-to-end settings-driven distributed placement (VERDICT round-1
+item 2): REAL Service processes — not bare pipeline objects — placed by
+``dist_mode`` from ONE settings dict, launched as torchrun would launch
+them (WORLD_SIZE/RANK env, gloo on CPU), exchanging frames over the
+collective paths the engine's dist loops use on RCCL/xGMI.
+
+Topologies covered:
+  fanout — rank 0 ingests from its engine socket, processes, broadcasts
+           its outputs; rank 1 consumes the broadcast and emits to its
+           own out_addr socket (reference multi_output as ONE collective)
+  stage  — rank 0 ingests + forwards P2P to rank 1, which emits
+  dp     — two NewValueDetector services learn DIFFERENT values, then a
+           collective dp_sync merges state so each knows the other's
+"""
+import multiprocessing as mp
+import os
+import time
+
+import pytest
+
+
+def _run_worker(rank, world, port, fn_name, tmp, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        os.environ["RANK"] = str(rank)
+        os.environ["WORLD_SIZE"] = str(world)
+        result = globals()[fn_name](rank, world, tmp)
+        q.put((rank, "ok", result))
+    except Exception:  # noqa: BLE001
+        import traceback
+
+        q.put((rank, "err", traceback.format_exc()))
+
+
+def _launch(fn_name, tmp, world=2, free_port=None, timeout=180):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [
+        ctx.Process(target=_run_worker, args=(r, world, free_port, fn_name, str(tmp), q))
+        for r in range(world)
+    ]
+    for p in procs:
+        p.start()
+    results = {}
+    try:
+        for _ in range(world):
+            rank, status, payload = q.get(timeout=timeout)
+            assert status == "ok", f"rank {rank} failed:\n{payload}"
+            results[rank] = payload
+    finally:
+        for p in procs:
+            p.join(timeout=30)
+            if p.is_alive():
+                p.terminate()
+    return results
+
+
+def _settings(tmp, rank_addr_tag, **kw):
+    from detectmateservice_amd import ServiceSettings
+
+    defaults = dict(
+        component_type="core",
+        engine_addr=f"ipc://{tmp}/{rank_addr_tag}-{{rank}}.ipc",
+        http_enabled=False,
+        log_dir=f"{tmp}/logs",
+        engine_recv_timeout=50,
+        engine_batch_linger_ms=5.0,
+        dist_backend="gloo",
+    )
+    defaults.update(kw)
+    return ServiceSettings(**defaults)
+
+
+# ---- worker bodies (module-level so spawn can pickle by name) -------------
+
+
+def _body_fanout(rank, world, tmp):
+    import threading
+
+    from detectmateservice_amd import Service
+    from detectmateservice_amd.engine.sockets import PairDialer, PairListener, RecvTimeout
+
+    # ONE settings shape for every rank — placeholders differentiate
+    settings = _settings(
+        tmp, "fan", dist_mode="fanout",
+        out_addr=[f"ipc://{tmp}/fan-out-{{rank}}.ipc"],
+    )
+    svc = Service(settings)
+    # placeholder substitution happened inside Service
+    assert f"fan-{rank}.ipc" in svc.settings.engine_addr
+    t = threading.Thread(target=svc.run, daemon=True)
+    t.start()
+    try:
+        if rank == 0:
+            # feed the head rank's engine socket with real frames
+            time.sleep(0.5)
+            feeder = PairDialer(svc.settings.engine_addr)
+            assert feeder.wait_connected(10.0)
+            for i in range(5):
+                assert feeder.send(b"frame-%d" % i, block=True)
+            time.sleep(3.0)  # keep broadcasting until sinks consumed
+            feeder.close()
+            return "fed"
+        # sink rank: collect what arrives on OUR out_addr
+        sink = PairListener(svc.settings.out_addr[0].replace("{rank}", str(rank)))
+        got = []
+        deadline = time.monotonic() + 20
+        while len(got) < 5 and time.monotonic() < deadline:
+            try:
+                got.append(sink.recv(timeout_ms=500))
+            except RecvTimeout:
+                continue
+        sink.close()
+        return sorted(got)
+    finally:
+        svc.shutdown()
+        t.join(timeout=10.0)
+
+
+def _body_stage(rank, world, tmp):
+    import threading
+
+    from detectmateservice_amd import Service
+    from detectmateservice_amd.engine.sockets import PairDialer, PairListener, RecvTimeout
+
+    settings = _settings(
+        tmp, "stg", dist_mode="stage",
+        out_addr=[f"ipc://{tmp}/stg-out.ipc"],
+    )
+    svc = Service(settings)
+    t = threading.Thread(target=svc.run, daemon=True)
+    t.start()
+    try:
+        if rank == 0:
+            time.sleep(0.5)
+            feeder = PairDialer(svc.settings.engine_addr)
+            assert feeder.wait_connected(10.0)
+            for i in range(4):
+                assert feeder.send(b"stage-%d" % i, block=True)
+            time.sleep(3.0)
+            feeder.close()
+            return "fed"
+        sink = PairListener(f"ipc://{tmp}/stg-out.ipc")
+        got = []
+        deadline = time.monotonic() + 20
+        while len(got) < 4 and time.monotonic() < deadline:
+            try:
+                got.append(sink.recv(timeout_ms=500))
+            except RecvTimeout:
+                continue
+        sink.close()
+        return sorted(got)
+    finally:
+        svc.shutdown()
+        t.join(timeout=10.0)
+
+
+def _body_dp_sync(rank, world, tmp):
+    import yaml
+
+    from detectmateservice_amd import Service
+    from detectmateservice_amd.schemas import ParserSchema
+
+    cfg_file = f"{tmp}/nv-{rank}.yaml"
+    with open(cfg_file, "w") as fh:
+        yaml.safe_dump({
+            "detectors": {"NewValueDetector": {
+                "method_type": "new_value_detector",
+                "data_use_training": 1,
+                "global": {"g": {"header_variables": [{"pos": "URL"}]}},
+            }}
+        }, fh)
+    settings = _settings(
+        tmp, "dp", dist_mode="dp", component_type="NewValueDetector",
+        config_file=cfg_file, engine_autostart=False,
+    )
+    svc = Service(settings)
+    try:
+
+        def frame(url, lid):
+            return ParserSchema(EventID=1, logID=lid,
+                                logFormatVariables={"URL": url}).serialize()
+
+        # each rank trains on ITS OWN value (1 training line), then
+        # detects: before sync, the OTHER rank's value is an anomaly
+        mine, other = f"/rank{rank}", f"/rank{1 - rank}"
+        svc.process_batch([frame(mine, "t")])
+        assert svc.process(frame(mine, "a")) is None       # known
+        assert svc.process(frame(other, "b")) is not None  # alert pre-sync
+        out = svc.dp_sync()                                # COLLECTIVE
+        assert out["synced"] is True and out["world"] == world
+        assert svc.process(frame(other, "c")) is None      # known post-sync
+        assert svc.process(frame("/evil", "d")) is not None
+        return "dp-ok"
+    finally:
+        svc.engine.close()
+
+
+# ---- tests ----------------------------------------------------------------
+
+
+def test_fanout_service_end_to_end(tmp_path, free_port):
+    results = _launch("_body_fanout", tmp_path, 2, free_port)
+    assert results[0] == "fed"
+    assert results[1] == [b"frame-%d" % i for i in range(5)]
+
+
+def test_stage_service_end_to_end(tmp_path, free_port):
+    results = _launch("_body_stage", tmp_path, 2, free_port)
+    assert results[0] == "fed"
+    assert results[1] == [b"stage-%d" % i for i in range(4)]
+
+
+def test_dp_sync_merges_newvalue_state(tmp_path, free_port):
+    results = _launch("_body_dp_sync", tmp_path, 2, free_port)
+    assert results[0] == results[1] == "dp-ok"
+
+
+def test_dist_mode_single_process_noop(tmp_path):
+    """dist_mode set but WORLD_SIZE unset: the Service runs standalone
+    (placeholders substituted with rank 0 / world 1)."""
+    from detectmateservice_amd import Service
+
+    settings = _settings(tmp_path, "solo", dist_mode="dp")
+    svc = Service(settings)
+    try:
+        assert svc.dist_ctx is None
+        assert "solo-0.ipc" in svc.settings.engine_addr
+        assert svc.process(b"x") == b"x"
+    finally:
+        svc.engine.close()
