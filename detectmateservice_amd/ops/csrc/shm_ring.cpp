@@ -30,6 +30,7 @@
 #include <cerrno>
 #include <cstring>
 #include <deque>
+#include <memory>
 #include <stdexcept>
 #include <string>
 #include <thread>
@@ -107,6 +108,37 @@ class ShmRing {
     if (map_) munmap(map_, map_bytes_);
   }
 
+  // GIL-free core (called from write_frames under gil_scoped_release and
+  // from ShmFeeder's C++ thread): writes raw[from..] until the ring is
+  // full, returns frames accepted.
+  int64_t write_raw(const std::vector<std::pair<const uint8_t*, size_t>>& raw,
+                    size_t from) {
+    int64_t accepted = 0;
+    uint32_t expect = 0;
+    int spins = 0;
+    while (!hdr_->wlock.compare_exchange_weak(expect, 1)) {
+      expect = 0;
+      if (++spins > 200000) return 0;  // holder died: drop, don't hang
+      if ((spins & 1023) == 0) std::this_thread::yield();
+    }
+    uint64_t head = hdr_->head.load(std::memory_order_relaxed);
+    const uint64_t tail = hdr_->tail.load(std::memory_order_acquire);
+    for (size_t i = from; i < raw.size(); ++i) {
+      const auto& f = raw[i];
+      const uint64_t need = 4 + f.second;
+      if (need > size_) break;  // oversize for this ring: drop
+      if (size_ - (head - tail) < need) break;  // full: drop the rest
+      const uint32_t len32 = (uint32_t)f.second;
+      put_bytes(head, (const uint8_t*)&len32, 4);  // native-endian u32
+      put_bytes(head + 4, f.first, f.second);
+      head += need;
+      ++accepted;
+    }
+    hdr_->head.store(head, std::memory_order_release);
+    hdr_->wlock.store(0, std::memory_order_release);
+    return accepted;
+  }
+
   int64_t write_frames(const std::vector<py::bytes>& frames) {
     std::vector<std::pair<const uint8_t*, size_t>> raw(frames.size());
     for (size_t i = 0; i < frames.size(); ++i) {
@@ -115,31 +147,10 @@ class ShmRing {
       PyBytes_AsStringAndSize(frames[i].ptr(), &p, &n);
       raw[i] = {(const uint8_t*)p, (size_t)n};
     }
-    int64_t accepted = 0;
+    int64_t accepted;
     {
       py::gil_scoped_release release;
-      // bounded spinlock (multi-producer fan-in)
-      uint32_t expect = 0;
-      int spins = 0;
-      while (!hdr_->wlock.compare_exchange_weak(expect, 1)) {
-        expect = 0;
-        if (++spins > 200000) return 0;  // holder died: drop, don't hang
-        if ((spins & 1023) == 0) std::this_thread::yield();
-      }
-      uint64_t head = hdr_->head.load(std::memory_order_relaxed);
-      const uint64_t tail = hdr_->tail.load(std::memory_order_acquire);
-      for (auto& f : raw) {
-        const uint64_t need = 4 + f.second;
-        if (need > size_) break;  // oversize for this ring: drop
-        if (size_ - (head - tail) < need) break;  // full: drop the rest
-        const uint32_t len32 = (uint32_t)f.second;
-        put_bytes(head, (const uint8_t*)&len32, 4);  // native-endian u32
-        put_bytes(head + 4, f.first, f.second);
-        head += need;
-        ++accepted;
-      }
-      hdr_->head.store(head, std::memory_order_release);
-      hdr_->wlock.store(0, std::memory_order_release);
+      accepted = write_raw(raw, 0);
     }
     return accepted;
   }
@@ -279,6 +290,92 @@ class ShmRing {
   uint64_t size_ = 0, mask_ = 0;
 };
 
+// ---------------------------------------------------------------------------
+// ShmFeeder: a C++ load generator (VERDICT round-1 item 3 — the Python
+// feeder thread was the measured bound in service mode at 5.4M lines/s).
+// Frames are copied into an owned pool at construction; start() spawns a
+// plain std::thread that cycles the pool into the ring with ZERO Python
+// involvement (no GIL, no interpreter) until `total` frames are written.
+// ---------------------------------------------------------------------------
+class ShmFeeder {
+ public:
+  ShmFeeder(const std::string& path, const std::vector<py::bytes>& frames,
+            int64_t ring_bytes)
+      : ring_(new ShmRing(path, ring_bytes, false)) {
+    pool_.reserve(frames.size());
+    for (auto& f : frames) {
+      char* p;
+      Py_ssize_t n;
+      PyBytes_AsStringAndSize(f.ptr(), &p, &n);
+      pool_.emplace_back(p, p + (size_t)n);
+    }
+    ptrs_.reserve(pool_.size());
+    for (auto& s : pool_)
+      ptrs_.push_back({(const uint8_t*)s.data(), s.size()});
+    if (ptrs_.empty()) throw std::runtime_error("empty frame pool");
+  }
+
+  ~ShmFeeder() {
+    stop_ = true;
+    if (th_.joinable()) th_.join();
+  }
+
+  void start(int64_t total) {
+    if (th_.joinable()) throw std::runtime_error("feeder already started");
+    stop_ = false;
+    sent_ = 0;
+    done_ = false;
+    th_ = std::thread([this, total] {
+      size_t i = 0;
+      std::vector<std::pair<const uint8_t*, size_t>> batch;
+      const int64_t CH = 4096;
+      int64_t sent = 0;
+      while (!stop_ && sent < total) {
+        batch.clear();
+        const int64_t want = std::min<int64_t>(CH, total - sent);
+        for (int64_t k = 0; k < want; ++k) {
+          batch.push_back(ptrs_[i]);
+          i = (i + 1) % ptrs_.size();
+        }
+        size_t off = 0;
+        while (off < batch.size() && !stop_) {
+          const int64_t n = ring_->write_raw(batch, off);
+          if (n <= 0) {
+            std::this_thread::sleep_for(std::chrono::microseconds(100));
+            continue;
+          }
+          off += (size_t)n;
+          sent += n;
+          sent_.store(sent, std::memory_order_relaxed);
+        }
+      }
+      done_ = true;
+    });
+  }
+
+  int64_t sent() const { return sent_.load(std::memory_order_relaxed); }
+  bool done() const { return done_.load(); }
+  void stop() { stop_ = true; }
+
+  void join(int timeout_ms) {
+    py::gil_scoped_release release;
+    const auto deadline = std::chrono::steady_clock::now() +
+                          std::chrono::milliseconds(timeout_ms);
+    while (!done_.load() && std::chrono::steady_clock::now() < deadline)
+      std::this_thread::sleep_for(std::chrono::milliseconds(1));
+    if (done_.load() && th_.joinable()) th_.join();
+  }
+
+ private:
+  std::unique_ptr<ShmRing> ring_;
+  std::vector<std::string> pool_;
+  std::vector<std::pair<const uint8_t*, size_t>> ptrs_;
+  std::thread th_;
+  std::atomic<bool> stop_{false};
+  std::atomic<bool> done_{false};
+  std::atomic<int64_t> sent_{0};
+};
+
 }  // namespace
 
 void register_shm_ring(py::module_& m) {
@@ -292,4 +389,14 @@ void register_shm_ring(py::module_& m) {
            py::arg("max_frames") = 4096, py::arg("timeout_ms") = 100,
            py::arg("max_len") = 256, py::arg("pin") = false)
       .def("pending", &ShmRing::pending);
+  py::class_<ShmFeeder>(m, "ShmFeeder")
+      .def(py::init<const std::string&, const std::vector<py::bytes>&,
+                    int64_t>(),
+           py::arg("path"), py::arg("frames"),
+           py::arg("ring_bytes") = 16 << 20)
+      .def("start", &ShmFeeder::start)
+      .def("sent", &ShmFeeder::sent)
+      .def("done", &ShmFeeder::done)
+      .def("stop", &ShmFeeder::stop)
+      .def("join", &ShmFeeder::join, py::arg("timeout_ms") = 60000);
 }

@@ -703,3 +703,35 @@ def test_shm_listener_drops_stale_ring_from_crashed_run(monkeypatch):
             fresh.close()
     finally:
         dialer.close()
+
+
+def test_shm_cpp_feeder(monkeypatch):
+    """ShmFeeder (C++ load generator, zero Python in the pump loop)
+    cycles its frame pool into the ring; the listener receives exactly
+    `total` frames in pool order."""
+    import uuid as uuid_mod
+
+    from detectmateservice_amd import ops
+
+    if not ops.have_extension():
+        pytest.skip("extension not built")
+    from detectmateservice_amd.engine.sockets import ShmListener, _shm_paths
+    from detectmateservice_amd.ops import _dmx_C
+
+    monkeypatch.setenv("DMX_SHM_RING_BYTES", str(1 << 20))
+    addr = f"shm:///dmx-feed-{uuid_mod.uuid4().hex[:8]}"
+    listener = ShmListener(addr)
+    try:
+        c2s, _ = _shm_paths(listener.addr)
+        pool = [b"pool-%d" % i for i in range(10)]
+        feeder = _dmx_C.ShmFeeder(c2s, pool, 1 << 20)
+        feeder.start(2500)
+        got = []
+        while len(got) < 2500:
+            got.extend(listener.recv_many(4096, 2000, linger_ms=5.0))
+        feeder.join(10000)
+        assert feeder.done() and feeder.sent() == 2500
+        assert got[:10] == pool and got[-1] == pool[2499 % 10]
+        assert len(got) == 2500
+    finally:
+        listener.close()

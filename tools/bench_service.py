@@ -246,11 +246,26 @@ def main():
         # detector input: we time until the parser has accepted all
         # frames AND the pipeline is idle (sink quiet + sockets drained).
         t0 = time.perf_counter()
-        sent_total = 0
-        while sent_total < args.lines:
-            todo = min(args.lines - sent_total, pool_n)
-            pump(frames[:todo])
-            sent_total += todo
+        if args.shm:
+            # C++ GIL-free load generator (ops/csrc/shm_ring.cpp
+            # ShmFeeder): the Python feeder thread was the measured
+            # bound at 5.4M lines/s (round-1 BASELINE.md); the pump now
+            # runs in a plain std::thread with zero interpreter work.
+            from detectmateservice_amd.engine.sockets import _shm_paths, _shm_ring_bytes
+            from detectmateservice_amd.ops import _dmx_C as _c
+
+            c2s_path, _ = _shm_paths(feeder.addr)
+            cpp_feeder = _c.ShmFeeder(c2s_path, frames, _shm_ring_bytes())
+            cpp_feeder.start(args.lines)
+            cpp_feeder.join(600_000)
+            assert cpp_feeder.done(), "C++ feeder timed out"
+            sent_total = cpp_feeder.sent()
+        else:
+            sent_total = 0
+            while sent_total < args.lines:
+                todo = min(args.lines - sent_total, pool_n)
+                pump(frames[:todo])
+                sent_total += todo
         pump_elapsed = time.perf_counter() - t0
         # wait until pipeline is idle: detector emits nothing for clean
         # traffic; send one marked anomalous line and wait for its alert
