@@ -66,6 +66,9 @@ def parse_args():
     p.add_argument("--chunk", type=int, default=8192,
                    help="frames per ring read (constant size so pinned "
                         "buffers come from the caching allocator)")
+    p.add_argument("--line-buffer", action="store_true",
+                   help="retain every ingested batch in the capacity-managed "
+                        "HBM line buffer (BASELINE config 5) during the run")
     p.add_argument("--device", default=None)
     return p.parse_args()
 
@@ -189,6 +192,15 @@ def main() -> None:
     )
     pipe = GpuPipeline(cfg, device=device)
 
+    line_buf = None
+    if args.line_buffer:
+        from detectmateservice_amd.line_buffer import GpuLineBuffer
+
+        # budget_fraction of FREE HBM after model/pools (288 GB part:
+        # >500M resident lines at max_len 256)
+        line_buf = GpuLineBuffer(max_len=args.max_len, device=device,
+                                 budget_fraction=0.5)
+
     gen = AuditLogGenerator(seed=1000 + rank, anomaly_rate=0.01)
 
     use_graph = [False]
@@ -240,6 +252,8 @@ def main() -> None:
                     lines, lens = item
                     dl = lines.to(device, non_blocking=True)
                     dn = lens.to(device, non_blocking=True)
+                    if line_buf is not None:
+                        line_buf.append(dl, dn)
                     outs.append(pipe.process_packed(dl, dn))
             if dist is not None:
                 # DP aggregation over RCCL/xGMI — the same helper the
@@ -268,6 +282,8 @@ def main() -> None:
 
         def step(i: int) -> None:
             lines, lens = pool_dev[i % len(pool_dev)]
+            if line_buf is not None:
+                line_buf.append(lines, lens)
             if use_graph[0]:
                 out = pipe.process_packed_graph(lines, lens)
             else:
@@ -361,6 +377,8 @@ def main() -> None:
                 "p99_batch_ms": round(p99_step * 1000.0, 3),
                 "transformer": not args.no_transformer,
                 "hip_graph": use_graph[0],
+                "line_buffer": (line_buf.watermark() if line_buf is not None
+                                else None),
                 "device": str(device),
             },
         }
