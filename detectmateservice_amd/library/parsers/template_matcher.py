@@ -46,6 +46,11 @@ class MatcherParserConfig(CoreConfig):
     #: templates (existing EventIDs are stable — new templates get higher
     #: ids). MI355X-first addition: the reference mines once at startup.
     auto_config_refit_lines: int = 0
+    #: template aging: a template that matched NOTHING for this many
+    #: observed lines is deactivated at the next drift refit (its
+    #: EventID is retired, never reused — downstream per-event state
+    #: stays valid). 0 disables aging.
+    template_max_idle_lines: int = 0
     log_format: Optional[str] = None
     time_format: Optional[str] = None
     params: Dict = {}
@@ -139,6 +144,10 @@ class MatcherParser(CoreComponent):
                     )
         self.templates = templates
         self._segments = [split_template(t) for t in templates]
+        #: aging state: EventIDs are positional and append-only, so aged
+        #: templates DEACTIVATE (mask) instead of being removed
+        self._active = [True] * len(templates)
+        self._idle = [0] * len(templates)
         self.parser_id = f"matcher_parser-{id(self):x}"
         # auto_config: buffer content lines until the mining threshold,
         # then derive templates (template_miner.py) and start matching.
@@ -192,12 +201,34 @@ class MatcherParser(CoreComponent):
         if len(self._drift_pending) >= lim:
             self.refit_templates()
 
+    def _observe_usage(self, event_ids, batch_lines: int) -> None:
+        """Aging bookkeeping: reset idle for templates that matched in
+        this batch, advance it for the rest."""
+        if self.config.template_max_idle_lines <= 0:
+            return
+        try:
+            ids = {int(v) for v in event_ids.unique().tolist() if v > 0}
+        except AttributeError:
+            ids = {v for v in event_ids if v > 0}
+        for i in range(len(self._idle)):
+            if self._active[i]:
+                self._idle[i] = 0 if (i + 1) in ids else self._idle[i] + batch_lines
+
     def refit_templates(self) -> List[str]:
         """Mine the drift buffer and append templates not already known.
         Existing templates keep their EventIDs (append-only)."""
         from .template_miner import TemplateMiner
         import torch
 
+        # aging: retire templates that matched nothing for too long
+        lim_idle = self.config.template_max_idle_lines
+        if lim_idle > 0:
+            aged = [i for i in range(len(self.templates))
+                    if self._active[i] and self._idle[i] > lim_idle]
+            for i in aged:
+                self._active[i] = False
+            if aged:
+                self._batch_matcher = None  # rebuild without retired rows
         pending, self._drift_pending = self._drift_pending, []
         if not pending:
             return []
@@ -210,6 +241,8 @@ class MatcherParser(CoreComponent):
         if new:
             self.templates = self.templates + new
             self._segments = [split_template(t) for t in self.templates]
+            self._active = self._active + [True] * len(new)
+            self._idle = self._idle + [0] * len(new)
             self._batch_matcher = None  # rebuild packed tables
         return new
 
@@ -227,6 +260,8 @@ class MatcherParser(CoreComponent):
         mined = miner.fit(pending)
         self.templates = mined
         self._segments = [split_template(t) for t in mined]
+        self._active = [True] * len(mined)
+        self._idle = [0] * len(mined)
         self._batch_matcher = None  # rebuild packed tables on next batch
         return mined
 
@@ -242,6 +277,8 @@ class MatcherParser(CoreComponent):
                     content = header[self._content_field]
         content = self._normalize(content)
         for event_id, segments in enumerate(self._segments, start=1):
+            if not self._active[event_id - 1]:
+                continue
             caps = match_template(content, segments)
             if caps is not None:
                 return header, event_id, self.templates[event_id - 1], caps
@@ -293,9 +330,11 @@ class MatcherParser(CoreComponent):
         now = int(time.time())
         out: List[Optional[bytes]] = []
         drift: List[str] = []
+        used_ids: List[int] = []
         for raw in frames:
             log = LogSchema.deserialize(raw)
             header, event_id, template, variables = self.parse_line(log.log)
+            used_ids.append(event_id)
             if event_id == -1 and self.config.auto_config_refit_lines > 0:
                 content = log.log
                 if self._format_re is not None:
@@ -317,6 +356,7 @@ class MatcherParser(CoreComponent):
                 parsedTimestamp=now,
             )
             out.append(parsed.serialize())
+        self._observe_usage(used_ids, len(frames))
         if drift:
             self._drift_observe(drift)
         return out
@@ -336,8 +376,16 @@ class MatcherParser(CoreComponent):
             if device is None:
                 device = "cuda" if torch.cuda.is_available() else "cpu"
             self._batch_device = device
+            # aged templates are masked out; the matcher indexes only
+            # ACTIVE templates and _ev_remap maps its 1-based ids back
+            # to the stable positional EventIDs
+            active_ids = [i for i in range(len(self.templates))
+                          if self._active[i]]
+            self._matcher_ids = active_ids
+            self._ev_remap = torch.tensor(
+                [0] + [i + 1 for i in active_ids], dtype=torch.int32)
             self._batch_matcher = ops.TemplateMatcher(
-                self.templates,
+                [self.templates[i] for i in active_ids],
                 log_format=self.config.log_format,
                 lowercase=self._lowercase,
                 device=device,
@@ -357,6 +405,15 @@ class MatcherParser(CoreComponent):
         dev_lens = lens.to(self._batch_device)
         match = matcher.match_packed(dev_lines, dev_lens)
         match_cpu = {k: v.cpu() for k, v in match.items()}
+        # remap matcher-local event ids to the stable positional ids
+        ev_local = match_cpu["event_id"]
+        if getattr(self, "_matcher_ids", None) is not None and (
+                len(self._matcher_ids) != len(self.templates)):
+            remapped = ev_local.clone()
+            pos = ev_local > 0
+            remapped[pos] = self._ev_remap[ev_local[pos].long()]
+            match_cpu["event_id"] = remapped
+        self._observe_usage(match_cpu["event_id"], len(frames))
         if self.config.auto_config_refit_lines > 0:
             ev = match_cpu["event_id"]
             idxs = (ev < 0).nonzero().flatten().tolist()[:256]

@@ -194,3 +194,46 @@ def test_drift_refit_batched_path():
     again = [LogSchema(logID="x", log="connection from 10.0.0.99 port 4099 accepted").serialize()] * 8
     outs = [ParserSchema.deserialize(o) for o in parser.process_batch(again)]
     assert all(o.EventID > 1 for o in outs)
+
+
+def test_template_aging_on_refit():
+    """A template that matches nothing for template_max_idle_lines is
+    retired at the next drift refit: its EventID stops being assigned
+    (never reused), live templates keep their positional ids, and newly
+    mined drift templates append after it."""
+    from detectmateservice_amd.library.parsers import MatcherParser
+    from detectmateservice_amd.schemas import LogSchema
+
+    parser = MatcherParser({
+        "method_type": "matcher_parser",
+        "templates": ["alive <*> ok", "deadwood <*> gone"],
+        "auto_config_refit_lines": 8,
+        "template_max_idle_lines": 10,
+    })
+
+    def frames(lines):
+        return [LogSchema(logID=str(i), log=l).serialize()
+                for i, l in enumerate(lines)]
+
+    import json
+
+    from detectmateservice_amd.schemas import ParserSchema
+
+    # template 2 matches once up front: its idle clock starts at 0
+    out = parser.process_batch(frames(["deadwood 1 gone"]))
+    assert ParserSchema.deserialize(out[0]).EventID == 2
+
+    # 12 lines that only exercise template 1 -> template 2 idles past 10
+    out = parser.process_batch(frames([f"alive {i} ok" for i in range(12)]))
+    assert all(ParserSchema.deserialize(o).EventID == 1 for o in out)
+
+    # 8 unmatched lines trigger the drift refit, which ages template 2
+    out = parser.process_batch(frames([f"novel thing {i} seen" for i in range(8)]))
+    assert parser._active == [True, False] + [True] * (len(parser.templates) - 2)
+    assert len(parser.templates) > 2  # drift mined something new
+
+    # retired id 2 no longer matches; id 1 still does
+    out = parser.process_batch(frames(["deadwood 9 gone", "alive 9 ok"]))
+    evs = [ParserSchema.deserialize(o).EventID for o in out]
+    assert evs[1] == 1
+    assert evs[0] != 2  # -1 or a new drift template, never the retired id
