@@ -279,6 +279,25 @@ def _ws_control(opcode: int, payload: bytes, mask: bool) -> bytes:
     return bytes(header) + payload
 
 
+def _ws_xor_mask(data: bytes, mkey: bytes) -> bytes:
+    """Vectorized WebSocket masking (the per-byte Python loop measured
+    ~20k lines/s end to end — tools/bench_transport.py ws): big-int XOR
+    for typical log-frame sizes (no per-call numpy overhead), numpy for
+    large payloads."""
+    n = len(data)
+    if n == 0:
+        return data
+    if n <= 4096:
+        key = mkey * ((n >> 2) + 1)
+        return (int.from_bytes(data, "little")
+                ^ int.from_bytes(key[:n], "little")).to_bytes(n, "little")
+    import numpy as _np
+
+    arr = _np.frombuffer(data, dtype=_np.uint8)
+    key = _np.frombuffer((mkey * ((n + 3) // 4))[:n], dtype=_np.uint8)
+    return (arr ^ key).tobytes()
+
+
 def _ws_encode(data: bytes, mask: bool) -> bytes:
     """One binary message frame (FIN=1, opcode 2); clients mask."""
     import os as _os
@@ -297,8 +316,7 @@ def _ws_encode(data: bytes, mask: bool) -> bytes:
     if mask:
         mkey = _os.urandom(4)
         header += mkey
-        masked = bytes(b ^ mkey[i % 4] for i, b in enumerate(data))
-        return bytes(header) + masked
+        return bytes(header) + _ws_xor_mask(data, mkey)
     return bytes(header) + data
 
 
@@ -309,10 +327,18 @@ class _WsFrameReader:
         self.sock = sock
         self.server_side = server_side
         self.buf = bytearray()
+        self._off = 0  # consumed-bytes cursor (del buf[:n] per frame is O(buffer))
         self._frag: Optional[bytearray] = None  # in-progress fragmented msg
 
+    def _avail(self) -> int:
+        return len(self.buf) - self._off
+
     def _fill(self, need: int) -> bool:
-        while len(self.buf) < need:
+        while self._avail() < need:
+            # compact lazily: only when the consumed prefix dominates
+            if self._off > 1 << 20 and self._off > len(self.buf) // 2:
+                del self.buf[: self._off]
+                self._off = 0
             try:
                 chunk = self.sock.recv(262144)
             except (ssl.SSLWantReadError, BlockingIOError):
@@ -327,7 +353,8 @@ class _WsFrameReader:
         while True:
             if not self._fill(2):
                 return out or None
-            b0, b1 = self.buf[0], self.buf[1]
+            o = self._off
+            b0, b1 = self.buf[o], self.buf[o + 1]
             opcode = b0 & 0x0F
             masked = bool(b1 & 0x80)
             n = b1 & 0x7F
@@ -335,27 +362,26 @@ class _WsFrameReader:
             if n == 126:
                 if not self._fill(4):
                     return out or None
-                n = int.from_bytes(self.buf[2:4], "big")
+                n = int.from_bytes(self.buf[o + 2:o + 4], "big")
                 pos = 4
             elif n == 127:
                 if not self._fill(10):
                     return out or None
-                n = int.from_bytes(self.buf[2:10], "big")
+                n = int.from_bytes(self.buf[o + 2:o + 10], "big")
                 pos = 10
             if n > MAX_FRAME_BYTES:
                 raise ValueError("oversize ws frame")
             need = pos + (4 if masked else 0) + n
             if not self._fill(need):
                 return out or None
+            o = self._off
             if masked:
-                mkey = bytes(self.buf[pos:pos + 4])
-                payload = bytes(
-                    b ^ mkey[i % 4]
-                    for i, b in enumerate(self.buf[pos + 4:need])
-                )
+                mkey = bytes(self.buf[o + pos:o + pos + 4])
+                payload = _ws_xor_mask(
+                    bytes(self.buf[o + pos + 4:o + need]), mkey)
             else:
-                payload = bytes(self.buf[pos:need])
-            del self.buf[:need]
+                payload = bytes(self.buf[o + pos:o + need])
+            self._off += need
             if opcode == 0x8:  # close
                 return out or None
             if opcode == 0x9:  # ping -> pong (server unmasked, client masked)
@@ -379,7 +405,10 @@ class _WsFrameReader:
                     self._frag = bytearray(payload)
                     continue
                 out.append(payload)
-                if not self.buf:
+                if not self._avail():
+                    if self._off:
+                        del self.buf[:]
+                        self._off = 0
                     return out
                 continue
             # pong / unknown control: ignore and keep parsing

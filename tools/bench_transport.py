@@ -17,15 +17,41 @@ from detectmateservice_amd.utils.synthetic import AuditLogGenerator
 
 
 def main():
+    # usage: bench_transport.py [lines] [scheme]   scheme in ipc|tcp|tls|ws
     n = int(sys.argv[1]) if len(sys.argv) > 1 else 400_000
+    scheme = sys.argv[2] if len(sys.argv) > 2 else "ipc"
     tmp = tempfile.mkdtemp(prefix="dmx-tr-")
-    addr = f"ipc://{tmp}/t-{uuid.uuid4().hex[:6]}.ipc"
+    tls_in = tls_out = None
+    if scheme == "ipc":
+        addr = f"ipc://{tmp}/t-{uuid.uuid4().hex[:6]}.ipc"
+    elif scheme in ("tcp", "ws"):
+        addr = f"{scheme}://127.0.0.1:0"
+    elif scheme == "tls":
+        import subprocess
+
+        from detectmateservice_amd.settings import TlsInputConfig, TlsOutputConfig
+
+        key, crt, pem = f"{tmp}/key.pem", f"{tmp}/crt.pem", f"{tmp}/ck.pem"
+        subprocess.run(
+            ["openssl", "req", "-x509", "-newkey", "rsa:2048", "-nodes",
+             "-keyout", key, "-out", crt, "-days", "1", "-subj",
+             "/CN=localhost", "-addext", "subjectAltName=DNS:localhost"],
+            check=True, capture_output=True)
+        with open(pem, "w") as fh:
+            fh.write(open(key).read() + open(crt).read())
+        tls_in = TlsInputConfig(cert_key_file=pem)
+        tls_out = TlsOutputConfig(ca_file=crt, server_name="localhost")
+        addr = "tls+tcp://127.0.0.1:0"
+    else:
+        raise SystemExit(f"unknown scheme {scheme}")
     gen = AuditLogGenerator(seed=7, anomaly_rate=0.0)
     frames = [LogSchema(logID=f"m{i}", log=gen.line()[0]).serialize()
               for i in range(n)]
     total_bytes = sum(len(f) for f in frames)
 
-    listener = PairListener(addr, buffer_size=8192)
+    listener = PairListener(addr, buffer_size=8192, tls_config=tls_in)
+    if scheme in ("tcp", "ws", "tls"):
+        addr = addr.replace(":0", f":{listener.bound_port}")
     got = [0]
     done = threading.Event()
 
@@ -41,7 +67,7 @@ def main():
 
     t = threading.Thread(target=drain, daemon=True)
     t.start()
-    feeder = PairDialer(addr, buffer_size=8192)
+    feeder = PairDialer(addr, buffer_size=8192, tls_config=tls_out)
     assert feeder.wait_connected(10.0)
     t0 = time.perf_counter()
     sent = 0
@@ -52,7 +78,7 @@ def main():
         sent += k
     done.wait(30.0)
     dt = time.perf_counter() - t0
-    print(f"transport only: {n/dt:,.0f} lines/s  "
+    print(f"[{scheme}] transport only: {n/dt:,.0f} lines/s  "
           f"({total_bytes/dt/1e6:.0f} MB/s)  recv={got[0]}")
     feeder.close()
     listener.close()
@@ -69,6 +95,13 @@ def main():
         dt = (time.perf_counter() - t0) / iters
         print(f"decode_log_batch(8192): {dt*1e3:.2f} ms "
               f"({8192/dt:,.0f} lines/s)")
+        # the TLS/ws packed-mode fallback converts byte frames per batch
+        t0 = time.perf_counter()
+        for _ in range(iters):
+            _dmx_C.decode_log_batch_packed(batch, 256, False)
+        dt = (time.perf_counter() - t0) / iters
+        print(f"decode_log_batch_packed(8192) [packed-fallback conversion]: "
+              f"{dt*1e3:.2f} ms ({8192/dt:,.0f} lines/s)")
 
 
 if __name__ == "__main__":
