@@ -245,6 +245,8 @@ def main():
         # the PARSER stage drain by timing the feed + drain of the
         # detector input: we time until the parser has accepted all
         # frames AND the pipeline is idle (sink quiet + sockets drained).
+        print(f"# t+0.0 starting timed pump ({args.lines} lines)",
+              file=sys.stderr, flush=True)
         t0 = time.perf_counter()
         if args.shm:
             # C++ GIL-free load generator (ops/csrc/shm_ring.cpp
@@ -257,8 +259,12 @@ def main():
             c2s_path, _ = _shm_paths(feeder.addr)
             cpp_feeder = _c.ShmFeeder(c2s_path, frames, _shm_ring_bytes())
             cpp_feeder.start(args.lines)
-            cpp_feeder.join(600_000)
-            assert cpp_feeder.done(), "C++ feeder timed out"
+            while not cpp_feeder.done():
+                cpp_feeder.join(5000)
+                print(f"# t+{time.perf_counter()-t0:.1f}s cpp feeder "
+                      f"sent={cpp_feeder.sent()}", file=sys.stderr, flush=True)
+                if time.perf_counter() - t0 > 240:
+                    raise SystemExit("C++ feeder stalled")
             sent_total = cpp_feeder.sent()
         else:
             sent_total = 0
