@@ -279,7 +279,12 @@ class ShmRing {
     const uint64_t off = pos & mask_;
     const size_t first = std::min<size_t>(n, size_ - off);
     std::memcpy(dst, data_ + off, first);
-    if (first < n) std::memcpy(dst, data_, n - first);
+    // wrap continuation APPENDS at dst+first: writing to dst overwrote
+    // the head bytes — a frame LENGTH straddling the ring boundary read
+    // as 0 and the consumer livelocked 4 bytes into the frame (found at
+    // exactly 13x ring-size tail during the 8M-line service bench);
+    // straddling payloads silently corrupted.
+    if (first < n) std::memcpy(dst + first, data_, n - first);
   }
 
   std::string path_;

@@ -838,3 +838,42 @@ def test_nng_golden_bytes_against_dialer():
     finally:
         dialer.close()
         srv.close()
+
+
+def test_shm_ring_wrap_straddling_reads(monkeypatch):
+    """Regression: ShmRing reads that straddle the ring boundary must
+    append the wrapped tail at dst+first (a straddling LENGTH prefix
+    used to read as 0 -> consumer livelock; straddling payloads
+    silently corrupted). Layout forces the second frame's length to
+    straddle byte ring_size-2."""
+    import uuid as uuid_mod
+
+    from detectmateservice_amd import ops
+
+    if not ops.have_extension():
+        pytest.skip("extension not built")
+    from detectmateservice_amd.ops import _dmx_C
+
+    ring_bytes = 1 << 16
+    path = f"/dev/shm/dmx-wrap-{uuid_mod.uuid4().hex[:8]}"
+    ring = _dmx_C.ShmRing(path, ring_bytes, True)
+    try:
+        # frame A fills the ring to 2 bytes short of the boundary for
+        # the next length prefix: 4 + lenA = ring - 2
+        lenA = ring_bytes - 4 - 2
+        a = bytes(range(256)) * (lenA // 256) + bytes(range(lenA % 256))
+        b = b"straddle-me-" * 20
+        assert ring.write_frames([a]) == 1
+        got = ring.read_batch(1, 1000)
+        assert bytes(got[0]) == a
+        # now head == tail == ring-2: frame B's length prefix straddles
+        assert ring.write_frames([b]) == 1
+        got = ring.read_batch(1, 1000)
+        assert len(got) == 1 and bytes(got[0]) == b
+        # packed variant exercises the same get_bytes path via scratch
+        assert ring.write_frames([a]) == 1
+        got = ring.read_batch(1, 1000)
+        assert bytes(got[0]) == a  # payload itself straddles now
+    finally:
+        import os as os_mod
+        os_mod.unlink(path)
