@@ -252,3 +252,111 @@ def test_sklearn_frequency_resolvable():
     for name in ("SklearnDetector", "FrequencyDetector"):
         path, _ = ComponentResolver().resolve(name)
         assert path.endswith(name)
+
+
+def test_tokenizer_detector_compression_surprise():
+    """TokenizerDetector: lines fitting the learned BPE structure score
+    low; structurally novel lines exceed the z-threshold and alert."""
+    from detectmateservice_amd.library.detectors.tokenizer_llm import (
+        TokenizerDetector,
+    )
+    from detectmateservice_amd.schemas import DetectorSchema, ParserSchema
+
+    det = TokenizerDetector({
+        "method_type": "tokenizer_detector",
+        "vocab_size": 512,
+        "z_threshold": 3.0,
+        "data_use_training": 0,
+    })
+    train = [
+        ParserSchema(logID=f"t{i}",
+                     log=f"type=SYSCALL msg=audit({i}.0:1): arch=c000003e "
+                         f"syscall=59 success=yes exit=0 pid={i}")
+        for i in range(400)
+    ]
+    det.train(train)
+    normal = ParserSchema(
+        logID="n", log="type=SYSCALL msg=audit(999.0:1): arch=c000003e "
+                       "syscall=59 success=yes exit=0 pid=999")
+    a = DetectorSchema()
+    assert det.detect(normal, a) is False
+    weird = ParserSchema(logID="w", log="\x01\x02zq9!jx@#kk%%&&~~``\x7f" * 6)
+    a = DetectorSchema()
+    assert det.detect(weird, a) is True
+    assert a.detectorType == "tokenizer_detector" and a.score > 3.0
+
+    # checkpoint roundtrip keeps the learned tokenizer + calibration
+    st = det.state_dict()
+    det2 = TokenizerDetector({"method_type": "tokenizer_detector"})
+    det2.load_state_dict(st)
+    a = DetectorSchema()
+    assert det2.detect(weird, a) is True
+
+
+def test_llm_assist_detector_enriches_alerts_via_stub():
+    """LLMAssistDetector: inner detector fires; the alert is enriched
+    through an OpenAI-compatible stub endpoint; LLM failure never
+    suppresses the alert."""
+    import http.server
+    import json as json_mod
+    import threading
+
+    from detectmateservice_amd.library.detectors.tokenizer_llm import (
+        LLMAssistDetector,
+    )
+    from detectmateservice_amd.schemas import DetectorSchema, ParserSchema
+
+    class Stub(http.server.BaseHTTPRequestHandler):
+        def do_POST(self):
+            n = int(self.headers["Content-Length"])
+            body = json_mod.loads(self.rfile.read(n))
+            assert body["messages"][0]["role"] == "user"
+            out = json_mod.dumps({"choices": [{"message": {
+                "content": "unseen account value"}}]}).encode()
+            self.send_response(200)
+            self.send_header("Content-Type", "application/json")
+            self.send_header("Content-Length", str(len(out)))
+            self.end_headers()
+            self.wfile.write(out)
+
+        def log_message(self, *a):  # quiet
+            pass
+
+    srv = http.server.HTTPServer(("127.0.0.1", 0), Stub)
+    t = threading.Thread(target=srv.serve_forever, daemon=True)
+    t.start()
+    try:
+        det = LLMAssistDetector({
+            "method_type": "llm_assist_detector",
+            "inner": "new_value_detector",
+            "llm_base_url": f"http://127.0.0.1:{srv.server_port}",
+            "data_use_training": 1,
+            "global": {"g": {"header_variables": [{"pos": "URL"}]}},
+        })
+        det.train([ParserSchema(EventID=1, logID="t",
+                                logFormatVariables={"URL": "/ok"})])
+        hit = ParserSchema(EventID=1, logID="x",
+                           logFormatVariables={"URL": "/evil"})
+        a = DetectorSchema()
+        assert det.detect(hit, a) is True
+        assert "LLM: unseen account value" in a.description
+        assert a.alertsObtain.get("llm_summary") == "unseen account value"
+    finally:
+        srv.shutdown()
+
+    # endpoint down: alert still fires, description unenriched
+    det2 = LLMAssistDetector({
+        "method_type": "llm_assist_detector",
+        "inner": "new_value_detector",
+        "llm_base_url": "http://127.0.0.1:1",  # refused
+        "llm_timeout_s": 0.3,
+        "data_use_training": 1,
+        "global": {"g": {"header_variables": [{"pos": "URL"}]}},
+    })
+    det2.train([ParserSchema(EventID=1, logID="t",
+                             logFormatVariables={"URL": "/ok"})])
+    a = DetectorSchema()
+    assert det2.detect(ParserSchema(EventID=1, logID="y",
+                                    logFormatVariables={"URL": "/bad"}),
+                       a) is True
+    assert "LLM:" not in (a.description or "")
