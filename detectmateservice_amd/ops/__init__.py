@@ -209,15 +209,16 @@ class TemplateMatcher:
             z = lambda *shape: torch.zeros(shape, dtype=torch.int32, device=lines.device)  # noqa: E731
             return {"event_id": z(0), "fmt_caps": z(0, self.max_fmt_caps, 2),
                     "n_fmt_caps": z(0), "caps": z(0, self.max_caps, 2),
-                    "n_caps": z(0)}
+                    "n_caps": z(0), "span_start": z(0), "span_end": z(0)}
         if lines.is_cuda:
-            ev, fc, nfc, caps, ncaps = _require_ext().template_match(
+            (ev, fc, nfc, caps, ncaps, sps, spe) = _require_ext().template_match(
                 lines, line_len, self.fmt_bytes, self.fmt_seg_off,
                 self.seg_bytes, self.seg_off, self.tpl_seg_start,
                 self.lowercase, self.max_fmt_caps, self.max_caps,
             )
             return {"event_id": ev, "fmt_caps": fc, "n_fmt_caps": nfc,
-                    "caps": caps, "n_caps": ncaps}
+                    "caps": caps, "n_caps": ncaps, "span_start": sps,
+                    "span_end": spe}
         if _C is not None:
             # C++ twin of the kernel (the pure-Python matcher measured
             # ~9.5k lines/s and bottlenecked CPU service mode)
@@ -226,9 +227,34 @@ class TemplateMatcher:
                 self.seg_bytes, self.seg_off, self.tpl_seg_start,
                 self.lowercase, self.max_fmt_caps, self.max_caps,
             )
-            return {"event_id": ev, "fmt_caps": fc, "n_fmt_caps": nfc,
-                    "caps": caps, "n_caps": ncaps}
-        return self._match_cpu(lines, line_len)
+            out = {"event_id": ev, "fmt_caps": fc, "n_fmt_caps": nfc,
+                   "caps": caps, "n_caps": ncaps}
+            self._add_spans(out, line_len)
+            return out
+        out = self._match_cpu(lines, line_len)
+        self._add_spans(out, line_len)
+        return out
+
+    @staticmethod
+    def _add_spans(match: dict, line_len: torch.Tensor) -> None:
+        """CPU-path twin of the kernel's span outputs: content span =
+        last header capture when the format matched, else [0, len)."""
+        nfc = match["n_fmt_caps"].long()
+        has_hdr = nfc > 0
+        last = (nfc - 1).clamp(min=0)
+        fc = match["fmt_caps"]
+        start = torch.where(
+            has_hdr,
+            fc.gather(1, last.view(-1, 1, 1).expand(-1, 1, 2))[:, 0, 0].long(),
+            torch.zeros_like(nfc),
+        )
+        end = torch.where(
+            has_hdr,
+            fc.gather(1, last.view(-1, 1, 1).expand(-1, 1, 2))[:, 0, 1].long(),
+            line_len.long(),
+        )
+        match["span_start"] = start.int()
+        match["span_end"] = end.int()
 
     # -- CPU reference (same algorithm, used for CI + kernel parity tests) --
     def _match_cpu(self, lines: torch.Tensor, line_len: torch.Tensor):

@@ -34,8 +34,8 @@ void dmx_launch_bert_fused_timed(const void*, const void*, const void*,
 void dmx_launch_template_match(const void*, const void*, int, int,
                                const void*, const void*, int, const void*,
                                int, const void*, const void*, int, int,
-                               void*, void*, void*, void*, void*, int, int,
-                               hipStream_t);
+                               void*, void*, void*, void*, void*, void*,
+                               void*, int, int, hipStream_t);
 void dmx_launch_watch_hashes(const void*, int, const void*, const void*,
                              const void*, int, const void*, const void*,
                              int, const void*, int, int, int, void*,
@@ -200,6 +200,8 @@ std::vector<torch::Tensor> template_match(
   auto n_fmt_caps = torch::zeros({B}, opts);
   auto caps = torch::zeros({B, max_caps, 2}, opts);
   auto n_caps = torch::zeros({B}, opts);
+  auto span_start = torch::empty({B}, opts);
+  auto span_end = torch::empty({B}, opts);
   dmx_launch_template_match(
       lines.data_ptr(), line_len.data_ptr(), (int)B, (int)max_len,
       fmt_bytes.numel() ? fmt_bytes.data_ptr() : nullptr,
@@ -207,8 +209,9 @@ std::vector<torch::Tensor> template_match(
       seg_bytes.data_ptr(), (int)seg_bytes.numel(), seg_off.data_ptr(),
       tpl_seg_start.data_ptr(), n_tpl, lower ? 1 : 0, event_id.data_ptr(),
       fmt_caps.data_ptr(), n_fmt_caps.data_ptr(), caps.data_ptr(),
-      n_caps.data_ptr(), (int)max_fmt_caps, (int)max_caps, cur_stream());
-  return {event_id, fmt_caps, n_fmt_caps, caps, n_caps};
+      n_caps.data_ptr(), span_start.data_ptr(), span_end.data_ptr(),
+      (int)max_fmt_caps, (int)max_caps, cur_stream());
+  return {event_id, fmt_caps, n_fmt_caps, caps, n_caps, span_start, span_end};
 }
 
 torch::Tensor bert_fused_probe(torch::Tensor lines, torch::Tensor start,

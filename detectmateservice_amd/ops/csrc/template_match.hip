@@ -100,6 +100,12 @@ void dmx_template_match(
     int* __restrict__ n_fmt_caps,   // [B]
     int* __restrict__ caps,         // [B, max_caps, 2]
     int* __restrict__ n_caps,       // [B]
+    // content span (what the transformer scorer consumes): the last
+    // header capture when the format matched, else [0, len) — computed
+    // here so the hot loop needs no gather/where chain of ~6 torch
+    // kernels per batch (profiles/r10)
+    int* __restrict__ span_start,   // [B]
+    int* __restrict__ span_end,     // [B]
     int max_fmt_caps, int max_caps) {
   const int wid = threadIdx.x / DMX_WAVE;
   const int lane = threadIdx.x % DMX_WAVE;
@@ -162,6 +168,10 @@ void dmx_template_match(
   if (lane == 0) {
     event_id[line_idx] = eid;
     n_caps[line_idx] = nc;
+    if (span_start) {
+      span_start[line_idx] = content_start;
+      span_end[line_idx] = content_end;
+    }
   }
 }
 
@@ -171,7 +181,8 @@ extern "C" void dmx_launch_template_match(
     const void* seg_bytes, int seg_bytes_len, const void* seg_off,
     const void* tpl_seg_start, int n_tpl, int lower,
     void* event_id, void* fmt_caps, void* n_fmt_caps, void* caps,
-    void* n_caps, int max_fmt_caps, int max_caps, hipStream_t stream) {
+    void* n_caps, void* span_start, void* span_end, int max_fmt_caps,
+    int max_caps, hipStream_t stream) {
   const int grid = (B + TM_WAVES - 1) / TM_WAVES;
   const size_t lds = ((seg_bytes_len + 15) & ~15) + TM_WAVES * TM_MAX_LINE;
   hipLaunchKernelGGL(dmx_template_match, dim3(grid), dim3(TM_WAVES * DMX_WAVE),
@@ -182,5 +193,6 @@ extern "C" void dmx_launch_template_match(
                      (const unsigned char*)seg_bytes, seg_bytes_len,
                      (const int*)seg_off, (const int*)tpl_seg_start, n_tpl,
                      lower, (int*)event_id, (int*)fmt_caps, (int*)n_fmt_caps,
-                     (int*)caps, (int*)n_caps, max_fmt_caps, max_caps);
+                     (int*)caps, (int*)n_caps, (int*)span_start,
+                     (int*)span_end, max_fmt_caps, max_caps);
 }

@@ -119,7 +119,11 @@ class GpuPipeline:
             hashes = ops.watch_hashes(lines, match, self.specs, self.config.lowercase)
             if train_upto > 0:
                 self.hashsets.insert(hashes[:train_upto])
-            if train_upto < B:
+            if train_upto == 0:
+                # steady state: probe output IS the result (a zeros +
+                # full-copy here cost 2 kernel launches per batch)
+                nv_unseen = self.hashsets.probe(hashes)
+            elif train_upto < B:
                 unseen_tail = self.hashsets.probe(hashes[train_upto:])
                 nv_unseen = torch.zeros(
                     (B, self.specs.shape[0]), dtype=torch.int32, device=lines.device
@@ -137,23 +141,11 @@ class GpuPipeline:
         if _rng:
             torch.cuda.nvtx.range_push("dmx::transformer")
         if self.model is not None:
-            # content span: last fmt capture when the header matched,
-            # else the whole line
-            nfc = match["n_fmt_caps"].long()
-            has_hdr = nfc > 0
-            last = (nfc - 1).clamp(min=0)
-            fc = match["fmt_caps"]
-            start = torch.where(
-                has_hdr,
-                fc.gather(1, last.view(-1, 1, 1).expand(-1, 1, 2))[:, 0, 0].long(),
-                torch.zeros_like(nfc),
-            )
-            end = torch.where(
-                has_hdr,
-                fc.gather(1, last.view(-1, 1, 1).expand(-1, 1, 2))[:, 0, 1].long(),
-                line_len.long(),
-            )
-            scores = self.model.score_spans(lines, start.int(), end.int())
+            # content span comes straight from the match kernel
+            # (span_start/span_end outputs — the former gather/where
+            # chain here was ~6 kernel launches per batch, profiles/r10)
+            scores = self.model.score_spans(
+                lines, match["span_start"], match["span_end"])
         if _rng:
             torch.cuda.nvtx.range_pop()
 
