@@ -42,18 +42,12 @@ typedef __attribute__((address_space(3))) float lds_float;
 #define BF_DH 64
 #define BF_FFN 512
 
-// LDS strides (elements, 16-B-aligned rows). Round-2 PMC found 6.1B
-// SQ_LDS_BANK_CONFLICT cycles (~16% of a CU's step time): the +8 pads
-// left every A-fragment ds_read_b128 ~2-way conflicted (stride mod 64
-// dwords = 4 collides (r,h)-pairs inside the 16-lane groups). +16 pads
-// (stride mod 64 dwords = 8) are conflict-free for the fragment
-// pattern (brute-force over the b128 lane groups; LN's short8v reads
-// go 2->3-way on x, a far smaller op count).
-#define XS (BF_H + 16)      // 144, x rows
-#define QKS (2 * BF_H + 16) // 272, Q|K rows and FFN-half rows
-#define VTS (BF_S + 16)     // 80, transposed-V and P rows
-#define O_OFF (BF_WAVES * 16 * VTS)      // 10240: attn-out after P tiles
-#define BUF_ELEMS (O_OFF + BF_S * XS)    // 19456 = P+O (> QK, FFN-half)
+// LDS strides (elements); +8 pads keep rows 16-B aligned and bank-spread
+#define XS (BF_H + 8)       // 136, x rows
+#define QKS (2 * BF_H + 8)  // 264, Q|K rows and FFN-half rows
+#define VTS (BF_S + 8)      // 72, transposed-V and P rows
+#define O_OFF (BF_WAVES * 16 * VTS)    // 9216: attn-out after the P tiles
+#define BUF_ELEMS (BF_S * QKS + 1024)  // 17920 = max(QK, P+O, FFN-half)
 
 // bf16 weight-blob element offsets (host packs identically: bert_tiny.py)
 #define WB_TOK 0
