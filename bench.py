@@ -63,7 +63,7 @@ def parse_args():
                         "off: device-resident line pool (GPU pipeline only)")
     p.add_argument("--shards", type=int, default=4,
                    help="ingest shards (ring + feeder + reader threads) per rank")
-    p.add_argument("--chunk", type=int, default=8192,
+    p.add_argument("--chunk", type=int, default=16384,
                    help="frames per ring read (constant size so pinned "
                         "buffers come from the caching allocator)")
     p.add_argument("--line-buffer", action="store_true",
