@@ -44,10 +44,18 @@ typedef __attribute__((address_space(3))) float lds_float;
 
 // LDS strides (elements); +8 pads keep rows 16-B aligned and bank-spread
 #define XS (BF_H + 8)       // 136, x rows
-#define QKS (2 * BF_H + 8)  // 264, Q|K rows and FFN-half rows
+// QKS at +16 (not +8): stride 136 dw = 8 mod 64 makes the FFN W2 /
+// attention A-fragment ds_read_b128 groups conflict-free (PMC: FFN's
+// 6.98B conflict cycles are 100% A-reads — probe ABL=2 zeroes them).
+// Q|K (64x272=17408) and FFN halves still fit inside BUF_ELEMS (17920,
+// sized by the P+O region), so this costs ZERO extra LDS. Measured
+// NEUTRAL on wall (7.30M both ways): in this latency-bound regime the
+// conflict cycles hide under the operand waits — kept because it is
+// free and removes them from every future profile.
+#define QKS (2 * BF_H + 16)  // 272, Q|K rows and FFN-half rows
 #define VTS (BF_S + 8)      // 72, transposed-V and P rows
 #define O_OFF (BF_WAVES * 16 * VTS)    // 9216: attn-out after the P tiles
-#define BUF_ELEMS (BF_S * QKS + 1024)  // 17920 = max(QK, P+O, FFN-half)
+#define BUF_ELEMS (O_OFF + BF_S * XS)  // 17920 = P+O (>= QK 17408)
 
 // bf16 weight-blob element offsets (host packs identically: bert_tiny.py)
 #define WB_TOK 0

@@ -363,12 +363,13 @@ class ShmFeeder {
   void stop() { stop_ = true; }
 
   void join(int timeout_ms) {
+    if (!th_.joinable()) return;  // never started
     py::gil_scoped_release release;
     const auto deadline = std::chrono::steady_clock::now() +
                           std::chrono::milliseconds(timeout_ms);
     while (!done_.load() && std::chrono::steady_clock::now() < deadline)
       std::this_thread::sleep_for(std::chrono::milliseconds(1));
-    if (done_.load() && th_.joinable()) th_.join();
+    if (done_.load()) th_.join();
   }
 
  private:
