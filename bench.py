@@ -229,7 +229,7 @@ def main() -> None:
                     LogSchema(logID=f"l{j}", log=gen.line()[0]).serialize()
                     for j in range(nlines)
                 ])
-            q: "queue.Queue" = queue.Queue(maxsize=24)
+            q: "queue.Queue" = queue.Queue(maxsize=8)
             feeders.append(ShardFeeder(ring, pool, nlines * n_total_steps))
             readers.append(ShardReader(ring, nlines, n_total_steps,
                                         args.chunk, args.max_len, q, use_gpu))
