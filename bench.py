@@ -240,6 +240,21 @@ def main() -> None:
         for r in readers:
             r.start()
 
+        # H2D copies ride a dedicated stream so chunk N+1's upload
+        # overlaps chunk N's kernels (events order the compute stream
+        # behind each chunk's copy); no-op on CPU
+        copy_stream = torch.cuda.Stream(device) if use_gpu else None
+
+        def _upload(lines, lens):
+            if copy_stream is None:
+                return lines.to(device), lens.to(device), None
+            with torch.cuda.stream(copy_stream):
+                dl = lines.to(device, non_blocking=True)
+                dn = lens.to(device, non_blocking=True)
+                ev = torch.cuda.Event()
+                ev.record(copy_stream)
+            return dl, dn, ev
+
         def step(i: int) -> None:
             outs = []
             for q in queues:
@@ -250,8 +265,9 @@ def main() -> None:
                     if item == "step_end":
                         break
                     lines, lens = item
-                    dl = lines.to(device, non_blocking=True)
-                    dn = lens.to(device, non_blocking=True)
+                    dl, dn, ev = _upload(lines, lens)
+                    if ev is not None:
+                        torch.cuda.current_stream(device).wait_event(ev)
                     if line_buf is not None:
                         line_buf.append(dl, dn)
                     outs.append(pipe.process_packed(dl, dn))
