@@ -268,3 +268,37 @@ def test_all_reduce_hashsets_multi_gpu():
 
     port = 29531
     tmp_mod.spawn(worker, args=(2, port), nprocs=2, join=True)
+
+
+@pytest.mark.gpu
+def test_rccl_collectives_execute_on_device():
+    """RCCL (torch.distributed backend "nccl" on ROCm) initializes and
+    executes real collectives on the MI355X. The builder's leases are
+    1 GPU, so world_size=1 is the largest RCCL group that can run here —
+    it still exercises communicator init + device collective calls
+    (broadcast, all_gather) through RCCL; the 2-GPU variant above covers
+    the multi-rank path when hardware allows."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    import torch.distributed as dist
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29533")
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    try:
+        from detectmateservice_amd.parallel import dist as dmx_dist
+
+        dev = torch.device("cuda", 0)
+        s = torch.tensor([3.0, 4.0], device=dev)
+        out = dmx_dist.all_gather_summaries(s)
+        assert out.shape == (1, 2) and out[0].tolist() == [3.0, 4.0]
+        lines = torch.full((4, 8), 9, dtype=torch.uint8, device=dev)
+        lens = torch.full((4,), 8, dtype=torch.int32, device=dev)
+        l2, n2 = dmx_dist.broadcast_packed(lines, lens, 0, dev)
+        assert torch.equal(l2, lines) and torch.equal(n2, lens)
+        frames = [b"rccl-frame-%d" % i for i in range(3)]
+        dmx_dist.broadcast_frames_src(frames, 0, dev)
+        # world=1: the src's own broadcast completes; a sink-side read
+        # needs a second rank (gated test above)
+    finally:
+        dist.destroy_process_group()
