@@ -327,6 +327,15 @@ def main() -> None:
     for i in range(args.warmup):
         step(i)
     sync()
+    if dist is not None and pipe.hashsets is not None and use_gpu:
+        # BASELINE config-4 semantics end-to-end: merge the data-parallel
+        # NewValue hash sets once after training — a REAL RCCL all-gather
+        # + batched insert over xGMI in the driver's multi-GPU runs
+        # (outside the timed region; parallel/dist.py::all_reduce_hashsets)
+        from detectmateservice_amd.parallel import dist as dmx_dist
+
+        dmx_dist.all_reduce_hashsets(pipe.hashsets.tables)
+        sync()
     if use_gpu and args.graph and ingest == "off":
         try:
             if pipe.enable_graph(args.batch):
