@@ -38,6 +38,10 @@ class DetectMateClient:
     def shutdown(self) -> dict:
         return requests.post(f"{self.url}/admin/shutdown", timeout=TIMEOUT_S).json()
 
+    def dp_sync(self) -> dict:
+        """Collective DP state merge (dist_mode "dp"); call on EVERY rank."""
+        return requests.post(f"{self.url}/admin/dp-sync", timeout=TIMEOUT_S).json()
+
     def checkpoint(self, path: str) -> dict:
         return requests.post(
             f"{self.url}/admin/checkpoint", json={"path": path}, timeout=TIMEOUT_S
@@ -65,7 +69,7 @@ def main(argv=None) -> int:
     )
     parser.add_argument("--url", default="http://127.0.0.1:8000")
     sub = parser.add_subparsers(dest="command", required=True)
-    for cmd in ("start", "stop", "status", "metrics", "shutdown"):
+    for cmd in ("start", "stop", "status", "metrics", "shutdown", "dp-sync"):
         sub.add_parser(cmd)
     rec = sub.add_parser("reconfigure")
     rec.add_argument("config_file")
@@ -84,6 +88,8 @@ def main(argv=None) -> int:
         elif args.command == "reconfigure":
             print(json.dumps(client.reconfigure(
                 args.config_file, args.persist, args.reload), indent=2))
+        elif args.command == "dp-sync":
+            print(json.dumps(client.dp_sync(), indent=2))
         elif args.command in ("checkpoint", "restore"):
             print(json.dumps(getattr(client, args.command)(args.path), indent=2))
         else:
