@@ -209,7 +209,7 @@ def main() -> None:
     if ingest == "shm":
         from detectmateservice_amd.ops import _dmx_C
 
-        shards = max(1, args.shards)
+        shards = max(1, min(args.shards, args.batch))  # no zero-line shards
         per_shard = args.batch // shards
         rem = args.batch - per_shard * shards
         shard_lines = [per_shard + (1 if i < rem else 0) for i in range(shards)]
