@@ -124,17 +124,22 @@ def all_reduce_hashsets(tables: torch.Tensor, group=None) -> None:
     world = dist.get_world_size(group)
     if world <= 1:
         return
-    gathered = [torch.empty_like(tables) for _ in range(world)]
-    dist.all_gather(gathered, tables, group=group)
     if not tables.is_cuda:  # pragma: no cover - CPU tables are python sets
         raise RuntimeError("all_reduce_hashsets expects device tensor tables")
+    # gloo cannot all_gather CUDA tensors (mixed gloo + GPU deployments:
+    # several services sharing one GPU): stage through host copies there
+    backend = dist.get_backend(group)
+    comm = tables if backend == "nccl" else tables.cpu()
+    gathered = [torch.empty_like(comm) for _ in range(world)]
+    dist.all_gather(gathered, comm, group=group)
     from ..ops import _dmx_C
 
     me = dist.get_rank(group)
     for r, other in enumerate(gathered):
         if r == me:
             continue
-        _dmx_C.hashset_insert(other.t().contiguous(), tables)
+        _dmx_C.hashset_insert(
+            other.t().contiguous().to(tables.device), tables)
 
 
 # ---------------------------------------------------------------------------
