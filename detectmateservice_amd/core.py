@@ -56,7 +56,8 @@ class Service:
 
             from .parallel import dist as dmx_dist
 
-            rank, world = dmx_dist.init_from_env(settings.dist_backend)
+            rank, world = dmx_dist.init_from_env(
+                settings.dist_backend, timeout_s=settings.dist_timeout_s)
             settings = settings.resolve_dist_placeholders(rank, world)
             if world > 1:
                 self.dist_ctx = SimpleNamespace(

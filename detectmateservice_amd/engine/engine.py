@@ -164,6 +164,11 @@ class Engine:
             return self._run_source_loop()
         if self.settings.engine_packed_mode and self._try_packed_loop():
             return
+        return self._run_socket_loop()
+
+    def _run_socket_loop(self) -> None:
+        """The plain batched frame loop (also the elastic-degradation
+        target when a dist collective path loses a peer)."""
         s = self.settings
         m = self.metrics
         self._log.info(
@@ -458,13 +463,27 @@ class Engine:
             "engine DIST loop started: mode=%s rank=%d/%d src=%d head=%s",
             mode, rank, world, src, head)
         if head:
-            self._dist_head_loop(dmx_dist, mode, rank, world, device)
+            degraded = self._dist_head_loop(dmx_dist, mode, rank, world,
+                                            device)
+            if degraded and not self._stop_event.is_set():
+                # elastic degradation (SURVEY §5.8 hard part): a dead
+                # peer broke the collective path — keep serving through
+                # the plain socket loop (out_addr fan-out, retry-then-
+                # drop) instead of going dark. Peers that come back
+                # rejoin at the next service restart (communicator
+                # reform is parallel/elastic.py's mechanism).
+                self._log.warning(
+                    "dist peer lost: degrading to the socket loop")
+                self.metrics.engine_dist_degraded.inc()
+                return self._run_socket_loop()
         else:
             self._dist_sink_loop(dmx_dist, mode, rank, world, src, device)
         self._log.info("engine dist loop exited")
 
-    def _dist_head_loop(self, dmx_dist, mode, rank, world, device) -> None:
+    def _dist_head_loop(self, dmx_dist, mode, rank, world, device) -> bool:
+        """Returns True when the collective path failed (degrade)."""
         s, m = self.settings, self.metrics
+        degraded = False
         while not self._stop_event.is_set():
             try:
                 frames = self._pair_sock.recv_many(
@@ -499,6 +518,7 @@ class Engine:
                     dmx_dist.send_frames(outs, rank + 1, device, flag)
             except RuntimeError as exc:
                 self._log.error("dist send failed (peer lost?): %s", exc)
+                degraded = True
                 break
             if outs:
                 nb = sum(len(o) for o in outs)
@@ -514,6 +534,7 @@ class Engine:
                                      dmx_dist.FRAME_SHUTDOWN)
         except RuntimeError:
             pass
+        return degraded
 
     def _dist_sink_loop(self, dmx_dist, mode, rank, world, src, device) -> None:
         m = self.metrics

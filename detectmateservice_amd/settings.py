@@ -188,6 +188,10 @@ class ServiceSettings(BaseModel):
     #: placeholders, substituted after the process group initializes.
     dist_mode: Optional[str] = None
     dist_src_rank: int = Field(default=0, ge=0)
+    #: collective timeout: a dead peer surfaces as an error after this
+    #: long, and the engine's dist loops DEGRADE to the socket path
+    #: (drop-don't-block posture on RCCL — SURVEY.md §5.8 hard part)
+    dist_timeout_s: float = Field(default=120.0, gt=0)
 
     @field_validator("dist_mode")
     @classmethod

@@ -77,6 +77,10 @@ class ServiceMetrics:
         self.engine_starts_total = get_counter(
             "engine_starts_total", "Number of engine starts"
         ).labels(**labels)
+        self.engine_dist_degraded = get_counter(
+            "engine_dist_degraded_total",
+            "Collective-path failures degraded to the socket loop",
+        ).labels(**labels)
         self.processing_duration_seconds = get_histogram(
             "processing_duration_seconds", "Per-line processing latency (s)"
         ).labels(**labels)

@@ -33,7 +33,9 @@ def _run_worker(rank, world, port, fn_name, tmp, q):
         q.put((rank, "err", traceback.format_exc()))
 
 
-def _launch(fn_name, tmp, world=2, free_port=None, timeout=180):
+def _launch(fn_name, tmp, world=2, free_port=None, timeout=180,
+            expect_results=None):
+    # expect_results < world when a rank dies on purpose (degrade test)
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     procs = [
@@ -44,7 +46,7 @@ def _launch(fn_name, tmp, world=2, free_port=None, timeout=180):
         p.start()
     results = {}
     try:
-        for _ in range(world):
+        for _ in range(world if expect_results is None else expect_results):
             rank, status, payload = q.get(timeout=timeout)
             assert status == "ok", f"rank {rank} failed:\n{payload}"
             results[rank] = payload
@@ -289,3 +291,65 @@ def test_stage_three_rank_chain(tmp_path, free_port):
     assert results[0] == "fed"
     assert results[1] == "relayed"
     assert results[2] == [b"c-%d" % i for i in range(3)]
+
+
+def _body_fanout_degrade(rank, world, tmp):
+    """The sink DIES mid-run; the head's collective times out and the
+    engine DEGRADES to the socket loop (drop-don't-block
+    elasticity must survive on the collective path)."""
+    import threading
+
+    from detectmateservice_amd import Service
+    from detectmateservice_amd.engine.sockets import PairDialer, PairListener, RecvTimeout
+
+    settings = _settings(
+        tmp, "deg", dist_mode="fanout",
+        out_addr=[f"ipc://{tmp}/deg-out.ipc"],
+        dist_timeout_s=6.0,
+    )
+    if rank == 1:
+        # sink comes up, participates in one round, then CRASHES (no
+        # shutdown sentinel, no clean exit of the collective)
+        svc = Service(settings)
+        t = threading.Thread(target=svc.run, daemon=True)
+        t.start()
+        time.sleep(2.5)
+        os._exit(0)  # hard death: the head's next broadcast must time out
+    svc = Service(settings)
+    t = threading.Thread(target=svc.run, daemon=True)
+    t.start()
+    try:
+        time.sleep(0.5)
+        feeder = PairDialer(svc.settings.engine_addr)
+        assert feeder.wait_connected(10.0)
+        sink = PairListener(f"ipc://{tmp}/deg-out.ipc")  # head's own out
+        # phase 1: healthy traffic
+        assert feeder.send(b"pre-death", block=True)
+        time.sleep(4.0)  # sink dies at ~2.5s; next broadcast times out
+        # phase 2: after degradation the head serves via its out_addr
+        got = None
+        deadline = time.monotonic() + 40
+        n = 0
+        while got is None and time.monotonic() < deadline:
+            feeder.send(b"post-death-%d" % n, block=False)
+            n += 1
+            try:
+                frame = sink.recv(timeout_ms=500)
+            except RecvTimeout:
+                continue
+            if frame.startswith(b"post-death"):
+                got = frame  # drain leftover pre-death frames
+        assert got is not None and got.startswith(b"post-death")
+        assert svc.metrics.engine_dist_degraded._value.get() >= 1
+        sink.close()
+        feeder.close()
+        return "degraded-and-serving"
+    finally:
+        svc.shutdown()
+        t.join(timeout=10.0)
+
+
+def test_fanout_head_degrades_to_socket_loop_on_peer_death(tmp_path, free_port):
+    results = _launch("_body_fanout_degrade", tmp_path, 2, free_port,
+                      timeout=120, expect_results=1)
+    assert results[0] == "degraded-and-serving"
