@@ -465,19 +465,19 @@ class Engine:
         if head:
             degraded = self._dist_head_loop(dmx_dist, mode, rank, world,
                                             device)
-            if degraded and not self._stop_event.is_set():
-                # elastic degradation (SURVEY §5.8 hard part): a dead
-                # peer broke the collective path — keep serving through
-                # the plain socket loop (out_addr fan-out, retry-then-
-                # drop) instead of going dark. Peers that come back
-                # rejoin at the next service restart (communicator
-                # reform is parallel/elastic.py's mechanism).
-                self._log.warning(
-                    "dist peer lost: degrading to the socket loop")
-                self.metrics.engine_dist_degraded.inc()
-                return self._run_socket_loop()
         else:
-            self._dist_sink_loop(dmx_dist, mode, rank, world, src, device)
+            degraded = self._dist_sink_loop(dmx_dist, mode, rank, world,
+                                            src, device)
+        if degraded and not self._stop_event.is_set():
+            # elastic degradation (SURVEY §5.8 hard part): a dead peer
+            # broke the collective path — keep serving through the plain
+            # socket loop (own engine_addr in, out_addr fan-out, retry-
+            # then-drop) instead of going dark. Peers that come back
+            # rejoin at the next service restart (communicator reform is
+            # parallel/elastic.py's mechanism).
+            self._log.warning("dist peer lost: degrading to the socket loop")
+            self.metrics.engine_dist_degraded.inc()
+            return self._run_socket_loop()
         self._log.info("engine dist loop exited")
 
     def _dist_head_loop(self, dmx_dist, mode, rank, world, device) -> bool:
@@ -536,9 +536,12 @@ class Engine:
             pass
         return degraded
 
-    def _dist_sink_loop(self, dmx_dist, mode, rank, world, src, device) -> None:
+    def _dist_sink_loop(self, dmx_dist, mode, rank, world, src,
+                        device) -> bool:
+        """Returns True when the collective path failed (degrade)."""
         m = self.metrics
         last = mode == "fanout" or rank == world - 1
+        degraded = False
         while not self._stop_event.is_set():
             try:
                 if mode == "fanout":
@@ -547,6 +550,7 @@ class Engine:
                     frames, flag = dmx_dist.recv_frames(rank - 1, device)
             except RuntimeError as exc:
                 self._log.error("dist recv failed (peer lost?): %s", exc)
+                degraded = True
                 break
             if flag == dmx_dist.FRAME_SHUTDOWN:
                 if mode == "stage" and rank + 1 < world:
@@ -579,6 +583,7 @@ class Engine:
                     dmx_dist.send_frames(outs, rank + 1, device, flag)
                 except RuntimeError as exc:
                     self._log.error("dist forward failed: %s", exc)
+                    degraded = True
                     break
                 if outs:
                     m.data_written_bytes_total.inc(sum(len(o) for o in outs))
@@ -586,6 +591,7 @@ class Engine:
             else:
                 for out in outs:
                     self._send_to_outputs(out)
+        return degraded
 
     def _run_source_loop(self) -> None:
         """Source mode: the component generates frames (reader services).

@@ -353,3 +353,57 @@ def test_fanout_head_degrades_to_socket_loop_on_peer_death(tmp_path, free_port):
     results = _launch("_body_fanout_degrade", tmp_path, 2, free_port,
                       timeout=120, expect_results=1)
     assert results[0] == "degraded-and-serving"
+
+
+def _body_fanout_sink_degrade(rank, world, tmp):
+    """The HEAD dies mid-run; the sink's broadcast fails and the sink
+    DEGRADES to its own socket loop — it keeps ingesting on its own
+    engine_addr and serving its out_addr (symmetric elasticity)."""
+    import threading
+
+    from detectmateservice_amd import Service
+    from detectmateservice_amd.engine.sockets import PairDialer, PairListener, RecvTimeout
+
+    settings = _settings(
+        tmp, "sdeg", dist_mode="fanout",
+        out_addr=[f"ipc://{tmp}/sdeg-out-{{rank}}.ipc"],
+        dist_timeout_s=6.0,
+    )
+    svc = Service(settings)
+    t = threading.Thread(target=svc.run, daemon=True)
+    t.start()
+    if rank == 0:
+        time.sleep(2.5)
+        os._exit(0)  # head hard death: sinks' broadcast must fail
+    try:
+        time.sleep(5.0)  # head dies at ~2.5s; the failed collective
+        # surfaces and the sink falls back to its socket loop
+        feeder = PairDialer(svc.settings.engine_addr)
+        assert feeder.wait_connected(15.0)
+        sink = PairListener(f"ipc://{tmp}/sdeg-out-1.ipc")
+        got = None
+        deadline = time.monotonic() + 40
+        n = 0
+        while got is None and time.monotonic() < deadline:
+            feeder.send(b"direct-%d" % n, block=False)
+            n += 1
+            try:
+                frame = sink.recv(timeout_ms=500)
+            except RecvTimeout:
+                continue
+            if frame.startswith(b"direct"):
+                got = frame
+        assert got is not None and got.startswith(b"direct")
+        assert svc.metrics.engine_dist_degraded._value.get() >= 1
+        sink.close()
+        feeder.close()
+        return "sink-degraded-and-serving"
+    finally:
+        svc.shutdown()
+        t.join(timeout=10.0)
+
+
+def test_fanout_sink_degrades_when_head_dies(tmp_path, free_port):
+    results = _launch("_body_fanout_sink_degrade", tmp_path, 2, free_port,
+                      timeout=120, expect_results=1)
+    assert results[1] == "sink-degraded-and-serving"
