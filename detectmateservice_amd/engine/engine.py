@@ -305,7 +305,7 @@ class Engine:
             s.engine_addr, max_len, pin, pipelined)
         stats_on = os.environ.get("DMX_ENGINE_STATS") == "1"
         st = {"n": 0, "frames": 0, "recv": 0.0, "proc": 0.0,
-              "last": time.perf_counter()}
+              "submit": 0.0, "drain": 0.0, "last": time.perf_counter()}
         prev = None          # in-flight token
         prev_meta = None     # (B, nbytes, t_submit, conn, conn_segs)
         reply_conn = getattr(self._pair_sock, "reply_conn", None)
@@ -341,6 +341,7 @@ class Engine:
             if prev is None:
                 return
             pB, pbytes, t_sub, pconn, psegs = prev_meta
+            t_d0 = time.perf_counter() if stats_on else 0.0
             try:
                 emit(collect(prev), pconn, psegs)
             except Exception as exc:  # noqa: BLE001
@@ -356,6 +357,7 @@ class Engine:
                 st["n"] += 1
                 st["frames"] += pB
                 st["proc"] += time.perf_counter() - t_sub
+                st["drain"] += time.perf_counter() - t_d0
             prev, prev_meta = None, None
 
         while not self._stop_event.is_set():
@@ -393,6 +395,8 @@ class Engine:
                 try:
                     prev = submit(lines, lens, blob, off)
                     prev_meta = (B, nbytes, t0, conn, segs)
+                    if stats_on:
+                        st["submit"] += time.perf_counter() - t0
                 except Exception as exc:  # noqa: BLE001
                     m.processing_errors_total.inc(B)
                     self._log.error(
@@ -427,11 +431,14 @@ class Engine:
                 if now - st["last"] > 2.0 and st["n"]:
                     self._log.info(
                         "[packed-stats] %d batches (%.0f fr/b): recv-wait "
-                        "%.1fms/b process %.1fms/b",
+                        "%.1fms/b process %.1fms/b "
+                        "(submit %.1f drain %.1f)",
                         st["n"], st["frames"] / st["n"],
                         st["recv"] * 1e3 / st["n"], st["proc"] * 1e3 / st["n"],
+                        st["submit"] * 1e3 / st["n"], st["drain"] * 1e3 / st["n"],
                     )
-                    st.update(n=0, frames=0, recv=0.0, proc=0.0, last=now)
+                    st.update(n=0, frames=0, recv=0.0, proc=0.0,
+                              submit=0.0, drain=0.0, last=now)
         drain_prev()
         self._log.info("engine packed loop exited")
         return True

@@ -13,7 +13,7 @@ import time
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 
-from detectmateservice_amd.pipeline import Pipeline, PipelineConfig
+from detectmateservice_amd.pipeline import GpuPipeline, PipelineConfig
 from detectmateservice_amd.utils.synthetic import (
     AUDIT_LOG_FORMAT,
     AUDIT_TEMPLATES,
@@ -23,7 +23,7 @@ from detectmateservice_amd import ops
 
 
 def main():
-    dev = "cuda"
+    dev = "cuda" if torch.cuda.is_available() else "cpu"
     cfg = PipelineConfig(
         templates=list(AUDIT_TEMPLATES),
         log_format=AUDIT_LOG_FORMAT,
@@ -33,7 +33,7 @@ def main():
         score_threshold=1.0e9,
         max_len=256,
     )
-    pipe = Pipeline(cfg, device=dev)
+    pipe = GpuPipeline(cfg, device=dev)
     gen = AuditLogGenerator(seed=7, anomaly_rate=0.0)
     maxB = 65536
     raw = [gen.line()[0].encode() for _ in range(8192)]
@@ -48,12 +48,12 @@ def main():
         l, n = lines[:B], lens[:B]
         for _ in range(3):
             pipe.process_packed(l, n)
-        torch.cuda.synchronize()
+        torch.cuda.synchronize() if torch.cuda.is_available() else None
         t0 = time.perf_counter()
         iters = 10
         for _ in range(iters):
             pipe.process_packed(l, n)
-        torch.cuda.synchronize()
+        torch.cuda.synchronize() if torch.cuda.is_available() else None
         dt = (time.perf_counter() - t0) / iters
         print(f"B={B:6d}  {dt*1e3:8.3f} ms/launch  {B/dt/1e6:7.2f} M lines/s",
               flush=True)
