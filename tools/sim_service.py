@@ -54,23 +54,28 @@ def main():
     ring = _dmx_C.ShmRing(path, RING, True)
     feeder = _dmx_C.ShmFeeder(path, frames, RING)
 
-    cfg = PipelineConfig(
-        templates=list(AUDIT_TEMPLATES), log_format=AUDIT_LOG_FORMAT,
-        watches=[{"kind": "variable", "pos": 5, "event": 1}],
-        train_lines=0, use_transformer=True, score_threshold=1.0e9,
-        max_len=256)
-    pipe = GpuPipeline(cfg, device=dev)
-    static_l = torch.randint(32, 127, (args.batch, 256), dtype=torch.uint8,
-                             device=dev)
-    static_n = torch.full((args.batch,), 150, dtype=torch.int32, device=dev)
+    pipe = static_l = static_n = None
+    if args.kernels:
+        cfg = PipelineConfig(
+            templates=list(AUDIT_TEMPLATES), log_format=AUDIT_LOG_FORMAT,
+            watches=[{"kind": "variable", "pos": 5, "event": 1}],
+            train_lines=0, use_transformer=True, score_threshold=1.0e9,
+            max_len=256)
+        pipe = GpuPipeline(cfg, device=dev)
+        static_l = torch.randint(32, 127, (args.batch, 256),
+                                 dtype=torch.uint8, device=dev)
+        static_n = torch.full((args.batch,), 150, dtype=torch.int32,
+                              device=dev)
 
     def submit(lines_cpu, lens_cpu):
         if args.h2d:
             dl = lines_cpu.to(dev, non_blocking=True)
             dn = lens_cpu.to(dev, non_blocking=True)
-        else:
+        elif args.kernels:
             B = lines_cpu.shape[0]
             dl, dn = static_l[:B], static_n[:B]
+        else:
+            return (None, lines_cpu)  # pure CPU loop (no CUDA at all)
         if not args.kernels:
             return (dl, lines_cpu)
         return (pipe.process_packed(dl, dn), lines_cpu)
@@ -79,7 +84,7 @@ def main():
         out, _held = token
         if args.kernels:
             bool(out["anomaly"].any())  # the engine's alert readback sync
-        else:
+        elif args.h2d:
             torch.cuda.synchronize()
 
     feeder.start(args.lines)
