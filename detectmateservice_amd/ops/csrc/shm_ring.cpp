@@ -29,6 +29,7 @@
 #include <chrono>
 #include <cerrno>
 #include <cstring>
+#include <cstdlib>
 #include <deque>
 #include <memory>
 #include <stdexcept>
@@ -188,73 +189,96 @@ class ShmRing {
   // mapped ring; no kernel copies, no Python objects per frame.
   py::tuple read_batch_packed(int max_frames, int timeout_ms, int max_len,
                               bool pin) {
-    std::vector<std::pair<const uint8_t*, size_t>> raw;
+    // CHUNKED in-place decode: one read call drains up to max_frames,
+    // but scans/decodes/releases the ring in bounded-byte chunks
+    // (kReadChunkBytes). Measured (profiles/r12_engine_batch_cliff.txt):
+    // a single in-place pass over >=5.4 MB of a producer-saturated ring
+    // degrades ~6x on the target host's memory system, collapsing
+    // engine batches >=32768; bounded chunks with early tail release
+    // keep the live footprint small and overlap the feeder's refill.
+    const size_t chunk_bytes = read_chunk_bytes();
     // Frames crossing the ring boundary are copied to scratch. A deque is
     // required (NOT vector<string>): raw holds .data() pointers into these
     // strings, and short strings store their bytes inline (SSO) — a
     // vector reallocation would move the string objects and dangle every
     // SSO pointer already recorded. Deque push_back never moves elements.
     std::deque<std::string> wrapped;
-    uint64_t tail0, tail;
+    auto lopts = torch::TensorOptions().dtype(torch::kUInt8);
+    if (pin) lopts = lopts.pinned_memory(true);
+    // constant-size alloc (cache-friendly for the pinned allocator);
+    // narrowed to the decoded row count at the end
+    torch::Tensor lines = torch::empty({(int64_t)max_frames, max_len}, lopts);
+    torch::Tensor lens = torch::zeros({(int64_t)max_frames}, torch::kInt32);
+    torch::Tensor ids_off =
+        torch::zeros({(int64_t)max_frames + 1}, torch::kInt32);
+    uint8_t* lbuf = lines.data_ptr<uint8_t>();
+    int32_t* lenp = lens.data_ptr<int32_t>();
+    int32_t* offp = ids_off.data_ptr<int32_t>();
+    std::string blob;
+    std::vector<std::pair<const uint8_t*, size_t>> raw;
+    std::vector<dmx_proto::LogSpan> spans;
+    int64_t rows = 0;
+    uint64_t consumed = 0;
     {
       py::gil_scoped_release release;
       wait_data(timeout_ms);
-      tail0 = tail = hdr_->tail.load(std::memory_order_relaxed);
-      const uint64_t head = hdr_->head.load(std::memory_order_acquire);
-      while ((int)raw.size() < max_frames && head - tail >= 4) {
-        uint32_t len = 0;
-        get_bytes(tail, (uint8_t*)&len, 4);
-        if (len > kMaxFrame || head - tail < 4 + (uint64_t)len) break;
-        const uint64_t off = (tail + 4) & mask_;
-        if (off + len <= size_) {
-          raw.emplace_back(data_ + off, (size_t)len);
-        } else {  // wraps: copy to scratch (rare: ~1 per ring lap)
-          wrapped.emplace_back();
-          wrapped.back().resize(len);
-          get_bytes(tail + 4, (uint8_t*)wrapped.back().data(), len);
-          raw.emplace_back((const uint8_t*)wrapped.back().data(), (size_t)len);
+      while (rows < (int64_t)max_frames) {
+        uint64_t tail = hdr_->tail.load(std::memory_order_relaxed);
+        const uint64_t head = hdr_->head.load(std::memory_order_acquire);
+        const uint64_t chunk0 = tail;
+        raw.clear();
+        while (rows + (int64_t)raw.size() < (int64_t)max_frames &&
+               head - tail >= 4 && tail - chunk0 < chunk_bytes) {
+          uint32_t len = 0;
+          get_bytes(tail, (uint8_t*)&len, 4);
+          if (len > kMaxFrame || head - tail < 4 + (uint64_t)len) break;
+          const uint64_t off = (tail + 4) & mask_;
+          if (off + len <= size_) {
+            raw.emplace_back(data_ + off, (size_t)len);
+          } else {  // wraps: copy to scratch (rare: ~1 per ring lap)
+            wrapped.emplace_back();
+            wrapped.back().resize(len);
+            get_bytes(tail + 4, (uint8_t*)wrapped.back().data(), len);
+            raw.emplace_back((const uint8_t*)wrapped.back().data(),
+                             (size_t)len);
+          }
+          tail += 4 + len;
         }
-        tail += 4 + len;
+        if (raw.empty()) break;  // ring drained: return what we have
+        const int64_t n = (int64_t)raw.size();
+        spans.assign(n, dmx_proto::LogSpan{});
+        dmx_proto::decode_log_core(raw, max_len, lbuf + rows * max_len,
+                                   lenp + rows, spans);
+        for (int64_t i = 0; i < n; ++i) {
+          offp[rows + i] = (int32_t)blob.size();
+          blob.append(spans[i].id, spans[i].id_len);
+        }
+        rows += n;
+        offp[rows] = (int32_t)blob.size();
+        consumed += tail - chunk0;
+        // spans/ids copied out: release this chunk's ring space NOW so
+        // the feeder refills while we decode the next chunk
+        hdr_->tail.store(tail, std::memory_order_release);
       }
     }
-    const int64_t B = (int64_t)raw.size();
-    if (B == 0) {
+    if (rows == 0) {
       auto opts = torch::TensorOptions().dtype(torch::kUInt8);
       return py::make_tuple(torch::zeros({0, max_len}, opts),
                             torch::zeros({0}, torch::kInt32), py::bytes(""),
                             torch::zeros({1}, torch::kInt32), 0);
     }
-    auto lopts = torch::TensorOptions().dtype(torch::kUInt8);
-    torch::Tensor lines;
-    if (pin) {  // constant-size pinned alloc (see frame_reader.cpp)
-      lopts = lopts.pinned_memory(true);
-      lines = torch::empty({(int64_t)max_frames, max_len}, lopts)
-                  .narrow(0, 0, B);
-    } else {
-      lines = torch::empty({B, max_len}, lopts);
-    }
-    auto lens = torch::zeros({B}, torch::kInt32);
-    auto ids_off = torch::zeros({B + 1}, torch::kInt32);
-    std::vector<dmx_proto::LogSpan> spans(B);
-    std::string blob;
-    {
-      py::gil_scoped_release release;
-      dmx_proto::decode_log_core(raw, max_len, lines.data_ptr<uint8_t>(),
-                                 lens.data_ptr<int32_t>(), spans);
-      int32_t* off = ids_off.data_ptr<int32_t>();
-      size_t total = 0;
-      for (int64_t i = 0; i < B; ++i) total += spans[i].id_len;
-      blob.reserve(total);
-      for (int64_t i = 0; i < B; ++i) {
-        off[i] = (int32_t)blob.size();
-        blob.append(spans[i].id, spans[i].id_len);
-      }
-      off[B] = (int32_t)blob.size();
-      // release the ring space only AFTER the decode read everything
-      hdr_->tail.store(tail, std::memory_order_release);
-    }
-    return py::make_tuple(lines, lens, py::bytes(blob), ids_off,
-                          (int64_t)(tail - tail0));
+    return py::make_tuple(lines.narrow(0, 0, rows), lens.narrow(0, 0, rows),
+                          py::bytes(blob), ids_off.narrow(0, 0, rows + 1),
+                          (int64_t)consumed);
+  }
+
+  static size_t read_chunk_bytes() {
+    static size_t v = [] {
+      const char* e = std::getenv("DMX_SHM_READ_CHUNK_BYTES");
+      long long n = e ? atoll(e) : 0;
+      return (size_t)(n > 0 ? n : (2ll << 20));
+    }();
+    return v;
   }
 
  private:
