@@ -203,14 +203,35 @@ class ShmRing {
     // vector reallocation would move the string objects and dangle every
     // SSO pointer already recorded. Deque push_back never moves elements.
     std::deque<std::string> wrapped;
-    auto lopts = torch::TensorOptions().dtype(torch::kUInt8);
-    if (pin) lopts = lopts.pinned_memory(true);
-    // constant-size alloc (cache-friendly for the pinned allocator);
-    // narrowed to the decoded row count at the end
-    torch::Tensor lines = torch::empty({(int64_t)max_frames, max_len}, lopts);
-    torch::Tensor lens = torch::zeros({(int64_t)max_frames}, torch::kInt32);
-    torch::Tensor ids_off =
-        torch::zeros({(int64_t)max_frames + 1}, torch::kInt32);
+    // PREALLOCATED rotating staging (kStagingDepth buffers): per-read
+    // torch::empty of the [max_frames, max_len] output re-pays
+    // allocator/page-fault cost every call — measured 10-17 ms/read at
+    // >=32768 frames on the target host (profiles/r12) with BOTH pinned
+    // and pageable allocs, while identical reads into reused buffers are
+    // ~2 ms. Contract: a returned batch is valid until kStagingDepth-1
+    // further reads (the engine's pipelined depth-1 holds at most 2).
+    if (stage_.empty() || stage_mf_ != max_frames || stage_ml_ != max_len ||
+        stage_pin_ != pin) {
+      stage_.clear();
+      auto so = torch::TensorOptions().dtype(torch::kUInt8);
+      if (pin) so = so.pinned_memory(true);
+      for (int i = 0; i < kStagingDepth; ++i) {
+        stage_.push_back({
+            torch::empty({(int64_t)max_frames, max_len}, so),
+            torch::zeros({(int64_t)max_frames}, torch::kInt32),
+            torch::zeros({(int64_t)max_frames + 1}, torch::kInt32),
+        });
+      }
+      stage_mf_ = max_frames;
+      stage_ml_ = max_len;
+      stage_pin_ = pin;
+      stage_i_ = 0;
+    }
+    Staging& sbuf = stage_[stage_i_];
+    stage_i_ = (stage_i_ + 1) % kStagingDepth;
+    torch::Tensor lines = sbuf.lines;
+    torch::Tensor lens = sbuf.lens;
+    torch::Tensor ids_off = sbuf.ids_off;
     uint8_t* lbuf = lines.data_ptr<uint8_t>();
     int32_t* lenp = lens.data_ptr<int32_t>();
     int32_t* offp = ids_off.data_ptr<int32_t>();
@@ -280,6 +301,14 @@ class ShmRing {
     }();
     return v;
   }
+
+  struct Staging {
+    torch::Tensor lines, lens, ids_off;
+  };
+  static constexpr int kStagingDepth = 4;
+  std::vector<Staging> stage_;
+  int stage_mf_ = -1, stage_ml_ = -1, stage_i_ = 0;
+  bool stage_pin_ = false;
 
  private:
   void wait_data(int timeout_ms) {  // GIL must be released by caller
