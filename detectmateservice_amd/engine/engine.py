@@ -303,15 +303,6 @@ class Engine:
         self._log.info(
             "engine PACKED loop started on %s (max_len=%d pin=%s pipelined=%s)",
             s.engine_addr, max_len, pin, pipelined)
-        if s.engine_batch_size >= 32768:
-            # measured host memory-system cliff: single in-place ring
-            # reads of >=32768 frames collapse throughput ~6x on the
-            # target boxes (profiles/r12_engine_batch_cliff.txt)
-            self._log.warning(
-                "engine_batch_size=%d is past the measured packed-read "
-                "cliff (>=32768); throughput is best at <=30720 "
-                "(see profiles/r12_engine_batch_cliff.txt)",
-                s.engine_batch_size)
         stats_on = os.environ.get("DMX_ENGINE_STATS") == "1"
         st = {"n": 0, "frames": 0, "recv": 0.0, "proc": 0.0,
               "submit": 0.0, "drain": 0.0, "last": time.perf_counter()}
