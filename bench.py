@@ -223,6 +223,9 @@ def main() -> None:
         for i, nlines in enumerate(shard_lines):
             path = f"/dev/shm/dmx-bench-{run_id}-r{rank}s{i}"
             ring = _dmx_C.ShmRing(path, ring_bytes, True)
+            # chunks sit in a depth-8 queue + upload/compute in flight:
+            # the ring's rotating staging must outlive them all
+            ring.set_staging_depth(12)
             # per-shard frame pool: distinct pre-serialized protobuf
             # batches (generation + serialization are reader work the
             # reference does outside the service too — the TIMED work is

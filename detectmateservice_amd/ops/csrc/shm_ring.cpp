@@ -203,19 +203,21 @@ class ShmRing {
     // vector reallocation would move the string objects and dangle every
     // SSO pointer already recorded. Deque push_back never moves elements.
     std::deque<std::string> wrapped;
-    // PREALLOCATED rotating staging (kStagingDepth buffers): per-read
+    // PREALLOCATED rotating staging (stage_depth_ buffers): per-read
     // torch::empty of the [max_frames, max_len] output re-pays
     // allocator/page-fault cost every call — measured 10-17 ms/read at
     // >=32768 frames on the target host (profiles/r12) with BOTH pinned
     // and pageable allocs, while identical reads into reused buffers are
-    // ~2 ms. Contract: a returned batch is valid until kStagingDepth-1
-    // further reads (the engine's pipelined depth-1 holds at most 2).
-    if (stage_.empty() || stage_mf_ != max_frames || stage_ml_ != max_len ||
+    // ~2 ms. Contract: a returned batch is valid until stage_depth_-1
+    // further reads (the engine's pipelined depth-1 holds at most 2;
+    // consumers that queue chunks deeper call set_staging_depth first).
+    if (stage_.empty() || (int)stage_.size() != stage_depth_ ||
+        stage_mf_ != max_frames || stage_ml_ != max_len ||
         stage_pin_ != pin) {
       stage_.clear();
       auto so = torch::TensorOptions().dtype(torch::kUInt8);
       if (pin) so = so.pinned_memory(true);
-      for (int i = 0; i < kStagingDepth; ++i) {
+      for (int i = 0; i < stage_depth_; ++i) {
         stage_.push_back({
             torch::empty({(int64_t)max_frames, max_len}, so),
             torch::zeros({(int64_t)max_frames}, torch::kInt32),
@@ -228,7 +230,7 @@ class ShmRing {
       stage_i_ = 0;
     }
     Staging& sbuf = stage_[stage_i_];
-    stage_i_ = (stage_i_ + 1) % kStagingDepth;
+    stage_i_ = (stage_i_ + 1) % stage_depth_;
     torch::Tensor lines = sbuf.lines;
     torch::Tensor lens = sbuf.lens;
     torch::Tensor ids_off = sbuf.ids_off;
@@ -305,10 +307,19 @@ class ShmRing {
   struct Staging {
     torch::Tensor lines, lens, ids_off;
   };
-  static constexpr int kStagingDepth = 4;
   std::vector<Staging> stage_;
+  int stage_depth_ = 4;
   int stage_mf_ = -1, stage_ml_ = -1, stage_i_ = 0;
   bool stage_pin_ = false;
+
+ public:
+  // Deepen the rotation when the consumer holds more than 2 returned
+  // batches at once (e.g. bench.py queues up to 8 chunks per shard).
+  void set_staging_depth(int depth) {
+    if (depth < 2) throw std::runtime_error("staging depth must be >= 2");
+    stage_depth_ = depth;
+    stage_.clear();  // rebuilt at the next read
+  }
 
  private:
   void wait_data(int timeout_ms) {  // GIL must be released by caller
@@ -446,7 +457,12 @@ void register_shm_ring(py::module_& m) {
            py::arg("timeout_ms") = 100)
       .def("read_batch_packed", &ShmRing::read_batch_packed,
            py::arg("max_frames") = 4096, py::arg("timeout_ms") = 100,
-           py::arg("max_len") = 256, py::arg("pin") = false)
+           py::arg("max_len") = 256, py::arg("pin") = false,
+           "Decode up to max_frames into a ROTATING preallocated buffer "
+           "set; the returned batch is overwritten after staging_depth-1 "
+           "further reads (default 4: safe to hold 3 batches)")
+      .def("set_staging_depth", &ShmRing::set_staging_depth,
+           py::arg("depth"))
       .def("pending", &ShmRing::pending);
   py::class_<ShmFeeder>(m, "ShmFeeder")
       .def(py::init<const std::string&, const std::vector<py::bytes>&,
