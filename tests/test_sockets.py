@@ -927,3 +927,63 @@ def test_shm_ring_randomized_wrap_integrity(monkeypatch):
     finally:
         import os as os_mod
         os_mod.unlink(path)
+
+
+def test_shm_ring_staging_rotation_and_chunked_decode(monkeypatch):
+    """The packed read decodes into ROTATING preallocated buffers: a
+    returned batch stays valid for staging_depth-1 further reads, and a
+    read larger than the 2 MB decode chunk (multi-chunk path, early
+    tail release) must still produce byte-exact rows in order."""
+    import uuid as uuid_mod
+
+    from detectmateservice_amd import ops
+    from detectmateservice_amd.schemas import LogSchema
+
+    if not ops.have_extension():
+        pytest.skip("extension not built")
+    from detectmateservice_amd.ops import _dmx_C
+
+    path = f"/dev/shm/dmx-stage-{uuid_mod.uuid4().hex[:8]}"
+    ring = _dmx_C.ShmRing(path, 64 << 20, True)
+    try:
+        # --- multi-chunk: 24k frames x ~190B > 2 MB chunk bound -------
+        frames = [LogSchema(logID=f"id{i}", log=f"line-{i:06d} " + "x" * 150)
+                  .serialize() for i in range(24000)]
+        assert ring.write_frames(frames) == 24000
+        lines, lens, blob, off, nb = ring.read_batch_packed(
+            24000, 1000, 256, False)
+        assert lines.shape[0] == 24000
+        for i in (0, 9999, 23999):  # rows spanning several chunks
+            row = bytes(lines[i, : int(lens[i])].numpy().tobytes())
+            assert row == f"line-{i:06d} ".encode() + b"x" * 150
+            lo, hi = int(off[i]), int(off[i + 1])
+            assert blob[lo:hi] == f"id{i}".encode()
+
+        # --- rotation contract: depth 4 => valid for 3 more reads -----
+        held = []
+        for k in range(4):
+            batch = [LogSchema(logID=f"b{k}", log=f"batch-{k}").serialize()
+                     for _ in range(8)]
+            assert ring.write_frames(batch) == 8
+            l, n, _, _, _ = ring.read_batch_packed(8, 1000, 256, False)
+            held.append((k, l, n))
+        # the FIRST batch was overwritten by the 4th read (same buffer);
+        # batches 1..3 must still be intact
+        for k, l, n in held[1:]:
+            row = bytes(l[0, : int(n[0])].numpy().tobytes())
+            assert row == f"batch-{k}".encode()
+        # set_staging_depth rebuilds: a deeper rotation keeps all 5
+        ring.set_staging_depth(6)
+        held = []
+        for k in range(5):
+            batch = [LogSchema(logID=f"c{k}", log=f"cycle-{k}").serialize()
+                     for _ in range(4)]
+            ring.write_frames(batch)
+            l, n, _, _, _ = ring.read_batch_packed(4, 1000, 256, False)
+            held.append((k, l, n))
+        for k, l, n in held:
+            row = bytes(l[0, : int(n[0])].numpy().tobytes())
+            assert row == f"cycle-{k}".encode()
+    finally:
+        import os as os_mod
+        os_mod.unlink(path)
