@@ -407,3 +407,58 @@ def test_fanout_sink_degrades_when_head_dies(tmp_path, free_port):
     results = _launch("_body_fanout_sink_degrade", tmp_path, 2, free_port,
                       timeout=120, expect_results=1)
     assert results[1] == "sink-degraded-and-serving"
+
+
+def _body_stage_head_degrade(rank, world, tmp):
+    """STAGE topology: the downstream rank dies; the head's P2P forward
+    fails and the head degrades to its socket loop (forward-failure
+    branch of _dist_head_loop/_dist_sink_loop)."""
+    import threading
+
+    from detectmateservice_amd import Service
+    from detectmateservice_amd.engine.sockets import PairDialer, PairListener, RecvTimeout
+
+    settings = _settings(
+        tmp, "stgdeg", dist_mode="stage",
+        out_addr=[f"ipc://{tmp}/stgdeg-out-{{rank}}.ipc"],
+        dist_timeout_s=6.0,
+    )
+    svc = Service(settings)
+    t = threading.Thread(target=svc.run, daemon=True)
+    t.start()
+    if rank == 1:
+        time.sleep(2.5)
+        os._exit(0)  # downstream hard death
+    try:
+        time.sleep(0.5)
+        feeder = PairDialer(svc.settings.engine_addr)
+        assert feeder.wait_connected(10.0)
+        sink = PairListener(f"ipc://{tmp}/stgdeg-out-0.ipc")
+        assert feeder.send(b"pre-death", block=True)
+        time.sleep(4.0)
+        got = None
+        deadline = time.monotonic() + 40
+        n = 0
+        while got is None and time.monotonic() < deadline:
+            feeder.send(b"post-death-%d" % n, block=False)
+            n += 1
+            try:
+                frame = sink.recv(timeout_ms=500)
+            except RecvTimeout:
+                continue
+            if frame.startswith(b"post-death"):
+                got = frame
+        assert got is not None
+        assert svc.metrics.engine_dist_degraded._value.get() >= 1
+        sink.close()
+        feeder.close()
+        return "stage-head-degraded"
+    finally:
+        svc.shutdown()
+        t.join(timeout=10.0)
+
+
+def test_stage_head_degrades_when_downstream_dies(tmp_path, free_port):
+    results = _launch("_body_stage_head_degrade", tmp_path, 2, free_port,
+                      timeout=120, expect_results=1)
+    assert results[0] == "stage-head-degraded"
