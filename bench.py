@@ -110,7 +110,7 @@ class ShardReader(threading.Thread):
     time (GIL-released C++ scan + protobuf decode, shm_ring.cpp)."""
 
     def __init__(self, ring, lines_per_step, n_steps, chunk, max_len, out_q,
-                 pin):
+                 pin, device=None, copy_stream=None):
         super().__init__(daemon=True)
         self.ring = ring
         self.lines_per_step = lines_per_step
@@ -119,11 +119,19 @@ class ShardReader(threading.Thread):
         self.max_len = max_len
         self.out_q = out_q
         self.pin = pin
+        self.device = device
+        self.copy_stream = copy_stream
 
     def run(self):
         # per-CHUNK queueing: the GPU starts on a step's first decoded
-        # chunk while this thread is still decoding the rest of it
+        # chunk while this thread is still decoding the rest of it.
+        # On GPU the H2D upload happens HERE (eagerly, on the shared
+        # copy stream) so every chunk of step N+1 lands on-device during
+        # step N's compute — the step boundary no longer exposes the
+        # last chunk's decode+copy tail.
         try:
+            if self.copy_stream is not None:
+                torch.cuda.set_device(self.device)
             for _ in range(self.n_steps):
                 got = 0
                 while got < self.lines_per_step:
@@ -133,7 +141,17 @@ class ShardReader(threading.Thread):
                     b = int(lines.shape[0])
                     if b == 0:
                         continue
-                    self.out_q.put((lines, lens))
+                    if self.copy_stream is not None:
+                        with torch.cuda.stream(self.copy_stream):
+                            dl = lines.to(self.device, non_blocking=True)
+                            dn = lens.to(self.device, non_blocking=True)
+                            ev = torch.cuda.Event()
+                            ev.record(self.copy_stream)
+                        # keep the host (staging) tensors referenced
+                        # until the consumer observed the copy event
+                        self.out_q.put((dl, dn, ev, lines, lens))
+                    else:
+                        self.out_q.put((lines, lens))
                     got += b
                 self.out_q.put("step_end")
         except Exception as exc:  # noqa: BLE001 - surface in the main loop
@@ -220,6 +238,10 @@ def main() -> None:
         rings, feeders, readers, queues = [], [], [], []
         ring_bytes = 64 << 20
         run_id = uuid.uuid4().hex[:8]
+        # H2D copies ride a dedicated stream, driven by the READER
+        # threads right after decode: chunk N+1's upload overlaps chunk
+        # N's kernels and the step boundary exposes no copy tail
+        copy_stream = torch.cuda.Stream(device) if use_gpu else None
         for i, nlines in enumerate(shard_lines):
             path = f"/dev/shm/dmx-bench-{run_id}-r{rank}s{i}"
             ring = _dmx_C.ShmRing(path, ring_bytes, True)
@@ -239,28 +261,15 @@ def main() -> None:
             q: "queue.Queue" = queue.Queue(maxsize=8)
             feeders.append(ShardFeeder(ring, pool, nlines * n_total_steps))
             readers.append(ShardReader(ring, nlines, n_total_steps,
-                                        args.chunk, args.max_len, q, use_gpu))
+                                        args.chunk, args.max_len, q, use_gpu,
+                                        device if use_gpu else None,
+                                        copy_stream))
             rings.append((ring, path))
             queues.append(q)
         for f in feeders:
             f.start()
         for r in readers:
             r.start()
-
-        # H2D copies ride a dedicated stream so chunk N+1's upload
-        # overlaps chunk N's kernels (events order the compute stream
-        # behind each chunk's copy); no-op on CPU
-        copy_stream = torch.cuda.Stream(device) if use_gpu else None
-
-        def _upload(lines, lens):
-            if copy_stream is None:
-                return lines.to(device), lens.to(device), None
-            with torch.cuda.stream(copy_stream):
-                dl = lines.to(device, non_blocking=True)
-                dn = lens.to(device, non_blocking=True)
-                ev = torch.cuda.Event()
-                ev.record(copy_stream)
-            return dl, dn, ev
 
         def step(i: int) -> None:
             outs = []
@@ -271,10 +280,12 @@ def main() -> None:
                         raise item
                     if item == "step_end":
                         break
-                    lines, lens = item
-                    dl, dn, ev = _upload(lines, lens)
-                    if ev is not None:
+                    if len(item) == 5:  # GPU: reader already uploaded
+                        dl, dn, ev, _hl, _hn = item
                         torch.cuda.current_stream(device).wait_event(ev)
+                    else:  # CPU fallback: host tensors
+                        lines, lens = item
+                        dl, dn = lines.to(device), lens.to(device)
                     if line_buf is not None:
                         line_buf.append(dl, dn)
                     outs.append(pipe.process_packed(dl, dn))
