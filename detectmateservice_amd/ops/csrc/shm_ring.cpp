@@ -25,6 +25,7 @@
 #include <sys/stat.h>
 #include <unistd.h>
 
+#include <algorithm>
 #include <atomic>
 #include <chrono>
 #include <cerrno>
@@ -211,20 +212,24 @@ class ShmRing {
     // ~2 ms. Contract: a returned batch is valid until stage_depth_-1
     // further reads (the engine's pipelined depth-1 holds at most 2;
     // consumers that queue chunks deeper call set_staging_depth first).
+    // sticky sizing: rebuild only when the request OUTGROWS the staging
+    // (or max_len/pin change) — consumers that alternate read sizes
+    // (e.g. a bench tail chunk) must not re-pay the allocation
     if (stage_.empty() || (int)stage_.size() != stage_depth_ ||
-        stage_mf_ != max_frames || stage_ml_ != max_len ||
+        stage_mf_ < max_frames || stage_ml_ != max_len ||
         stage_pin_ != pin) {
+      const int mf = std::max(max_frames, stage_mf_);
       stage_.clear();
       auto so = torch::TensorOptions().dtype(torch::kUInt8);
       if (pin) so = so.pinned_memory(true);
       for (int i = 0; i < stage_depth_; ++i) {
         stage_.push_back({
-            torch::empty({(int64_t)max_frames, max_len}, so),
-            torch::zeros({(int64_t)max_frames}, torch::kInt32),
-            torch::zeros({(int64_t)max_frames + 1}, torch::kInt32),
+            torch::empty({(int64_t)mf, max_len}, so),
+            torch::zeros({(int64_t)mf}, torch::kInt32),
+            torch::zeros({(int64_t)mf + 1}, torch::kInt32),
         });
       }
-      stage_mf_ = max_frames;
+      stage_mf_ = mf;
       stage_ml_ = max_len;
       stage_pin_ = pin;
       stage_i_ = 0;
