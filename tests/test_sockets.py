@@ -987,3 +987,39 @@ def test_shm_ring_staging_rotation_and_chunked_decode(monkeypatch):
     finally:
         import os as os_mod
         os_mod.unlink(path)
+
+
+def test_shm_ring_staging_growth_and_depth_validation(monkeypatch):
+    """Sticky staging: a smaller read after a bigger one must NOT
+    rebuild (alternating bench tail chunks), a bigger one grows the
+    buffers; set_staging_depth rejects depth < 2."""
+    import uuid as uuid_mod
+
+    from detectmateservice_amd import ops
+    from detectmateservice_amd.schemas import LogSchema
+
+    if not ops.have_extension():
+        pytest.skip("extension not built")
+    from detectmateservice_amd.ops import _dmx_C
+
+    path = f"/dev/shm/dmx-grow-{uuid_mod.uuid4().hex[:8]}"
+    ring = _dmx_C.ShmRing(path, 1 << 22, True)
+    try:
+        def rt(n, mf):
+            batch = [LogSchema(logID=f"g{i}", log=f"grow-{i}").serialize()
+                     for i in range(n)]
+            assert ring.write_frames(batch) == n
+            l, ln, _, _, _ = ring.read_batch_packed(mf, 1000, 128, False)
+            assert l.shape[0] == n
+            assert bytes(l[n - 1, : int(ln[n - 1])].numpy().tobytes()) \
+                == f"grow-{n-1}".encode()
+
+        rt(64, 1024)   # allocate at 1024
+        rt(16, 64)     # smaller mf: reuses (no rebuild), correctness holds
+        rt(128, 4096)  # grows
+        rt(8, 16)      # shrunk request against the grown buffers
+        with pytest.raises(RuntimeError):
+            ring.set_staging_depth(1)
+    finally:
+        import os as os_mod
+        os_mod.unlink(path)
